@@ -1,0 +1,132 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: managed Llama-3-8B training throughput on MI355X.
+
+Measures the metric BASELINE.json names — tokens/sec of Llama-3-8B at
+1/2/4/8 MI355X GPUs (DDP over RCCL/xGMI) on synthetic data with
+random-init weights.
+
+Single GPU:   python bench.py --gpus 1 --steps 10 --warmup 3
+Multi GPU:    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                  --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Rank 0 prints exactly ONE JSON line on stdout.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+
+def log(msg: str) -> None:
+    print(msg, file=sys.stderr, flush=True)
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=10)
+    ap.add_argument("--warmup", type=int, default=3)
+    ap.add_argument("--model", default=None,
+                    help="model config (default: llama3-8b on GPU, "
+                         "llama-tiny on CPU)")
+    ap.add_argument("--micro-batch", type=int, default=1)
+    ap.add_argument("--grad-accum", type=int, default=4)
+    ap.add_argument("--seq-len", type=int, default=4096)
+    ap.add_argument("--checkpoint-activations", action="store_true")
+    ap.add_argument("--bucket-mb", type=int, default=128)
+    args = ap.parse_args()
+
+    import torch
+    import torch.distributed as dist
+
+    from trainingjob_operator_amd.parallel import dist_ctx
+    from trainingjob_operator_amd.training import TrainConfig, Trainer
+
+    on_gpu = torch.cuda.is_available()
+    model = args.model or ("llama3-8b" if on_gpu else "llama-tiny")
+    seq_len = args.seq_len if on_gpu else min(args.seq_len, 128)
+
+    ctx = dist_ctx.from_env()
+    if "RANK" not in os.environ and args.gpus == 1:
+        ctx.world_size = 1
+    dist_ctx.init_process_group(ctx)
+    n_gpus = ctx.world_size if ctx.is_distributed else 1
+
+    cfg = TrainConfig(
+        model=model,
+        micro_batch=args.micro_batch,
+        grad_accum=args.grad_accum,
+        seq_len=seq_len,
+        checkpoint_activations=args.checkpoint_activations,
+        bucket_bytes=args.bucket_mb << 20,
+    )
+    log(f"[bench] rank {ctx.rank}/{n_gpus} model={model} "
+        f"mb={cfg.micro_batch} ga={cfg.grad_accum} seq={cfg.seq_len} "
+        f"device={'cuda:%d' % ctx.local_rank if on_gpu else 'cpu'}")
+
+    t_build = time.perf_counter()
+    trainer = Trainer(cfg, ctx)
+    log(f"[bench] built in {time.perf_counter() - t_build:.1f}s "
+        f"({cfg.model_config.n_params / 1e9:.2f}B params)")
+
+    for i in range(args.warmup):
+        loss = trainer.train_step()
+        log(f"[bench] warmup {i}: loss={loss.item():.4f}")
+
+    if ctx.is_distributed:
+        dist.barrier()
+    if on_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        trainer.train_step()
+    if on_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    if ctx.is_distributed:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=trainer.device if on_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+        dist.barrier()
+
+    tokens_per_step_rank = cfg.tokens_per_step_per_rank()
+    total_tokens = tokens_per_step_rank * n_gpus * args.steps
+    tokens_per_sec = total_tokens / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if ctx.rank == 0:
+        result = {
+            "metric": "tokens_per_sec_llama3_8b" if model == "llama3-8b"
+                      else f"tokens_per_sec_{model}",
+            "value": round(tokens_per_sec, 1),
+            "unit": "tokens/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 1),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": model,
+                "global_batch": cfg.micro_batch * cfg.grad_accum * n_gpus,
+                "seq_len": cfg.seq_len,
+                "parallelism": f"dp{n_gpus}",
+                "micro_batch": cfg.micro_batch,
+                "grad_accum": cfg.grad_accum,
+                "checkpoint_activations": cfg.checkpoint_activations,
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    dist_ctx.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,112 @@
+"""Multi-process DDP correctness over gloo (world_size 2, CPU).
+
+Validates the distributed path the driver's 8-GPU scaling bench exercises:
+bucketed SUM all-reduce over the flat grad buffer + 1/world pre-scale in
+the optimizer must keep ranks bit-identical and match a single-process run
+on the combined batch.
+"""
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from trainingjob_operator_amd.parallel.dist_ctx import DistContext
+
+
+def _free_port():
+    import socket
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _worker(rank, world, port, results):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from trainingjob_operator_amd.training import TrainConfig, Trainer
+        ctx = DistContext(rank=rank, world_size=world, backend="gloo")
+        cfg = TrainConfig(model="llama-tiny", micro_batch=1, grad_accum=2,
+                          seq_len=32, lr=1e-3, bucket_bytes=4096)
+        trainer = Trainer(cfg, ctx)
+        for _ in range(3):
+            loss = trainer.train_step()
+        # ranks must agree on params after synced steps
+        flat = trainer.store.flat_param.float()
+        other = flat.clone()
+        dist.broadcast(other, src=0)
+        assert torch.equal(flat, other), f"rank {rank} diverged"
+        results[rank] = flat.sum().item()
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_ddp_two_ranks_stay_in_sync():
+    port = _free_port()
+    world = 2
+    with mp.Manager() as mgr:
+        results = mgr.dict()
+        mp.spawn(_worker, args=(world, port, results), nprocs=world,
+                 join=True)
+        assert len(results) == world
+        assert results[0] == results[1]
+
+
+def _worker_vs_single(rank, world, port, results):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from trainingjob_operator_amd.parallel.flat import FlatParamStore
+        from trainingjob_operator_amd.parallel.ddp import DDPEngine
+        from trainingjob_operator_amd.models.config import LLAMA_TINY
+        from trainingjob_operator_amd.models.llama import LlamaModel
+        torch.manual_seed(7)
+        model = LlamaModel(LLAMA_TINY)
+        store = FlatParamStore(model)
+        ddp = DDPEngine(store, bucket_bytes=2048)
+        # per-rank distinct batch
+        g = torch.Generator().manual_seed(100 + rank)
+        tokens = torch.randint(0, LLAMA_TINY.vocab_size, (1, 16), generator=g)
+        loss = model(tokens, tokens)
+        loss.backward()
+        ddp.finish_backward()
+        grads = store.flat_grad.float() * ddp.grad_pre_scale
+        results[rank] = grads
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_ddp_grads_match_single_process_average():
+    port = _free_port()
+    world = 2
+    with mp.Manager() as mgr:
+        results = mgr.dict()
+        mp.spawn(_worker_vs_single, args=(world, port, results), nprocs=world,
+                 join=True)
+        dist_grads = results[0]
+
+    # single-process equivalent: average of both ranks' grads
+    from trainingjob_operator_amd.parallel.flat import FlatParamStore
+    from trainingjob_operator_amd.models.config import LLAMA_TINY
+    from trainingjob_operator_amd.models.llama import LlamaModel
+    acc = None
+    for rank in range(world):
+        torch.manual_seed(7)
+        model = LlamaModel(LLAMA_TINY)
+        store = FlatParamStore(model)
+        g = torch.Generator().manual_seed(100 + rank)
+        tokens = torch.randint(0, LLAMA_TINY.vocab_size, (1, 16), generator=g)
+        model(tokens, tokens).backward()
+        acc = store.flat_grad.float() if acc is None \
+            else acc + store.flat_grad.float()
+    expected = acc / world
+    # bf16 all-reduce rounding: tolerance scaled to grad magnitude
+    assert torch.allclose(dist_grads, expected, atol=5e-3, rtol=5e-2)

@@ -1,0 +1,69 @@
+"""End-to-end training engine tests on CPU (tiny Llama, reference op path)."""
+import torch
+
+from trainingjob_operator_amd.models.config import LLAMA_TINY
+from trainingjob_operator_amd.models.llama import LlamaModel
+from trainingjob_operator_amd.parallel.flat import FlatParamStore
+from trainingjob_operator_amd.training import TrainConfig, Trainer
+
+torch.manual_seed(0)
+
+
+def test_model_forward_shapes():
+    model = LlamaModel(LLAMA_TINY)
+    tokens = torch.randint(0, LLAMA_TINY.vocab_size, (2, 16))
+    logits = model(tokens)
+    assert logits.shape == (2, 16, LLAMA_TINY.vocab_size)
+    loss = model(tokens, tokens)
+    assert loss.dim() == 0 and torch.isfinite(loss)
+
+
+def test_param_count_matches_formula():
+    model = LlamaModel(LLAMA_TINY)
+    n = sum(p.numel() for p in model.parameters())
+    assert n == LLAMA_TINY.n_params
+
+
+def test_flat_store_aliases_params_and_grads():
+    model = LlamaModel(LLAMA_TINY)
+    store = FlatParamStore(model)
+    # params are views into the flat buffer
+    base = store.flat_param.data_ptr()
+    end = base + store.flat_param.numel() * store.flat_param.element_size()
+    for p in model.parameters():
+        assert base <= p.data_ptr() < end
+    # grads accumulate into the flat grad buffer via hooks
+    tokens = torch.randint(0, LLAMA_TINY.vocab_size, (2, 16))
+    loss = model(tokens.to(torch.long), tokens)
+    loss.backward()
+    assert store.flat_grad.abs().sum() > 0
+    for p in model.parameters():
+        assert p.grad is None  # folded into flat and released
+    # accumulation across two backwards adds up
+    g1 = store.flat_grad.clone()
+    loss = model(tokens, tokens)
+    loss.backward()
+    assert torch.allclose(store.flat_grad.float(), (g1.float() * 2), atol=2e-1)
+
+
+def test_trainer_loss_decreases():
+    cfg = TrainConfig(model="llama-tiny", micro_batch=2, grad_accum=2,
+                      seq_len=32, lr=1e-3, clip_grad_norm=1.0)
+    trainer = Trainer(cfg)
+    losses = [trainer.train_step().item() for _ in range(8)]
+    assert all(torch.isfinite(torch.tensor(l)) for l in losses)
+    # random data: loss should move toward uniform-vocab entropy, i.e. drop
+    # from the random-init value
+    assert losses[-1] < losses[0]
+
+
+def test_optimizer_updates_params():
+    cfg = TrainConfig(model="llama-tiny", micro_batch=1, grad_accum=1,
+                      seq_len=16)
+    trainer = Trainer(cfg)
+    before = trainer.store.flat_param.clone()
+    trainer.train_step()
+    assert not torch.equal(before, trainer.store.flat_param)
+    # fp32 master tracks the bf16 copy
+    assert torch.allclose(trainer.opt.p32.to(torch.bfloat16),
+                          trainer.store.flat_param)

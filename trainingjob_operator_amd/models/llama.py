@@ -1,0 +1,153 @@
+"""Llama-3 family, MI355X-first.
+
+Design (per /opt/skills/guides/MI355X_MICROARCH.md):
+  * bf16 weights/activations; plain projections go through hipBLASLt via
+    nn.functional.linear (library GEMMs belong to the library — the fused
+    non-GEMM hot ops are hand-written CDNA4 kernels in ops/).
+  * residual-add + RMSNorm fused to one HBM round trip (ops.fused_rmsnorm);
+    SwiGLU and RoPE fused likewise; loss is a fused streaming CE over the
+    [T, 128256] logits.
+  * attention = torch SDPA (flash path on ROCm) with grouped-query KV.
+  * activations optionally checkpointed per block; with 288 GB HBM3E per
+    GPU the default keeps activations resident (checkpointing off) — flip
+    it on only for long-sequence configs.
+
+This is worker-side compute the reference operator never had (it launches
+opaque containers — SURVEY.md §2.3); it is the workload the operator's
+benchmark jobs run and what bench.py measures.
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..ops import apply_rope, fused_cross_entropy, fused_rmsnorm, make_inv_freq, swiglu
+from .config import LlamaConfig
+
+
+class Attention(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        H = cfg.hidden_size
+        self.q_proj = nn.Linear(H, cfg.num_heads * cfg.head_dim, bias=False)
+        self.k_proj = nn.Linear(H, cfg.num_kv_heads * cfg.head_dim, bias=False)
+        self.v_proj = nn.Linear(H, cfg.num_kv_heads * cfg.head_dim, bias=False)
+        self.o_proj = nn.Linear(cfg.num_heads * cfg.head_dim, H, bias=False)
+
+    def forward(self, x: torch.Tensor, inv_freq: torch.Tensor) -> torch.Tensor:
+        B, S, H = x.shape
+        cfg = self.cfg
+        q = self.q_proj(x).view(B, S, cfg.num_heads, cfg.head_dim)
+        k = self.k_proj(x).view(B, S, cfg.num_kv_heads, cfg.head_dim)
+        v = self.v_proj(x).view(B, S, cfg.num_kv_heads, cfg.head_dim)
+        q = apply_rope(q, inv_freq, S)
+        k = apply_rope(k, inv_freq, S)
+        q = q.transpose(1, 2)  # [B, nh, S, D]
+        k = k.transpose(1, 2)
+        v = v.transpose(1, 2)
+        groups = cfg.num_heads // cfg.num_kv_heads
+        if groups > 1:
+            k = k.repeat_interleave(groups, dim=1)
+            v = v.repeat_interleave(groups, dim=1)
+        o = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+        o = o.transpose(1, 2).reshape(B, S, cfg.num_heads * cfg.head_dim)
+        return self.o_proj(o)
+
+
+class MLP(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        H, FF = cfg.hidden_size, cfg.intermediate_size
+        self.gate_proj = nn.Linear(H, FF, bias=False)
+        self.up_proj = nn.Linear(H, FF, bias=False)
+        self.down_proj = nn.Linear(FF, H, bias=False)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.down_proj(swiglu(self.gate_proj(x), self.up_proj(x)))
+
+
+class Block(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.attn = Attention(cfg)
+        self.mlp = MLP(cfg)
+        self.input_norm_weight = nn.Parameter(torch.ones(cfg.hidden_size))
+        self.post_attn_norm_weight = nn.Parameter(torch.ones(cfg.hidden_size))
+
+    def forward(self, x: torch.Tensor, residual: Optional[torch.Tensor],
+                inv_freq: torch.Tensor):
+        """Pre-norm block over a carried residual stream:
+        (x = branch output, residual = running stream)."""
+        normed, residual = fused_rmsnorm(x, self.input_norm_weight, residual,
+                                         self.cfg.norm_eps)
+        attn_out = self.attn(normed, inv_freq)
+        normed, residual = fused_rmsnorm(attn_out, self.post_attn_norm_weight,
+                                         residual, self.cfg.norm_eps)
+        mlp_out = self.mlp(normed)
+        return mlp_out, residual
+
+
+class LlamaModel(nn.Module):
+    def __init__(self, cfg: LlamaConfig, checkpoint_activations: bool = False):
+        super().__init__()
+        self.cfg = cfg
+        self.checkpoint_activations = checkpoint_activations
+        self.embed = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+        self.blocks = nn.ModuleList(Block(cfg) for _ in range(cfg.num_layers))
+        self.final_norm_weight = nn.Parameter(torch.ones(cfg.hidden_size))
+        self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False)
+        if cfg.tie_embeddings:
+            self.lm_head.weight = self.embed.weight
+        self.register_buffer(
+            "inv_freq", make_inv_freq(cfg.head_dim, cfg.rope_theta),
+            persistent=False)
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        std = self.cfg.init_std
+        for mod in self.modules():
+            if isinstance(mod, (nn.Linear, nn.Embedding)):
+                nn.init.normal_(mod.weight, mean=0.0, std=std)
+        for name, p in self.named_parameters():
+            if name.endswith("norm_weight"):
+                nn.init.ones_(p)
+        # scaled init for output projections (depth-aware, GPT-2 style)
+        scale = 1.0 / math.sqrt(2 * self.cfg.num_layers)
+        for blk in self.blocks:
+            nn.init.normal_(blk.attn.o_proj.weight, std=std * scale)
+            nn.init.normal_(blk.mlp.down_proj.weight, std=std * scale)
+
+    def forward_hidden(self, tokens: torch.Tensor) -> torch.Tensor:
+        """tokens [B, S] -> final normed hidden [B, S, H]."""
+        x = self.embed(tokens)
+        residual = None
+        for blk in self.blocks:
+            if self.checkpoint_activations and self.training:
+                if residual is None:
+                    residual = torch.zeros_like(x)
+                x, residual = torch.utils.checkpoint.checkpoint(
+                    blk, x, residual, self.inv_freq, use_reentrant=False)
+            else:
+                x, residual = blk(x, residual, self.inv_freq)
+        normed, _ = fused_rmsnorm(x, self.final_norm_weight, residual,
+                                  self.cfg.norm_eps)
+        return normed
+
+    def forward(self, tokens: torch.Tensor,
+                targets: Optional[torch.Tensor] = None):
+        """Returns loss (scalar, fp32) when targets given, else logits."""
+        hidden = self.forward_hidden(tokens)
+        logits = self.lm_head(hidden)
+        if targets is None:
+            return logits
+        T = logits.shape[0] * logits.shape[1]
+        per_tok = fused_cross_entropy(
+            logits.reshape(T, -1).contiguous(), targets.reshape(T))
+        n_valid = (targets.reshape(T) != -100).sum().clamp(min=1)
+        return per_tok.sum() / n_valid

@@ -1,0 +1,223 @@
+"""Autograd-wrapped ops: HIP/CDNA4 kernels on GPU, PyTorch reference on CPU.
+
+GPU dispatch is strict: if a tensor is on a HIP device and the in-tree
+libhipops.so is missing, these ops raise instead of silently running eager
+PyTorch (the native path must be the one that runs on a GPU box).
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from . import native, reference
+
+
+def _hip():
+    return native.load(require=True)
+
+
+def _ptr(t: Optional[torch.Tensor]):
+    return None if t is None else t.data_ptr()
+
+
+# ---------------------------------------------------------------------------
+# Fused residual-add + RMSNorm
+# ---------------------------------------------------------------------------
+
+class _FusedRMSNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, weight: torch.Tensor,
+                residual: Optional[torch.Tensor], eps: float):
+        H = x.shape[-1]
+        rows = x.numel() // H
+        if x.is_cuda:
+            assert x.dtype == torch.bfloat16 and x.is_contiguous()
+            lib = _hip()
+            y = torch.empty_like(x)
+            res_out = torch.empty_like(x) if residual is not None else None
+            rrms = torch.empty(rows, dtype=torch.float32, device=x.device)
+            lib.rmsnorm_fwd(native.stream_ptr(), _ptr(x), _ptr(residual),
+                            _ptr(weight), _ptr(y), _ptr(res_out), _ptr(rrms),
+                            rows, H, eps)
+        else:
+            y, res_out, rrms = reference.rmsnorm_fwd(x, residual, weight, eps)
+        saved_res = res_out if res_out is not None else x
+        ctx.save_for_backward(saved_res, weight, rrms)
+        ctx.has_residual = residual is not None
+        ctx.eps = eps
+        return (y, res_out) if residual is not None else (y, x)
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor, dres: Optional[torch.Tensor]):
+        res_out, weight, rrms = ctx.saved_tensors
+        H = res_out.shape[-1]
+        rows = res_out.numel() // H
+        if dy.is_cuda:
+            lib = _hip()
+            dy = dy.contiguous()
+            dx = torch.empty_like(res_out)
+            P = lib.rmsnorm_dw_partials()
+            dw_partial = torch.zeros(P, H, dtype=torch.float32, device=dy.device)
+            lib.rmsnorm_bwd(native.stream_ptr(), _ptr(dy), _ptr(res_out),
+                            _ptr(weight), _ptr(rrms), _ptr(dx),
+                            _ptr(dw_partial), rows, H)
+            dw32 = torch.zeros(H, dtype=torch.float32, device=dy.device)
+            lib.rmsnorm_dw_reduce(native.stream_ptr(), _ptr(dw_partial), None,
+                                  _ptr(dw32), H)
+            dw = dw32.to(weight.dtype)
+        else:
+            dx, dw32 = reference.rmsnorm_bwd(dy, res_out, weight, rrms)
+            dw = dw32.to(weight.dtype)
+        # res_out = x + residual, y = norm(res_out): both x and residual
+        # receive dx plus any gradient that arrived through res_out directly.
+        if dres is not None:
+            dx = dx + dres
+        if ctx.has_residual:
+            return dx, dw, dx, None
+        return dx, dw, None, None
+
+
+def fused_rmsnorm(x: torch.Tensor, weight: torch.Tensor,
+                  residual: Optional[torch.Tensor] = None,
+                  eps: float = 1e-5) -> Tuple[torch.Tensor, torch.Tensor]:
+    """y = RMSNorm(x + residual) * weight.
+
+    Returns (y, new_residual): new_residual = x + residual (the pre-norm
+    stream the next block consumes); when residual is None it is x itself.
+    One HBM round-trip on GPU (fused add+norm; reference operator has no
+    compute at all — SURVEY.md §2.3).
+    """
+    return _FusedRMSNorm.apply(x, weight, residual, eps)
+
+
+# ---------------------------------------------------------------------------
+# RoPE
+# ---------------------------------------------------------------------------
+
+class _Rope(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, inv_freq: torch.Tensor, seq_len: int):
+        ctx.seq_len = seq_len
+        ctx.save_for_backward(inv_freq)
+        return _rope_run(x, inv_freq, seq_len, 1.0)
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        (inv_freq,) = ctx.saved_tensors
+        return _rope_run(dy.contiguous(), inv_freq, ctx.seq_len, -1.0), None, None
+
+
+def _rope_run(x: torch.Tensor, inv_freq: torch.Tensor, seq_len: int,
+              sign: float) -> torch.Tensor:
+    orig_shape = x.shape
+    D = x.shape[-1]
+    n_heads = x.shape[-2]
+    xt = x.reshape(-1, n_heads, D)
+    if x.is_cuda:
+        assert x.dtype == torch.bfloat16 and xt.is_contiguous()
+        lib = _hip()
+        out = torch.empty_like(xt)
+        lib.rope(native.stream_ptr(), _ptr(xt), _ptr(out), _ptr(inv_freq),
+                 xt.shape[0], n_heads, seq_len, D, sign)
+    else:
+        out = reference.rope_rotate(xt, inv_freq, seq_len, sign)
+    return out.reshape(orig_shape)
+
+
+def apply_rope(x: torch.Tensor, inv_freq: torch.Tensor,
+               seq_len: int) -> torch.Tensor:
+    """Neox-style rotary embedding over [..., S, n_heads, D] (token dim folded:
+    position = flat_token_index % seq_len)."""
+    return _Rope.apply(x, inv_freq, seq_len)
+
+
+def make_inv_freq(head_dim: int, theta: float = 500000.0,
+                  device="cpu") -> torch.Tensor:
+    return 1.0 / (theta ** (torch.arange(0, head_dim, 2,
+                                         dtype=torch.float32,
+                                         device=device) / head_dim))
+
+
+# ---------------------------------------------------------------------------
+# SwiGLU
+# ---------------------------------------------------------------------------
+
+class _SwiGLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, g: torch.Tensor, u: torch.Tensor):
+        ctx.save_for_backward(g, u)
+        if g.is_cuda:
+            assert g.dtype == torch.bfloat16 and g.is_contiguous()
+            lib = _hip()
+            out = torch.empty_like(g)
+            lib.swiglu_fwd(native.stream_ptr(), _ptr(g), _ptr(u), _ptr(out),
+                           g.numel())
+            return out
+        return reference.swiglu_fwd(g, u)
+
+    @staticmethod
+    def backward(ctx, dout: torch.Tensor):
+        g, u = ctx.saved_tensors
+        if g.is_cuda:
+            lib = _hip()
+            dout = dout.contiguous()
+            dg = torch.empty_like(g)
+            du = torch.empty_like(u)
+            lib.swiglu_bwd(native.stream_ptr(), _ptr(dout), _ptr(g), _ptr(u),
+                           _ptr(dg), _ptr(du), g.numel())
+            return dg, du
+        return reference.swiglu_bwd(dout, g, u)
+
+
+def swiglu(g: torch.Tensor, u: torch.Tensor) -> torch.Tensor:
+    """silu(g) * u, fused (backward recomputes silu from g)."""
+    return _SwiGLU.apply(g, u)
+
+
+# ---------------------------------------------------------------------------
+# Fused cross-entropy
+# ---------------------------------------------------------------------------
+
+class _FusedCE(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits: torch.Tensor, targets: torch.Tensor,
+                ignore_index: int):
+        T, V = logits.shape
+        if logits.is_cuda:
+            assert logits.dtype == torch.bfloat16 and logits.is_contiguous()
+            assert V % 8 == 0
+            lib = _hip()
+            t32 = targets.to(torch.int32).contiguous()
+            lse = torch.empty(T, dtype=torch.float32, device=logits.device)
+            loss = torch.empty(T, dtype=torch.float32, device=logits.device)
+            lib.ce_fwd(native.stream_ptr(), _ptr(logits), _ptr(t32), _ptr(lse),
+                       _ptr(loss), T, V, ignore_index)
+        else:
+            loss, lse = reference.ce_fwd(logits, targets, ignore_index)
+            t32 = targets.to(torch.int32)
+        ctx.save_for_backward(logits, t32, lse)
+        ctx.ignore_index = ignore_index
+        return loss
+
+    @staticmethod
+    def backward(ctx, gout: torch.Tensor):
+        logits, t32, lse = ctx.saved_tensors
+        T, V = logits.shape
+        gscale = gout.to(torch.float32).contiguous()
+        if logits.is_cuda:
+            lib = _hip()
+            dlogits = torch.empty_like(logits)
+            lib.ce_bwd(native.stream_ptr(), _ptr(logits), _ptr(t32), _ptr(lse),
+                       _ptr(gscale), _ptr(dlogits), T, V, ctx.ignore_index)
+        else:
+            dlogits = reference.ce_bwd(logits, t32, lse, gscale,
+                                       ctx.ignore_index)
+        return dlogits, None, None
+
+
+def fused_cross_entropy(logits: torch.Tensor, targets: torch.Tensor,
+                        ignore_index: int = -100) -> torch.Tensor:
+    """Per-token CE loss vector (fp32, [T]); rows with ignore_index get 0.
+    Mean it over the valid-token count for the training loss."""
+    return _FusedCE.apply(logits, targets, ignore_index)

@@ -1,0 +1,83 @@
+"""ctypes bindings to the in-tree gfx950 HIP library (libhipops.so).
+
+The library is built by ``build_ops()`` (one direct hipcc invocation,
+``--offload-arch=gfx950`` — no hipify, no compatibility layers) and loaded
+from the package directory so the ``.so`` travels with the repo snapshot.
+
+Policy: on a GPU box these bindings are REQUIRED — ops fail loudly if the
+extension is missing rather than silently falling back to eager PyTorch.
+"""
+from __future__ import annotations
+
+import ctypes
+import os
+import subprocess
+from typing import Optional
+
+_PKG_DIR = os.path.dirname(os.path.abspath(__file__))
+LIB_PATH = os.path.join(_PKG_DIR, "libhipops.so")
+SRC_PATH = os.path.join(_PKG_DIR, "hip", "ops.hip")
+
+_lib: Optional[ctypes.CDLL] = None
+
+
+class HipOpsUnavailable(RuntimeError):
+    pass
+
+
+def build_ops(verbose: bool = False) -> str:
+    """Compile ops.hip -> libhipops.so in-tree for gfx950."""
+    cmd = [
+        "hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17",
+        "-shared", "-fPIC", SRC_PATH, "-o", LIB_PATH,
+    ]
+    if verbose:
+        print("+", " ".join(cmd))
+    subprocess.run(cmd, check=True)
+    return LIB_PATH
+
+
+def _sig(fn, argtypes, restype=None):
+    fn.argtypes = argtypes
+    fn.restype = restype
+
+
+def load(require: bool = True) -> Optional[ctypes.CDLL]:
+    """Load (and memoize) the HIP ops library."""
+    global _lib
+    if _lib is not None:
+        return _lib
+    if not os.path.exists(LIB_PATH):
+        if require:
+            raise HipOpsUnavailable(
+                f"{LIB_PATH} not built — run trainingjob_operator_amd.ops."
+                "native.build_ops() (or __graft_entry__.build())")
+        return None
+    lib = ctypes.CDLL(LIB_PATH)
+    vp, l, i, f = ctypes.c_void_p, ctypes.c_long, ctypes.c_int, ctypes.c_float
+    _sig(lib.hipops_arch_check, [], i)
+    _sig(lib.rmsnorm_fwd, [vp, vp, vp, vp, vp, vp, vp, l, i, f])
+    _sig(lib.rmsnorm_bwd, [vp, vp, vp, vp, vp, vp, vp, l, i])
+    _sig(lib.rmsnorm_dw_partials, [], i)
+    _sig(lib.rmsnorm_dw_reduce, [vp, vp, vp, vp, i])
+    _sig(lib.rope, [vp, vp, vp, vp, l, i, i, i, f])
+    _sig(lib.swiglu_fwd, [vp, vp, vp, vp, l])
+    _sig(lib.swiglu_bwd, [vp, vp, vp, vp, vp, vp, l])
+    _sig(lib.ce_fwd, [vp, vp, vp, vp, vp, l, i, i])
+    _sig(lib.ce_bwd, [vp, vp, vp, vp, vp, vp, l, i, i])
+    _sig(lib.l2normsq, [vp, vp, l, vp, i, vp])
+    _sig(lib.adamw_step, [vp, vp, vp, vp, vp, vp, vp, l,
+                          f, f, f, f, f, f, f, f, f])
+    assert lib.hipops_arch_check() == 950
+    _lib = lib
+    return lib
+
+
+def available() -> bool:
+    return os.path.exists(LIB_PATH)
+
+
+def stream_ptr() -> ctypes.c_void_p:
+    """Current torch HIP stream as a raw handle."""
+    import torch
+    return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)

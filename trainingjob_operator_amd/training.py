@@ -1,0 +1,107 @@
+"""Training engine: model build, synthetic data, fused step loop.
+
+This is the workload the operator's AITrainingJob pods run (launcher ->
+Trainer) and what bench.py measures standalone. DDP is the in-house
+bucketed engine over RCCL/xGMI (parallel/ddp.py), the optimizer the fused
+flat AdamW (optim.py).
+"""
+from __future__ import annotations
+
+from contextlib import nullcontext
+from dataclasses import dataclass, field
+from typing import Iterator, Optional, Tuple
+
+import torch
+
+from .models.config import CONFIGS, LlamaConfig
+from .models.llama import LlamaModel
+from .optim import FlatAdamW
+from .ops import make_inv_freq
+from .parallel.ddp import DDPEngine
+from .parallel.dist_ctx import DistContext
+from .parallel.flat import FlatParamStore
+
+
+@dataclass
+class TrainConfig:
+    model: str = "llama3-8b"
+    micro_batch: int = 1
+    grad_accum: int = 4
+    seq_len: int = 4096
+    lr: float = 3e-4
+    weight_decay: float = 0.1
+    betas: Tuple[float, float] = (0.9, 0.95)
+    clip_grad_norm: float = 1.0
+    checkpoint_activations: bool = False
+    bucket_bytes: int = 128 << 20
+    seed: int = 1234
+
+    @property
+    def model_config(self) -> LlamaConfig:
+        return CONFIGS[self.model]
+
+    def tokens_per_step_per_rank(self) -> int:
+        return self.micro_batch * self.grad_accum * self.seq_len
+
+
+def build_model(cfg: LlamaConfig, device: torch.device,
+                checkpoint_activations: bool = False) -> LlamaModel:
+    with torch.device(device):
+        model = LlamaModel(cfg, checkpoint_activations=checkpoint_activations)
+    model = model.to(torch.bfloat16)
+    # RoPE frequencies must stay fp32 (precision of angles at long context)
+    model.inv_freq = make_inv_freq(cfg.head_dim, cfg.rope_theta,
+                                   device=device)
+    return model
+
+
+def synthetic_batches(cfg: TrainConfig, device: torch.device,
+                      rank: int = 0) -> Iterator[Tuple[torch.Tensor, torch.Tensor]]:
+    """Deterministic-per-rank synthetic LM batches (no network for real data)."""
+    g = torch.Generator(device="cpu").manual_seed(cfg.seed * 1000 + rank)
+    V = cfg.model_config.vocab_size
+    while True:
+        tokens = torch.randint(0, V, (cfg.micro_batch, cfg.seq_len + 1),
+                               generator=g)
+        tokens = tokens.to(device, non_blocking=True)
+        yield tokens[:, :-1].contiguous(), tokens[:, 1:].contiguous()
+
+
+class Trainer:
+    def __init__(self, cfg: TrainConfig, ctx: Optional[DistContext] = None,
+                 device: Optional[torch.device] = None):
+        self.cfg = cfg
+        self.ctx = ctx or DistContext()
+        if device is None:
+            device = torch.device(
+                f"cuda:{self.ctx.local_rank}" if torch.cuda.is_available()
+                else "cpu")
+        self.device = device
+        torch.manual_seed(cfg.seed)  # identical init on every rank
+        self.model = build_model(cfg.model_config, device,
+                                 cfg.checkpoint_activations)
+        self.store = FlatParamStore(self.model, device=device)
+        self.ddp = DDPEngine(self.store, bucket_bytes=cfg.bucket_bytes)
+        self.opt = FlatAdamW(self.store, lr=cfg.lr, betas=cfg.betas,
+                             weight_decay=cfg.weight_decay,
+                             clip_grad_norm=cfg.clip_grad_norm)
+        self.data = synthetic_batches(cfg, device, self.ctx.rank)
+        self.step_count = 0
+
+    def train_step(self) -> torch.Tensor:
+        """One optimizer step = grad_accum micro-batches; returns the last
+        micro-batch loss (device tensor, not synced)."""
+        cfg = self.cfg
+        loss = None
+        for micro in range(cfg.grad_accum):
+            tokens, targets = next(self.data)
+            sync = micro == cfg.grad_accum - 1
+            with (nullcontext() if sync else self.ddp.no_sync()):
+                loss = self.model(tokens, targets)
+                # scale so accumulated grads average over micro-batches
+                (loss / cfg.grad_accum).backward()
+        self.ddp.finish_backward()
+        self.opt.step(grad_pre_scale=self.ddp.grad_pre_scale)
+        self.opt.zero_grad()
+        self.step_count += 1
+        return loss.detach()
