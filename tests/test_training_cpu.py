@@ -37,8 +37,12 @@ def test_flat_store_aliases_params_and_grads():
     loss = model(tokens.to(torch.long), tokens)
     loss.backward()
     assert store.flat_grad.abs().sum() > 0
+    gbase = store.flat_grad.data_ptr()
+    gend = gbase + store.flat_grad.numel() * store.flat_grad.element_size()
     for p in model.parameters():
-        assert p.grad is None  # folded into flat and released
+        # grads are primed flat views: accumulation lands in the flat buffer
+        assert p.grad is not None
+        assert gbase <= p.grad.data_ptr() < gend
     # accumulation across two backwards adds up
     g1 = store.flat_grad.clone()
     loss = model(tokens, tokens)

@@ -56,23 +56,27 @@ class _FusedRMSNorm(torch.autograd.Function):
         if dy.is_cuda:
             lib = _hip()
             dy = dy.contiguous()
+            if dres is not None:
+                dres = dres.contiguous()
             dx = torch.empty_like(res_out)
-            P = lib.rmsnorm_dw_partials()
-            dw_partial = torch.zeros(P, H, dtype=torch.float32, device=dy.device)
+            P = lib.rmsnorm_bwd_partials(rows)
+            dw_partial = torch.empty(P, H, dtype=torch.float32,
+                                     device=dy.device)
+            # dres (grad via the residual stream) is fused into dx in-kernel
             lib.rmsnorm_bwd(native.stream_ptr(), _ptr(dy), _ptr(res_out),
-                            _ptr(weight), _ptr(rrms), _ptr(dx),
+                            _ptr(weight), _ptr(rrms), _ptr(dres), _ptr(dx),
                             _ptr(dw_partial), rows, H)
-            dw32 = torch.zeros(H, dtype=torch.float32, device=dy.device)
-            lib.rmsnorm_dw_reduce(native.stream_ptr(), _ptr(dw_partial), None,
+            dw32 = torch.empty(H, dtype=torch.float32, device=dy.device)
+            lib.rmsnorm_dw_reduce(native.stream_ptr(), _ptr(dw_partial), P,
                                   _ptr(dw32), H)
             dw = dw32.to(weight.dtype)
         else:
             dx, dw32 = reference.rmsnorm_bwd(dy, res_out, weight, rrms)
             dw = dw32.to(weight.dtype)
-        # res_out = x + residual, y = norm(res_out): both x and residual
-        # receive dx plus any gradient that arrived through res_out directly.
-        if dres is not None:
-            dx = dx + dres
+            # res_out = x + residual: both receive dx plus any gradient that
+            # arrived through res_out directly.
+            if dres is not None:
+                dx = dx + dres
         if ctx.has_residual:
             return dx, dw, dx, None
         return dx, dw, None, None

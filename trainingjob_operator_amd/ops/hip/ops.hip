@@ -135,73 +135,96 @@ __global__ void rmsnorm_fwd_kernel(
 // RMSNorm backward.
 //   xhat = res_out * rrms
 //   dxhat = dy * w
-//   dx = rrms * (dxhat - xhat * mean(dxhat * xhat))
-//   dw_partial[row % P] += dy * xhat        (fp32, atomic; reduced separately)
+//   dx = rrms * (dxhat - xhat * mean(dxhat * xhat)) + dres   (dres optional:
+//        the gradient arriving through the residual stream, fused here to
+//        save a full elementwise add pass)
+//   dw_partial[block] = sum over the block's rows of dy * xhat
+//        (plain fp32 stores — no atomics; a tiny reduce kernel finishes it.
+//         The atomic version measured 433 us/call on MI355X, 17x the fwd.)
+// Each block processes DW_ROWS rows serially, keeping its dw accumulator in
+// registers across them.
 // ===========================================================================
 
-#define DW_PARTIALS 256
+#define DW_ROWS 4
 
 template <int NITER>
 __global__ void rmsnorm_bwd_kernel(
     const uint4* __restrict__ dy, const uint4* __restrict__ res_out,
     const uint4* __restrict__ w, const float* __restrict__ rrms_in,
-    uint4* __restrict__ dx, float* __restrict__ dw_partial, int H8) {
+    const uint4* __restrict__ dres, uint4* __restrict__ dx,
+    float* __restrict__ dw_partial, long n_rows, int H8) {
   __shared__ float lds[WAVES_PER_BLOCK];
-  const long row = blockIdx.x;
-  const uint4* dyr = dy + row * H8;
-  const uint4* xr = res_out + row * H8;
-  uint4* dxr = dx + row * H8;
-  float* dwp = dw_partial + (row % DW_PARTIALS) * (long)(H8 * 8);
-  const float rrms = rrms_in[row];
-
-  float xv[NITER][8], dxh[NITER][8];
-  float dot = 0.f;
+  const int H = H8 * 8;
+  float wv[NITER][8];
 #pragma unroll
   for (int it = 0; it < NITER; ++it) {
-    const int idx = it * BLOCK + threadIdx.x;
-    BF8 a; a.v = xr[idx];
-    BF8 d; d.v = dyr[idx];
-    BF8 wv; wv.v = w[idx];
+    BF8 a; a.v = w[it * BLOCK + threadIdx.x];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      xv[it][j] = bf2f(a.h[j]);
-      dxh[it][j] = bf2f(d.h[j]) * bf2f(wv.h[j]);
-      dot += dxh[it][j] * xv[it][j];
+    for (int j = 0; j < 8; ++j) wv[it][j] = bf2f(a.h[j]);
+  }
+  float dwacc[NITER][8] = {};
+
+  const long row0 = (long)blockIdx.x * DW_ROWS;
+  for (long row = row0; row < row0 + DW_ROWS && row < n_rows; ++row) {
+    const uint4* dyr = dy + row * H8;
+    const uint4* xr = res_out + row * H8;
+    uint4* dxr = dx + row * H8;
+    const float rrms = rrms_in[row];
+
+    float xv[NITER][8], dxh[NITER][8], dyv[NITER][8];
+    float dot = 0.f;
+#pragma unroll
+    for (int it = 0; it < NITER; ++it) {
+      const int idx = it * BLOCK + threadIdx.x;
+      BF8 a; a.v = xr[idx];
+      BF8 d; d.v = dyr[idx];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        xv[it][j] = bf2f(a.h[j]);
+        dyv[it][j] = bf2f(d.h[j]);
+        dxh[it][j] = dyv[it][j] * wv[it][j];
+        dot += dxh[it][j] * xv[it][j];
+      }
     }
-    // dw contribution: dy * xhat = dy * x * rrms (fp32 partials, reduced later)
+    const float rr = rrms;
+    const float m = block_reduce_sum(dot, lds) * rr * rr / (float)H;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      atomicAdd(&dwp[idx * 8 + j], bf2f(d.h[j]) * xv[it][j] * rrms);
+    for (int it = 0; it < NITER; ++it) {
+      const int idx = it * BLOCK + threadIdx.x;
+      BF8 o;
+      if (dres) {
+        BF8 dr; dr.v = dres[row * H8 + idx];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          o.h[j] = f2bf(rr * (dxh[it][j] - xv[it][j] * m) + bf2f(dr.h[j]));
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          o.h[j] = f2bf(rr * (dxh[it][j] - xv[it][j] * m));
+      }
+      dxr[idx] = o.v;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dwacc[it][j] += dyv[it][j] * xv[it][j] * rr;
     }
   }
-  const int H = H8 * 8;
-  const float m = block_reduce_sum(dot, lds) * rrms * rrms / (float)H;
+  float* dwp = dw_partial + (long)blockIdx.x * H;
 #pragma unroll
   for (int it = 0; it < NITER; ++it) {
     const int idx = it * BLOCK + threadIdx.x;
-    BF8 o;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) o.h[j] = f2bf(rrms * (dxh[it][j] - xv[it][j] * m));
-    dxr[idx] = o.v;
+    for (int j = 0; j < 8; ++j) dwp[idx * 8 + j] = dwacc[it][j];
   }
 }
 
 __global__ void rmsnorm_dw_reduce_kernel(const float* __restrict__ dw_partial,
-                                         uint4* __restrict__ dw_accum_bf16,
-                                         float* __restrict__ dw_accum_f32,
-                                         int H) {
+                                         int n_partials,
+                                         float* __restrict__ dw_out, int H) {
   // one thread per output element (H threads total over grid)
   const int h = blockIdx.x * blockDim.x + threadIdx.x;
   if (h >= H) return;
   float acc = 0.f;
-  for (int p = 0; p < DW_PARTIALS; ++p) acc += dw_partial[(long)p * H + h];
-  if (dw_accum_f32) {
-    dw_accum_f32[h] += acc;
-  } else {
-    // accumulate into an existing bf16 grad buffer
-    u16* out = reinterpret_cast<u16*>(dw_accum_bf16);
-    out[h] = f2bf(bf2f(out[h]) + acc);
-  }
+  for (int p = 0; p < n_partials; ++p) acc += dw_partial[(long)p * H + h];
+  dw_out[h] = acc;
 }
 
 // ===========================================================================
@@ -494,16 +517,21 @@ void rmsnorm_fwd(void* stream, const void* x, const void* res_in, const void* w,
 #undef CASE
 }
 
+long rmsnorm_bwd_partials(long n_rows) {
+  return (n_rows + DW_ROWS - 1) / DW_ROWS;
+}
+
 void rmsnorm_bwd(void* stream, const void* dy, const void* res_out,
-                 const void* w, const void* rrms, void* dx, void* dw_partial,
-                 long n_rows, int H) {
+                 const void* w, const void* rrms, const void* dres, void* dx,
+                 void* dw_partial, long n_rows, int H) {
   const int H8 = H / 8;
   const int niter = H8 / BLOCK;
-  dim3 grid((unsigned)n_rows), block(BLOCK);
+  dim3 grid((unsigned)rmsnorm_bwd_partials(n_rows)), block(BLOCK);
 #define CASE(N) \
   case N: hipLaunchKernelGGL((rmsnorm_bwd_kernel<N>), grid, block, 0, STREAM, \
       (const uint4*)dy, (const uint4*)res_out, (const uint4*)w, \
-      (const float*)rrms, (uint4*)dx, (float*)dw_partial, H8); break;
+      (const float*)rrms, (const uint4*)dres, (uint4*)dx, (float*)dw_partial, \
+      n_rows, H8); break;
   switch (niter) {
     CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
     default:
@@ -512,13 +540,11 @@ void rmsnorm_bwd(void* stream, const void* dy, const void* res_out,
 #undef CASE
 }
 
-int rmsnorm_dw_partials() { return DW_PARTIALS; }
-
-void rmsnorm_dw_reduce(void* stream, const void* dw_partial, void* dw_bf16,
+void rmsnorm_dw_reduce(void* stream, const void* dw_partial, long n_partials,
                        void* dw_f32, int H) {
   dim3 grid((H + BLOCK - 1) / BLOCK), block(BLOCK);
   hipLaunchKernelGGL(rmsnorm_dw_reduce_kernel, grid, block, 0, STREAM,
-                     (const float*)dw_partial, (uint4*)dw_bf16, (float*)dw_f32,
+                     (const float*)dw_partial, (int)n_partials, (float*)dw_f32,
                      H);
 }
 

@@ -57,9 +57,9 @@ def load(require: bool = True) -> Optional[ctypes.CDLL]:
     vp, l, i, f = ctypes.c_void_p, ctypes.c_long, ctypes.c_int, ctypes.c_float
     _sig(lib.hipops_arch_check, [], i)
     _sig(lib.rmsnorm_fwd, [vp, vp, vp, vp, vp, vp, vp, l, i, f])
-    _sig(lib.rmsnorm_bwd, [vp, vp, vp, vp, vp, vp, vp, l, i])
-    _sig(lib.rmsnorm_dw_partials, [], i)
-    _sig(lib.rmsnorm_dw_reduce, [vp, vp, vp, vp, i])
+    _sig(lib.rmsnorm_bwd, [vp, vp, vp, vp, vp, vp, vp, vp, l, i])
+    _sig(lib.rmsnorm_bwd_partials, [l], l)
+    _sig(lib.rmsnorm_dw_reduce, [vp, vp, l, vp, i])
     _sig(lib.rope, [vp, vp, vp, vp, l, i, i, i, f])
     _sig(lib.swiglu_fwd, [vp, vp, vp, vp, l])
     _sig(lib.swiglu_bwd, [vp, vp, vp, vp, vp, vp, l])
