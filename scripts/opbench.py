@@ -1,8 +1,11 @@
 #!/usr/bin/env python3
 """Per-op microbenchmarks at Llama-3-8B shapes; reports achieved GB/s against
 the ~6.3 TB/s MI355X streaming ceiling (HBM-bound ops should be close)."""
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 

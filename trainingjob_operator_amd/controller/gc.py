@@ -1,0 +1,83 @@
+"""Garbage collector: cluster-wide sweep for orphaned/stuck EDL pods
+(reference: pkg/controller/garbage_collection.go:20-106)."""
+from __future__ import annotations
+
+import logging
+from typing import Optional
+
+from ..api import constants as C
+from ..kube import objects as ko
+from ..kube.client import ApiError, KubeApi
+from ..utils.k8stime import parse_time
+
+log = logging.getLogger(__name__)
+
+
+class GarbageCollector:
+    def __init__(self, api: KubeApi, namespace: str = ""):
+        self.api = api
+        self.namespace = namespace or None
+
+    def clean_garbage_pods(self, now: float) -> int:
+        """One sweep; returns number of pods deleted.
+
+        Deletes (a) EDL pods whose deletionTimestamp expired (stuck
+        terminating — force delete), (b) pods whose owning AITrainingJob no
+        longer exists, unless the pod sits on a dead node with deletion
+        still pending (reference: garbage_collection.go:36-76,91-106).
+        """
+        deleted = 0
+        node_ready = self._node_ready()
+        for pod in self.api.list_pods(self.namespace):
+            labels = ko.labels_of(pod)
+            if C.LABEL_GROUP_NAME not in labels:
+                continue  # not ours (gc:45-48)
+            ns, name = ko.namespace_of(pod), ko.name_of(pod)
+            dts = pod["metadata"].get("deletionTimestamp")
+            if dts is not None:
+                ts = parse_time(dts)
+                if ts is not None and now > ts:
+                    self._force_delete(ns, name)
+                    deleted += 1
+                continue
+            ref = ko.controller_ref(pod)
+            if ref is None or ref.get("kind") != C.CRD_KIND:
+                continue
+            if self._job_exists(ns, ref.get("name", "")):
+                continue
+            node = ko.pod_node(pod)
+            if node and node not in node_ready and dts is not None:
+                continue  # dead node, deletion pending (gc:91-106)
+            self._force_delete(ns, name)
+            deleted += 1
+        if deleted:
+            log.info("gc: deleted %d orphaned pods", deleted)
+        return deleted
+
+    def _job_exists(self, namespace: str, name: str) -> bool:
+        try:
+            self.api.get_job(namespace, name)
+            return True
+        except ApiError as e:
+            if e.not_found:
+                return False
+            raise
+
+    def _force_delete(self, namespace: str, name: str) -> None:
+        try:
+            self.api.delete_pod(namespace, name, grace_period=0)
+        except ApiError as e:
+            if not e.not_found:
+                raise
+
+    def _node_ready(self):
+        ready = {}
+        try:
+            for node in self.api.list_nodes():
+                for cond in (node.get("status") or {}).get("conditions") or []:
+                    if cond.get("type") == "Ready" and \
+                            cond.get("status") == "True":
+                        ready[node["metadata"]["name"]] = True
+        except ApiError:
+            pass
+        return ready

@@ -1,0 +1,331 @@
+"""In-memory fake Kubernetes API for tests.
+
+The moral equivalent of the reference's generated-but-unused fake clientset
+(reference: pkg/client/clientset/versioned/fake/) plus the
+cluster-simulation helpers the reference never had: drive pod phases,
+container statuses, node readiness, and scheduling from test code, and
+assert on the recorded action log.
+"""
+from __future__ import annotations
+
+import copy
+import itertools
+import queue
+import threading
+import time
+from typing import Dict, Iterator, List, Optional, Tuple
+
+from ..api import constants as C
+from ..utils.k8stime import format_time
+from .client import ApiError, KubeApi
+from .objects import labels_of, matches_selector, meta
+
+
+class FakeKubeApi(KubeApi):
+    def __init__(self):
+        self._lock = threading.RLock()
+        self.pods: Dict[Tuple[str, str], dict] = {}
+        self.services: Dict[Tuple[str, str], dict] = {}
+        self.nodes: Dict[str, dict] = {}
+        self.jobs: Dict[Tuple[str, str], dict] = {}
+        self.leases: Dict[Tuple[str, str], dict] = {}
+        self.events: List[dict] = []
+        self.crds: Dict[str, dict] = {}
+        self.actions: List[tuple] = []
+        self._uid = itertools.count(1)
+        self._rv = itertools.count(1)
+        self._watchers: List[queue.Queue] = []
+        # simulation knobs
+        self.auto_schedule = True   # bind new pods to a node immediately
+        self.default_node = "node-0"
+        self.add_node("node-0", ready=True)
+
+    # -- internals --------------------------------------------------------
+    def _record(self, *action):
+        self.actions.append(action)
+
+    def _stamp(self, obj: dict, namespace: str, kind: str):
+        m = meta(obj)
+        m["namespace"] = namespace
+        m.setdefault("uid", f"uid-{next(self._uid)}")
+        m["resourceVersion"] = str(next(self._rv))
+        m.setdefault("creationTimestamp", format_time())
+        obj["kind"] = kind
+
+    def _emit(self, evt_type: str, kind: str, obj: dict):
+        for q in list(self._watchers):
+            q.put((evt_type, kind, copy.deepcopy(obj)))
+
+    # -- pods -------------------------------------------------------------
+    def create_pod(self, namespace, pod):
+        with self._lock:
+            pod = copy.deepcopy(pod)
+            name = meta(pod)["name"]
+            key = (namespace, name)
+            if key in self.pods:
+                raise ApiError(409, f"pod {name} exists")
+            self._stamp(pod, namespace, "Pod")
+            pod.setdefault("status", {})["phase"] = "Pending"
+            if self.auto_schedule and self.default_node:
+                pod["spec"]["nodeName"] = self.default_node
+                pod["status"]["startTime"] = format_time()
+            self.pods[key] = pod
+            self._record("create", "pod", namespace, name)
+            self._emit("ADDED", "pod", pod)
+            return copy.deepcopy(pod)
+
+    def get_pod(self, namespace, name):
+        with self._lock:
+            key = (namespace, name)
+            if key not in self.pods:
+                raise ApiError(404, f"pod {name}")
+            return copy.deepcopy(self.pods[key])
+
+    def list_pods(self, namespace=None, selector=None):
+        with self._lock:
+            out = []
+            for (ns, _), pod in self.pods.items():
+                if namespace and ns != namespace:
+                    continue
+                if selector and not matches_selector(pod, selector):
+                    continue
+                out.append(copy.deepcopy(pod))
+            return out
+
+    def delete_pod(self, namespace, name, grace_period=None):
+        with self._lock:
+            key = (namespace, name)
+            if key not in self.pods:
+                raise ApiError(404, f"pod {name}")
+            pod = self.pods.pop(key)
+            self._record("delete", "pod", namespace, name, grace_period)
+            self._emit("DELETED", "pod", pod)
+
+    # -- services ---------------------------------------------------------
+    def create_service(self, namespace, svc):
+        with self._lock:
+            svc = copy.deepcopy(svc)
+            name = meta(svc)["name"]
+            key = (namespace, name)
+            if key in self.services:
+                raise ApiError(409, f"service {name} exists")
+            self._stamp(svc, namespace, "Service")
+            self.services[key] = svc
+            self._record("create", "service", namespace, name)
+            self._emit("ADDED", "service", svc)
+            return copy.deepcopy(svc)
+
+    def list_services(self, namespace=None, selector=None):
+        with self._lock:
+            out = []
+            for (ns, _), svc in self.services.items():
+                if namespace and ns != namespace:
+                    continue
+                if selector and not matches_selector(svc, selector):
+                    continue
+                out.append(copy.deepcopy(svc))
+            return out
+
+    def delete_service(self, namespace, name):
+        with self._lock:
+            key = (namespace, name)
+            if key not in self.services:
+                raise ApiError(404, f"service {name}")
+            svc = self.services.pop(key)
+            self._record("delete", "service", namespace, name)
+            self._emit("DELETED", "service", svc)
+
+    # -- nodes ------------------------------------------------------------
+    def list_nodes(self):
+        with self._lock:
+            return [copy.deepcopy(n) for n in self.nodes.values()]
+
+    # -- jobs -------------------------------------------------------------
+    def ensure_crd(self, crd_manifest):
+        with self._lock:
+            self.crds[meta(crd_manifest).get("name", "crd")] = crd_manifest
+            self._record("ensure", "crd", meta(crd_manifest).get("name"))
+
+    def create_job(self, namespace, job_dict):
+        with self._lock:
+            job_dict = copy.deepcopy(job_dict)
+            name = meta(job_dict)["name"]
+            key = (namespace, name)
+            if key in self.jobs:
+                raise ApiError(409, f"job {name} exists")
+            self._stamp(job_dict, namespace, C.CRD_KIND)
+            self.jobs[key] = job_dict
+            self._emit("ADDED", "job", job_dict)
+            return copy.deepcopy(job_dict)
+
+    def get_job(self, namespace, name):
+        with self._lock:
+            key = (namespace, name)
+            if key not in self.jobs:
+                raise ApiError(404, f"job {name}")
+            return copy.deepcopy(self.jobs[key])
+
+    def list_jobs(self, namespace=None):
+        with self._lock:
+            return [copy.deepcopy(j) for (ns, _), j in self.jobs.items()
+                    if namespace is None or ns == namespace]
+
+    def update_job(self, namespace, name, job):
+        with self._lock:
+            key = (namespace, name)
+            if key not in self.jobs:
+                raise ApiError(404, f"job {name}")
+            cur = self.jobs[key]
+            new_rv = job.get("metadata", {}).get("resourceVersion")
+            if new_rv and new_rv != cur["metadata"]["resourceVersion"]:
+                raise ApiError(409, "resourceVersion conflict")
+            job = copy.deepcopy(job)
+            meta(job)["resourceVersion"] = str(next(self._rv))
+            meta(job)["namespace"] = namespace
+            meta(job).setdefault("uid", cur["metadata"]["uid"])
+            self.jobs[key] = job
+            self._record("update", "job", namespace, name)
+            self._emit("MODIFIED", "job", job)
+            return copy.deepcopy(job)
+
+    def update_job_status(self, namespace, name, job):
+        return self.update_job(namespace, name, job)
+
+    def delete_job(self, namespace, name):
+        with self._lock:
+            key = (namespace, name)
+            if key not in self.jobs:
+                raise ApiError(404, f"job {name}")
+            j = self.jobs.pop(key)
+            self._record("delete", "job", namespace, name)
+            self._emit("DELETED", "job", j)
+
+    # -- events / leases --------------------------------------------------
+    def create_event(self, namespace, event):
+        with self._lock:
+            self.events.append(event)
+
+    def get_lease(self, namespace, name):
+        with self._lock:
+            key = (namespace, name)
+            if key not in self.leases:
+                raise ApiError(404, f"lease {name}")
+            return copy.deepcopy(self.leases[key])
+
+    def create_lease(self, namespace, lease):
+        with self._lock:
+            key = (namespace, meta(lease)["name"])
+            if key in self.leases:
+                raise ApiError(409, "lease exists")
+            self._stamp(lease, namespace, "Lease")
+            self.leases[key] = copy.deepcopy(lease)
+            return copy.deepcopy(lease)
+
+    def update_lease(self, namespace, name, lease):
+        with self._lock:
+            key = (namespace, name)
+            if key not in self.leases:
+                raise ApiError(404, f"lease {name}")
+            self.leases[key] = copy.deepcopy(lease)
+            return copy.deepcopy(lease)
+
+    # -- watches ----------------------------------------------------------
+    def _watch_kind(self, kind: str, stop: threading.Event) -> Iterator[tuple]:
+        q: queue.Queue = queue.Queue()
+        self._watchers.append(q)
+        try:
+            while not stop.is_set():
+                try:
+                    evt_type, k, obj = q.get(timeout=0.2)
+                except queue.Empty:
+                    continue
+                if k == kind:
+                    yield evt_type, obj
+        finally:
+            self._watchers.remove(q)
+
+    def watch_pods(self, namespace, stop):
+        return self._watch_kind("pod", stop)
+
+    def watch_services(self, namespace, stop):
+        return self._watch_kind("service", stop)
+
+    def watch_jobs(self, namespace, stop):
+        return self._watch_kind("job", stop)
+
+    # ======================================================================
+    # Simulation helpers (test-side cluster behavior)
+    # ======================================================================
+
+    def add_node(self, name: str, ready: bool = True):
+        with self._lock:
+            self.nodes[name] = {
+                "metadata": {"name": name},
+                "status": {"conditions": [
+                    {"type": "Ready",
+                     "status": "True" if ready else "False"}]},
+            }
+
+    def set_node_ready(self, name: str, ready: bool):
+        self.add_node(name, ready)
+
+    def remove_node(self, name: str):
+        with self._lock:
+            self.nodes.pop(name, None)
+
+    def _pod(self, namespace, name):
+        key = (namespace, name)
+        if key not in self.pods:
+            raise KeyError(f"no pod {key}")
+        return self.pods[key]
+
+    def set_pod_phase(self, namespace: str, name: str, phase: str,
+                      exit_code: Optional[int] = None,
+                      waiting_reason: Optional[str] = None):
+        """Drive a pod through its lifecycle, synthesizing container
+        statuses for the aitj-* containers."""
+        with self._lock:
+            pod = self._pod(namespace, name)
+            status = pod.setdefault("status", {})
+            status["phase"] = phase
+            cstatuses = []
+            for c in (pod.get("spec") or {}).get("containers") or []:
+                cname = c.get("name", "")
+                if phase == "Running":
+                    state = {"running": {"startedAt": format_time()}}
+                elif phase == "Succeeded":
+                    state = {"terminated": {"exitCode": 0}}
+                elif phase == "Failed":
+                    state = {"terminated": {
+                        "exitCode": exit_code if exit_code is not None else 1,
+                        "reason": "Error"}}
+                elif waiting_reason:
+                    state = {"waiting": {"reason": waiting_reason}}
+                else:
+                    state = {"waiting": {"reason": "ContainerCreating"}}
+                cstatuses.append({"name": cname, "state": state})
+            status["containerStatuses"] = cstatuses
+            status.setdefault("startTime", format_time())
+            pod["metadata"]["resourceVersion"] = str(next(self._rv))
+            self._emit("MODIFIED", "pod", pod)
+
+    def set_all_pods_phase(self, namespace: str, phase: str, **kw):
+        for (ns, name) in list(self.pods):
+            if ns == namespace:
+                self.set_pod_phase(ns, name, phase, **kw)
+
+    def bind_pod(self, namespace: str, name: str, node: str):
+        with self._lock:
+            pod = self._pod(namespace, name)
+            pod["spec"]["nodeName"] = node
+            self._emit("MODIFIED", "pod", pod)
+
+    def pod_names(self, namespace: Optional[str] = None) -> List[str]:
+        with self._lock:
+            return sorted(n for (ns, n) in self.pods
+                          if namespace is None or ns == namespace)
+
+    def service_names(self, namespace: Optional[str] = None) -> List[str]:
+        with self._lock:
+            return sorted(n for (ns, n) in self.services
+                          if namespace is None or ns == namespace)
