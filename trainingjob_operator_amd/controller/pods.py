@@ -60,13 +60,23 @@ def out_of_range_pods(pods: List[dict], replicas: int) -> List[dict]:
 
 
 def node_ready_map(api: KubeApi) -> Dict[str, bool]:
-    """reference: pod.go:439-455 (map contains only Ready nodes)."""
+    """Map of healthy nodes (absent == failed). Extends the reference's
+    NodeReady check (pod.go:439-455) with the node agent's GPU-granular
+    ``EDLGPUHealthy`` condition: a node whose agent reports GPU-lost / ECC /
+    thermal / xGMI-down is treated as failed even while kubelet-Ready, so
+    pods there hit the NodeFail phase + restart policies."""
     ready = {}
     for node in api.list_nodes():
+        is_ready = False
+        gpu_healthy = True
         for cond in (node.get("status") or {}).get("conditions") or []:
             if cond.get("type") == "Ready" and cond.get("status") == "True":
-                ready[node["metadata"]["name"]] = True
-                break
+                is_ready = True
+            if cond.get("type") == "EDLGPUHealthy" and \
+                    cond.get("status") == "False":
+                gpu_healthy = False
+        if is_ready and gpu_healthy:
+            ready[node["metadata"]["name"]] = True
     return ready
 
 

@@ -61,6 +61,9 @@ class KubeApi:
 
     # nodes --------------------------------------------------------------
     def list_nodes(self) -> List[dict]: ...
+    def get_node(self, name: str) -> dict: ...
+    def patch_node_status(self, name: str, status_patch: dict) -> None: ...
+    def annotate_node(self, name: str, annotations: Dict[str, str]) -> None: ...
 
     # custom resources (aitrainingjobs) ----------------------------------
     def ensure_crd(self, crd_manifest: dict) -> None: ...
@@ -186,6 +189,17 @@ class RealKubeApi(KubeApi):
     # -- nodes ------------------------------------------------------------
     def list_nodes(self):
         return self._req("GET", "/api/v1/nodes")["items"]
+
+    def get_node(self, name):
+        return self._req("GET", f"/api/v1/nodes/{name}")
+
+    def patch_node_status(self, name, status_patch):
+        self._req("PATCH", f"/api/v1/nodes/{name}/status",
+                  {"status": status_patch})
+
+    def annotate_node(self, name, annotations):
+        self._req("PATCH", f"/api/v1/nodes/{name}",
+                  {"metadata": {"annotations": annotations}})
 
     # -- custom resources -------------------------------------------------
     def ensure_crd(self, crd_manifest):

@@ -140,6 +140,37 @@ class FakeKubeApi(KubeApi):
         with self._lock:
             return [copy.deepcopy(n) for n in self.nodes.values()]
 
+    def get_node(self, name):
+        with self._lock:
+            if name not in self.nodes:
+                raise ApiError(404, f"node {name}")
+            return copy.deepcopy(self.nodes[name])
+
+    def patch_node_status(self, name, status_patch):
+        with self._lock:
+            if name not in self.nodes:
+                raise ApiError(404, f"node {name}")
+            node = self.nodes[name]
+            status = node.setdefault("status", {})
+            for key, val in status_patch.items():
+                if key == "conditions":
+                    conds = {c["type"]: c
+                             for c in status.get("conditions") or []}
+                    for c in val:
+                        conds[c["type"]] = c
+                    status["conditions"] = list(conds.values())
+                else:
+                    status[key] = copy.deepcopy(val)
+            self._record("patch", "node-status", name)
+
+    def annotate_node(self, name, annotations):
+        with self._lock:
+            if name not in self.nodes:
+                raise ApiError(404, f"node {name}")
+            anns = self.nodes[name].setdefault("metadata", {}) \
+                .setdefault("annotations", {})
+            anns.update(annotations)
+
     # -- jobs -------------------------------------------------------------
     def ensure_crd(self, crd_manifest):
         with self._lock:

@@ -1,0 +1,95 @@
+"""In-pod worker entrypoint.
+
+The operator injects the rendezvous env (controller/envinject.py); this
+launcher reads it, forms the RCCL world, restores the latest checkpoint
+(fault restart and elastic resize both arrive here as fresh processes with
+a new TRAININGJOB_RENDEZVOUS_EPOCH / WORLD_SIZE), trains with periodic
+async checkpoints, and exits 0 on completion — the exit code is the
+operator's completion signal (SURVEY.md §1 'contract with the container').
+
+    python -m trainingjob_operator_amd.launcher.main \
+        --model llama3-8b --steps 1000 --ckpt-dir /ckpt
+"""
+from __future__ import annotations
+
+import argparse
+import logging
+import os
+import signal
+import sys
+import time
+
+log = logging.getLogger("launcher")
+
+
+def main(argv=None) -> int:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--model", default="llama3-8b")
+    ap.add_argument("--steps", type=int, default=1000,
+                    help="total optimizer steps (across restarts)")
+    ap.add_argument("--micro-batch", type=int, default=1)
+    ap.add_argument("--grad-accum", type=int, default=4)
+    ap.add_argument("--seq-len", type=int, default=4096)
+    ap.add_argument("--lr", type=float, default=3e-4)
+    ap.add_argument("--ckpt-dir", default=os.environ.get(
+        "TRAININGJOB_CKPT_DIR", "/tmp/aitj-ckpt"))
+    ap.add_argument("--ckpt-every", type=int, default=50)
+    ap.add_argument("--log-every", type=int, default=10)
+    ap.add_argument("--checkpoint-activations", action="store_true")
+    args = ap.parse_args(argv)
+
+    logging.basicConfig(
+        level=logging.INFO,
+        format="%(asctime)s %(name)s [%(levelname)s] %(message)s")
+
+    import torch
+
+    from ..parallel import dist_ctx
+    from ..training import TrainConfig, Trainer
+    from .checkpoint import Checkpointer
+
+    ctx = dist_ctx.from_env()
+    epoch = os.environ.get("TRAININGJOB_RENDEZVOUS_EPOCH", "0")
+    restart = os.environ.get("TRAININGJOB_REPLICA_RESTARTCOUNT", "0")
+    log.info("rank %d/%d epoch=%s restart=%s master=%s:%s",
+             ctx.rank, ctx.world_size, epoch, restart,
+             ctx.master_addr, ctx.master_port)
+    t_start = time.time()
+    dist_ctx.init_process_group(ctx)
+
+    cfg = TrainConfig(
+        model=args.model, micro_batch=args.micro_batch,
+        grad_accum=args.grad_accum, seq_len=args.seq_len, lr=args.lr,
+        checkpoint_activations=args.checkpoint_activations)
+    trainer = Trainer(cfg, ctx)
+    ckpt = Checkpointer(args.ckpt_dir)
+    resumed = ckpt.load_latest(trainer)
+    if resumed is not None:
+        log.info("resumed from step %d (world=%d)", resumed, ctx.world_size)
+    log.info("rejoin latency: %.2fs from exec to first step",
+             time.time() - t_start)
+
+    stop_requested = {"flag": False}
+
+    def on_term(signum, frame):
+        stop_requested["flag"] = True
+    signal.signal(signal.SIGTERM, on_term)
+
+    while trainer.step_count < args.steps and not stop_requested["flag"]:
+        loss = trainer.train_step()
+        step = trainer.step_count
+        if step % args.log_every == 0 and ctx.is_rank0:
+            log.info("step %d loss %.4f", step, loss.item())
+        if step % args.ckpt_every == 0 and ctx.is_rank0:
+            ckpt.save_async(trainer)
+
+    if ctx.is_rank0:
+        ckpt.save_async(trainer, blocking=True)
+    ckpt.wait()
+    dist_ctx.destroy_process_group()
+    log.info("done at step %d", trainer.step_count)
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
