@@ -24,6 +24,31 @@ def log(msg: str) -> None:
     print(msg, file=sys.stderr, flush=True)
 
 
+TUNABLEOP_CSV = os.path.join(
+    os.path.dirname(os.path.abspath(__file__)),
+    "trainingjob_operator_amd", "data", "tunableop_gfx950.csv")
+
+
+def stage_tunableop() -> None:
+    """Use the committed hipBLASLt/rocBLAS GEMM tuning results when present
+    (tuned once on MI355X; torch appends the device ordinal to the
+    filename, so stage one copy per potential device). Must run before
+    torch is imported."""
+    if os.environ.get("PYTORCH_TUNABLEOP_ENABLED"):
+        return  # user controls it
+    if not os.path.exists(TUNABLEOP_CSV):
+        return
+    import shutil
+    import tempfile
+    d = tempfile.mkdtemp(prefix="tunableop_")
+    for i in range(16):
+        shutil.copy(TUNABLEOP_CSV, os.path.join(d, f"tunableop{i}.csv"))
+    os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
+    os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
+    os.environ["PYTORCH_TUNABLEOP_FILENAME"] = os.path.join(d, "tunableop.csv")
+    os.environ["PYTORCH_TUNABLEOP_VERBOSE"] = "0"
+
+
 def main() -> None:
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -39,6 +64,7 @@ def main() -> None:
     ap.add_argument("--bucket-mb", type=int, default=128)
     args = ap.parse_args()
 
+    stage_tunableop()
     import torch
     import torch.distributed as dist
 
