@@ -62,6 +62,8 @@ def main() -> None:
     ap.add_argument("--seq-len", type=int, default=4096)
     ap.add_argument("--checkpoint-activations", action="store_true")
     ap.add_argument("--bucket-mb", type=int, default=128)
+    ap.add_argument("--no-graphs", action="store_true",
+                    help="disable hipGraph step capture")
     args = ap.parse_args()
 
     stage_tunableop()
@@ -88,6 +90,7 @@ def main() -> None:
         seq_len=seq_len,
         checkpoint_activations=args.checkpoint_activations,
         bucket_bytes=args.bucket_mb << 20,
+        use_graphs=on_gpu and not args.no_graphs,
     )
     log(f"[bench] rank {ctx.rank}/{n_gpus} model={model} "
         f"mb={cfg.micro_batch} ga={cfg.grad_accum} seq={cfg.seq_len} "
@@ -99,8 +102,17 @@ def main() -> None:
         f"({cfg.model_config.n_params / 1e9:.2f}B params)")
 
     for i in range(args.warmup):
-        loss = trainer.train_step()
-        log(f"[bench] warmup {i}: loss={loss.item():.4f}")
+        try:
+            loss = trainer.train_step()
+        except Exception as e:
+            if cfg.use_graphs and trainer._graph is None:
+                log(f"[bench] graph capture failed ({e!r}); eager fallback")
+                cfg.use_graphs = False
+                loss = trainer.train_step()
+            else:
+                raise
+        log(f"[bench] warmup {i}: loss={loss.item():.4f} "
+            f"graphs={trainer._graph is not None}")
 
     if ctx.is_distributed:
         dist.barrier()

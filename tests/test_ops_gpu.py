@@ -141,7 +141,7 @@ def test_adamw_matches_reference():
                  normsq.data_ptr())
     lib.adamw_step(stream_ptr(), p32.data_ptr(), m.data_ptr(), v.data_ptr(),
                    grad.data_ptr(), pb.data_ptr(), normsq.data_ptr(), n,
-                   lr, b1, b2, eps, wd, 1 - b1, 1 - b2, clip, pre)
+                   lr, b1, b2, eps, wd, 1 - b1, 1 - b2, clip, pre, None)
     torch.cuda.synchronize()
     # CPU reference
     p32r = p0.clone()

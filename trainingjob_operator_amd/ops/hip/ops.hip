@@ -443,9 +443,14 @@ __global__ void adamw_kernel(float* __restrict__ p32, float* __restrict__ m,
                              const float* __restrict__ normsq, long n4,
                              float lr, float beta1, float beta2, float eps,
                              float weight_decay, float bc1, float bc2,
-                             float clip, float pre_scale) {
+                             float clip, float pre_scale,
+                             const float* __restrict__ bc_dev) {
   // pre_scale folds the DDP 1/world_size average into the update (grads are
   // SUM-all-reduced); the clip compares against the POST-scale norm.
+  // bc_dev (optional, [bc1, bc2] in device memory) overrides the host
+  // bias-correction args so a hipGraph-captured step stays step-correct on
+  // replay.
+  if (bc_dev) { bc1 = bc_dev[0]; bc2 = bc_dev[1]; }
   float gscale = pre_scale;
   if (normsq) {
     const float norm = sqrtf(*normsq) * pre_scale;
@@ -606,13 +611,14 @@ void l2normsq(void* stream, const void* grad, long n, void* partials,
 void adamw_step(void* stream, void* p32, void* m, void* v, const void* grad,
                 void* p_bf16, const void* normsq, long n, float lr, float beta1,
                 float beta2, float eps, float weight_decay, float bc1,
-                float bc2, float clip, float pre_scale) {
+                float bc2, float clip, float pre_scale, const void* bc_dev) {
   const long n4 = n / 4;
   dim3 grid(elementwise_grid(n4)), block(BLOCK);
   hipLaunchKernelGGL(adamw_kernel, grid, block, 0, STREAM, (float*)p32,
                      (float*)m, (float*)v, (const uint2*)grad, (uint2*)p_bf16,
                      (const float*)normsq, n4, lr, beta1, beta2, eps,
-                     weight_decay, bc1, bc2, clip, pre_scale);
+                     weight_decay, bc1, bc2, clip, pre_scale,
+                     (const float*)bc_dev);
 }
 
 }  // extern "C"

@@ -67,7 +67,7 @@ def load(require: bool = True) -> Optional[ctypes.CDLL]:
     _sig(lib.ce_bwd, [vp, vp, vp, vp, vp, vp, l, i, i])
     _sig(lib.l2normsq, [vp, vp, l, vp, i, vp])
     _sig(lib.adamw_step, [vp, vp, vp, vp, vp, vp, vp, l,
-                          f, f, f, f, f, f, f, f, f])
+                          f, f, f, f, f, f, f, f, f, vp])
     assert lib.hipops_arch_check() == 950
     _lib = lib
     return lib

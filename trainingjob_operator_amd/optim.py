@@ -37,18 +37,43 @@ class FlatAdamW:
             self._partials = torch.empty(_N_PARTIALS, dtype=torch.float32,
                                          device=dev)
             self._normsq = torch.empty(1, dtype=torch.float32, device=dev)
+            # bias-correction terms in device memory so a hipGraph-captured
+            # step stays correct across replays (the kernel reads them)
+            self._bc = torch.ones(2, dtype=torch.float32, device=dev)
         else:
             self._partials = None
             self._normsq = None
+            self._bc = None
+
+    def update_bias_correction(self, step_count: Optional[int] = None) -> None:
+        """Refresh the device-side [bc1, bc2]. Called inside eager step();
+        a graph-replay driver calls it (plus bump_step) BEFORE replay."""
+        if self._bc is None:
+            return
+        t = self.step_count if step_count is None else step_count
+        self._bc.copy_(torch.tensor(
+            [1.0 - self.beta1 ** t, 1.0 - self.beta2 ** t],
+            dtype=torch.float32))
+
+    def bump_step(self) -> int:
+        self.step_count += 1
+        return self.step_count
 
     @torch.no_grad()
-    def step(self, grad_pre_scale: float = 1.0) -> None:
-        self.step_count += 1
+    def step(self, grad_pre_scale: float = 1.0,
+             in_graph_capture: bool = False) -> None:
+        """One fused update. With in_graph_capture=True the step counter and
+        bc buffer are managed by the caller (bump_step +
+        update_bias_correction before each replay)."""
+        if not in_graph_capture:
+            self.bump_step()
         st = self.store
         n = st.total
-        bc1 = 1.0 - self.beta1 ** self.step_count
-        bc2 = 1.0 - self.beta2 ** self.step_count
+        bc1 = 1.0 - self.beta1 ** max(self.step_count, 1)
+        bc2 = 1.0 - self.beta2 ** max(self.step_count, 1)
         if st.flat_param.is_cuda:
+            if not in_graph_capture:
+                self.update_bias_correction()
             lib = native.load(require=True)
             sp = native.stream_ptr()
             normsq_ptr = None
@@ -62,7 +87,7 @@ class FlatAdamW:
                            st.flat_param.data_ptr(), normsq_ptr, n, self.lr,
                            self.beta1, self.beta2, self.eps,
                            self.weight_decay, bc1, bc2, self.clip,
-                           grad_pre_scale)
+                           grad_pre_scale, self._bc.data_ptr())
         else:
             normsq = None
             if self.clip > 0.0:
