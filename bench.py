@@ -66,6 +66,11 @@ def main() -> None:
                     help="disable hipGraph step capture")
     args = ap.parse_args()
 
+    if os.environ.get("BENCH_DEBUG"):
+        import faulthandler
+        faulthandler.dump_traceback_later(
+            int(os.environ.get("BENCH_DEBUG_TIMEOUT", "150")), exit=True)
+
     stage_tunableop()
     import torch
     import torch.distributed as dist
@@ -121,6 +126,8 @@ def main() -> None:
     t0 = time.perf_counter()
     for i in range(args.steps):
         trainer.train_step()
+        if os.environ.get("BENCH_DEBUG"):
+            log(f"[bench] step {i} queued")
     if on_gpu:
         torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
