@@ -104,6 +104,20 @@ def main():
             x.grad = None
         capture_and_replay(fn)
 
+    elif which == "trainer":
+        from trainingjob_operator_amd.training import TrainConfig, Trainer
+        model = sys.argv[3] if len(sys.argv) > 3 else "llama-smoke"
+        ga = int(sys.argv[4]) if len(sys.argv) > 4 else 1
+        cfg = TrainConfig(model=model, micro_batch=1, grad_accum=ga,
+                          seq_len=S, use_graphs=True)
+        trainer = Trainer(cfg)
+        for i in range(6):
+            loss = trainer.train_step()
+            torch.cuda.synchronize()
+            print(f"step {i} ok loss={loss.item():.3f}", flush=True)
+        print(f"trainer {model} S={S} ga={ga}: ALL OK")
+        return
+
     elif which == "embed":
         emb = torch.nn.Embedding(128256, 4096).to(DEV).bfloat16()
         tokens = torch.randint(0, 128256, (1, S), device=DEV)
