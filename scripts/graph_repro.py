@@ -104,6 +104,25 @@ def main():
             x.grad = None
         capture_and_replay(fn)
 
+    elif which == "sdpa_gqa":
+        # smoke-model attention shape: GQA repeat_interleave + rope
+        q = torch.randn(1, S, 16, 128, device=DEV, dtype=torch.bfloat16,
+                        requires_grad=True)
+        kv = torch.randn(1, S, 8, 128, device=DEV, dtype=torch.bfloat16,
+                         requires_grad=True)
+        inv_freq = make_inv_freq(128, 500000.0, device=DEV)
+
+        def fn():
+            qr = apply_rope(q, inv_freq, S).transpose(1, 2)
+            kr = apply_rope(kv, inv_freq, S).transpose(1, 2)
+            vr = kv.transpose(1, 2)
+            kr = kr.repeat_interleave(2, dim=1)
+            vr = vr.repeat_interleave(2, dim=1)
+            o = F.scaled_dot_product_attention(qr, kr, vr, is_causal=True)
+            o.sum().backward()
+            q.grad = kv.grad = None
+        capture_and_replay(fn)
+
     elif which == "trainer":
         from trainingjob_operator_amd.training import TrainConfig, Trainer
         model = sys.argv[3] if len(sys.argv) > 3 else "llama-smoke"
