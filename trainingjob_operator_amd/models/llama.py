@@ -19,11 +19,27 @@ benchmark jobs run and what bench.py measures.
 from __future__ import annotations
 
 import math
+import os
 from typing import Optional
 
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
+
+
+def _sdpa(q, k, v):
+    """SDPA with a backend override knob (AITJ_SDPA_BACKEND =
+    flash|efficient|math|auto) — used to pin a hipGraph-replay-safe
+    backend and for debugging."""
+    backend = os.environ.get("AITJ_SDPA_BACKEND", "auto")
+    if backend == "auto" or not q.is_cuda:
+        return F.scaled_dot_product_attention(q, k, v, is_causal=True)
+    from torch.nn.attention import SDPBackend, sdpa_kernel
+    mapping = {"flash": SDPBackend.FLASH_ATTENTION,
+               "efficient": SDPBackend.EFFICIENT_ATTENTION,
+               "math": SDPBackend.MATH}
+    with sdpa_kernel([mapping[backend]]):
+        return F.scaled_dot_product_attention(q, k, v, is_causal=True)
 
 from ..ops import apply_rope, fused_cross_entropy, fused_rmsnorm, make_inv_freq, swiglu
 from .config import LlamaConfig
@@ -54,7 +70,7 @@ class Attention(nn.Module):
         if groups > 1:
             k = k.repeat_interleave(groups, dim=1)
             v = v.repeat_interleave(groups, dim=1)
-        o = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+        o = _sdpa(q, k, v)
         o = o.transpose(1, 2).reshape(B, S, cfg.num_heads * cfg.head_dim)
         return self.o_proj(o)
 
