@@ -66,7 +66,7 @@ class _FusedRMSNorm(torch.autograd.Function):
             lib.rmsnorm_bwd(native.stream_ptr(), _ptr(dy), _ptr(res_out),
                             _ptr(weight), _ptr(rrms), _ptr(dres), _ptr(dx),
                             _ptr(dw_partial), rows, H)
-            dw32 = torch.empty(H, dtype=torch.float32, device=dy.device)
+            dw32 = torch.zeros(H, dtype=torch.float32, device=dy.device)
             lib.rmsnorm_dw_reduce(native.stream_ptr(), _ptr(dw_partial), P,
                                   _ptr(dw32), H)
             dw = dw32.to(weight.dtype)

@@ -218,13 +218,20 @@ __global__ void rmsnorm_bwd_kernel(
 
 __global__ void rmsnorm_dw_reduce_kernel(const float* __restrict__ dw_partial,
                                          int n_partials,
-                                         float* __restrict__ dw_out, int H) {
-  // one thread per output element (H threads total over grid)
+                                         float* __restrict__ dw_out, int H,
+                                         int rows_per_block) {
+  // 2D grid: x tiles the H columns, y tiles the partial rows. Each block
+  // sums its row chunk for its 256 columns and atomically folds into
+  // dw_out (zero-initialized). The 1D one-thread-per-column version had
+  // only H/256 blocks and measured 66 GB/s / 242 us per call on MI355X.
   const int h = blockIdx.x * blockDim.x + threadIdx.x;
   if (h >= H) return;
+  const int p0 = blockIdx.y * rows_per_block;
+  int p1 = p0 + rows_per_block;
+  if (p1 > n_partials) p1 = n_partials;
   float acc = 0.f;
-  for (int p = 0; p < n_partials; ++p) acc += dw_partial[(long)p * H + h];
-  dw_out[h] = acc;
+  for (int p = p0; p < p1; ++p) acc += dw_partial[(long)p * H + h];
+  atomicAdd(&dw_out[h], acc);
 }
 
 // ===========================================================================
@@ -547,10 +554,15 @@ void rmsnorm_bwd(void* stream, const void* dy, const void* res_out,
 
 void rmsnorm_dw_reduce(void* stream, const void* dw_partial, long n_partials,
                        void* dw_f32, int H) {
-  dim3 grid((H + BLOCK - 1) / BLOCK), block(BLOCK);
+  // aim for ~1024 blocks to fill 256 CUs
+  int ny = (int)((1024 + (H / BLOCK) - 1) / (H / BLOCK));
+  if (ny < 1) ny = 1;
+  if (ny > n_partials) ny = (int)n_partials;
+  const int rows_per_block = (int)((n_partials + ny - 1) / ny);
+  dim3 grid((H + BLOCK - 1) / BLOCK, ny), block(BLOCK);
   hipLaunchKernelGGL(rmsnorm_dw_reduce_kernel, grid, block, 0, STREAM,
                      (const float*)dw_partial, (int)n_partials, (float*)dw_f32,
-                     H);
+                     H, rows_per_block);
 }
 
 void rope(void* stream, const void* x, void* out, const void* inv_freq,
