@@ -32,6 +32,10 @@ def _sdpa(q, k, v):
     flash|efficient|math|auto) — used to pin a hipGraph-replay-safe
     backend and for debugging."""
     backend = os.environ.get("AITJ_SDPA_BACKEND", "auto")
+    if backend == "native" and q.is_cuda:
+        from ..ops.attention import flash_attention
+        return flash_attention(q.contiguous() if q.stride(-1) != 1 else q,
+                               k, v)
     if backend == "auto" or not q.is_cuda:
         return F.scaled_dot_product_attention(q, k, v, is_causal=True)
     from torch.nn.attention import SDPBackend, sdpa_kernel

@@ -1,0 +1,82 @@
+"""Native MFMA flash-attention forward (hip/attention.hip) with the aten
+flash backward for training.
+
+The forward emits exactly what aten::_scaled_dot_product_flash_attention_
+backward consumes (O + logsumexp at matching scale semantics), so the
+hand-written CDNA4 forward drops into autograd with the library backward.
+Enable via AITJ_SDPA_BACKEND=native (models/llama.py _sdpa).
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+
+from . import native
+
+
+def _supported(q: torch.Tensor, k: torch.Tensor) -> bool:
+    B, H, S, D = q.shape
+    return (q.dtype == torch.bfloat16 and D == 128 and S % 64 == 0
+            and q.stride(-1) == 1 and k.stride(-1) == 1
+            and k.shape[1] == H)
+
+
+class _NativeFlashAttention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, scale):
+        B, H, S, D = q.shape
+        lib = native.load(require=True)
+        out = torch.empty(B, H, S, D, dtype=torch.bfloat16, device=q.device)
+        lse = torch.empty(B, H, S, dtype=torch.float32, device=q.device)
+        lib.attn_fwd(
+            native.stream_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+            out.data_ptr(), lse.data_ptr(),
+            q.stride(0), q.stride(1), q.stride(2),
+            k.stride(0), k.stride(1), k.stride(2),
+            v.stride(0), v.stride(1), v.stride(2),
+            B, H, S, scale)
+        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.scale = scale
+        return out
+
+    @staticmethod
+    def backward(ctx, gout):
+        q, k, v, out, lse = ctx.saved_tensors
+        S = q.shape[2]
+        empty_i = torch.empty(0, dtype=torch.int64, device=q.device)
+        dq, dk, dv = torch.ops.aten._scaled_dot_product_flash_attention_backward(
+            gout.contiguous(), q, k, v, out, lse,
+            empty_i, empty_i, S, S, 0.0, True,
+            empty_i, empty_i, scale=ctx.scale)
+        return dq, dk, dv, None
+
+
+def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                    scale: Optional[float] = None) -> torch.Tensor:
+    """Causal flash attention over [B, H, S, D=128] bf16 (S % 64 == 0)."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    assert _supported(q, k), (
+        f"native flash attention needs bf16, D=128, S%64==0, d-contiguous "
+        f"strides; got {q.shape} {q.dtype}")
+    return _NativeFlashAttention.apply(q, k, v, scale)
+
+
+def flash_attention_fwd_only(q, k, v, scale=None):
+    """Forward-only entry returning (out, lse) for tests/benchmarks."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    B, H, S, D = q.shape
+    lib = native.load(require=True)
+    out = torch.empty(B, H, S, D, dtype=torch.bfloat16, device=q.device)
+    lse = torch.empty(B, H, S, dtype=torch.float32, device=q.device)
+    lib.attn_fwd(
+        native.stream_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+        out.data_ptr(), lse.data_ptr(),
+        q.stride(0), q.stride(1), q.stride(2),
+        k.stride(0), k.stride(1), k.stride(2),
+        v.stride(0), v.stride(1), v.stride(2),
+        B, H, S, scale)
+    return out, lse

@@ -1,0 +1,212 @@
+// Hand-written CDNA4 (gfx950) flash-attention FORWARD for MI355X.
+//
+// Causal, bf16, head_dim 128, arbitrary [B, H, S, D] strides (S % 64 == 0).
+// One 4-wave workgroup owns one (batch, head, 64-query block); K/V tiles
+// stream through LDS; Q fragments stay in registers; online softmax in
+// fp32; P takes one per-wave LDS round trip to re-shape from the MFMA C
+// layout to the A layout. Emits O and the logsumexp rows the aten flash
+// backward consumes (torch.ops.aten._scaled_dot_product_flash_attention_
+// backward), so training uses this forward + the library backward.
+//
+// MFMA: v_mfma_f32_16x16x32_bf16 per-wave tiles (layouts verified on
+// silicon by mfma_probe in ops.hip / tests/test_ops_gpu.py):
+//   A[16x32]: lane l -> A[l & 15][(l >> 4) * 8 + j]
+//   B[32x16]: lane l -> B[(l >> 4) * 8 + j][l & 15]
+//   C[16x16]: lane l, reg r -> C[(l >> 4) * 4 + r][l & 15]
+//
+// Replaces the aotriton attn_fwd (442 us at B1 H32 S4096 D128 causal,
+// ~310 TF) — see profiles/ for the measured comparison.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <cstdint>
+
+typedef unsigned short u16;
+typedef unsigned int u32;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+#define ATTN_BM 64
+#define ATTN_BN 64
+#define ATTN_D 128
+#define ATTN_WAVES 4
+
+__device__ __forceinline__ float attn_bf2f(u16 b) {
+  union { u32 u; float f; } c;
+  c.u = ((u32)b) << 16;
+  return c.f;
+}
+
+__device__ __forceinline__ u16 attn_f2bf(float f) {
+  __hip_bfloat16_raw r = __float2bfloat16(f);
+  return r.x;
+}
+
+__global__ __launch_bounds__(256) void attn_fwd_kernel(
+    const u16* __restrict__ q, const u16* __restrict__ k,
+    const u16* __restrict__ v, u16* __restrict__ out,
+    float* __restrict__ lse,
+    long q_sb, long q_sh, long q_ss,
+    long k_sb, long k_sh, long k_ss,
+    long v_sb, long v_sh, long v_ss,
+    int n_heads, int S, float scale) {
+  const int qb = blockIdx.x;
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int low = lane & 15;   // A/C row | B/C col within a 16-tile
+  const int kg = lane >> 4;    // lane group (k chunk | C row group)
+
+  __shared__ u16 ldsK[ATTN_BN * ATTN_D];
+  __shared__ u16 ldsV[ATTN_BN * ATTN_D];
+  __shared__ u16 ldsP[ATTN_WAVES][16 * ATTN_BN];
+
+  // ---- load this wave's Q fragments (rows wid*16 .. +15) ----
+  const u16* qbase = q + (long)b * q_sb + (long)h * q_sh
+                     + (long)(qb * ATTN_BM) * q_ss;
+  union { bf16x8 v; uint4 u; } qfrag[4];
+#pragma unroll
+  for (int ks = 0; ks < 4; ++ks) {
+    qfrag[ks].u = *reinterpret_cast<const uint4*>(
+        qbase + (long)(wid * 16 + low) * q_ss + ks * 32 + kg * 8);
+  }
+
+  float m_run[4], l_run[4];
+  float oacc[8][4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) { m_run[r] = -INFINITY; l_run[r] = 0.f; }
+#pragma unroll
+  for (int ct = 0; ct < 8; ++ct)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) oacc[ct][r] = 0.f;
+
+  const u16* kbase = k + (long)b * k_sb + (long)h * k_sh;
+  const u16* vbase = v + (long)b * v_sb + (long)h * v_sh;
+  const int kv_end = (qb + 1) * ATTN_BM;  // causal upper bound (<= S)
+
+  for (int kv0 = 0; kv0 < kv_end; kv0 += ATTN_BN) {
+    // ---- stage K/V tiles: 1024 uint4 each, 4 per thread ----
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+      const int idx = t * 256 + threadIdx.x;
+      const int r = idx >> 4;            // kv row within tile
+      const int c8 = (idx & 15) * 8;     // 8-elem column chunk
+      *reinterpret_cast<uint4*>(&ldsK[r * ATTN_D + c8]) =
+          *reinterpret_cast<const uint4*>(
+              kbase + (long)(kv0 + r) * k_ss + c8);
+      *reinterpret_cast<uint4*>(&ldsV[r * ATTN_D + c8]) =
+          *reinterpret_cast<const uint4*>(
+              vbase + (long)(kv0 + r) * v_ss + c8);
+    }
+    __syncthreads();
+
+    // ---- S = scale * (Q @ K^T), 16x64 strip per wave ----
+    f32x4 sacc[4];
+#pragma unroll
+    for (int ct = 0; ct < 4; ++ct) {
+      f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < 4; ++ks) {
+        bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+            &ldsK[(ct * 16 + low) * ATTN_D + ks * 32 + kg * 8]);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[ks].v, bfrag,
+                                                      acc, 0, 0, 0);
+      }
+      sacc[ct] = acc;
+    }
+
+    // ---- causal mask + online softmax (state per reg = per C row) ----
+    const int qrow0 = qb * ATTN_BM + wid * 16 + kg * 4;  // + r
+    float mx[4], alpha[4], psum[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      mx[r] = -INFINITY;
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        const int col = kv0 + ct * 16 + low;
+        float s = sacc[ct][r] * scale;
+        if (col > qrow0 + r) s = -INFINITY;
+        sacc[ct][r] = s;
+        mx[r] = fmaxf(mx[r], s);
+      }
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        mx[r] = fmaxf(mx[r], __shfl_xor(mx[r], off, 64));
+      const float mnew = fmaxf(m_run[r], mx[r]);
+      alpha[r] = (m_run[r] == -INFINITY) ? 0.f : __expf(m_run[r] - mnew);
+      m_run[r] = mnew;
+      psum[r] = 0.f;
+#pragma unroll
+      for (int ct = 0; ct < 4; ++ct) {
+        const float p = (sacc[ct][r] == -INFINITY)
+                            ? 0.f : __expf(sacc[ct][r] - mnew);
+        sacc[ct][r] = p;
+        psum[r] += p;
+        ldsP[wid][(kg * 4 + r) * ATTN_BN + ct * 16 + low] = attn_f2bf(p);
+      }
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        psum[r] += __shfl_xor(psum[r], off, 64);
+      l_run[r] = l_run[r] * alpha[r] + psum[r];
+#pragma unroll
+      for (int ct2 = 0; ct2 < 8; ++ct2) oacc[ct2][r] *= alpha[r];
+    }
+    __syncthreads();  // P strips visible; K/V reads done before restage
+
+    // ---- O += P @ V ----
+#pragma unroll
+    for (int ct2 = 0; ct2 < 8; ++ct2) {
+      f32x4 acc = {oacc[ct2][0], oacc[ct2][1], oacc[ct2][2], oacc[ct2][3]};
+#pragma unroll
+      for (int ks2 = 0; ks2 < 2; ++ks2) {
+        bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
+            &ldsP[wid][low * ATTN_BN + ks2 * 32 + kg * 8]);
+        union { bf16x8 v; u16 h[8]; } bfrag;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          bfrag.h[j] =
+              ldsV[(ks2 * 32 + kg * 8 + j) * ATTN_D + ct2 * 16 + low];
+        }
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag.v, acc,
+                                                      0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r) oacc[ct2][r] = acc[r];
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: normalize, write O [B,H,S,D] contiguous + LSE ----
+  u16* obase = out + (((long)b * n_heads + h) * S + qb * ATTN_BM
+                      + wid * 16) * ATTN_D;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const float inv_l = 1.0f / l_run[r];
+#pragma unroll
+    for (int ct2 = 0; ct2 < 8; ++ct2) {
+      obase[(long)(kg * 4 + r) * ATTN_D + ct2 * 16 + low] =
+          attn_f2bf(oacc[ct2][r] * inv_l);
+    }
+  }
+  if (low == 0) {
+    float* lbase = lse + ((long)b * n_heads + h) * S + qb * ATTN_BM
+                   + wid * 16 + kg * 4;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) lbase[r] = m_run[r] + __logf(l_run[r]);
+  }
+}
+
+extern "C" void attn_fwd(void* stream, const void* q, const void* k,
+                         const void* v, void* out, void* lse,
+                         long q_sb, long q_sh, long q_ss,
+                         long k_sb, long k_sh, long k_ss,
+                         long v_sb, long v_sh, long v_ss,
+                         int batch, int n_heads, int S, float scale) {
+  dim3 grid(S / ATTN_BM, n_heads, batch), block(256);
+  hipLaunchKernelGGL(attn_fwd_kernel, grid, block, 0,
+                     reinterpret_cast<hipStream_t>(stream),
+                     (const u16*)q, (const u16*)k, (const u16*)v, (u16*)out,
+                     (float*)lse, q_sb, q_sh, q_ss, k_sb, k_sh, k_ss,
+                     v_sb, v_sh, v_ss, n_heads, S, scale);
+}

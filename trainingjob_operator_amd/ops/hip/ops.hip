@@ -492,6 +492,37 @@ __global__ void adamw_kernel(float* __restrict__ p32, float* __restrict__ m,
 }
 
 // ===========================================================================
+// MFMA layout probe (test-only): computes C = A @ B with one
+// v_mfma_f32_16x16x32_bf16 per wave under the layout assumptions the
+// attention kernel builds on:
+//   A[16x32]: lane l holds A[l & 15][(l >> 4) * 8 + j], j = 0..7
+//   B[32x16]: lane l holds B[(l >> 4) * 8 + j][l & 15]
+//   C[16x16]: lane l reg r holds C[(l >> 4) * 4 + r][l & 15]
+// Verified by tests/test_ops_gpu.py with asymmetric inputs (guide G9).
+// ===========================================================================
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+__global__ void mfma_probe_kernel(const u16* __restrict__ A,
+                                  const u16* __restrict__ B,
+                                  float* __restrict__ C) {
+  const int lane = threadIdx.x & 63;
+  const int row = lane & 15;
+  const int kg = lane >> 4;
+  union { bf16x8 v; u16 h[8]; } a, b;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    a.h[j] = A[row * 32 + kg * 8 + j];        // A[row][k]
+    b.h[j] = B[(kg * 8 + j) * 16 + row];      // B[k][col], col = lane&15
+  }
+  f32x4 c = {0.f, 0.f, 0.f, 0.f};
+  c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, c, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) C[(kg * 4 + r) * 16 + row] = c[r];
+}
+
+// ===========================================================================
 // extern "C" launchers (called from Python via ctypes with the torch stream)
 // ===========================================================================
 
@@ -509,6 +540,11 @@ extern "C" {
 int hipops_arch_check() {
   // host-side marker so Python can verify the lib loaded
   return 950;
+}
+
+void mfma_probe(void* stream, const void* A, const void* B, void* C) {
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, STREAM,
+                     (const u16*)A, (const u16*)B, (float*)C);
 }
 
 void rmsnorm_fwd(void* stream, const void* x, const void* res_in, const void* w,

@@ -17,6 +17,7 @@ from typing import Optional
 _PKG_DIR = os.path.dirname(os.path.abspath(__file__))
 LIB_PATH = os.path.join(_PKG_DIR, "libhipops.so")
 SRC_PATH = os.path.join(_PKG_DIR, "hip", "ops.hip")
+ATTN_SRC_PATH = os.path.join(_PKG_DIR, "hip", "attention.hip")
 
 _lib: Optional[ctypes.CDLL] = None
 
@@ -29,7 +30,7 @@ def build_ops(verbose: bool = False) -> str:
     """Compile ops.hip -> libhipops.so in-tree for gfx950."""
     cmd = [
         "hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17",
-        "-shared", "-fPIC", SRC_PATH, "-o", LIB_PATH,
+        "-shared", "-fPIC", SRC_PATH, ATTN_SRC_PATH, "-o", LIB_PATH,
     ]
     if verbose:
         print("+", " ".join(cmd))
@@ -68,6 +69,9 @@ def load(require: bool = True) -> Optional[ctypes.CDLL]:
     _sig(lib.l2normsq, [vp, vp, l, vp, i, vp])
     _sig(lib.adamw_step, [vp, vp, vp, vp, vp, vp, vp, l,
                           f, f, f, f, f, f, f, f, f, vp])
+    _sig(lib.mfma_probe, [vp, vp, vp, vp])
+    _sig(lib.attn_fwd, [vp, vp, vp, vp, vp, vp,
+                        l, l, l, l, l, l, l, l, l, i, i, i, f])
     assert lib.hipops_arch_check() == 950
     _lib = lib
     return lib
