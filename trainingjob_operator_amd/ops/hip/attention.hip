@@ -14,8 +14,14 @@
 //   B[32x16]: lane l -> B[(l >> 4) * 8 + j][l & 15]
 //   C[16x16]: lane l, reg r -> C[(l >> 4) * 4 + r][l & 15]
 //
-// Replaces the aotriton attn_fwd (442 us at B1 H32 S4096 D128 causal,
-// ~310 TF) — see profiles/ for the measured comparison.
+// LDS layout (the v1 linear layouts measured 1113 us vs aotriton's 442 —
+// every 256-B-stride row put a 16-lane read group on one bank):
+//   * K tile [64][128] with the guide's T2 XOR swizzle
+//     (byte ^= (row & 15) << 4): ds_read_b128 B-fragments conflict-free.
+//   * V stored TRANSPOSED [128][64+4] (pad 4 elems): the P@V B-fragment
+//     becomes one contiguous ds_read_b128 per lane instead of 8 scalar
+//     same-bank u16 reads; the pad staggers banks.
+//   * P strip [16][64] with byte ^= (row & 7) << 4.
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
@@ -30,16 +36,24 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 #define ATTN_BN 64
 #define ATTN_D 128
 #define ATTN_WAVES 4
-
-__device__ __forceinline__ float attn_bf2f(u16 b) {
-  union { u32 u; float f; } c;
-  c.u = ((u32)b) << 16;
-  return c.f;
-}
+// Transposed-V row pitch: 72 elems = 144 B keeps every ds_read_b128 16-B
+// aligned (row*144 % 16 == 0) and spreads the 16-lane read group over
+// distinct banks (36 dwords/row, gcd(36,64)=4 -> 16 distinct slots).
+#define VT_PITCH (ATTN_BN + 8)
 
 __device__ __forceinline__ u16 attn_f2bf(float f) {
   __hip_bfloat16_raw r = __float2bfloat16(f);
   return r.x;
+}
+
+// T2 swizzle for the K tile: 16-byte slot index XORed with row & 15.
+__device__ __forceinline__ int k_byte(int row, int col_elem) {
+  return (row * ATTN_D * 2 + col_elem * 2) ^ ((row & 15) << 4);
+}
+
+// P strip swizzle (row length 128 B = 8 slots): XOR with row & 7.
+__device__ __forceinline__ int p_byte(int row, int col_elem) {
+  return (row * ATTN_BN * 2 + col_elem * 2) ^ ((row & 7) << 4);
 }
 
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
@@ -58,9 +72,9 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int low = lane & 15;   // A/C row | B/C col within a 16-tile
   const int kg = lane >> 4;    // lane group (k chunk | C row group)
 
-  __shared__ u16 ldsK[ATTN_BN * ATTN_D];
-  __shared__ u16 ldsV[ATTN_BN * ATTN_D];
-  __shared__ u16 ldsP[ATTN_WAVES][16 * ATTN_BN];
+  __shared__ u16 ldsK[ATTN_BN * ATTN_D];          // swizzled rows
+  __shared__ u16 ldsVt[ATTN_D * VT_PITCH];        // transposed + padded
+  __shared__ u16 ldsP[ATTN_WAVES][16 * ATTN_BN];  // swizzled rows
 
   // ---- load this wave's Q fragments (rows wid*16 .. +15) ----
   const u16* qbase = q + (long)b * q_sb + (long)h * q_sh
@@ -84,20 +98,25 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const u16* kbase = k + (long)b * k_sb + (long)h * k_sh;
   const u16* vbase = v + (long)b * v_sb + (long)h * v_sh;
   const int kv_end = (qb + 1) * ATTN_BM;  // causal upper bound (<= S)
+  char* ldsKb = reinterpret_cast<char*>(ldsK);
+  char* ldsPb = reinterpret_cast<char*>(ldsP[wid]);
 
   for (int kv0 = 0; kv0 < kv_end; kv0 += ATTN_BN) {
-    // ---- stage K/V tiles: 1024 uint4 each, 4 per thread ----
+    // ---- stage K (swizzled rows) and V (transposed, padded) ----
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
       const int idx = t * 256 + threadIdx.x;
       const int r = idx >> 4;            // kv row within tile
       const int c8 = (idx & 15) * 8;     // 8-elem column chunk
-      *reinterpret_cast<uint4*>(&ldsK[r * ATTN_D + c8]) =
+      *reinterpret_cast<uint4*>(&ldsKb[k_byte(r, c8)]) =
           *reinterpret_cast<const uint4*>(
               kbase + (long)(kv0 + r) * k_ss + c8);
-      *reinterpret_cast<uint4*>(&ldsV[r * ATTN_D + c8]) =
-          *reinterpret_cast<const uint4*>(
-              vbase + (long)(kv0 + r) * v_ss + c8);
+      union { uint4 u; u16 h[8]; } vv;
+      vv.u = *reinterpret_cast<const uint4*>(
+          vbase + (long)(kv0 + r) * v_ss + c8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        ldsVt[(c8 + j) * VT_PITCH + r] = vv.h[j];
     }
     __syncthreads();
 
@@ -109,7 +128,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
       for (int ks = 0; ks < 4; ++ks) {
         bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
-            &ldsK[(ct * 16 + low) * ATTN_D + ks * 32 + kg * 8]);
+            &ldsKb[k_byte(ct * 16 + low, ks * 32 + kg * 8)]);
         acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[ks].v, bfrag,
                                                       acc, 0, 0, 0);
       }
@@ -143,7 +162,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
                             ? 0.f : __expf(sacc[ct][r] - mnew);
         sacc[ct][r] = p;
         psum[r] += p;
-        ldsP[wid][(kg * 4 + r) * ATTN_BN + ct * 16 + low] = attn_f2bf(p);
+        *reinterpret_cast<u16*>(
+            &ldsPb[p_byte(kg * 4 + r, ct * 16 + low)]) = attn_f2bf(p);
       }
 #pragma unroll
       for (int off = 1; off < 16; off <<= 1)
@@ -161,14 +181,11 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
       for (int ks2 = 0; ks2 < 2; ++ks2) {
         bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
-            &ldsP[wid][low * ATTN_BN + ks2 * 32 + kg * 8]);
-        union { bf16x8 v; u16 h[8]; } bfrag;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          bfrag.h[j] =
-              ldsV[(ks2 * 32 + kg * 8 + j) * ATTN_D + ct2 * 16 + low];
-        }
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag.v, acc,
+            &ldsPb[p_byte(low, ks2 * 32 + kg * 8)]);
+        // transposed V: B[k][col] = ldsVt[col][k] — contiguous 8 k's
+        bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+            &ldsVt[(ct2 * 16 + low) * VT_PITCH + ks2 * 32 + kg * 8]);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc,
                                                       0, 0, 0);
       }
 #pragma unroll
