@@ -45,11 +45,12 @@ class _NativeFlashAttention(torch.autograd.Function):
     def backward(ctx, gout):
         q, k, v, out, lse = ctx.saved_tensors
         S = q.shape[2]
-        cum = torch.empty(0, dtype=torch.int32, device=q.device)
+        # None -> undefined Tensor: selects the dense (non-varlen) path in
+        # the aten flash backward (empty tensors select varlen and fail).
         philox = torch.empty(0, dtype=torch.int64, device=q.device)
         dq, dk, dv = torch.ops.aten._scaled_dot_product_flash_attention_backward(
             gout.contiguous(), q, k, v, out, lse,
-            cum, cum, S, S, 0.0, True,
+            None, None, S, S, 0.0, True,
             philox, philox, scale=ctx.scale)
         return dq, dk, dv, None
 
