@@ -62,6 +62,7 @@ class DDPEngine:
         for b in self.buckets:
             for pid in b.param_ids:
                 self._bucket_of[pid] = b
+        self._next_launch = 0
         self._reset_pending()
         store.on_param_grad_ready(self._param_ready)
 
@@ -69,6 +70,7 @@ class DDPEngine:
         for b in self.buckets:
             b.pending = set(b.param_ids)
             b.work = None
+        self._next_launch = 0
 
     # -- hooks ------------------------------------------------------------
     def _param_ready(self, p) -> None:
@@ -76,8 +78,17 @@ class DDPEngine:
             return
         b = self._bucket_of[id(p)]
         b.pending.discard(id(p))
-        if not b.pending:
-            self._launch(b)
+        self._launch_ready_in_order()
+
+    def _launch_ready_in_order(self) -> None:
+        """Launch ready buckets strictly in index order so every rank
+        enqueues RCCL collectives in the same sequence (a readiness-order
+        launch can deadlock NCCL/RCCL if autograd hook timing differs
+        across ranks — same discipline as torch DDP)."""
+        while (self._next_launch < len(self.buckets)
+               and not self.buckets[self._next_launch].pending):
+            self._launch(self.buckets[self._next_launch])
+            self._next_launch += 1
 
     def _launch(self, b: _Bucket) -> None:
         flat = self.store.flat_grad[b.start:b.end]
@@ -89,9 +100,11 @@ class DDPEngine:
         """Flush un-fired buckets (params without grads) and wait for all
         in-flight reductions. Call after loss.backward() on the sync step."""
         if self.world_size > 1 and self.require_sync:
-            for b in self.buckets:
-                if b.work is None:
-                    self._launch(b)
+            # flush: any bucket whose params never produced grads counts as
+            # ready (its flat region holds zeros), keeping launch order
+            for b in self.buckets[self._next_launch:]:
+                b.pending.clear()
+            self._launch_ready_in_order()
             for b in self.buckets:
                 if b.work is not None:
                     b.work.wait()
