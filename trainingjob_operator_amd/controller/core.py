@@ -42,6 +42,8 @@ class TrainingJobController:
         self.status_engine = StatusEngine(api, self.recorder, self._enqueue)
         self.gc = GarbageCollector(api, self.options.namespace)
         self.metrics = metrics
+        from ..utils.tracing import tracer
+        self.trace = tracer("controller")
         self._stop = threading.Event()
 
     # ------------------------------------------------------------------
@@ -85,10 +87,17 @@ class TrainingJobController:
         # gate on active phases (controller.go:298-304)
         if job.status.phase not in ACTIVE_PHASES:
             return
+        phase_before = job.status.phase
         self.reconcile(job, now)
         if self.metrics:
             self.metrics.observe_sync(time.perf_counter() - t0,
                                       job.status.phase)
+        self.trace.event("sync", job=key,
+                         duration_ms=round((time.perf_counter() - t0) * 1e3, 3),
+                         phase=job.status.phase)
+        if job.status.phase != phase_before:
+            self.trace.event("phase_transition", job=key,
+                             from_phase=phase_before, to=job.status.phase)
 
     # ------------------------------------------------------------------
     # reconcile (reference: controller.go:314-388)

@@ -75,13 +75,25 @@ def main(argv=None) -> int:
         stop_requested["flag"] = True
     signal.signal(signal.SIGTERM, on_term)
 
+    from ..utils.tracing import tracer
+    trace = tracer("worker")
+    tokens_per_step = cfg.tokens_per_step_per_rank() * ctx.world_size
+    t_last = time.time()
     while trainer.step_count < args.steps and not stop_requested["flag"]:
         loss = trainer.train_step()
         step = trainer.step_count
         if step % args.log_every == 0 and ctx.is_rank0:
-            log.info("step %d loss %.4f", step, loss.item())
+            now_t = time.time()
+            tps = tokens_per_step * args.log_every / max(now_t - t_last, 1e-9)
+            t_last = now_t
+            log.info("step %d loss %.4f tokens/s %.0f", step, loss.item(),
+                     tps)
+            trace.event("train_step", step=step, loss=round(loss.item(), 4),
+                        tokens_per_sec=round(tps, 1),
+                        world_size=ctx.world_size)
         if step % args.ckpt_every == 0 and ctx.is_rank0:
             ckpt.save_async(trainer)
+            trace.event("checkpoint", step=step)
 
     if ctx.is_rank0:
         ckpt.save_async(trainer, blocking=True)
