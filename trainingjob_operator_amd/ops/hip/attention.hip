@@ -15,12 +15,16 @@
 //   C[16x16]: lane l, reg r -> C[(l >> 4) * 4 + r][l & 15]
 //
 // LDS layout (the v1 linear layouts measured 1113 us vs aotriton's 442 —
-// every 256-B-stride row put a 16-lane read group on one bank):
+// every 256-B-stride row put a 16-lane read group on one bank; v2's
+// transposed-V writes were 16-way write-conflicted, SQ_LDS_BANK_CONFLICT
+// = 35% of wave cycles):
 //   * K tile [64][128] with the guide's T2 XOR swizzle
 //     (byte ^= (row & 15) << 4): ds_read_b128 B-fragments conflict-free.
-//   * V stored TRANSPOSED [128][64+4] (pad 4 elems): the P@V B-fragment
-//     becomes one contiguous ds_read_b128 per lane instead of 8 scalar
-//     same-bank u16 reads; the pad staggers banks.
+//   * V in the 4x16 subtiled image the hardware transpose read wants:
+//     element (kv, d) at ((d/16)*16 + kv/4)*64 + (kv%4)*16 + (d%16).
+//     Staging writes stay contiguous uint4 (one per thread/iter); the
+//     P@V B-fragment is two ds_read_b64_tr_b16 per 16-col tile (each
+//     gathers 4 kv rows of this lane's column at 32-B stride).
 //   * P strip [16][64] with byte ^= (row & 7) << 4.
 
 #include <hip/hip_runtime.h>
@@ -36,10 +40,10 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 #define ATTN_BN 64
 #define ATTN_D 128
 #define ATTN_WAVES 4
-// Transposed-V row pitch: 72 elems = 144 B keeps every ds_read_b128 16-B
-// aligned (row*144 % 16 == 0) and spreads the 16-lane read group over
-// distinct banks (36 dwords/row, gcd(36,64)=4 -> 16 distinct slots).
-#define VT_PITCH (ATTN_BN + 8)
+// V subtile image index (elements): 8 d-blocks x 16 kv-blocks x (4x16).
+__device__ __forceinline__ int v_img_elem(int kv, int d) {
+  return (((d >> 4) << 4) + (kv >> 2)) * 64 + ((kv & 3) << 4) + (d & 15);
+}
 
 __device__ __forceinline__ u16 attn_f2bf(float f) {
   __hip_bfloat16_raw r = __float2bfloat16(f);
@@ -73,7 +77,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int kg = lane >> 4;    // lane group (k chunk | C row group)
 
   __shared__ u16 ldsK[ATTN_BN * ATTN_D];          // swizzled rows
-  __shared__ u16 ldsVt[ATTN_D * VT_PITCH];        // transposed + padded
+  __shared__ u16 ldsV[ATTN_BN * ATTN_D];          // 4x16 subtiled image
   __shared__ u16 ldsP[ATTN_WAVES][16 * ATTN_BN];  // swizzled rows
 
   // ---- load this wave's Q fragments (rows wid*16 .. +15) ----
@@ -111,12 +115,10 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       *reinterpret_cast<uint4*>(&ldsKb[k_byte(r, c8)]) =
           *reinterpret_cast<const uint4*>(
               kbase + (long)(kv0 + r) * k_ss + c8);
-      union { uint4 u; u16 h[8]; } vv;
-      vv.u = *reinterpret_cast<const uint4*>(
-          vbase + (long)(kv0 + r) * v_ss + c8);
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        ldsVt[(c8 + j) * VT_PITCH + r] = vv.h[j];
+      // subtiled image keeps the 8-elem chunk contiguous (c8 % 16 in {0,8})
+      *reinterpret_cast<uint4*>(&ldsV[v_img_elem(r, c8)]) =
+          *reinterpret_cast<const uint4*>(
+              vbase + (long)(kv0 + r) * v_ss + c8);
     }
     __syncthreads();
 
@@ -175,6 +177,10 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     __syncthreads();  // P strips visible; K/V reads done before restage
 
     // ---- O += P @ V ----
+    // B-fragment via hardware transpose read: lane l gathers the 4 kv rows
+    // of its own column (32-B stride in the subtiled image). The two reads
+    // + their wait live in ONE asm statement so hipcc's (absent) asm
+    // bookkeeping can't consume the destinations early (guide §5.7 form i).
 #pragma unroll
     for (int ct2 = 0; ct2 < 8; ++ct2) {
       f32x4 acc = {oacc[ct2][0], oacc[ct2][1], oacc[ct2][2], oacc[ct2][3]};
@@ -182,10 +188,22 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       for (int ks2 = 0; ks2 < 2; ++ks2) {
         bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
             &ldsPb[p_byte(low, ks2 * 32 + kg * 8)]);
-        // transposed V: B[k][col] = ldsVt[col][k] — contiguous 8 k's
-        bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
-            &ldsVt[(ct2 * 16 + low) * VT_PITCH + ks2 * 32 + kg * 8]);
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc,
+        union { unsigned long long q[2]; bf16x8 v; } bfrag;
+        // AS(3) pointer == 32-bit LDS byte offset for the ds instruction
+        const __attribute__((address_space(3))) u16* p3 =
+            (const __attribute__((address_space(3))) u16*)(
+                &ldsV[(ct2 * 16 + ks2 * 8 + kg * 2) * 64 + low]);
+        unsigned long long lo, hi;
+        asm volatile(
+            "ds_read_b64_tr_b16 %0, %2\n\t"
+            "ds_read_b64_tr_b16 %1, %2 offset:128\n\t"
+            "s_waitcnt lgkmcnt(0)"
+            : "=&v"(lo), "=&v"(hi)
+            : "v"(p3)
+            : "memory");
+        bfrag.q[0] = lo;
+        bfrag.q[1] = hi;
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag.v, acc,
                                                       0, 0, 0);
       }
 #pragma unroll
