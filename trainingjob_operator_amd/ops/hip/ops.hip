@@ -523,6 +523,34 @@ __global__ void mfma_probe_kernel(const u16* __restrict__ A,
 }
 
 // ===========================================================================
+// ds_read_b64_tr_b16 semantics probe (test-only): LDS holds raw u16 == its
+// element index; each lane issues one transpose-read at a mode-dependent
+// address and dumps its 4 raw results, revealing the exact lane/elem ->
+// LDS-offset mapping on silicon.
+// ===========================================================================
+
+__global__ void tr_probe_kernel(u16* __restrict__ out, int base_elems,
+                                int mode) {
+  __shared__ u16 lds[4096];
+  for (int i = threadIdx.x; i < 4096; i += 64) lds[i] = (u16)i;
+  __syncthreads();
+  const int lane = threadIdx.x & 63;
+  int addr_elem = base_elems;
+  if (mode == 1) addr_elem += lane;            // per-lane elem stride
+  else if (mode == 2) addr_elem += (lane & 15);  // column-only offset
+  else if (mode == 3) addr_elem += (lane & 15) + (lane >> 4) * 64;
+  const __attribute__((address_space(3))) u16* p3 =
+      (const __attribute__((address_space(3))) u16*)(&lds[addr_elem]);
+  unsigned long long lo;
+  asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+               : "=&v"(lo) : "v"(p3) : "memory");
+  union { unsigned long long q; u16 h[4]; } u;
+  u.q = lo;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) out[lane * 4 + j] = u.h[j];
+}
+
+// ===========================================================================
 // extern "C" launchers (called from Python via ctypes with the torch stream)
 // ===========================================================================
 
@@ -545,6 +573,11 @@ int hipops_arch_check() {
 void mfma_probe(void* stream, const void* A, const void* B, void* C) {
   hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, STREAM,
                      (const u16*)A, (const u16*)B, (float*)C);
+}
+
+void tr_probe(void* stream, void* out, int base_elems, int mode) {
+  hipLaunchKernelGGL(tr_probe_kernel, dim3(1), dim3(64), 0, STREAM,
+                     (u16*)out, base_elems, mode);
 }
 
 void rmsnorm_fwd(void* stream, const void* x, const void* res_in, const void* w,

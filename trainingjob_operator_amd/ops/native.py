@@ -85,3 +85,18 @@ def stream_ptr() -> ctypes.c_void_p:
     """Current torch HIP stream as a raw handle."""
     import torch
     return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+
+def tr_probe_maps(device="cuda"):
+    """Dump ds_read_b64_tr_b16 lane/elem -> LDS element mappings."""
+    import torch
+    lib = load()
+    lib.tr_probe.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                             ctypes.c_int, ctypes.c_int]
+    results = {}
+    for mode in (0, 1, 2, 3):
+        out = torch.empty(64, 4, dtype=torch.int16, device=device)
+        lib.tr_probe(stream_ptr(), ctypes.c_void_p(out.data_ptr()), 0, mode)
+        torch.cuda.synchronize()
+        results[mode] = out.to(torch.int32).cpu() & 0xFFFF
+    return results
