@@ -40,11 +40,6 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 #define ATTN_BN 64
 #define ATTN_D 128
 #define ATTN_WAVES 4
-// V subtile image index (elements): 8 d-blocks x 16 kv-blocks x (4x16).
-__device__ __forceinline__ int v_img_elem(int kv, int d) {
-  return (((d >> 4) << 4) + (kv >> 2)) * 64 + ((kv & 3) << 4) + (d & 15);
-}
-
 __device__ __forceinline__ u16 attn_f2bf(float f) {
   __hip_bfloat16_raw r = __float2bfloat16(f);
   return r.x;
@@ -77,7 +72,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int kg = lane >> 4;    // lane group (k chunk | C row group)
 
   __shared__ u16 ldsK[ATTN_BN * ATTN_D];          // swizzled rows
-  __shared__ u16 ldsV[ATTN_BN * ATTN_D];          // 4x16 subtiled image
+  __shared__ u16 ldsV[ATTN_BN * ATTN_D];          // swizzled rows (like K)
   __shared__ u16 ldsP[ATTN_WAVES][16 * ATTN_BN];  // swizzled rows
 
   // ---- load this wave's Q fragments (rows wid*16 .. +15) ----
@@ -103,6 +98,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const u16* vbase = v + (long)b * v_sb + (long)h * v_sh;
   const int kv_end = (qb + 1) * ATTN_BM;  // causal upper bound (<= S)
   char* ldsKb = reinterpret_cast<char*>(ldsK);
+  char* ldsVb = reinterpret_cast<char*>(ldsV);
   char* ldsPb = reinterpret_cast<char*>(ldsP[wid]);
 
   for (int kv0 = 0; kv0 < kv_end; kv0 += ATTN_BN) {
@@ -115,8 +111,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       *reinterpret_cast<uint4*>(&ldsKb[k_byte(r, c8)]) =
           *reinterpret_cast<const uint4*>(
               kbase + (long)(kv0 + r) * k_ss + c8);
-      // subtiled image keeps the 8-elem chunk contiguous (c8 % 16 in {0,8})
-      *reinterpret_cast<uint4*>(&ldsV[v_img_elem(r, c8)]) =
+      *reinterpret_cast<uint4*>(&ldsVb[k_byte(r, c8)]) =
           *reinterpret_cast<const uint4*>(
               vbase + (long)(kv0 + r) * v_ss + c8);
     }
@@ -177,10 +172,13 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     __syncthreads();  // P strips visible; K/V reads done before restage
 
     // ---- O += P @ V ----
-    // B-fragment via hardware transpose read: lane l gathers the 4 kv rows
-    // of its own column (32-B stride in the subtiled image). The two reads
-    // + their wait live in ONE asm statement so hipcc's (absent) asm
-    // bookkeeping can't consume the destinations early (guide §5.7 form i).
+    // B-fragment: 8 scalar u16 reads down this lane's column of the
+    // T2-swizzled V rows. The swizzle makes the 8 reads land on 8 distinct
+    // bank sets (2-way worst case: the kg half-pairs) instead of the
+    // linear layout's all-on-one-bank (v1: 35% of cycles in conflicts).
+    // (ds_read_b64_tr_b16 was probed on silicon — tests/test_ops_gpu.py
+    // tr_probe — and delivers only 16 distinct values per 16-lane group,
+    // so it cannot feed this fragment shape.)
 #pragma unroll
     for (int ct2 = 0; ct2 < 8; ++ct2) {
       f32x4 acc = {oacc[ct2][0], oacc[ct2][1], oacc[ct2][2], oacc[ct2][3]};
@@ -188,21 +186,13 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       for (int ks2 = 0; ks2 < 2; ++ks2) {
         bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
             &ldsPb[p_byte(low, ks2 * 32 + kg * 8)]);
-        union { unsigned long long q[2]; bf16x8 v; } bfrag;
-        // AS(3) pointer == 32-bit LDS byte offset for the ds instruction
-        const __attribute__((address_space(3))) u16* p3 =
-            (const __attribute__((address_space(3))) u16*)(
-                &ldsV[(ct2 * 16 + ks2 * 8 + kg * 2) * 64 + low]);
-        unsigned long long lo, hi;
-        asm volatile(
-            "ds_read_b64_tr_b16 %0, %2\n\t"
-            "ds_read_b64_tr_b16 %1, %2 offset:128\n\t"
-            "s_waitcnt lgkmcnt(0)"
-            : "=&v"(lo), "=&v"(hi)
-            : "v"(p3)
-            : "memory");
-        bfrag.q[0] = lo;
-        bfrag.q[1] = hi;
+        union { bf16x8 v; u16 h[8]; } bfrag;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int kv = ks2 * 32 + kg * 8 + j;
+          bfrag.h[j] = *reinterpret_cast<const u16*>(
+              &ldsVb[k_byte(kv, ct2 * 16 + low)]);
+        }
         acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag.v, acc,
                                                       0, 0, 0);
       }
