@@ -18,13 +18,11 @@
 // every 256-B-stride row put a 16-lane read group on one bank; v2's
 // transposed-V writes were 16-way write-conflicted, SQ_LDS_BANK_CONFLICT
 // = 35% of wave cycles):
-//   * K tile [64][128] with the guide's T2 XOR swizzle
-//     (byte ^= (row & 15) << 4): ds_read_b128 B-fragments conflict-free.
-//   * V in the 4x16 subtiled image the hardware transpose read wants:
-//     element (kv, d) at ((d/16)*16 + kv/4)*64 + (kv%4)*16 + (d%16).
-//     Staging writes stay contiguous uint4 (one per thread/iter); the
-//     P@V B-fragment is two ds_read_b64_tr_b16 per 16-col tile (each
-//     gathers 4 kv rows of this lane's column at 32-B stride).
+//   * K and V tiles [64][128] with the guide's T2 XOR swizzle
+//     (byte ^= (row & 15) << 4): staging writes are contiguous uint4; the
+//     QK^T B-fragment is a conflict-free ds_read_b128, the P@V B-fragment
+//     8 scalar u16 column reads spread over distinct bank sets by the
+//     same swizzle (2-way worst case).
 //   * P strip [16][64] with byte ^= (row & 7) << 4.
 
 #include <hip/hip_runtime.h>
@@ -102,7 +100,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   char* ldsPb = reinterpret_cast<char*>(ldsP[wid]);
 
   for (int kv0 = 0; kv0 < kv_end; kv0 += ATTN_BN) {
-    // ---- stage K (swizzled rows) and V (transposed, padded) ----
+    // ---- stage K and V tiles (swizzled rows, contiguous uint4) ----
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
       const int idx = t * 256 + threadIdx.x;
