@@ -27,23 +27,29 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 
-def _sdpa(q, k, v):
+def _sdpa(q, k, v, enable_gqa=False):
     """SDPA with a backend override knob (AITJ_SDPA_BACKEND =
-    flash|efficient|math|auto) — used to pin a hipGraph-replay-safe
-    backend and for debugging."""
+    flash|efficient|math|native|auto) — pins a backend for debugging or
+    selects the hand-written CDNA4 forward (ops/attention.py)."""
     backend = os.environ.get("AITJ_SDPA_BACKEND", "auto")
     if backend == "native" and q.is_cuda:
         from ..ops.attention import flash_attention
+        groups = q.shape[1] // k.shape[1]
+        if groups > 1:  # the native kernel wants equal head counts
+            k = k.repeat_interleave(groups, dim=1)
+            v = v.repeat_interleave(groups, dim=1)
         return flash_attention(q.contiguous() if q.stride(-1) != 1 else q,
                                k, v)
     if backend == "auto" or not q.is_cuda:
-        return F.scaled_dot_product_attention(q, k, v, is_causal=True)
+        return F.scaled_dot_product_attention(q, k, v, is_causal=True,
+                                              enable_gqa=enable_gqa)
     from torch.nn.attention import SDPBackend, sdpa_kernel
     mapping = {"flash": SDPBackend.FLASH_ATTENTION,
                "efficient": SDPBackend.EFFICIENT_ATTENTION,
                "math": SDPBackend.MATH}
     with sdpa_kernel([mapping[backend]]):
-        return F.scaled_dot_product_attention(q, k, v, is_causal=True)
+        return F.scaled_dot_product_attention(q, k, v, is_causal=True,
+                                              enable_gqa=enable_gqa)
 
 from ..ops import apply_rope, fused_cross_entropy, fused_rmsnorm, make_inv_freq, swiglu
 from .config import LlamaConfig
@@ -70,11 +76,9 @@ class Attention(nn.Module):
         q = q.transpose(1, 2)  # [B, nh, S, D]
         k = k.transpose(1, 2)
         v = v.transpose(1, 2)
-        groups = cfg.num_heads // cfg.num_kv_heads
-        if groups > 1:
-            k = k.repeat_interleave(groups, dim=1)
-            v = v.repeat_interleave(groups, dim=1)
-        o = _sdpa(q, k, v)
+        # grouped-query KV handled inside SDPA (enable_gqa) — avoids
+        # materializing the repeated K/V (~100 MB/layer/direction at 8B)
+        o = _sdpa(q, k, v, enable_gqa=cfg.num_heads != cfg.num_kv_heads)
         o = o.transpose(1, 2).reshape(B, S, cfg.num_heads * cfg.head_dim)
         return self.o_proj(o)
 
