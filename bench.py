@@ -57,8 +57,8 @@ def main() -> None:
     ap.add_argument("--model", default=None,
                     help="model config (default: llama3-8b on GPU, "
                          "llama-tiny on CPU)")
-    ap.add_argument("--micro-batch", type=int, default=1)
-    ap.add_argument("--grad-accum", type=int, default=4)
+    ap.add_argument("--micro-batch", type=int, default=2)
+    ap.add_argument("--grad-accum", type=int, default=2)
     ap.add_argument("--seq-len", type=int, default=4096)
     ap.add_argument("--checkpoint-activations", action="store_true")
     ap.add_argument("--bucket-mb", type=int, default=128)

@@ -77,8 +77,17 @@ class Attention(nn.Module):
         k = k.transpose(1, 2)
         v = v.transpose(1, 2)
         # grouped-query KV handled inside SDPA (enable_gqa) — avoids
-        # materializing the repeated K/V (~100 MB/layer/direction at 8B)
-        o = _sdpa(q, k, v, enable_gqa=cfg.num_heads != cfg.num_kv_heads)
+        # materializing the repeated K/V (~100 MB/layer/direction at 8B).
+        # AITJ_DISABLE_GQA=1 falls back to explicit repeat (A/B knob).
+        if cfg.num_heads != cfg.num_kv_heads and \
+                os.environ.get("AITJ_DISABLE_GQA"):
+            groups = cfg.num_heads // cfg.num_kv_heads
+            k = k.repeat_interleave(groups, dim=1)
+            v = v.repeat_interleave(groups, dim=1)
+            o = _sdpa(q, k, v)
+        else:
+            o = _sdpa(q, k, v,
+                      enable_gqa=cfg.num_heads != cfg.num_kv_heads)
         o = o.transpose(1, 2).reshape(B, S, cfg.num_heads * cfg.head_dim)
         return self.o_proj(o)
 
