@@ -58,6 +58,20 @@ def test_ddp_two_ranks_stay_in_sync():
         assert results[0] == results[1]
 
 
+@pytest.mark.timeout(300)
+def test_ddp_three_ranks_stay_in_sync():
+    """Odd world size: exercises the 1/world pre-scale and the strict
+    index-ordered bucket launches with non-power-of-two participation."""
+    port = _free_port()
+    world = 3
+    with mp.Manager() as mgr:
+        results = mgr.dict()
+        mp.spawn(_worker, args=(world, port, results), nprocs=world,
+                 join=True)
+        assert len(results) == world
+        assert results[0] == results[1] == results[2]
+
+
 def _worker_vs_single(rank, world, port, results):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
