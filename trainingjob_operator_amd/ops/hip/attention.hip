@@ -18,11 +18,12 @@
 // every 256-B-stride row put a 16-lane read group on one bank; v2's
 // transposed-V writes were 16-way write-conflicted, SQ_LDS_BANK_CONFLICT
 // = 35% of wave cycles):
-//   * K and V tiles [64][128] with the guide's T2 XOR swizzle
-//     (byte ^= (row & 15) << 4): staging writes are contiguous uint4; the
-//     QK^T B-fragment is a conflict-free ds_read_b128, the P@V B-fragment
-//     8 scalar u16 column reads spread over distinct bank sets by the
-//     same swizzle (2-way worst case).
+//   * K tile [64][128] with the guide's T2 XOR swizzle
+//     (byte ^= (row & 15) << 4): contiguous uint4 staging writes and
+//     conflict-free ds_read_b128 QK^T B-fragments.
+//   * V transposed [d][kv] with a (d>>3)-keyed slot XOR (see vt_byte):
+//     <=2-way scatter writes, single conflict-free b128 P@V B-fragment
+//     reads (v4's scalar-read variant was VALU-bound on address math).
 //   * P strip [16][64] with byte ^= (row & 7) << 4.
 
 #include <hip/hip_runtime.h>
@@ -48,6 +49,19 @@ __device__ __forceinline__ int k_byte(int row, int col_elem) {
   return (row * ATTN_D * 2 + col_elem * 2) ^ ((row & 15) << 4);
 }
 
+// Transposed V image [d][kv] with a (d>>3)-keyed slot XOR.
+// Writes scatter one u16 per d (16 lanes share a kv row but have distinct
+// c8 -> distinct (d>>3) -> distinct banks, <=2-way on the r halves);
+// reads are one contiguous b128 of 8 kv for this lane's d (the XOR is
+// constant across the 16-byte run and 36*low already spreads the lane
+// group over distinct dword banks). Pitch 144 B keeps b128 16-B aligned;
+// the XOR (<=240 B) may cross row ends, so the buffer carries 256 B of
+// tail padding.
+#define VT_PITCH_B 144
+__device__ __forceinline__ int vt_byte(int d, int kv) {
+  return (d * VT_PITCH_B + kv * 2) ^ (((d >> 3) & 15) << 4);
+}
+
 // P strip swizzle (row length 128 B = 8 slots): XOR with row & 7.
 __device__ __forceinline__ int p_byte(int row, int col_elem) {
   return (row * ATTN_BN * 2 + col_elem * 2) ^ ((row & 7) << 4);
@@ -69,9 +83,9 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int low = lane & 15;   // A/C row | B/C col within a 16-tile
   const int kg = lane >> 4;    // lane group (k chunk | C row group)
 
-  __shared__ u16 ldsK[ATTN_BN * ATTN_D];          // swizzled rows
-  __shared__ u16 ldsV[ATTN_BN * ATTN_D];          // swizzled rows (like K)
-  __shared__ u16 ldsP[ATTN_WAVES][16 * ATTN_BN];  // swizzled rows
+  __shared__ u16 ldsK[ATTN_BN * ATTN_D];              // swizzled rows
+  __shared__ u16 ldsV[ATTN_D * (VT_PITCH_B / 2) + 128];  // transposed image
+  __shared__ u16 ldsP[ATTN_WAVES][16 * ATTN_BN];      // swizzled rows
 
   // ---- load this wave's Q fragments (rows wid*16 .. +15) ----
   const u16* qbase = q + (long)b * q_sb + (long)h * q_sh
@@ -109,9 +123,12 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       *reinterpret_cast<uint4*>(&ldsKb[k_byte(r, c8)]) =
           *reinterpret_cast<const uint4*>(
               kbase + (long)(kv0 + r) * k_ss + c8);
-      *reinterpret_cast<uint4*>(&ldsVb[k_byte(r, c8)]) =
-          *reinterpret_cast<const uint4*>(
-              vbase + (long)(kv0 + r) * v_ss + c8);
+      union { uint4 u; u16 h[8]; } vv;
+      vv.u = *reinterpret_cast<const uint4*>(
+          vbase + (long)(kv0 + r) * v_ss + c8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        *reinterpret_cast<u16*>(&ldsVb[vt_byte(c8 + j, r)]) = vv.h[j];
     }
     __syncthreads();
 
@@ -170,13 +187,12 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     __syncthreads();  // P strips visible; K/V reads done before restage
 
     // ---- O += P @ V ----
-    // B-fragment: 8 scalar u16 reads down this lane's column of the
-    // T2-swizzled V rows. The swizzle makes the 8 reads land on 8 distinct
-    // bank sets (2-way worst case: the kg half-pairs) instead of the
-    // linear layout's all-on-one-bank (v1: 35% of cycles in conflicts).
-    // (ds_read_b64_tr_b16 was probed on silicon — tests/test_ops_gpu.py
-    // tr_probe — and delivers only 16 distinct values per 16-lane group,
-    // so it cannot feed this fragment shape.)
+    // B-fragment: ONE contiguous b128 read of 8 kv for this lane's column
+    // from the transposed V image. v4's 8 scalar reads per fragment cost
+    // ~14 VALU per MFMA in address math (SQ_INSTS_VALU 118.8M) — the wide
+    // read removes that. (ds_read_b64_tr_b16 was probed on silicon —
+    // tests/test_ops_gpu.py tr_probe — and delivers only 16 distinct
+    // values per 16-lane group, so it cannot feed this fragment shape.)
 #pragma unroll
     for (int ct2 = 0; ct2 < 8; ++ct2) {
       f32x4 acc = {oacc[ct2][0], oacc[ct2][1], oacc[ct2][2], oacc[ct2][3]};
@@ -184,14 +200,9 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       for (int ks2 = 0; ks2 < 2; ++ks2) {
         bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
             &ldsPb[p_byte(low, ks2 * 32 + kg * 8)]);
-        union { bf16x8 v; u16 h[8]; } bfrag;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int kv = ks2 * 32 + kg * 8 + j;
-          bfrag.h[j] = *reinterpret_cast<const u16*>(
-              &ldsVb[k_byte(kv, ct2 * 16 + low)]);
-        }
-        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag.v, acc,
+        bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(
+            &ldsVb[vt_byte(ct2 * 16 + low, ks2 * 32 + kg * 8)]);
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc,
                                                       0, 0, 0);
       }
 #pragma unroll
