@@ -83,9 +83,11 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int low = lane & 15;   // A/C row | B/C col within a 16-tile
   const int kg = lane >> 4;    // lane group (k chunk | C row group)
 
-  __shared__ u16 ldsK[ATTN_BN * ATTN_D];              // swizzled rows
-  __shared__ u16 ldsV[ATTN_D * (VT_PITCH_B / 2) + 128];  // transposed image
-  __shared__ u16 ldsP[ATTN_WAVES][16 * ATTN_BN];      // swizzled rows
+  // K/V double-buffered: stage tile t+1 while computing tile t, ONE
+  // barrier per iteration (P is wave-private, needs none).
+  __shared__ u16 ldsK[2][ATTN_BN * ATTN_D];
+  __shared__ u16 ldsV[2][ATTN_D * (VT_PITCH_B / 2) + 128];
+  __shared__ u16 ldsP[ATTN_WAVES][16 * ATTN_BN];
 
   // ---- load this wave's Q fragments (rows wid*16 .. +15) ----
   const u16* qbase = q + (long)b * q_sb + (long)h * q_sh
@@ -109,28 +111,37 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const u16* kbase = k + (long)b * k_sb + (long)h * k_sh;
   const u16* vbase = v + (long)b * v_sb + (long)h * v_sh;
   const int kv_end = (qb + 1) * ATTN_BM;  // causal upper bound (<= S)
-  char* ldsKb = reinterpret_cast<char*>(ldsK);
-  char* ldsVb = reinterpret_cast<char*>(ldsV);
   char* ldsPb = reinterpret_cast<char*>(ldsP[wid]);
 
+#define STAGE_KV(buf, kv0)                                                  \
+  _Pragma("unroll")                                                         \
+  for (int t = 0; t < 4; ++t) {                                             \
+    const int idx = t * 256 + threadIdx.x;                                  \
+    const int rr = idx >> 4;                                                \
+    const int c8 = (idx & 15) * 8;                                          \
+    *reinterpret_cast<uint4*>(                                              \
+        &reinterpret_cast<char*>(ldsK[buf])[k_byte(rr, c8)]) =              \
+        *reinterpret_cast<const uint4*>(                                    \
+            kbase + (long)((kv0) + rr) * k_ss + c8);                        \
+    union { uint4 u; u16 h[8]; } vv;                                        \
+    vv.u = *reinterpret_cast<const uint4*>(                                 \
+        vbase + (long)((kv0) + rr) * v_ss + c8);                            \
+    _Pragma("unroll")                                                       \
+    for (int j = 0; j < 8; ++j)                                             \
+      *reinterpret_cast<u16*>(&reinterpret_cast<char*>(                     \
+          ldsV[buf])[vt_byte(c8 + j, rr)]) = vv.h[j];                       \
+  }
+
+  STAGE_KV(0, 0);
+  __syncthreads();
+  int cur = 0;
   for (int kv0 = 0; kv0 < kv_end; kv0 += ATTN_BN) {
-    // ---- stage K and V tiles (swizzled rows, contiguous uint4) ----
-#pragma unroll
-    for (int t = 0; t < 4; ++t) {
-      const int idx = t * 256 + threadIdx.x;
-      const int r = idx >> 4;            // kv row within tile
-      const int c8 = (idx & 15) * 8;     // 8-elem column chunk
-      *reinterpret_cast<uint4*>(&ldsKb[k_byte(r, c8)]) =
-          *reinterpret_cast<const uint4*>(
-              kbase + (long)(kv0 + r) * k_ss + c8);
-      union { uint4 u; u16 h[8]; } vv;
-      vv.u = *reinterpret_cast<const uint4*>(
-          vbase + (long)(kv0 + r) * v_ss + c8);
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        *reinterpret_cast<u16*>(&ldsVb[vt_byte(c8 + j, r)]) = vv.h[j];
+    // ---- stage the NEXT tile into the other buffer (overlaps compute) --
+    if (kv0 + ATTN_BN < kv_end) {
+      STAGE_KV(cur ^ 1, kv0 + ATTN_BN);
     }
-    __syncthreads();
+    char* ldsKb = reinterpret_cast<char*>(ldsK[cur]);
+    char* ldsVb = reinterpret_cast<char*>(ldsV[cur]);
 
     // ---- S = scale * (Q @ K^T), 16x64 strip per wave ----
     f32x4 sacc[4];
@@ -184,7 +195,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
       for (int ct2 = 0; ct2 < 8; ++ct2) oacc[ct2][r] *= alpha[r];
     }
-    __syncthreads();  // P strips visible; K/V reads done before restage
+    // no barrier: P is wave-private (in-order LDS pipe within a wave)
 
     // ---- O += P @ V ----
     // B-fragment: ONE contiguous b128 read of 8 kv for this lane's column
@@ -208,7 +219,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) oacc[ct2][r] = acc[r];
     }
-    __syncthreads();
+    __syncthreads();  // everyone done with buf[cur] before it is restaged
+    cur ^= 1;
   }
 
   // ---- epilogue: normalize, write O [B,H,S,D] contiguous + LSE ----
