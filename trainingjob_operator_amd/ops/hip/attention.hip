@@ -113,32 +113,41 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int kv_end = (qb + 1) * ATTN_BM;  // causal upper bound (<= S)
   char* ldsPb = reinterpret_cast<char*>(ldsP[wid]);
 
-#define STAGE_KV(buf, kv0)                                                  \
+  // T14 split staging: ISSUE the next tile's global loads at iteration
+  // start (into registers), WRITE them to LDS after this tile's compute —
+  // HBM latency hides under the MFMAs instead of stalling the loop head.
+  const int stage_r = threadIdx.x >> 2;            // thread's kv row
+  const int stage_c8 = (threadIdx.x & 3) * 8;      // base column chunk
+#define LOAD_KV(kreg, vreg, kv0)                                            \
   _Pragma("unroll")                                                         \
   for (int t = 0; t < 4; ++t) {                                             \
-    const int idx = t * 256 + threadIdx.x;                                  \
-    const int rr = idx >> 4;                                                \
-    const int c8 = (idx & 15) * 8;                                          \
-    *reinterpret_cast<uint4*>(                                              \
-        &reinterpret_cast<char*>(ldsK[buf])[k_byte(rr, c8)]) =              \
-        *reinterpret_cast<const uint4*>(                                    \
-            kbase + (long)((kv0) + rr) * k_ss + c8);                        \
+    kreg[t] = *reinterpret_cast<const uint4*>(                              \
+        kbase + (long)((kv0) + stage_r) * k_ss + stage_c8 + t * 32);        \
+    vreg[t] = *reinterpret_cast<const uint4*>(                              \
+        vbase + (long)((kv0) + stage_r) * v_ss + stage_c8 + t * 32);        \
+  }
+#define WRITE_KV(buf, kreg, vreg)                                           \
+  _Pragma("unroll")                                                         \
+  for (int t = 0; t < 4; ++t) {                                             \
+    *reinterpret_cast<uint4*>(&reinterpret_cast<char*>(                     \
+        ldsK[buf])[k_byte(stage_r, stage_c8 + t * 32)]) = kreg[t];          \
     union { uint4 u; u16 h[8]; } vv;                                        \
-    vv.u = *reinterpret_cast<const uint4*>(                                 \
-        vbase + (long)((kv0) + rr) * v_ss + c8);                            \
+    vv.u = vreg[t];                                                         \
     _Pragma("unroll")                                                       \
     for (int j = 0; j < 8; ++j)                                             \
       *reinterpret_cast<u16*>(&reinterpret_cast<char*>(                     \
-          ldsV[buf])[vt_byte(c8 + j, rr)]) = vv.h[j];                       \
+          ldsV[buf])[vt_byte(stage_c8 + t * 32 + j, stage_r)]) = vv.h[j];   \
   }
 
-  STAGE_KV(0, 0);
+  uint4 kreg[4], vreg[4];
+  LOAD_KV(kreg, vreg, 0);
+  WRITE_KV(0, kreg, vreg);
   __syncthreads();
   int cur = 0;
   for (int kv0 = 0; kv0 < kv_end; kv0 += ATTN_BN) {
-    // ---- stage the NEXT tile into the other buffer (overlaps compute) --
-    if (kv0 + ATTN_BN < kv_end) {
-      STAGE_KV(cur ^ 1, kv0 + ATTN_BN);
+    const bool prefetch = kv0 + ATTN_BN < kv_end;
+    if (prefetch) {
+      LOAD_KV(kreg, vreg, kv0 + ATTN_BN);  // issue only; consumed at end
     }
     char* ldsKb = reinterpret_cast<char*>(ldsK[cur]);
     char* ldsVb = reinterpret_cast<char*>(ldsV[cur]);
@@ -218,6 +227,9 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       }
 #pragma unroll
       for (int r = 0; r < 4; ++r) oacc[ct2][r] = acc[r];
+    }
+    if (prefetch) {
+      WRITE_KV(cur ^ 1, kreg, vreg);  // loads have had the whole tile to land
     }
     __syncthreads();  // everyone done with buf[cur] before it is restaged
     cur ^= 1;
