@@ -139,15 +139,21 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
           ldsV[buf])[vt_byte(stage_c8 + t * 32 + j, stage_r)]) = vv.h[j];   \
   }
 
+  // Guide T14 order: regs hold tile t+1 (loaded a FULL iteration ago, so
+  // the write never stalls); write it right after the barrier, then
+  // immediately re-issue the loads for tile t+2.
   uint4 kreg[4], vreg[4];
   LOAD_KV(kreg, vreg, 0);
   WRITE_KV(0, kreg, vreg);
+  if (ATTN_BN < kv_end) LOAD_KV(kreg, vreg, ATTN_BN);
   __syncthreads();
   int cur = 0;
   for (int kv0 = 0; kv0 < kv_end; kv0 += ATTN_BN) {
-    const bool prefetch = kv0 + ATTN_BN < kv_end;
-    if (prefetch) {
-      LOAD_KV(kreg, vreg, kv0 + ATTN_BN);  // issue only; consumed at end
+    if (kv0 + ATTN_BN < kv_end) {
+      WRITE_KV(cur ^ 1, kreg, vreg);             // tile t+1 -> other buffer
+      if (kv0 + 2 * ATTN_BN < kv_end) {
+        LOAD_KV(kreg, vreg, kv0 + 2 * ATTN_BN);  // issue tile t+2
+      }
     }
     char* ldsKb = reinterpret_cast<char*>(ldsK[cur]);
     char* ldsVb = reinterpret_cast<char*>(ldsV[cur]);
@@ -227,9 +233,6 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       }
 #pragma unroll
       for (int r = 0; r < 4; ++r) oacc[ct2][r] = acc[r];
-    }
-    if (prefetch) {
-      WRITE_KV(cur ^ 1, kreg, vreg);  // loads have had the whole tile to land
     }
     __syncthreads();  // everyone done with buf[cur] before it is restaged
     cur ^= 1;
