@@ -83,11 +83,9 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int low = lane & 15;   // A/C row | B/C col within a 16-tile
   const int kg = lane >> 4;    // lane group (k chunk | C row group)
 
-  // K/V double-buffered: stage tile t+1 while computing tile t, ONE
-  // barrier per iteration (P is wave-private, needs none).
-  __shared__ u16 ldsK[2][ATTN_BN * ATTN_D];
-  __shared__ u16 ldsV[2][ATTN_D * (VT_PITCH_B / 2) + 128];
-  __shared__ u16 ldsP[ATTN_WAVES][16 * ATTN_BN];
+  __shared__ u16 ldsK[ATTN_BN * ATTN_D];              // swizzled rows
+  __shared__ u16 ldsV[ATTN_D * (VT_PITCH_B / 2) + 128];  // transposed image
+  __shared__ u16 ldsP[ATTN_WAVES][16 * ATTN_BN];      // swizzled rows
 
   // ---- load this wave's Q fragments (rows wid*16 .. +15) ----
   const u16* qbase = q + (long)b * q_sb + (long)h * q_sh
@@ -111,52 +109,28 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const u16* kbase = k + (long)b * k_sb + (long)h * k_sh;
   const u16* vbase = v + (long)b * v_sb + (long)h * v_sh;
   const int kv_end = (qb + 1) * ATTN_BM;  // causal upper bound (<= S)
+  char* ldsKb = reinterpret_cast<char*>(ldsK);
+  char* ldsVb = reinterpret_cast<char*>(ldsV);
   char* ldsPb = reinterpret_cast<char*>(ldsP[wid]);
 
-  // T14 split staging: ISSUE the next tile's global loads at iteration
-  // start (into registers), WRITE them to LDS after this tile's compute —
-  // HBM latency hides under the MFMAs instead of stalling the loop head.
-  const int stage_r = threadIdx.x >> 2;            // thread's kv row
-  const int stage_c8 = (threadIdx.x & 3) * 8;      // base column chunk
-#define LOAD_KV(kreg, vreg, kv0)                                            \
-  _Pragma("unroll")                                                         \
-  for (int t = 0; t < 4; ++t) {                                             \
-    kreg[t] = *reinterpret_cast<const uint4*>(                              \
-        kbase + (long)((kv0) + stage_r) * k_ss + stage_c8 + t * 32);        \
-    vreg[t] = *reinterpret_cast<const uint4*>(                              \
-        vbase + (long)((kv0) + stage_r) * v_ss + stage_c8 + t * 32);        \
-  }
-#define WRITE_KV(buf, kreg, vreg)                                           \
-  _Pragma("unroll")                                                         \
-  for (int t = 0; t < 4; ++t) {                                             \
-    *reinterpret_cast<uint4*>(&reinterpret_cast<char*>(                     \
-        ldsK[buf])[k_byte(stage_r, stage_c8 + t * 32)]) = kreg[t];          \
-    union { uint4 u; u16 h[8]; } vv;                                        \
-    vv.u = vreg[t];                                                         \
-    _Pragma("unroll")                                                       \
-    for (int j = 0; j < 8; ++j)                                             \
-      *reinterpret_cast<u16*>(&reinterpret_cast<char*>(                     \
-          ldsV[buf])[vt_byte(stage_c8 + t * 32 + j, stage_r)]) = vv.h[j];   \
-  }
-
-  // Guide T14 order: regs hold tile t+1 (loaded a FULL iteration ago, so
-  // the write never stalls); write it right after the barrier, then
-  // immediately re-issue the loads for tile t+2.
-  uint4 kreg[4], vreg[4];
-  LOAD_KV(kreg, vreg, 0);
-  WRITE_KV(0, kreg, vreg);
-  if (ATTN_BN < kv_end) LOAD_KV(kreg, vreg, ATTN_BN);
-  __syncthreads();
-  int cur = 0;
   for (int kv0 = 0; kv0 < kv_end; kv0 += ATTN_BN) {
-    if (kv0 + ATTN_BN < kv_end) {
-      WRITE_KV(cur ^ 1, kreg, vreg);             // tile t+1 -> other buffer
-      if (kv0 + 2 * ATTN_BN < kv_end) {
-        LOAD_KV(kreg, vreg, kv0 + 2 * ATTN_BN);  // issue tile t+2
-      }
+    // ---- stage K and V tiles (swizzled rows, contiguous uint4) ----
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+      const int idx = t * 256 + threadIdx.x;
+      const int r = idx >> 4;            // kv row within tile
+      const int c8 = (idx & 15) * 8;     // 8-elem column chunk
+      *reinterpret_cast<uint4*>(&ldsKb[k_byte(r, c8)]) =
+          *reinterpret_cast<const uint4*>(
+              kbase + (long)(kv0 + r) * k_ss + c8);
+      union { uint4 u; u16 h[8]; } vv;
+      vv.u = *reinterpret_cast<const uint4*>(
+          vbase + (long)(kv0 + r) * v_ss + c8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        *reinterpret_cast<u16*>(&ldsVb[vt_byte(c8 + j, r)]) = vv.h[j];
     }
-    char* ldsKb = reinterpret_cast<char*>(ldsK[cur]);
-    char* ldsVb = reinterpret_cast<char*>(ldsV[cur]);
+    __syncthreads();
 
     // ---- S = scale * (Q @ K^T), 16x64 strip per wave ----
     f32x4 sacc[4];
@@ -210,7 +184,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
       for (int ct2 = 0; ct2 < 8; ++ct2) oacc[ct2][r] *= alpha[r];
     }
-    // no barrier: P is wave-private (in-order LDS pipe within a wave)
+    __syncthreads();  // P strips visible; K/V reads done before restage
 
     // ---- O += P @ V ----
     // B-fragment: ONE contiguous b128 read of 8 kv for this lane's column
@@ -234,8 +208,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) oacc[ct2][r] = acc[r];
     }
-    __syncthreads();  // everyone done with buf[cur] before it is restaged
-    cur ^= 1;
+    __syncthreads();
   }
 
   // ---- epilogue: normalize, write O [B,H,S,D] contiguous + LSE ----
