@@ -16,7 +16,6 @@ from __future__ import annotations
 
 import json
 import os
-import ssl
 import threading
 from typing import Any, Callable, Dict, Iterator, List, Optional
 

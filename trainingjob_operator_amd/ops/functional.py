@@ -45,7 +45,6 @@ class _FusedRMSNorm(torch.autograd.Function):
         saved_res = res_out if res_out is not None else x
         ctx.save_for_backward(saved_res, weight, rrms)
         ctx.has_residual = residual is not None
-        ctx.eps = eps
         return (y, res_out) if residual is not None else (y, x)
 
     @staticmethod
