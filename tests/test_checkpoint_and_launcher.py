@@ -1,0 +1,92 @@
+"""Checkpointer unit tests + launcher env-contract fallbacks."""
+import os
+
+import torch
+
+from trainingjob_operator_amd.launcher.checkpoint import Checkpointer
+from trainingjob_operator_amd.parallel import dist_ctx
+from trainingjob_operator_amd.training import TrainConfig, Trainer
+
+
+def make_trainer():
+    cfg = TrainConfig(model="llama-tiny", micro_batch=1, grad_accum=1,
+                      seq_len=16, lr=1e-3)
+    return Trainer(cfg)
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    t1 = make_trainer()
+    for _ in range(3):
+        t1.train_step()
+    ckpt = Checkpointer(str(tmp_path))
+    path = ckpt.save_async(t1, blocking=True)
+    assert os.path.exists(path)
+
+    t2 = make_trainer()
+    assert not torch.equal(t2.store.flat_param, t1.store.flat_param)
+    step = ckpt.load_latest(t2)
+    assert step == 3
+    assert torch.equal(t2.store.flat_param, t1.store.flat_param)
+    assert torch.equal(t2.opt.m, t1.opt.m)
+    assert t2.opt.step_count == 3
+    # resumed trainer keeps training
+    t2.train_step()
+    assert t2.opt.step_count == 4
+
+
+def test_checkpoint_prune_keeps_latest(tmp_path):
+    t = make_trainer()
+    ckpt = Checkpointer(str(tmp_path), keep=2)
+    for _ in range(4):
+        t.train_step()
+        ckpt.save_async(t, blocking=True)
+    ckpts = ckpt.list_checkpoints()
+    assert len(ckpts) == 2
+    assert ckpt.latest().endswith("00000004.pt")
+
+
+def test_checkpoint_async_then_wait(tmp_path):
+    t = make_trainer()
+    ckpt = Checkpointer(str(tmp_path))
+    path = ckpt.save_async(t, blocking=False)
+    ckpt.wait()
+    assert os.path.exists(path)
+    assert not os.path.exists(path + ".tmp")
+
+
+def test_load_latest_empty_dir(tmp_path):
+    t = make_trainer()
+    assert Checkpointer(str(tmp_path)).load_latest(t) is None
+
+
+def test_dist_ctx_torchrun_env(monkeypatch):
+    for k in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR",
+              "MASTER_PORT", "TRAININGJOB_REPLICA_NAME",
+              "TRAININGJOB_REPLICA_INDEX", "TRAINER_INSTANCES",
+              "TRAINER_INSTANCES_NUM"):
+        monkeypatch.delenv(k, raising=False)
+    monkeypatch.setenv("RANK", "3")
+    monkeypatch.setenv("WORLD_SIZE", "8")
+    monkeypatch.setenv("LOCAL_RANK", "3")
+    monkeypatch.setenv("MASTER_ADDR", "10.0.0.1")
+    monkeypatch.setenv("MASTER_PORT", "29500")
+    ctx = dist_ctx.from_env()
+    assert (ctx.rank, ctx.world_size, ctx.local_rank) == (3, 8, 3)
+    assert ctx.master_addr == "10.0.0.1" and ctx.master_port == 29500
+
+
+def test_dist_ctx_reference_contract_fallback(monkeypatch):
+    """The operator's reference env alone (no torch-style vars) must be
+    enough to bootstrap."""
+    for k in ("RANK", "WORLD_SIZE", "LOCAL_RANK", "MASTER_ADDR",
+              "MASTER_PORT"):
+        monkeypatch.delenv(k, raising=False)
+    monkeypatch.setenv("TRAININGJOB_REPLICA_NAME", "trainer")
+    monkeypatch.setenv("TRAININGJOB_REPLICA_INDEX", "2")
+    monkeypatch.setenv("TRAINER_INSTANCES_NUM", "4")
+    monkeypatch.setenv("TRAINER_INSTANCES",
+                       "j-trainer-0.ns,j-trainer-1.ns,j-trainer-2.ns,"
+                       "j-trainer-3.ns")
+    ctx = dist_ctx.from_env()
+    assert (ctx.rank, ctx.world_size) == (2, 4)
+    assert ctx.master_addr == "j-trainer-0.ns"
