@@ -1,0 +1,62 @@
+"""CLI arg-parsing smoke tests + multi-role env contract details."""
+import subprocess
+import sys
+
+from trainingjob_operator_amd.api import constants as C
+from trainingjob_operator_amd.api.defaults import set_defaults
+from trainingjob_operator_amd.api.types import AITrainingJob
+from trainingjob_operator_amd.controller.envinject import render_env
+
+
+def _help(module):
+    out = subprocess.run([sys.executable, "-m", module, "--help"],
+                         capture_output=True, text=True, timeout=60)
+    assert out.returncode == 0, out.stderr
+    return out.stdout
+
+
+def test_operator_server_help():
+    text = _help("trainingjob_operator_amd.controller.server")
+    for flag in ("--namespace", "--resync-period", "--thread-num",
+                 "--creating-restart-period", "--enable-creating-failed",
+                 "--metrics-port"):
+        assert flag in text
+
+
+def test_launcher_help():
+    text = _help("trainingjob_operator_amd.launcher.main")
+    for flag in ("--model", "--steps", "--ckpt-dir", "--seq-len"):
+        assert flag in text
+
+
+def test_node_agent_help():
+    text = _help("trainingjob_operator_amd.agent.node_agent")
+    assert "--node-name" in text
+
+
+def test_multirole_env_contract():
+    """PS/worker jobs: every pod sees BOTH roles' host lists (reference
+    pod.go:553-598 iterates all replicaSpecs) but a role-LOCAL torch world."""
+    job = set_defaults(AITrainingJob.from_dict({
+        "metadata": {"name": "ps", "namespace": "ml"},
+        "spec": {"replicaSpecs": {
+            "pserver": {"replicas": 2, "template": {"spec": {"containers": [
+                {"name": "aitj-ps",
+                 "ports": [{"name": "aitj-grpc", "containerPort": 2222}]}]}}},
+            "worker": {"replicas": 3, "template": {"spec": {"containers": [
+                {"name": "aitj-w",
+                 "ports": [{"name": "aitj-grpc", "containerPort": 2223}]}]}}},
+        }},
+    }))
+    env = {e["name"]: e["value"]
+           for e in render_env(job, "worker", index=1, restart_count=0)}
+    # cross-role visibility (reference contract)
+    assert env["PSERVER_INSTANCES"] == "ps-pserver-0.ml,ps-pserver-1.ml"
+    assert env["PSERVER_HOSTS"] == "ps-pserver-0.ml:2222,ps-pserver-1.ml:2222"
+    assert env["WORKER_INSTANCES_NUM"] == "3"
+    assert env["WORKER_PORTS"] == "2223"
+    # role-local torch world (MI355X extension)
+    assert env["WORLD_SIZE"] == "3"          # worker role size, not 5
+    assert env["RANK"] == "1"
+    assert env["MASTER_ADDR"] == "ps-worker-0.ml"
+    assert env["MASTER_PORT"] == "2223"      # the role's own aitj port
