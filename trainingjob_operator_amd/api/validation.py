@@ -124,6 +124,7 @@ def validate(job: AITrainingJob) -> List[str]:
                 "only aitj-* containers participate in status tracking"
             )
 
+    errors.extend(validate_hbm_sizing(job))
     return errors
 
 
@@ -131,6 +132,32 @@ def validate_or_raise(job: AITrainingJob) -> None:
     errs = validate(job)
     if errs:
         raise ValidationError(errs)
+
+
+def validate_hbm_sizing(job: AITrainingJob) -> List[str]:
+    """HBM-aware admission (SURVEY.md §2.3 'no GPU awareness'): when the job
+    declares its model size (annotations ``elasticdeeplearning.ai/model`` or
+    ``.../model-params``) and a replica requests amd.com/gpu, reject specs
+    whose per-GPU DP training state cannot fit MI355X's 288 GB."""
+    from . import sizing
+
+    errors: List[str] = []
+    n_params = sizing.declared_params(job)
+    if n_params is None:
+        return errors
+    est = sizing.estimate_training_bytes(n_params)
+    for rtype, rs in job.spec.replica_specs.items():
+        gpus = gpus_requested(rs)
+        if gpus <= 0:
+            continue
+        if not sizing.fits_per_gpu(est, gpus):
+            need = sizing.min_gpus_for(est)
+            errors.append(
+                f"spec.replicaSpecs[{rtype}]: model of {n_params:,} params "
+                f"needs ~{est.total_gb:.0f} GB of HBM per replica but the "
+                f"pod requests {gpus} x 288 GB GPU(s); request at least "
+                f"{need} amd.com/gpu (or shard the model)")
+    return errors
 
 
 def gpus_requested(replica_spec) -> int:
