@@ -133,3 +133,20 @@ def test_adamw_matches_torch():
                              eps, wd, step)
     # bf16 grads introduce small drift vs fp32 grads — loose tolerance
     assert torch.allclose(p32, pt.detach(), atol=5e-3)
+
+
+def test_swiglu_packed_matches_unpacked():
+    from trainingjob_operator_amd.ops import swiglu_packed
+    T, F = 8, 64
+    gu = torch.randn(T, 2 * F, requires_grad=True)
+    out = swiglu_packed(gu)
+    g, u = gu.detach().split([F, F], dim=-1)
+    ref = torch.nn.functional.silu(g) * u
+    assert torch.allclose(out, ref, atol=1e-5)
+    dy = torch.randn_like(out)
+    (out * dy).sum().backward()
+    ga = g.clone().requires_grad_()
+    ua = u.clone().requires_grad_()
+    ((torch.nn.functional.silu(ga) * ua) * dy).sum().backward()
+    assert torch.allclose(gu.grad[:, :F], ga.grad, atol=1e-5)
+    assert torch.allclose(gu.grad[:, F:], ua.grad, atol=1e-5)

@@ -167,3 +167,21 @@ def test_smoke_training_step_gpu():
     l1 = trainer.train_step().item()
     assert l0 == l0 and l1 == l1  # finite
     torch.cuda.synchronize()
+
+
+def test_swiglu_packed_matches_reference():
+    from trainingjob_operator_amd.ops import swiglu_packed
+    T, F = 512, 14336
+    gu = _mk((T, 2 * F), scale=2.0).requires_grad_()
+    out = swiglu_packed(gu)
+    g, u = gu.detach().float().cpu().split([F, F], dim=-1)
+    ref = reference.swiglu_fwd(g.contiguous(), u.contiguous())
+    assert torch.allclose(out.float().cpu(), ref, atol=BF16_ATOL, rtol=1e-2)
+    dy = _mk((T, F))
+    (out.float() * dy.float()).sum().backward()
+    dg_ref, du_ref = reference.swiglu_bwd(dy.float().cpu(), g.contiguous(),
+                                          u.contiguous())
+    assert torch.allclose(gu.grad[:, :F].float().cpu(), dg_ref, atol=5e-2,
+                          rtol=5e-2)
+    assert torch.allclose(gu.grad[:, F:].float().cpu(), du_ref, atol=5e-2,
+                          rtol=5e-2)

@@ -51,7 +51,8 @@ def _sdpa(q, k, v, enable_gqa=False):
         return F.scaled_dot_product_attention(q, k, v, is_causal=True,
                                               enable_gqa=enable_gqa)
 
-from ..ops import apply_rope, fused_cross_entropy, fused_rmsnorm, make_inv_freq, swiglu
+from ..ops import (apply_rope, fused_cross_entropy, fused_rmsnorm,
+                   make_inv_freq, swiglu_packed)
 from .config import LlamaConfig
 
 
@@ -60,17 +61,23 @@ class Attention(nn.Module):
         super().__init__()
         self.cfg = cfg
         H = cfg.hidden_size
-        self.q_proj = nn.Linear(H, cfg.num_heads * cfg.head_dim, bias=False)
-        self.k_proj = nn.Linear(H, cfg.num_kv_heads * cfg.head_dim, bias=False)
-        self.v_proj = nn.Linear(H, cfg.num_kv_heads * cfg.head_dim, bias=False)
-        self.o_proj = nn.Linear(cfg.num_heads * cfg.head_dim, H, bias=False)
+        self.q_size = cfg.num_heads * cfg.head_dim
+        self.kv_size = cfg.num_kv_heads * cfg.head_dim
+        # fused QKV: one hipBLASLt GEMM instead of three (the k/v GEMMs are
+        # skinny at GQA ratios and underutilize the chip on their own)
+        self.qkv_proj = nn.Linear(H, self.q_size + 2 * self.kv_size,
+                                  bias=False)
+        self.o_proj = nn.Linear(self.q_size, H, bias=False)
 
     def forward(self, x: torch.Tensor, inv_freq: torch.Tensor) -> torch.Tensor:
         B, S, H = x.shape
         cfg = self.cfg
-        q = self.q_proj(x).view(B, S, cfg.num_heads, cfg.head_dim)
-        k = self.k_proj(x).view(B, S, cfg.num_kv_heads, cfg.head_dim)
-        v = self.v_proj(x).view(B, S, cfg.num_kv_heads, cfg.head_dim)
+        qkv = self.qkv_proj(x)
+        q, k, v = qkv.split([self.q_size, self.kv_size, self.kv_size],
+                            dim=-1)
+        q = q.reshape(B, S, cfg.num_heads, cfg.head_dim)
+        k = k.reshape(B, S, cfg.num_kv_heads, cfg.head_dim)
+        v = v.reshape(B, S, cfg.num_kv_heads, cfg.head_dim)
         q = apply_rope(q, inv_freq, S)
         k = apply_rope(k, inv_freq, S)
         q = q.transpose(1, 2)  # [B, nh, S, D]
@@ -96,12 +103,13 @@ class MLP(nn.Module):
     def __init__(self, cfg: LlamaConfig):
         super().__init__()
         H, FF = cfg.hidden_size, cfg.intermediate_size
-        self.gate_proj = nn.Linear(H, FF, bias=False)
-        self.up_proj = nn.Linear(H, FF, bias=False)
+        self.ff = FF
+        # fused gate+up: one GEMM feeding the fused SwiGLU kernel
+        self.gate_up_proj = nn.Linear(H, 2 * FF, bias=False)
         self.down_proj = nn.Linear(FF, H, bias=False)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.down_proj(swiglu(self.gate_proj(x), self.up_proj(x)))
+        return self.down_proj(swiglu_packed(self.gate_up_proj(x)))
 
 
 class Block(nn.Module):

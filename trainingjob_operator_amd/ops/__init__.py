@@ -11,6 +11,7 @@ from .functional import (
     fused_rmsnorm,
     make_inv_freq,
     swiglu,
+    swiglu_packed,
 )
 from .native import HipOpsUnavailable, available, build_ops, load
 

@@ -178,6 +178,50 @@ def swiglu(g: torch.Tensor, u: torch.Tensor) -> torch.Tensor:
     return _SwiGLU.apply(g, u)
 
 
+class _SwiGLUPacked(torch.autograd.Function):
+    """Consumes the fused gate_up GEMM output [..., 2F] directly (rows are
+    [gate | up]); backward emits the packed [.., 2F] gradient so the GEMM
+    backward needs no concat."""
+
+    @staticmethod
+    def forward(ctx, gu: torch.Tensor):
+        F2 = gu.shape[-1]
+        F = F2 // 2
+        rows = gu.numel() // F2
+        ctx.save_for_backward(gu)
+        if gu.is_cuda:
+            assert gu.dtype == torch.bfloat16 and gu.is_contiguous()
+            lib = _hip()
+            out = torch.empty(*gu.shape[:-1], F, dtype=gu.dtype,
+                              device=gu.device)
+            lib.swiglu_packed_fwd(native.stream_ptr(), _ptr(gu), _ptr(out),
+                                  rows, F)
+            return out
+        g, u = gu.split([F, F], dim=-1)
+        return reference.swiglu_fwd(g.contiguous(), u.contiguous())
+
+    @staticmethod
+    def backward(ctx, dout: torch.Tensor):
+        (gu,) = ctx.saved_tensors
+        F = gu.shape[-1] // 2
+        rows = gu.numel() // (2 * F)
+        if gu.is_cuda:
+            lib = _hip()
+            dout = dout.contiguous()
+            dgu = torch.empty_like(gu)
+            lib.swiglu_packed_bwd(native.stream_ptr(), _ptr(dout), _ptr(gu),
+                                  _ptr(dgu), rows, F)
+            return dgu
+        g, u = gu.split([F, F], dim=-1)
+        dg, du = reference.swiglu_bwd(dout, g.contiguous(), u.contiguous())
+        return torch.cat([dg, du], dim=-1)
+
+
+def swiglu_packed(gu: torch.Tensor) -> torch.Tensor:
+    """SwiGLU over packed [..., 2F] rows ([gate | up]) from a fused GEMM."""
+    return _SwiGLUPacked.apply(gu)
+
+
 # ---------------------------------------------------------------------------
 # Fused cross-entropy
 # ---------------------------------------------------------------------------

@@ -298,6 +298,55 @@ __global__ void swiglu_fwd_kernel(const uint4* __restrict__ g,
   }
 }
 
+// Packed variant: gu rows are [gate(F) | up(F)] straight from the fused
+// gate_up GEMM; out rows are [F]. Backward writes the packed dgu, which
+// feeds the GEMM backward with no concat.
+__global__ void swiglu_packed_fwd_kernel(const uint4* __restrict__ gu,
+                                         uint4* __restrict__ out,
+                                         long n_rows, int F8) {
+  const long total = n_rows * F8;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * (long)blockDim.x) {
+    const long row = i / F8;
+    const int col = (int)(i % F8);
+    BF8 gv, uv, ov;
+    gv.v = gu[row * 2 * F8 + col];
+    uv.v = gu[row * 2 * F8 + F8 + col];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float gf = bf2f(gv.h[j]);
+      ov.h[j] = f2bf(gf * sigmoidf_fast(gf) * bf2f(uv.h[j]));
+    }
+    out[i] = ov.v;
+  }
+}
+
+__global__ void swiglu_packed_bwd_kernel(const uint4* __restrict__ dout,
+                                         const uint4* __restrict__ gu,
+                                         uint4* __restrict__ dgu,
+                                         long n_rows, int F8) {
+  const long total = n_rows * F8;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
+       i += gridDim.x * (long)blockDim.x) {
+    const long row = i / F8;
+    const int col = (int)(i % F8);
+    BF8 dov, gv, uv, dgv, duv;
+    dov.v = dout[i];
+    gv.v = gu[row * 2 * F8 + col];
+    uv.v = gu[row * 2 * F8 + F8 + col];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float dof = bf2f(dov.h[j]);
+      const float gf = bf2f(gv.h[j]);
+      const float sg = sigmoidf_fast(gf);
+      duv.h[j] = f2bf(dof * gf * sg);
+      dgv.h[j] = f2bf(dof * bf2f(uv.h[j]) * sg * (1.f + gf * (1.f - sg)));
+    }
+    dgu[row * 2 * F8 + col] = dgv.v;
+    dgu[row * 2 * F8 + F8 + col] = duv.v;
+  }
+}
+
 __global__ void swiglu_bwd_kernel(const uint4* __restrict__ dout,
                                   const uint4* __restrict__ g,
                                   const uint4* __restrict__ u,
@@ -649,6 +698,23 @@ void swiglu_fwd(void* stream, const void* g, const void* u, void* out, long n) {
   dim3 grid(elementwise_grid(n8)), block(BLOCK);
   hipLaunchKernelGGL(swiglu_fwd_kernel, grid, block, 0, STREAM,
                      (const uint4*)g, (const uint4*)u, (uint4*)out, n8);
+}
+
+void swiglu_packed_fwd(void* stream, const void* gu, void* out, long n_rows,
+                       int F) {
+  const int F8 = F / 8;
+  dim3 grid(elementwise_grid(n_rows * (long)F8)), block(BLOCK);
+  hipLaunchKernelGGL(swiglu_packed_fwd_kernel, grid, block, 0, STREAM,
+                     (const uint4*)gu, (uint4*)out, n_rows, F8);
+}
+
+void swiglu_packed_bwd(void* stream, const void* dout, const void* gu,
+                       void* dgu, long n_rows, int F) {
+  const int F8 = F / 8;
+  dim3 grid(elementwise_grid(n_rows * (long)F8)), block(BLOCK);
+  hipLaunchKernelGGL(swiglu_packed_bwd_kernel, grid, block, 0, STREAM,
+                     (const uint4*)dout, (const uint4*)gu, (uint4*)dgu,
+                     n_rows, F8);
 }
 
 void swiglu_bwd(void* stream, const void* dout, const void* g, const void* u,

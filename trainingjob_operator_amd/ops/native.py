@@ -64,6 +64,8 @@ def load(require: bool = True) -> Optional[ctypes.CDLL]:
     _sig(lib.rope, [vp, vp, vp, vp, l, i, i, i, f])
     _sig(lib.swiglu_fwd, [vp, vp, vp, vp, l])
     _sig(lib.swiglu_bwd, [vp, vp, vp, vp, vp, vp, l])
+    _sig(lib.swiglu_packed_fwd, [vp, vp, vp, l, i])
+    _sig(lib.swiglu_packed_bwd, [vp, vp, vp, vp, l, i])
     _sig(lib.ce_fwd, [vp, vp, vp, vp, vp, l, i, i])
     _sig(lib.ce_bwd, [vp, vp, vp, vp, vp, vp, l, i, i])
     _sig(lib.l2normsq, [vp, vp, l, vp, i, vp])
