@@ -75,8 +75,9 @@ class Attention(nn.Module):
         qkv = self.qkv_proj(x)
         q, k, v = qkv.split([self.q_size, self.kv_size, self.kv_size],
                             dim=-1)
-        q = q.reshape(B, S, cfg.num_heads, cfg.head_dim)
-        k = k.reshape(B, S, cfg.num_kv_heads, cfg.head_dim)
+        # q/k go through the RoPE kernel, which wants dense rows
+        q = q.reshape(B, S, cfg.num_heads, cfg.head_dim).contiguous()
+        k = k.reshape(B, S, cfg.num_kv_heads, cfg.head_dim).contiguous()
         v = v.reshape(B, S, cfg.num_kv_heads, cfg.head_dim)
         q = apply_rope(q, inv_freq, S)
         k = apply_rope(k, inv_freq, S)
