@@ -90,3 +90,40 @@ def test_dist_ctx_reference_contract_fallback(monkeypatch):
     ctx = dist_ctx.from_env()
     assert (ctx.rank, ctx.world_size) == (2, 4)
     assert ctx.master_addr == "j-trainer-0.ns"
+
+
+def test_worker_metrics_endpoint(tmp_path):
+    """Launcher with --metrics-port exposes tokens/s on /metrics."""
+    import subprocess, sys, time, urllib.request, socket
+    s = socket.socket(); s.bind(("127.0.0.1", 0))
+    mport = s.getsockname()[1]; s.close()
+    s = socket.socket(); s.bind(("127.0.0.1", 0))
+    dport = s.getsockname()[1]; s.close()
+    env = dict(os.environ, RANK="0", WORLD_SIZE="1",
+               MASTER_ADDR="127.0.0.1", MASTER_PORT=str(dport))
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "trainingjob_operator_amd.launcher.main",
+         "--model", "llama-tiny", "--steps", "6", "--seq-len", "16",
+         "--grad-accum", "1", "--log-every", "1", "--ckpt-every", "100",
+         "--ckpt-dir", str(tmp_path), "--metrics-port", str(mport)],
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
+    try:
+        body = ""
+        deadline = time.monotonic() + 90
+        while time.monotonic() < deadline:
+            try:
+                body = urllib.request.urlopen(
+                    f"http://127.0.0.1:{mport}/metrics", timeout=2).read() \
+                    .decode()
+                if "aitj_worker_tokens_per_sec" in body and \
+                        "aitj_worker_step" in body:
+                    break
+            except OSError:
+                pass
+            if proc.poll() is not None:
+                break
+            time.sleep(0.3)
+        assert "aitj_worker_tokens_per_sec" in body, proc.stdout.read()[-800:]
+    finally:
+        proc.kill()
+        proc.wait(timeout=30)

@@ -36,6 +36,9 @@ def main(argv=None) -> int:
     ap.add_argument("--ckpt-every", type=int, default=50)
     ap.add_argument("--log-every", type=int, default=10)
     ap.add_argument("--checkpoint-activations", action="store_true")
+    ap.add_argument("--metrics-port", type=int, default=int(os.environ.get(
+        "TRAININGJOB_METRICS_PORT", "0")),
+        help="expose Prometheus worker metrics (tokens/s, step time, loss)")
     args = ap.parse_args(argv)
 
     logging.basicConfig(
@@ -77,6 +80,10 @@ def main(argv=None) -> int:
 
     from ..utils.tracing import tracer
     trace = tracer("worker")
+    metrics = None
+    if args.metrics_port and ctx.is_rank0:
+        from .worker_metrics import WorkerMetrics
+        metrics = WorkerMetrics(args.metrics_port)
     tokens_per_step = cfg.tokens_per_step_per_rank() * ctx.world_size
     t_last = time.time()
     while trainer.step_count < args.steps and not stop_requested["flag"]:
@@ -91,6 +98,8 @@ def main(argv=None) -> int:
             trace.event("train_step", step=step, loss=round(loss.item(), 4),
                         tokens_per_sec=round(tps, 1),
                         world_size=ctx.world_size)
+            if metrics:
+                metrics.observe(step, loss.item(), tps)
         if step % args.ckpt_every == 0 and ctx.is_rank0:
             ckpt.save_async(trainer)
             trace.event("checkpoint", step=step)
