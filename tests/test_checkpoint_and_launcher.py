@@ -103,8 +103,8 @@ def test_worker_metrics_endpoint(tmp_path):
                MASTER_ADDR="127.0.0.1", MASTER_PORT=str(dport))
     proc = subprocess.Popen(
         [sys.executable, "-m", "trainingjob_operator_amd.launcher.main",
-         "--model", "llama-tiny", "--steps", "6", "--seq-len", "16",
-         "--grad-accum", "1", "--log-every", "1", "--ckpt-every", "100",
+         "--model", "llama-tiny", "--steps", "100000", "--seq-len", "16",
+         "--grad-accum", "1", "--log-every", "1", "--ckpt-every", "1000000",
          "--ckpt-dir", str(tmp_path), "--metrics-port", str(mport)],
         env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True)
     try:
