@@ -75,7 +75,7 @@ def check_invariants(api: FakeKubeApi, history):
     return job
 
 
-@pytest.mark.parametrize("seed", [1, 7, 42, 1234, 9999])
+@pytest.mark.parametrize("seed", [1, 7, 42, 77, 123, 1234, 4242, 9999])
 def test_torture(seed):
     rng = random.Random(seed)
     api = FakeKubeApi()
