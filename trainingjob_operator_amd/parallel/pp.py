@@ -1,0 +1,201 @@
+"""Pipeline parallelism: Llama stage partitioning + a GPipe schedule over
+torch.distributed P2P (RCCL send/recv on GPU, gloo in CPU tests).
+
+Completes the DP/TP/PP triple the launcher can form from the operator's
+injected env (BASELINE.json north star). The flagship 8B bench stays pure
+DP (one replica fits a 288 GB MI355X); PP is for models that outgrow one
+GPU. On one 8xMI355X node every stage boundary is a direct xGMI link, so
+the P2P activations ride point-to-point bandwidth (~153 GB/s/link) without
+touching collectives.
+
+Pipeline boundary payload: the pre-norm Llama block carries TWO tensors
+(branch output x and the running residual stream — models/llama.py Block),
+so each hop sends/recvs the pair.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from ..models.config import LlamaConfig
+from ..models.llama import Block, LlamaModel
+from ..ops import fused_cross_entropy, fused_rmsnorm
+
+
+def partition_layers(num_layers: int, n_stages: int) -> List[range]:
+    """Contiguous, balanced layer ranges per stage."""
+    base = num_layers // n_stages
+    extra = num_layers % n_stages
+    out = []
+    start = 0
+    for s in range(n_stages):
+        n = base + (1 if s < extra else 0)
+        out.append(range(start, start + n))
+        start += n
+    return out
+
+
+class LlamaStage(nn.Module):
+    """One pipeline stage of a LlamaModel: first stage owns the embedding,
+    last owns final norm + lm_head + loss. Built as views onto a full
+    model's modules so checkpoints/state are shared with the unsharded
+    layout (each PP rank instantiates only its slice in production via
+    `from_config`)."""
+
+    def __init__(self, cfg: LlamaConfig, layers: List[Block],
+                 embed: Optional[nn.Embedding],
+                 final_norm_weight: Optional[nn.Parameter],
+                 lm_head: Optional[nn.Linear], inv_freq: torch.Tensor):
+        super().__init__()
+        self.cfg = cfg
+        self.embed = embed
+        self.blocks = nn.ModuleList(layers)
+        self.lm_head = lm_head
+        if final_norm_weight is not None:
+            self.final_norm_weight = final_norm_weight
+        else:
+            self.final_norm_weight = None
+        self.register_buffer("inv_freq", inv_freq, persistent=False)
+
+    @property
+    def is_first(self) -> bool:
+        return self.embed is not None
+
+    @property
+    def is_last(self) -> bool:
+        return self.lm_head is not None
+
+    @classmethod
+    def from_model(cls, model: LlamaModel, stage: int,
+                   n_stages: int) -> "LlamaStage":
+        parts = partition_layers(model.cfg.num_layers, n_stages)
+        return cls(
+            model.cfg,
+            [model.blocks[i] for i in parts[stage]],
+            model.embed if stage == 0 else None,
+            model.final_norm_weight if stage == n_stages - 1 else None,
+            model.lm_head if stage == n_stages - 1 else None,
+            model.inv_freq,
+        )
+
+    @classmethod
+    def from_config(cls, cfg: LlamaConfig, stage: int, n_stages: int,
+                    device=None) -> "LlamaStage":
+        """Standalone stage (only this slice's parameters exist)."""
+        with torch.device(device or "cpu"):
+            model = LlamaModel(cfg)
+        return cls.from_model(model, stage, n_stages)
+
+    def forward(self, x: torch.Tensor,
+                residual: Optional[torch.Tensor],
+                targets: Optional[torch.Tensor] = None):
+        """First stage: x is the token ids. Last stage (with targets):
+        returns the per-token loss vector. Middle: returns (x, residual)."""
+        if self.is_first:
+            x = self.embed(x)
+            residual = None
+        for blk in self.blocks:
+            x, residual = blk(x, residual, self.inv_freq)
+        if not self.is_last:
+            return x, residual
+        normed, _ = fused_rmsnorm(x, self.final_norm_weight, residual,
+                                  self.cfg.norm_eps)
+        logits = self.lm_head(normed)
+        if targets is None:
+            return logits
+        T = logits.shape[0] * logits.shape[1]
+        per_tok = fused_cross_entropy(
+            logits.reshape(T, -1).contiguous(), targets.reshape(T))
+        n_valid = (targets.reshape(T) != -100).sum().clamp(min=1)
+        return per_tok.sum() / n_valid
+
+
+class GPipeSchedule:
+    """Fill-drain (GPipe) schedule: all micro-batch forwards, then all
+    backwards in reverse — simple, correct, and bubble-bounded by
+    (stages-1)/micro_batches. (1F1B is a round-2 refinement.)"""
+
+    def __init__(self, stage: LlamaStage, stage_idx: int, n_stages: int,
+                 group=None, device=None):
+        self.stage = stage
+        self.idx = stage_idx
+        self.n = n_stages
+        self.group = group
+        self.device = device or "cpu"
+        ranks = (list(range(dist.get_world_size(group)))
+                 if dist.is_initialized() else [0])
+        self.prev_rank = ranks[stage_idx - 1] if stage_idx > 0 else None
+        self.next_rank = ranks[stage_idx + 1] \
+            if stage_idx < n_stages - 1 else None
+
+    # -- P2P helpers ------------------------------------------------------
+    def _send(self, t: torch.Tensor, dst: int):
+        dist.send(t.detach().contiguous(), dst=dst, group=self.group)
+
+    def _recv(self, shape, dtype) -> torch.Tensor:
+        t = torch.empty(*shape, dtype=dtype, device=self.device)
+        dist.recv(t, src=self.prev_rank, group=self.group)
+        return t
+
+    # -- one optimizer-step's worth of micro-batches ----------------------
+    def step(self, micro_batches, hidden_shape,
+             act_dtype=torch.float32) -> Optional[torch.Tensor]:
+        """micro_batches: list of (tokens, targets) on EVERY rank (only the
+        ranks that need them use them: tokens at stage 0, targets at the
+        last). hidden_shape = (mb, seq, hidden). Returns the mean loss on
+        the last stage (None elsewhere). Gradients accumulate into the
+        stage parameters; the caller runs its optimizer."""
+        saved = []
+        n_micro = len(micro_batches)
+        loss_total = None
+
+        for tokens, targets in micro_batches:
+            if self.is_first_stage:
+                out = self.stage(tokens, None,
+                                 targets if self.is_last_stage else None)
+                x_in = res_in = None
+            else:
+                x_in = self._recv(hidden_shape, act_dtype).requires_grad_()
+                res_in = self._recv(hidden_shape, act_dtype).requires_grad_()
+                out = self.stage(x_in, res_in,
+                                 targets if self.is_last_stage else None)
+            if self.is_last_stage:
+                loss = out / n_micro
+                loss_total = loss.detach() if loss_total is None \
+                    else loss_total + loss.detach()
+                saved.append((x_in, res_in, loss, None))
+            else:
+                x_out, res_out = out
+                self._send(x_out, self.next_rank)
+                self._send(res_out, self.next_rank)
+                saved.append((x_in, res_in, x_out, res_out))
+
+        for x_in, res_in, a, b in reversed(saved):
+            if self.is_last_stage:
+                a.backward()  # loss already scaled by 1/n_micro
+            else:
+                dx = self._recv_grad(a)
+                dres = self._recv_grad(b)
+                torch.autograd.backward((a, b), (dx, dres))
+            if not self.is_first_stage:
+                self._send(x_in.grad, self.prev_rank)
+                self._send(res_in.grad, self.prev_rank)
+
+        return loss_total  # sum of (loss/n_micro) == mean over micro-batches
+
+    def _recv_grad(self, like: torch.Tensor) -> torch.Tensor:
+        t = torch.empty_like(like)
+        dist.recv(t, src=self.next_rank, group=self.group)
+        return t
+
+    @property
+    def is_first_stage(self) -> bool:
+        return self.idx == 0
+
+    @property
+    def is_last_stage(self) -> bool:
+        return self.idx == self.n - 1
