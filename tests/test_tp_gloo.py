@@ -104,3 +104,65 @@ def test_topology_grid_4ranks():
         mp.spawn(_topo_worker, args=(world, port, results), nprocs=world,
                  join=True)
         assert dict(results) == {0: (0, 0), 1: (0, 1), 2: (1, 0), 3: (1, 1)}
+
+
+def _tp_llama_worker(rank, world, port, results):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from trainingjob_operator_amd.models.config import LLAMA_TINY
+        from trainingjob_operator_amd.models.llama import LlamaModel
+        from trainingjob_operator_amd.parallel.tp import shard_from
+        from trainingjob_operator_amd.parallel.tp_llama import TPLlamaModel
+        cfg = LLAMA_TINY
+        torch.manual_seed(33)
+        full = LlamaModel(cfg)
+        tp_model = TPLlamaModel(cfg, group=None)
+        tp_model.shard_from_full(full)
+
+        g = torch.Generator().manual_seed(9)
+        tokens = torch.randint(0, cfg.vocab_size, (2, 16), generator=g)
+        loss = tp_model(tokens, tokens)
+        loss.backward()
+        ref_loss = full(tokens, tokens)
+        ref_loss.backward()
+        assert torch.allclose(loss, ref_loss, atol=1e-5), \
+            (loss.item(), ref_loss.item())
+
+        # sharded grads match slices of the full model's grads
+        b = tp_model.blocks[0]
+        fb = full.blocks[0]
+        q_size = cfg.num_heads * cfg.head_dim
+        kv = cfg.num_kv_heads * cfg.head_dim
+        wq_g, wk_g, _ = fb.attn.qkv_proj.weight.grad.split(
+            [q_size, kv, kv], dim=0)
+        assert torch.allclose(b.attn.q_proj.weight.grad,
+                              shard_from(wq_g, 0, None), atol=1e-4)
+        assert torch.allclose(b.attn.o_proj.weight.grad,
+                              shard_from(fb.attn.o_proj.weight.grad, 1, None),
+                              atol=1e-4)
+        wg_g, wu_g = fb.mlp.gate_up_proj.weight.grad.chunk(2, dim=0)
+        assert torch.allclose(b.mlp.gate_proj.weight.grad,
+                              shard_from(wg_g, 0, None), atol=1e-4)
+        assert torch.allclose(b.mlp.down_proj.weight.grad,
+                              shard_from(fb.mlp.down_proj.weight.grad, 1,
+                                         None), atol=1e-4)
+        # replicated pieces see the full gradient on every rank
+        assert torch.allclose(tp_model.final_norm_weight.grad,
+                              full.final_norm_weight.grad, atol=1e-4)
+        results[rank] = float(loss)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_tp_llama_matches_unsharded():
+    port = _free_port()
+    world = 2
+    with mp.Manager() as mgr:
+        results = mgr.dict()
+        mp.spawn(_tp_llama_worker, args=(world, port, results), nprocs=world,
+                 join=True)
+        assert len(results) == world
+        assert results[0] == pytest.approx(results[1])
