@@ -37,11 +37,18 @@ class _Bucket:
 class DDPEngine:
     def __init__(self, store: FlatParamStore,
                  process_group: Optional[object] = None,
-                 bucket_bytes: int = 128 << 20):
+                 bucket_bytes: int = 128 << 20,
+                 world_size: Optional[int] = None):
         self.store = store
         self.group = process_group
-        self.world_size = dist.get_world_size(process_group) \
-            if dist.is_initialized() else 1
+        if world_size is not None:
+            # explicit DP degree (a DPxTP grid passes its dp_size: with
+            # dp == 1 the group is None, which would otherwise read as the
+            # whole world)
+            self.world_size = world_size
+        else:
+            self.world_size = dist.get_world_size(process_group) \
+                if dist.is_initialized() else 1
         self.require_sync = True
         bucket_elems = max(1, bucket_bytes // 2)  # bf16
 

@@ -39,6 +39,10 @@ class TrainConfig:
     # AdamW) as ONE hipGraph: removes the per-kernel launch gaps that
     # measured ~18% of step wall time on MI355X
     use_graphs: bool = False
+    # tensor-parallel degree (world splits into dp x tp; tp ranks are
+    # contiguous -> adjacent GPUs over xGMI). 1 = pure DP (the flagship
+    # 8B config — one replica fits a 288 GB MI355X).
+    tp_size: int = 1
 
     @property
     def model_config(self) -> LlamaConfig:
@@ -82,14 +86,30 @@ class Trainer:
                 else "cpu")
         self.device = device
         torch.manual_seed(cfg.seed)  # identical init on every rank
-        self.model = build_model(cfg.model_config, device,
-                                 cfg.checkpoint_activations)
+        from .parallel.groups import build_topology
+        self.topo = build_topology(cfg.tp_size)
+        if cfg.tp_size > 1:
+            from .parallel.tp_llama import TPLlamaModel
+            full = build_model(cfg.model_config, device,
+                               cfg.checkpoint_activations)
+            self.model = TPLlamaModel(cfg.model_config,
+                                      group=self.topo.tp_group).to(
+                full.embed.weight.dtype).to(device)
+            self.model.inv_freq = full.inv_freq
+            self.model.shard_from_full(full)
+            del full
+        else:
+            self.model = build_model(cfg.model_config, device,
+                                     cfg.checkpoint_activations)
         self.store = FlatParamStore(self.model, device=device)
-        self.ddp = DDPEngine(self.store, bucket_bytes=cfg.bucket_bytes)
+        self.ddp = DDPEngine(self.store, process_group=self.topo.dp_group,
+                             bucket_bytes=cfg.bucket_bytes,
+                             world_size=self.topo.dp_size)
         self.opt = FlatAdamW(self.store, lr=cfg.lr, betas=cfg.betas,
                              weight_decay=cfg.weight_decay,
                              clip_grad_norm=cfg.clip_grad_norm)
-        self.data = synthetic_batches(cfg, device, self.ctx.rank)
+        # tp peers train on the SAME data (they hold shards of one replica)
+        self.data = synthetic_batches(cfg, device, self.topo.dp_rank)
         self.step_count = 0
         self._graph = None
         self._static_batches = None
