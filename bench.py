@@ -62,6 +62,8 @@ def main() -> None:
     ap.add_argument("--seq-len", type=int, default=4096)
     ap.add_argument("--checkpoint-activations", action="store_true")
     ap.add_argument("--bucket-mb", type=int, default=128)
+    ap.add_argument("--tp", type=int, default=1,
+                    help="tensor-parallel degree (world = dp x tp)")
     ap.add_argument("--graphs", action="store_true",
                     help="EXPERIMENTAL: hipGraph step capture (replay of "
                          "large-seq graphs currently faults in the ROCm "
@@ -98,6 +100,7 @@ def main() -> None:
         checkpoint_activations=args.checkpoint_activations,
         bucket_bytes=args.bucket_mb << 20,
         use_graphs=on_gpu and args.graphs,
+        tp_size=args.tp,
     )
     log(f"[bench] rank {ctx.rank}/{n_gpus} model={model} "
         f"mb={cfg.micro_batch} ga={cfg.grad_accum} seq={cfg.seq_len} "
@@ -140,8 +143,10 @@ def main() -> None:
         elapsed = float(t.item())
         dist.barrier()
 
+    # with TP, tp ranks share one replica: whole-job tokens count replicas
+    n_replicas = n_gpus // max(args.tp, 1)
     tokens_per_step_rank = cfg.tokens_per_step_per_rank()
-    total_tokens = tokens_per_step_rank * n_gpus * args.steps
+    total_tokens = tokens_per_step_rank * n_replicas * args.steps
     tokens_per_sec = total_tokens / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
@@ -164,7 +169,8 @@ def main() -> None:
                 "model": model,
                 "global_batch": cfg.micro_batch * cfg.grad_accum * n_gpus,
                 "seq_len": cfg.seq_len,
-                "parallelism": f"dp{n_gpus}",
+                "parallelism": (f"dp{n_gpus // max(args.tp, 1)}"
+                                + (f"tp{args.tp}" if args.tp > 1 else "")),
                 "micro_batch": cfg.micro_batch,
                 "grad_accum": cfg.grad_accum,
                 "checkpoint_activations": cfg.checkpoint_activations,
