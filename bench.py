@@ -167,7 +167,8 @@ def main() -> None:
             "data": "synthetic",
             "config": {
                 "model": model,
-                "global_batch": cfg.micro_batch * cfg.grad_accum * n_gpus,
+                "global_batch": (cfg.micro_batch * cfg.grad_accum
+                                 * (n_gpus // max(args.tp, 1))),
                 "seq_len": cfg.seq_len,
                 "parallelism": (f"dp{n_gpus // max(args.tp, 1)}"
                                 + (f"tp{args.tp}" if args.tp > 1 else "")),
