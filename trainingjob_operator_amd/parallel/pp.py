@@ -199,3 +199,46 @@ class GPipeSchedule:
     @property
     def is_last_stage(self) -> bool:
         return self.idx == self.n - 1
+
+
+class PPTrainer:
+    """Minimal pipeline trainer: one LlamaStage per rank, per-stage flat
+    param/grad store + fused AdamW, GPipe schedule per optimizer step.
+    Composes with the launcher's env bootstrap exactly like the DP Trainer
+    (training.py); DPxPP grids come from parallel/groups.py in round 2."""
+
+    def __init__(self, cfg, stage_idx: int, n_stages: int, device=None,
+                 act_dtype=None):
+        from ..models.config import CONFIGS
+        from ..parallel.flat import FlatParamStore
+        from ..optim import FlatAdamW
+        from ..training import synthetic_batches
+
+        self.cfg = cfg
+        mcfg = CONFIGS[cfg.model]
+        self.device = torch.device(device or "cpu")
+        torch.manual_seed(cfg.seed)
+        self.stage = LlamaStage.from_config(mcfg, stage_idx, n_stages,
+                                            device=self.device)
+        # the flat store keeps parameters in bf16; boundary activations
+        # travel in the same dtype
+        self.act_dtype = act_dtype or torch.bfloat16
+        self.sched = GPipeSchedule(self.stage, stage_idx, n_stages,
+                                   device=self.device)
+        self.store = FlatParamStore(self.stage, device=self.device)
+        self.opt = FlatAdamW(self.store, lr=cfg.lr, betas=cfg.betas,
+                             weight_decay=cfg.weight_decay,
+                             clip_grad_norm=cfg.clip_grad_norm)
+        # every pp rank draws the same data; only the roles that need
+        # tokens/targets use them
+        self.data = synthetic_batches(cfg, self.device, rank=0)
+        self.hidden_shape = (cfg.micro_batch, cfg.seq_len, mcfg.hidden_size)
+        self.step_count = 0
+
+    def train_step(self):
+        micros = [next(self.data) for _ in range(self.cfg.grad_accum)]
+        loss = self.sched.step(micros, self.hidden_shape, self.act_dtype)
+        self.opt.step()
+        self.opt.zero_grad()
+        self.step_count += 1
+        return loss
