@@ -201,6 +201,77 @@ class GPipeSchedule:
         return self.idx == self.n - 1
 
 
+class OneFOneBSchedule(GPipeSchedule):
+    """1F1B: after a (stages - stage_idx - 1)-deep warmup, every forward is
+    immediately followed by the OLDEST outstanding backward, so at most
+    warmup+1 micro-batches of activations are ever live (GPipe keeps all
+    of them). Sends are async (isend) so the canonical 1F1B orderings
+    cannot rendezvous-deadlock; receives stay blocking."""
+
+    def __init__(self, *args, **kw):
+        super().__init__(*args, **kw)
+        self.peak_live = 0  # observability/tests: max outstanding micros
+
+    def _send(self, t: torch.Tensor, dst: int):  # async override
+        w = dist.isend(t.detach().contiguous(), dst=dst, group=self.group)
+        self._send_works.append(w)
+
+    def step(self, micro_batches, hidden_shape,
+             act_dtype=torch.float32) -> Optional[torch.Tensor]:
+        self._send_works = []
+        self.peak_live = 0
+        n_micro = len(micro_batches)
+        warmup = min(self.n - self.idx - 1, n_micro)
+        outstanding = []
+        loss_total = None
+
+        def forward(mb):
+            tokens, targets = mb
+            if self.is_first_stage:
+                out = self.stage(tokens, None,
+                                 targets if self.is_last_stage else None)
+                x_in = res_in = None
+            else:
+                x_in = self._recv(hidden_shape, act_dtype).requires_grad_()
+                res_in = self._recv(hidden_shape, act_dtype).requires_grad_()
+                out = self.stage(x_in, res_in,
+                                 targets if self.is_last_stage else None)
+            if self.is_last_stage:
+                outstanding.append((x_in, res_in, out / n_micro, None))
+            else:
+                x_out, res_out = out
+                self._send(x_out, self.next_rank)
+                self._send(res_out, self.next_rank)
+                outstanding.append((x_in, res_in, x_out, res_out))
+            self.peak_live = max(self.peak_live, len(outstanding))
+
+        def backward():
+            nonlocal loss_total
+            x_in, res_in, a, b = outstanding.pop(0)
+            if self.is_last_stage:
+                loss_total = a.detach() if loss_total is None \
+                    else loss_total + a.detach()
+                a.backward()
+            else:
+                dx = self._recv_grad(a)
+                dres = self._recv_grad(b)
+                torch.autograd.backward((a, b), (dx, dres))
+            if not self.is_first_stage:
+                self._send(x_in.grad, self.prev_rank)
+                self._send(res_in.grad, self.prev_rank)
+
+        for i in range(warmup):
+            forward(micro_batches[i])
+        for i in range(warmup, n_micro):
+            forward(micro_batches[i])
+            backward()
+        while outstanding:
+            backward()
+        for w in self._send_works:
+            w.wait()
+        return loss_total
+
+
 class PPTrainer:
     """Minimal pipeline trainer: one LlamaStage per rank, per-stage flat
     param/grad store + fused AdamW, GPipe schedule per optimizer step.
