@@ -209,3 +209,50 @@ def test_trainer_dp2tp2_matches_dp2(tmp_path):
     # and each dp row matches the corresponding pure-DP rank
     assert load(2, 0) == pytest.approx(load(1, 0), abs=3e-2)
     assert load(2, 2) == pytest.approx(load(1, 1), abs=3e-2)
+
+
+def _vp_ce_worker(rank, world, port, outdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from trainingjob_operator_amd.parallel.vocab_parallel import (
+            vocab_parallel_cross_entropy,
+        )
+        torch.manual_seed(13)
+        T, V = 24, 64
+        full = torch.randn(T, V)
+        targets = torch.randint(0, V, (T,))
+        targets[3] = -100
+        shard = V // world
+        start = rank * shard
+        local = full[:, start:start + shard].clone().requires_grad_()
+        loss_vec = vocab_parallel_cross_entropy(local, targets, start,
+                                                start + shard)
+        n_valid = (targets != -100).sum()
+        loss = loss_vec.sum() / n_valid
+        loss.backward()
+
+        fref = full.clone().requires_grad_()
+        ref = torch.nn.functional.cross_entropy(fref, targets,
+                                                ignore_index=-100)
+        ref.backward()
+        assert torch.allclose(loss, ref, atol=1e-5), (loss, ref)
+        assert torch.allclose(local.grad,
+                              fref.grad[:, start:start + shard], atol=1e-5)
+        import json
+        with open(os.path.join(outdir, f"vp{rank}.json"), "w") as f:
+            json.dump(float(loss), f)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_vocab_parallel_ce_matches_full(tmp_path):
+    import json
+    port = _free_port()
+    mp.spawn(_vp_ce_worker, args=(2, port, str(tmp_path)), nprocs=2,
+             join=True)
+    a = json.load(open(os.path.join(str(tmp_path), "vp0.json")))
+    b = json.load(open(os.path.join(str(tmp_path), "vp1.json")))
+    assert a == pytest.approx(b)
