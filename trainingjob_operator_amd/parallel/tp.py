@@ -90,18 +90,25 @@ class ColumnParallelLinear(nn.Module):
     out_features/tp rows and computes its slice of Y."""
 
     def __init__(self, in_features: int, out_features: int, group=None,
-                 gather_output: bool = False, dtype=None):
+                 gather_output: bool = False, dtype=None,
+                 copy_input: bool = True):
         super().__init__()
         self.group = group
         n = _group_size(group)
         assert out_features % n == 0, (out_features, n)
         self.out_per_rank = out_features // n
         self.gather_output = gather_output
+        # copy_input=False under sequence parallelism: the seq all-gather
+        # seam already sums the input gradient across the group, so the
+        # f-op's backward all-reduce would double-count (Megatron drops it
+        # the same way when sequence_parallel is on).
+        self.copy_input = copy_input
         self.weight = nn.Parameter(
             torch.empty(self.out_per_rank, in_features, dtype=dtype))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        x = _CopyToTP.apply(x, self.group)
+        if self.copy_input:
+            x = _CopyToTP.apply(x, self.group)
         y = nn.functional.linear(x, self.weight)
         if self.gather_output:
             y = _GatherFromTP.apply(y, self.group)
@@ -114,17 +121,22 @@ class RowParallelLinear(nn.Module):
     products all-reduce into the full Y."""
 
     def __init__(self, in_features: int, out_features: int, group=None,
-                 dtype=None):
+                 dtype=None, reduce_output: bool = True):
         super().__init__()
         self.group = group
         n = _group_size(group)
         assert in_features % n == 0, (in_features, n)
         self.in_per_rank = in_features // n
+        # reduce_output=False under sequence parallelism: the partial sums
+        # flow into a seq reduce-scatter instead of the all-reduce.
+        self.reduce_output = reduce_output
         self.weight = nn.Parameter(
             torch.empty(out_features, self.in_per_rank, dtype=dtype))
 
     def forward(self, x_shard: torch.Tensor) -> torch.Tensor:
         partial = nn.functional.linear(x_shard, self.weight)
+        if not self.reduce_output:
+            return partial
         return _ReduceFromTP.apply(partial, self.group)
 
 

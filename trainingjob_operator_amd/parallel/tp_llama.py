@@ -27,7 +27,8 @@ from .tp import ColumnParallelLinear, RowParallelLinear, _group_size, shard_from
 
 
 class TPAttention(nn.Module):
-    def __init__(self, cfg: LlamaConfig, group=None):
+    def __init__(self, cfg: LlamaConfig, group=None,
+                 sequence_parallel: bool = False):
         super().__init__()
         self.cfg = cfg
         n = _group_size(group)
@@ -37,10 +38,15 @@ class TPAttention(nn.Module):
         self.n_local_kv = cfg.num_kv_heads // n
         H = cfg.hidden_size
         D = cfg.head_dim
-        self.q_proj = ColumnParallelLinear(H, cfg.num_heads * D, group)
-        self.k_proj = ColumnParallelLinear(H, cfg.num_kv_heads * D, group)
-        self.v_proj = ColumnParallelLinear(H, cfg.num_kv_heads * D, group)
-        self.o_proj = RowParallelLinear(cfg.num_heads * D, H, group)
+        ci = not sequence_parallel
+        self.q_proj = ColumnParallelLinear(H, cfg.num_heads * D, group,
+                                           copy_input=ci)
+        self.k_proj = ColumnParallelLinear(H, cfg.num_kv_heads * D, group,
+                                           copy_input=ci)
+        self.v_proj = ColumnParallelLinear(H, cfg.num_kv_heads * D, group,
+                                           copy_input=ci)
+        self.o_proj = RowParallelLinear(cfg.num_heads * D, H, group,
+                                        reduce_output=ci)
 
     def forward(self, x, inv_freq):
         B, S, _ = x.shape
@@ -57,14 +63,19 @@ class TPAttention(nn.Module):
 
 
 class TPMLP(nn.Module):
-    def __init__(self, cfg: LlamaConfig, group=None):
+    def __init__(self, cfg: LlamaConfig, group=None,
+                 sequence_parallel: bool = False):
         super().__init__()
+        ci = not sequence_parallel
         self.gate_proj = ColumnParallelLinear(cfg.hidden_size,
-                                              cfg.intermediate_size, group)
+                                              cfg.intermediate_size, group,
+                                              copy_input=ci)
         self.up_proj = ColumnParallelLinear(cfg.hidden_size,
-                                            cfg.intermediate_size, group)
+                                            cfg.intermediate_size, group,
+                                            copy_input=ci)
         self.down_proj = RowParallelLinear(cfg.intermediate_size,
-                                           cfg.hidden_size, group)
+                                           cfg.hidden_size, group,
+                                           reduce_output=ci)
 
     def forward(self, x):
         return self.down_proj(swiglu(self.gate_proj(x).contiguous(),
