@@ -1,0 +1,218 @@
+"""Expert parallelism: a dropless top-k MoE MLP with token dispatch over
+all-to-all.
+
+On RCCL the dispatch is ONE all_to_all_single each way — the collective the
+briefing's xGMI topology favors (7 direct p2p links per MI355X; all-to-all
+rides every link instead of serializing on a ring). Gloo (CPU tests) has no
+alltoall, so a pairwise isend/irecv emulation with identical semantics
+backs the same autograd seam.
+
+Experts are sharded over the EP group (contiguous blocks: rank r owns
+experts [r*E/ep, (r+1)*E/ep)); routing is exact/dropless — variable split
+sizes are exchanged first, no capacity factor, no token dropping.
+gloo-verified against a single-process MoE in tests/test_ep_gloo.py.
+"""
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from ..ops import swiglu
+from .tp import _group_size
+
+
+def _backend_has_alltoall(group) -> bool:
+    if not dist.is_initialized():
+        return False
+    return dist.get_backend(group) in ("nccl",)
+
+
+def _a2a_raw(inp: torch.Tensor, in_splits: List[int], out_splits: List[int],
+             group) -> torch.Tensor:
+    """all_to_all_single with uneven splits; P2P emulation where the
+    backend lacks alltoall (gloo)."""
+    n = _group_size(group)
+    out = torch.empty(sum(out_splits), *inp.shape[1:], dtype=inp.dtype,
+                      device=inp.device)
+    if n == 1:
+        out.copy_(inp)
+        return out
+    if _backend_has_alltoall(group):
+        dist.all_to_all_single(out, inp.contiguous(),
+                               output_split_sizes=out_splits,
+                               input_split_sizes=in_splits, group=group)
+        return out
+    rank = dist.get_rank(group)
+    in_chunks = list(inp.split(in_splits, dim=0))
+    out_chunks = list(out.split(out_splits, dim=0))
+    reqs = []
+    for peer in range(n):
+        if peer == rank:
+            out_chunks[peer].copy_(in_chunks[peer])
+            continue
+        if in_splits[peer] > 0:
+            reqs.append(dist.isend(in_chunks[peer].contiguous(), dst=peer,
+                                   group=group))
+        if out_splits[peer] > 0:
+            reqs.append(dist.irecv(out_chunks[peer], src=peer, group=group))
+    for r in reqs:
+        r.wait()
+    return out
+
+
+class _AllToAll(torch.autograd.Function):
+    """Differentiable token exchange: backward is the reverse exchange."""
+
+    @staticmethod
+    def forward(ctx, x, in_splits, out_splits, group):
+        ctx.in_splits = in_splits
+        ctx.out_splits = out_splits
+        ctx.group = group
+        return _a2a_raw(x, in_splits, out_splits, group)
+
+    @staticmethod
+    def backward(ctx, g):
+        return (_a2a_raw(g.contiguous(), ctx.out_splits, ctx.in_splits,
+                         ctx.group), None, None, None)
+
+
+class Expert(nn.Module):
+    """One SwiGLU expert (same shape family as the dense MLP)."""
+
+    def __init__(self, hidden: int, ff: int):
+        super().__init__()
+        self.gate_proj = nn.Linear(hidden, ff, bias=False)
+        self.up_proj = nn.Linear(hidden, ff, bias=False)
+        self.down_proj = nn.Linear(ff, hidden, bias=False)
+
+    def forward(self, x):
+        return self.down_proj(swiglu(self.gate_proj(x).contiguous(),
+                                     self.up_proj(x).contiguous()))
+
+
+class MoEMLP(nn.Module):
+    """Dropless top-k mixture-of-experts MLP, experts sharded over the EP
+    group. With ep_size == 1 it is a plain (single-process) MoE — the
+    reference the EP tests compare against."""
+
+    def __init__(self, hidden: int, ff: int, n_experts: int, top_k: int = 2,
+                 group=None):
+        super().__init__()
+        self.group = group
+        self.n_experts = n_experts
+        self.top_k = top_k
+        ep = _group_size(group)
+        assert n_experts % ep == 0, (n_experts, ep)
+        self.experts_per_rank = n_experts // ep
+        self.router = nn.Linear(hidden, n_experts, bias=False)
+        self.experts = nn.ModuleList(
+            Expert(hidden, ff) for _ in range(self.experts_per_rank))
+
+    def _expert_owner_splits(self, counts: torch.Tensor) -> List[int]:
+        """counts per expert [E] -> tokens destined per EP rank."""
+        per_rank = counts.view(-1, self.experts_per_rank).sum(dim=1)
+        return [int(c) for c in per_rank]
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        orig_shape = x.shape
+        H = orig_shape[-1]
+        xt = x.reshape(-1, H)
+        T = xt.shape[0]
+
+        logits = self.router(xt.float())
+        probs = torch.softmax(logits, dim=-1)
+        gates, idx = probs.topk(self.top_k, dim=-1)          # [T, k]
+        gates = gates / gates.sum(dim=-1, keepdim=True)
+
+        # flatten (token, k) pairs and sort by expert id — experts are
+        # contiguous per rank, so this is also sorted by destination rank
+        # AND by local expert within each destination
+        flat_e = idx.reshape(-1)                              # [T*k]
+        flat_t = (torch.arange(T, device=x.device)
+                  .repeat_interleave(self.top_k))
+        order = torch.argsort(flat_e, stable=True)
+        send_tokens = xt[flat_t[order]]
+        counts = torch.bincount(flat_e, minlength=self.n_experts)
+        in_splits = self._expert_owner_splits(counts)
+
+        # exchange split sizes: every rank needs how much each peer sends it
+        ep = _group_size(self.group)
+        if ep > 1:
+            # per-expert counts from every source rank (device tensors so
+            # the same code runs over gloo/cpu and rccl/gpu)
+            ecounts = [torch.zeros(self.n_experts, dtype=torch.long,
+                                   device=x.device) for _ in range(ep)]
+            dist.all_gather(ecounts, counts.to(torch.long), group=self.group)
+            r = dist.get_rank(self.group)
+            out_splits = [
+                int(c.view(ep, self.experts_per_rank)[r].sum())
+                for c in ecounts]
+            e0 = r * self.experts_per_rank
+            per_src = torch.stack(
+                [c[e0:e0 + self.experts_per_rank] for c in ecounts])
+        else:
+            out_splits = in_splits
+            per_src = counts.view(1, -1)
+
+        recv = _AllToAll.apply(send_tokens, in_splits, out_splits,
+                               self.group)
+
+        # recv layout: source-rank major, each source's block sorted by
+        # local expert. Permute to expert-major (contiguous run per
+        # expert), apply each expert to its slice, permute back — all via
+        # differentiable index ops.
+        sel_per_expert = []
+        src_offsets = torch.cat([
+            torch.zeros(1, dtype=torch.long, device=recv.device),
+            per_src.sum(dim=1).cumsum(0)[:-1]])
+        for le in range(self.experts_per_rank):
+            parts = []
+            for src in range(per_src.shape[0]):
+                base = int(src_offsets[src] + per_src[src, :le].sum())
+                parts.append(torch.arange(base,
+                                          base + int(per_src[src, le]),
+                                          device=recv.device))
+            sel_per_expert.append(
+                torch.cat(parts) if parts else
+                torch.empty(0, dtype=torch.long, device=recv.device))
+        perm = torch.cat(sel_per_expert)
+        grouped = recv[perm]
+        seg_sizes = [int(s.numel()) for s in sel_per_expert]
+        seg_out = []
+        off = 0
+        for le, size in enumerate(seg_sizes):
+            if size:
+                seg_out.append(self.experts[le](grouped[off:off + size]))
+            off += size
+        expert_out = (torch.cat(seg_out) if seg_out
+                      else grouped[:0])
+        inv_perm = torch.empty_like(perm)
+        inv_perm[perm] = torch.arange(perm.numel(), device=perm.device)
+        outputs = expert_out[inv_perm]
+
+        back = _AllToAll.apply(outputs, out_splits, in_splits, self.group)
+
+        # undo the sort, apply gates, combine top-k
+        inv = torch.empty_like(order)
+        inv[order] = torch.arange(order.numel(), device=order.device)
+        pair_out = back[inv]                                  # [T*k, H]
+        gated = pair_out * gates.reshape(-1, 1).to(pair_out.dtype)
+        y = gated.reshape(T, self.top_k, H).sum(dim=1)
+        return y.reshape(orig_shape)
+
+    @torch.no_grad()
+    def shard_from_full(self, full: "MoEMLP") -> None:
+        """Take this rank's expert block (and the replicated router) from a
+        single-process MoE with the full expert list."""
+        self.router.weight.copy_(full.router.weight)
+        r = dist.get_rank(self.group) if (
+            dist.is_initialized() and _group_size(self.group) > 1) else 0
+        base = r * self.experts_per_rank
+        for i, ex in enumerate(self.experts):
+            src = full.experts[base + i]
+            ex.gate_proj.weight.copy_(src.gate_proj.weight)
+            ex.up_proj.weight.copy_(src.up_proj.weight)
+            ex.down_proj.weight.copy_(src.down_proj.weight)
