@@ -112,3 +112,23 @@ def test_ep_two_ranks_match_single(tmp_path):
     a = json.load(open(os.path.join(str(tmp_path), "ep0.json")))
     b = json.load(open(os.path.join(str(tmp_path), "ep1.json")))
     assert a == a and b == b  # finite, both ranks asserted internally
+
+
+def test_moe_llama_trains():
+    from trainingjob_operator_amd.models.moe_llama import (
+        MOE_TINY, MoELlamaModel,
+    )
+    torch.manual_seed(2)
+    model = MoELlamaModel(MOE_TINY)
+    opt = torch.optim.AdamW(model.parameters(), lr=1e-3)
+    g = torch.Generator().manual_seed(3)
+    losses = []
+    for _ in range(8):
+        tokens = torch.randint(0, MOE_TINY.vocab_size, (2, 32), generator=g)
+        loss = model(tokens, tokens)
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        losses.append(loss.item())
+    assert all(l == l for l in losses)
+    assert losses[-1] < losses[0]
