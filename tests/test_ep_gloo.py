@@ -120,10 +120,10 @@ def test_moe_llama_trains():
     )
     torch.manual_seed(2)
     model = MoELlamaModel(MOE_TINY)
-    opt = torch.optim.AdamW(model.parameters(), lr=1e-3)
+    opt = torch.optim.AdamW(model.parameters(), lr=2e-3)
     g = torch.Generator().manual_seed(3)
     losses = []
-    for _ in range(8):
+    for _ in range(20):
         tokens = torch.randint(0, MOE_TINY.vocab_size, (2, 32), generator=g)
         loss = model(tokens, tokens)
         loss.backward()
@@ -131,4 +131,5 @@ def test_moe_llama_trains():
         opt.zero_grad()
         losses.append(loss.item())
     assert all(l == l for l in losses)
-    assert losses[-1] < losses[0]
+    # noisy routing at toy scale: compare window means
+    assert sum(losses[-5:]) / 5 < sum(losses[:5]) / 5
