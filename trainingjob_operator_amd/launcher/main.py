@@ -36,6 +36,9 @@ def main(argv=None) -> int:
     ap.add_argument("--ckpt-every", type=int, default=50)
     ap.add_argument("--log-every", type=int, default=10)
     ap.add_argument("--checkpoint-activations", action="store_true")
+    ap.add_argument("--tp", type=int, default=int(os.environ.get(
+        "TRAININGJOB_TP_SIZE", "1")),
+        help="tensor-parallel degree (world = dp x tp; tp ranks adjacent)")
     ap.add_argument("--metrics-port", type=int, default=int(os.environ.get(
         "TRAININGJOB_METRICS_PORT", "0")),
         help="expose Prometheus worker metrics (tokens/s, step time, loss)")
@@ -63,7 +66,8 @@ def main(argv=None) -> int:
     cfg = TrainConfig(
         model=args.model, micro_batch=args.micro_batch,
         grad_accum=args.grad_accum, seq_len=args.seq_len, lr=args.lr,
-        checkpoint_activations=args.checkpoint_activations)
+        checkpoint_activations=args.checkpoint_activations,
+        tp_size=args.tp)
     trainer = Trainer(cfg, ctx)
     ckpt = Checkpointer(args.ckpt_dir)
     resumed = ckpt.load_latest(trainer)
