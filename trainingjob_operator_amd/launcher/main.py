@@ -88,7 +88,9 @@ def main(argv=None) -> int:
     if args.metrics_port and ctx.is_rank0:
         from .worker_metrics import WorkerMetrics
         metrics = WorkerMetrics(args.metrics_port)
-    tokens_per_step = cfg.tokens_per_step_per_rank() * ctx.world_size
+    # tp ranks share one replica: whole-job tokens count DP replicas
+    tokens_per_step = (cfg.tokens_per_step_per_rank()
+                       * (ctx.world_size // max(args.tp, 1)))
     t_last = time.time()
     while trainer.step_count < args.steps and not stop_requested["flag"]:
         loss = trainer.train_step()
