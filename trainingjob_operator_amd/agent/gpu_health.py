@@ -18,7 +18,7 @@ import json
 import os
 import subprocess
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 FAKE_ENV = "AITJ_FAKE_GPU_HEALTH"
 

@@ -13,7 +13,6 @@ import argparse
 import logging
 import os
 import threading
-import time
 
 from ..api import constants as C
 from ..kube.client import KubeApi, RealKubeApi

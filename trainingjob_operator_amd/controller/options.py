@@ -6,8 +6,7 @@ Flag names kept where they map onto the reference's
 from __future__ import annotations
 
 import argparse
-from dataclasses import dataclass, field
-from typing import Optional
+from dataclasses import dataclass
 
 from ..policy.engine import CreatingFailurePolicy
 

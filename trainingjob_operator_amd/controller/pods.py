@@ -7,8 +7,7 @@ from __future__ import annotations
 
 import copy
 import logging
-from dataclasses import dataclass
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 from ..api import constants as C
 from ..api.types import (

@@ -6,7 +6,7 @@ import logging
 from typing import List
 
 from ..api import constants as C
-from ..api.types import AITrainingJob, Phase, gen_general_name
+from ..api.types import AITrainingJob, gen_general_name
 from ..kube import objects as ko
 from ..kube.client import ApiError, KubeApi
 from .envinject import ports_from_spec

@@ -7,11 +7,10 @@ blind Update() x5 on the main resource.
 from __future__ import annotations
 
 import logging
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 from ..api.types import (
-    AITrainingJob, CleanPodPolicy, ENDING_PHASES, EndingPolicy, Phase,
-    RestartScope,
+    AITrainingJob, CleanPodPolicy, ENDING_PHASES, Phase, RestartScope,
 )
 from ..kube import objects as ko
 from ..kube.client import ApiError, KubeApi

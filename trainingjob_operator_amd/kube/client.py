@@ -17,7 +17,7 @@ from __future__ import annotations
 import json
 import os
 import threading
-from typing import Any, Callable, Dict, Iterator, List, Optional
+from typing import Any, Dict, Iterator, List, Optional
 
 from ..api import constants as C
 

@@ -12,13 +12,12 @@ import copy
 import itertools
 import queue
 import threading
-import time
 from typing import Dict, Iterator, List, Optional, Tuple
 
 from ..api import constants as C
 from ..utils.k8stime import format_time
 from .client import ApiError, KubeApi
-from .objects import labels_of, matches_selector, meta
+from .objects import matches_selector, meta
 
 
 class FakeKubeApi(KubeApi):

@@ -6,7 +6,7 @@ needs.
 """
 from __future__ import annotations
 
-from typing import Any, Dict, List, Optional
+from typing import Dict, List, Optional
 
 from ..api import constants as C
 

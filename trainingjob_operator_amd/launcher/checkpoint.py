@@ -14,7 +14,6 @@ file lands (HBM -> host -> disk).
 from __future__ import annotations
 
 import os
-import shutil
 import threading
 import time
 from typing import Optional

@@ -48,8 +48,6 @@ def main(argv=None) -> int:
         level=logging.INFO,
         format="%(asctime)s %(name)s [%(levelname)s] %(message)s")
 
-    import torch
-
     from ..parallel import dist_ctx
     from ..training import TrainConfig, Trainer
     from .checkpoint import Checkpointer

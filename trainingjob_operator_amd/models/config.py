@@ -5,7 +5,7 @@ Llama-3-8B, 8 replicas x 1 amd.com/gpu on one MI355X node").
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 
 
 @dataclass

@@ -18,7 +18,6 @@ from contextlib import contextmanager
 from dataclasses import dataclass, field
 from typing import List, Optional
 
-import torch
 import torch.distributed as dist
 
 from .flat import FlatParamStore

@@ -14,7 +14,7 @@ gloo-verified against a single-process MoE in tests/test_ep_gloo.py.
 """
 from __future__ import annotations
 
-from typing import List, Optional
+from typing import List
 
 import torch
 import torch.distributed as dist

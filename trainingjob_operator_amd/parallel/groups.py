@@ -12,7 +12,7 @@ contiguity keeps NUMA/host affinity aligned too).
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import List, Optional
+from typing import Optional
 
 import torch.distributed as dist
 

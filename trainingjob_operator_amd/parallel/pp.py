@@ -14,7 +14,6 @@ so each hop sends/recvs the pair.
 """
 from __future__ import annotations
 
-from dataclasses import dataclass
 from typing import List, Optional
 
 import torch

@@ -14,8 +14,6 @@ tests/test_tp_gloo.py.
 """
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.distributed as dist
 import torch.nn as nn

@@ -15,8 +15,6 @@ is also how the gloo tests prove exact fwd/bwd parity
 """
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.nn as nn
 

@@ -14,8 +14,6 @@ with softmax_local = exp(local - lse) — no further communication.
 """
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.distributed as dist
 import torch.nn as nn

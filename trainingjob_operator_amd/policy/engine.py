@@ -16,8 +16,8 @@ Pods are plain dicts in corev1 shape.
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Tuple
+from dataclasses import dataclass
+from typing import Dict, List, Optional
 
 from ..api.constants import CONTAINER_PREFIX, ERROR_CONTAINER_STATUS
 from ..api.types import (
