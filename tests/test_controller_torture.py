@@ -122,3 +122,77 @@ def test_torture(seed):
     for _ in range(10):
         tc.sync_once(f"{NS}/torture")
         check_invariants(api, history)
+
+
+def make_multirole_job(rng):
+    return {
+        "apiVersion": C.API_VERSION, "kind": C.CRD_KIND,
+        "metadata": {"name": "torture", "namespace": NS},
+        "spec": {
+            "restartingExitCode": "137,128",
+            "completePolicy": "Any",
+            "replicaSpecs": {
+                "pserver": {
+                    "replicas": rng.randint(1, 2),
+                    "restartPolicy": "OnFailure",
+                    "restartScope": "Replica",
+                    "restartLimit": 3,
+                    "completePolicy": "None",
+                    "template": {"spec": {"containers": [{
+                        "name": "aitj-ps",
+                        "ports": [{"name": "aitj-p",
+                                   "containerPort": 6000}]}]}},
+                },
+                "worker": {
+                    "replicas": rng.randint(1, 3),
+                    "restartPolicy": rng.choice(["ExitCode", "Always"]),
+                    "restartScope": rng.choice(["All", "Pod"]),
+                    "restartLimit": 3,
+                    "template": {"spec": {"containers": [{
+                        "name": "aitj-w",
+                        "ports": [{"name": "aitj-p",
+                                   "containerPort": 5000}]}]}},
+                },
+            },
+        },
+    }
+
+
+@pytest.mark.parametrize("seed", [5, 55, 555])
+def test_torture_multirole(seed):
+    rng = random.Random(seed)
+    api = FakeKubeApi()
+    tc = TrainingJobController(api, OperatorOptions())
+    api.create_job(NS, make_multirole_job(rng))
+    history = {}
+
+    for step in range(100):
+        action = rng.random()
+        pods = api.pod_names(NS)
+        try:
+            if action < 0.4 and pods:
+                api.set_pod_phase(NS, rng.choice(pods), rng.choice(
+                    ["Running", "Running", "Succeeded"]))
+            elif action < 0.55 and pods:
+                api.set_pod_phase(NS, rng.choice(pods), "Failed",
+                                  exit_code=rng.choice([137, 1]))
+            elif action < 0.6:
+                api.set_node_ready("node-0", rng.random() < 0.85)
+        except KeyError:
+            pass
+        tc.sync_once(f"{NS}/torture")
+        job = AITrainingJob.from_dict(api.get_job(NS, "torture"))
+        assert job.status.phase in ALL_PHASES
+        # per-role restart counts monotonic
+        for rt in ("pserver", "worker"):
+            rc = job.status.restart_counts.get(rt, 0)
+            assert rc >= history.get(rt, 0)
+            history[rt] = rc
+        if history.get("terminal"):
+            assert job.status.phase == history["terminal"]
+        from trainingjob_operator_amd.api.types import ENDING_PHASES as EP
+        if job.status.phase in EP:
+            history["terminal"] = job.status.phase
+    api.set_node_ready("node-0", True)
+    for _ in range(6):
+        tc.sync_once(f"{NS}/torture")
