@@ -196,3 +196,28 @@ def test_1f1b_matches_reference_with_bounded_memory(tmp_path):
                                             f"r{r}.json")))["peak"]
              for r in range(2)}
     assert peaks == {0: 2, 1: 1}
+
+
+def _pp_launcher_worker(rank, world, port, ckdir):
+    os.environ.update({
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+        "RANK": str(rank), "WORLD_SIZE": str(world),
+        "LOCAL_RANK": str(rank),
+    })
+    from trainingjob_operator_amd.launcher.main import main
+    rc = main(["--model", "llama-tiny", "--steps", "4", "--seq-len", "32",
+               "--grad-accum", "2", "--micro-batch", "1",
+               "--ckpt-every", "2", "--log-every", "1",
+               "--ckpt-dir", ckdir, "--pp", str(world)])
+    assert rc == 0
+
+
+@pytest.mark.timeout(600)
+def test_launcher_pp_mode(tmp_path):
+    port = _free_port()
+    mp.spawn(_pp_launcher_worker, args=(2, port, str(tmp_path)), nprocs=2,
+             join=True)
+    # every stage checkpointed its slice
+    for r in range(2):
+        names = os.listdir(os.path.join(str(tmp_path), f"stage{r}"))
+        assert any(n.startswith("ckpt_step") for n in names), names

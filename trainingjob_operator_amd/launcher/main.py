@@ -39,6 +39,9 @@ def main(argv=None) -> int:
     ap.add_argument("--tp", type=int, default=int(os.environ.get(
         "TRAININGJOB_TP_SIZE", "1")),
         help="tensor-parallel degree (world = dp x tp; tp ranks adjacent)")
+    ap.add_argument("--pp", type=int, default=int(os.environ.get(
+        "TRAININGJOB_PP_SIZE", "1")),
+        help="pipeline-parallel stages (world must equal pp; rank = stage)")
     ap.add_argument("--metrics-port", type=int, default=int(os.environ.get(
         "TRAININGJOB_METRICS_PORT", "0")),
         help="expose Prometheus worker metrics (tokens/s, step time, loss)")
@@ -66,8 +69,18 @@ def main(argv=None) -> int:
         grad_accum=args.grad_accum, seq_len=args.seq_len, lr=args.lr,
         checkpoint_activations=args.checkpoint_activations,
         tp_size=args.tp)
-    trainer = Trainer(cfg, ctx)
-    ckpt = Checkpointer(args.ckpt_dir)
+    if args.pp > 1:
+        assert args.tp == 1, "DPxTPxPP grids are roadmap; use tp or pp"
+        assert ctx.world_size == args.pp, \
+            f"pp={args.pp} needs world=={args.pp} (got {ctx.world_size})"
+        from ..parallel.pp import PPTrainer
+        trainer = PPTrainer(cfg, stage_idx=ctx.rank, n_stages=args.pp)
+        # per-stage checkpoint streams (each rank owns its stage slice)
+        ckpt = Checkpointer(os.path.join(args.ckpt_dir,
+                                         f"stage{ctx.rank}"))
+    else:
+        trainer = Trainer(cfg, ctx)
+        ckpt = Checkpointer(args.ckpt_dir)
     resumed = ckpt.load_latest(trainer)
     if resumed is not None:
         log.info("resumed from step %d (world=%d)", resumed, ctx.world_size)
@@ -93,6 +106,10 @@ def main(argv=None) -> int:
     while trainer.step_count < args.steps and not stop_requested["flag"]:
         loss = trainer.train_step()
         step = trainer.step_count
+        if loss is None:  # non-last pipeline stages produce no loss
+            if step % args.ckpt_every == 0:
+                ckpt.save_async(trainer)
+            continue
         if step % args.log_every == 0 and ctx.is_rank0:
             now_t = time.time()
             tps = tokens_per_step * args.log_every / max(now_t - t_last, 1e-9)
@@ -104,11 +121,11 @@ def main(argv=None) -> int:
                         world_size=ctx.world_size)
             if metrics:
                 metrics.observe(step, loss.item(), tps)
-        if step % args.ckpt_every == 0 and ctx.is_rank0:
+        if step % args.ckpt_every == 0 and (ctx.is_rank0 or args.pp > 1):
             ckpt.save_async(trainer)
             trace.event("checkpoint", step=step)
 
-    if ctx.is_rank0:
+    if ctx.is_rank0 or args.pp > 1:
         ckpt.save_async(trainer, blocking=True)
     ckpt.wait()
     dist_ctx.destroy_process_group()
