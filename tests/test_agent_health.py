@@ -107,3 +107,45 @@ def test_gpu_unhealthy_drives_nodefail_restart(monkeypatch):
         api.get_job(NS, "j")).status.phase == Phase.RESTARTING
     tc.sync_once(f"{NS}/j")
     assert len(api.pod_names(NS)) == 2
+
+
+def test_probe_amd_smi_parsing(monkeypatch):
+    from trainingjob_operator_amd.agent import gpu_health as gh
+    monkeypatch.delenv(gh.FAKE_ENV, raising=False)
+
+    def fake_run_json(cmd, timeout=10.0):
+        if cmd[0] == "amd-smi":
+            return [
+                {"gpu": 0, "temperature": {"edge": {"value": 55}},
+                 "ecc": {"total_uncorrectable_count": 0}},
+                {"gpu": 1, "temperature": {"edge": {"value": 90}},
+                 "ecc": {"total_uncorrectable_count": 2}},
+            ]
+        return None
+
+    monkeypatch.setattr(gh, "_run_json", fake_run_json)
+    report = gh.probe(expected=2)
+    assert len(report.gpus) == 2
+    assert report.gpus[0].healthy
+    assert not report.gpus[1].healthy
+    assert "uncorrectable" in report.gpus[1].message
+    assert not report.healthy
+
+
+def test_probe_rocm_smi_fallback(monkeypatch):
+    from trainingjob_operator_amd.agent import gpu_health as gh
+    monkeypatch.delenv(gh.FAKE_ENV, raising=False)
+
+    def fake_run_json(cmd, timeout=10.0):
+        if cmd[0] == "amd-smi":
+            return None  # not installed
+        if cmd[0] == "rocm-smi":
+            return {"card0": {"Temperature (Sensor edge) (C)": "62.0"},
+                    "card1": {"Temperature (Sensor junction) (C)": "115.0"}}
+        return None
+
+    monkeypatch.setattr(gh, "_run_json", fake_run_json)
+    report = gh.probe()
+    assert len(report.gpus) == 2
+    assert report.gpus[0].healthy
+    assert not report.gpus[1].healthy  # 115C >= limit

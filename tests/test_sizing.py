@@ -65,3 +65,17 @@ def test_admission_rejects_oversized_model():
 
 def test_no_declaration_no_check():
     assert validate(_job()) == []
+
+
+def test_estimate_derives_shape_when_unspecified():
+    est = sizing.estimate_training_bytes(8_000_000_000)
+    # derived hidden/layers must land in a sane transformer regime
+    assert est.total_bytes > 8_000_000_000 * 16  # at least the state bytes
+    assert sizing.min_gpus_for(est) >= 1
+
+
+def test_declared_params_precedence():
+    j = _job(params_ann=123, model_ann="llama3-8b")
+    assert sizing.declared_params(j) == 123  # count wins over name
+    assert sizing.declared_params(_job(model_ann="nope")) is None
+    assert sizing.declared_params(_job()) is None

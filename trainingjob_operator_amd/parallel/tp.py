@@ -71,7 +71,6 @@ class _GatherFromTP(torch.autograd.Function):
         x = x.contiguous()
         parts = [torch.empty_like(x) for _ in range(n)]
         dist.all_gather(parts, x, group=group)
-        parts[dist.get_rank(group)] = x  # keep autograd-visible storage
         return torch.cat(parts, dim=-1)
 
     @staticmethod
