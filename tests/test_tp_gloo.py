@@ -305,3 +305,29 @@ def test_sp_llama_matches_unsharded(tmp_path):
     a = json.load(open(os.path.join(str(tmp_path), "sp0.json")))
     b = json.load(open(os.path.join(str(tmp_path), "sp1.json")))
     assert a == pytest.approx(b)
+
+
+def _tp_ckpt_worker(rank, world, port, ckdir):
+    os.environ.update({
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+        "RANK": str(rank), "WORLD_SIZE": str(world),
+        "LOCAL_RANK": str(rank),
+    })
+    from trainingjob_operator_amd.launcher.main import main
+    rc = main(["--model", "llama-tiny", "--steps", "4", "--seq-len", "32",
+               "--grad-accum", "1", "--micro-batch", "1",
+               "--ckpt-every", "2", "--log-every", "1",
+               "--ckpt-dir", ckdir, "--tp", str(world)])
+    assert rc == 0
+
+
+@pytest.mark.timeout(600)
+def test_launcher_tp_checkpoints_per_shard(tmp_path):
+    port = _free_port()
+    mp.spawn(_tp_ckpt_worker, args=(2, port, str(tmp_path)), nprocs=2,
+             join=True)
+    # one checkpoint stream PER TP SHARD (rank0-only saving would corrupt
+    # resumes by restoring rank0's shard everywhere)
+    for r in range(2):
+        names = os.listdir(os.path.join(str(tmp_path), f"tp{r}"))
+        assert any(n.startswith("ckpt_step") for n in names), (r, names)

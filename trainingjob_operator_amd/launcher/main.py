@@ -22,6 +22,16 @@ import time
 log = logging.getLogger("launcher")
 
 
+def _saves_ckpt(args, ctx, trainer) -> bool:
+    """Who writes checkpoints: DP -> rank 0; TP -> the dp_rank==0 replica
+    (every tp shard, its own stream); PP -> every stage."""
+    if args.pp > 1:
+        return True
+    if args.tp > 1:
+        return trainer.topo.dp_rank == 0
+    return ctx.is_rank0
+
+
 def main(argv=None) -> int:
     ap = argparse.ArgumentParser()
     ap.add_argument("--model", default="llama3-8b")
@@ -80,7 +90,13 @@ def main(argv=None) -> int:
                                          f"stage{ctx.rank}"))
     else:
         trainer = Trainer(cfg, ctx)
-        ckpt = Checkpointer(args.ckpt_dir)
+        if args.tp > 1:
+            # each tp rank owns a distinct shard: per-tp-rank checkpoint
+            # streams, written by the dp_rank==0 replica
+            ckpt = Checkpointer(os.path.join(
+                args.ckpt_dir, f"tp{trainer.topo.tp_rank}"))
+        else:
+            ckpt = Checkpointer(args.ckpt_dir)
     resumed = ckpt.load_latest(trainer)
     if resumed is not None:
         log.info("resumed from step %d (world=%d)", resumed, ctx.world_size)
@@ -121,11 +137,11 @@ def main(argv=None) -> int:
                         world_size=ctx.world_size)
             if metrics:
                 metrics.observe(step, loss.item(), tps)
-        if step % args.ckpt_every == 0 and (ctx.is_rank0 or args.pp > 1):
+        if step % args.ckpt_every == 0 and _saves_ckpt(args, ctx, trainer):
             ckpt.save_async(trainer)
             trace.event("checkpoint", step=step)
 
-    if ctx.is_rank0 or args.pp > 1:
+    if _saves_ckpt(args, ctx, trainer):
         ckpt.save_async(trainer, blocking=True)
     ckpt.wait()
     dist_ctx.destroy_process_group()
