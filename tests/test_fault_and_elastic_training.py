@@ -48,3 +48,32 @@ def test_elastic_resize_resumes_at_new_world(tmp_path):
     assert codes == [0]
     assert sup.report.final_step >= 10
     assert sup._epoch == 1
+
+
+@pytest.mark.timeout(600)
+def test_fault_then_resize_sequence(tmp_path):
+    """BASELINE configs 3+4 combined: train at world 2, SIGKILL -> restart,
+    then elastic resize 2 -> 1, finishing from the same checkpoints."""
+    sup = LocalSupervisor(world_size=2, ckpt_dir=str(tmp_path),
+                          total_steps=14, master_port=29871)
+    sup.start_world()
+    deadline = time.monotonic() + 120
+    while sup.report.final_step < 3 and time.monotonic() < deadline:
+        time.sleep(0.2)
+    assert sup.report.final_step >= 3
+
+    # fault: kill rank 0, world restarts at same size
+    kill_t = sup.kill_rank(0)
+    sup.restart_world(kill_t)
+    deadline = time.monotonic() + 120
+    while sup.report.final_step < 7 and time.monotonic() < deadline:
+        time.sleep(0.2)
+    assert sup.report.restarts == 1
+    assert sup.report.p50_rejoin is not None
+
+    # elastic resize down to 1 and run to completion
+    sup.restart_world(time.monotonic(), new_world_size=1)
+    codes = sup.wait(timeout=300)
+    assert codes == [0]
+    assert sup.report.final_step >= 14
+    assert sup._epoch == 1
