@@ -177,3 +177,16 @@ def test_validation_catches_errors():
 def test_gen_general_name():
     # reference: trainingjob.go:12-15
     assert gen_general_name("job", "trainer", 0) == "job-trainer-0"
+
+
+def test_k8stime_roundtrip():
+    from trainingjob_operator_amd.utils.k8stime import format_time, parse_time
+    t = 1700000000.0
+    assert parse_time(format_time(t)) == t
+    assert parse_time("2023-11-14T22:13:20Z") == t
+    assert parse_time("2023-11-14T22:13:20.500Z") == t + 0.5
+    assert parse_time("2023-11-14T23:13:20+01:00") == t
+    assert parse_time("2023-11-14T21:13:20-01:00") == t
+    assert parse_time(None) is None
+    assert parse_time("") is None
+    assert parse_time(12.5) == 12.5

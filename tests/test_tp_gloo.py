@@ -331,3 +331,41 @@ def test_launcher_tp_checkpoints_per_shard(tmp_path):
     for r in range(2):
         names = os.listdir(os.path.join(str(tmp_path), f"tp{r}"))
         assert any(n.startswith("ckpt_step") for n in names), (r, names)
+
+
+def _grid_worker(rank, world, port, outdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        import json
+        from trainingjob_operator_amd.parallel.groups import build_grid
+        topo = build_grid(tp_size=2, pp_size=2)  # world 8 -> dp2 x pp2 x tp2
+        assert (topo.dp_size, topo.pp_size, topo.tp_size) == (2, 2, 2)
+        assert topo.rank == ((topo.dp_rank * 2 + topo.pp_rank) * 2
+                             + topo.tp_rank)
+        # each axis group sums exactly its members' ranks
+        import torch as t
+        for group, size in ((topo.tp_group, 2), (topo.pp_group, 2),
+                            (topo.dp_group, 2)):
+            v = t.tensor([float(rank)])
+            dist.all_reduce(v, group=group)
+            members = dist.get_process_group_ranks(group)
+            assert len(members) == size and rank in members
+            assert v.item() == float(sum(members)), (rank, members, v)
+        with open(os.path.join(outdir, f"g{rank}.json"), "w") as f:
+            json.dump([topo.dp_rank, topo.pp_rank, topo.tp_rank], f)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_grid_topology_8ranks(tmp_path):
+    import json
+    port = _free_port()
+    mp.spawn(_grid_worker, args=(8, port, str(tmp_path)), nprocs=8,
+             join=True)
+    coords = {r: tuple(json.load(open(os.path.join(str(tmp_path),
+                                                   f"g{r}.json"))))
+              for r in range(8)}
+    assert len(set(coords.values())) == 8  # bijective rank <-> coordinate

@@ -62,3 +62,69 @@ def build_topology(tp_size: int = 1) -> ParallelTopology:
     # tp_size == 1: dp_group None means the default (whole-world) group
     return ParallelTopology(world, rank, dp_size, tp_size, dp_rank, tp_rank,
                             dp_group, tp_group)
+
+
+@dataclass
+class GridTopology:
+    """General DP x PP x TP grid (EP reuses one of the axes — usually EP
+    == DP for MoE — so three axes suffice). Rank layout, innermost last:
+
+        rank = ((dp_rank * pp_size) + pp_rank) * tp_size + tp_rank
+
+    tp ranks stay contiguous (adjacent GPUs over xGMI); a pp stage's
+    neighbors are +/- tp_size apart (still one xGMI hop on an 8-GPU node).
+    """
+    world_size: int
+    rank: int
+    dp_size: int
+    pp_size: int
+    tp_size: int
+    dp_rank: int
+    pp_rank: int
+    tp_rank: int
+    dp_group: Optional[object]
+    pp_group: Optional[object]
+    tp_group: Optional[object]
+
+
+def build_grid(tp_size: int = 1, pp_size: int = 1) -> GridTopology:
+    """Split the world into dp x pp x tp; every rank must call with the
+    same sizes (new_group is collective, created in a fixed global order).
+    """
+    if not dist.is_initialized():
+        return GridTopology(1, 0, 1, max(pp_size, 1), max(tp_size, 1),
+                            0, 0, 0, None, None, None)
+    world = dist.get_world_size()
+    rank = dist.get_rank()
+    if world % (tp_size * pp_size) != 0:
+        raise ValueError(
+            f"world {world} not divisible by tp*pp={tp_size * pp_size}")
+    dp_size = world // (tp_size * pp_size)
+    tp_rank = rank % tp_size
+    pp_rank = (rank // tp_size) % pp_size
+    dp_rank = rank // (tp_size * pp_size)
+
+    def _mk(axis_groups):
+        mine = None
+        for ranks in axis_groups:
+            g = dist.new_group(ranks)
+            if rank in ranks:
+                mine = g
+        return mine
+
+    tp_group = pp_group = dp_group = None
+    if tp_size > 1:
+        tp_group = _mk([[(d * pp_size + p) * tp_size + t
+                         for t in range(tp_size)]
+                        for d in range(dp_size) for p in range(pp_size)])
+    if pp_size > 1:
+        pp_group = _mk([[(d * pp_size + p) * tp_size + t
+                         for p in range(pp_size)]
+                        for d in range(dp_size) for t in range(tp_size)])
+    if dp_size > 1 and (tp_size > 1 or pp_size > 1):
+        dp_group = _mk([[(d * pp_size + p) * tp_size + t
+                         for d in range(dp_size)]
+                        for p in range(pp_size) for t in range(tp_size)])
+    return GridTopology(world, rank, dp_size, pp_size, tp_size,
+                        dp_rank, pp_rank, tp_rank,
+                        dp_group, pp_group, tp_group)
