@@ -278,7 +278,7 @@ class PPTrainer:
     (training.py); DPxPP grids come from parallel/groups.py in round 2."""
 
     def __init__(self, cfg, stage_idx: int, n_stages: int, device=None,
-                 act_dtype=None):
+                 act_dtype=None, schedule: str = "1f1b"):
         from ..models.config import CONFIGS
         from ..parallel.flat import FlatParamStore
         from ..optim import FlatAdamW
@@ -293,8 +293,10 @@ class PPTrainer:
         # the flat store keeps parameters in bf16; boundary activations
         # travel in the same dtype
         self.act_dtype = act_dtype or torch.bfloat16
-        self.sched = GPipeSchedule(self.stage, stage_idx, n_stages,
-                                   device=self.device)
+        sched_cls = {"gpipe": GPipeSchedule,
+                     "1f1b": OneFOneBSchedule}[schedule]
+        self.sched = sched_cls(self.stage, stage_idx, n_stages,
+                               device=self.device)
         self.store = FlatParamStore(self.stage, device=self.device)
         self.opt = FlatAdamW(self.store, lr=cfg.lr, betas=cfg.betas,
                              weight_decay=cfg.weight_decay,
