@@ -223,26 +223,27 @@ def test_launcher_pp_mode(tmp_path):
         assert any(n.startswith("ckpt_step") for n in names), names
 
 
+def _gpipe_sched_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from trainingjob_operator_amd.parallel.pp import PPTrainer
+        from trainingjob_operator_amd.training import TrainConfig
+        cfg = TrainConfig(model="llama-tiny", micro_batch=1,
+                          grad_accum=3, seq_len=32, lr=1e-3)
+        tr = PPTrainer(cfg, stage_idx=rank, n_stages=world,
+                       schedule="gpipe")
+        for _ in range(3):
+            tr.train_step()
+        assert tr.step_count == 3
+    finally:
+        dist.destroy_process_group()
+
+
 @pytest.mark.timeout(600)
-def test_pp_trainer_gpipe_schedule(tmp_path):
-    """Both schedules drive the trainer; GPipe selected explicitly."""
-    port = _free_port()
-
-    def worker(rank, world, port, outdir):
-        os.environ["MASTER_ADDR"] = "127.0.0.1"
-        os.environ["MASTER_PORT"] = str(port)
-        dist.init_process_group("gloo", rank=rank, world_size=world)
-        try:
-            from trainingjob_operator_amd.parallel.pp import PPTrainer
-            from trainingjob_operator_amd.training import TrainConfig
-            cfg = TrainConfig(model="llama-tiny", micro_batch=1,
-                              grad_accum=3, seq_len=32, lr=1e-3)
-            tr = PPTrainer(cfg, stage_idx=rank, n_stages=world,
-                           schedule="gpipe")
-            for _ in range(3):
-                tr.train_step()
-            assert tr.step_count == 3
-        finally:
-            dist.destroy_process_group()
-
-    mp.spawn(worker, args=(2, port, str(tmp_path)), nprocs=2, join=True)
+def test_pp_trainer_gpipe_schedule():
+    """Both schedules drive the trainer; GPipe selected explicitly
+    (1F1B is the PPTrainer default, covered by test_pp_trainer_trains)."""
+    mp.spawn(_gpipe_sched_worker, args=(2, _free_port()), nprocs=2,
+             join=True)
