@@ -52,6 +52,10 @@ def main(argv=None) -> int:
     ap.add_argument("--pp", type=int, default=int(os.environ.get(
         "TRAININGJOB_PP_SIZE", "1")),
         help="pipeline-parallel stages (world must equal pp; rank = stage)")
+    ap.add_argument("--pp-schedule", choices=("1f1b", "gpipe"),
+                    default=os.environ.get("TRAININGJOB_PP_SCHEDULE", "1f1b"),
+                    help="pipeline schedule: 1f1b bounds live microbatches "
+                         "per stage; gpipe holds all of them")
     ap.add_argument("--metrics-port", type=int, default=int(os.environ.get(
         "TRAININGJOB_METRICS_PORT", "0")),
         help="expose Prometheus worker metrics (tokens/s, step time, loss)")
@@ -84,7 +88,8 @@ def main(argv=None) -> int:
         assert ctx.world_size == args.pp, \
             f"pp={args.pp} needs world=={args.pp} (got {ctx.world_size})"
         from ..parallel.pp import PPTrainer
-        trainer = PPTrainer(cfg, stage_idx=ctx.rank, n_stages=args.pp)
+        trainer = PPTrainer(cfg, stage_idx=ctx.rank, n_stages=args.pp,
+                            schedule=args.pp_schedule)
         # per-stage checkpoint streams (each rank owns its stage slice)
         ckpt = Checkpointer(os.path.join(args.ckpt_dir,
                                          f"stage{ctx.rank}"))
