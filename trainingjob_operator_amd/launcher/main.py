@@ -26,7 +26,7 @@ def _saves_ckpt(args, ctx, trainer) -> bool:
     """Who writes checkpoints: DP -> rank 0; TP -> the dp_rank==0 replica
     (every tp shard, its own stream); PP -> every stage."""
     if args.pp > 1:
-        return True
+        return trainer.dp_rank == 0   # every stage, dp-replica 0 only
     if args.tp > 1:
         return trainer.topo.dp_rank == 0
     return ctx.is_rank0
@@ -85,14 +85,16 @@ def main(argv=None) -> int:
         tp_size=args.tp)
     if args.pp > 1:
         assert args.tp == 1, "DPxTPxPP grids are roadmap; use tp or pp"
-        assert ctx.world_size == args.pp, \
-            f"pp={args.pp} needs world=={args.pp} (got {ctx.world_size})"
+        assert ctx.world_size % args.pp == 0, \
+            f"world {ctx.world_size} not divisible by pp={args.pp}"
+        from ..parallel.groups import build_grid
         from ..parallel.pp import PPTrainer
-        trainer = PPTrainer(cfg, stage_idx=ctx.rank, n_stages=args.pp,
-                            schedule=args.pp_schedule)
-        # per-stage checkpoint streams (each rank owns its stage slice)
+        grid = build_grid(tp_size=1, pp_size=args.pp)
+        trainer = PPTrainer(cfg, schedule=args.pp_schedule, grid=grid)
+        # per-stage checkpoint streams (each stage slice its own stream,
+        # written by the dp_rank==0 replica — see _saves_ckpt)
         ckpt = Checkpointer(os.path.join(args.ckpt_dir,
-                                         f"stage{ctx.rank}"))
+                                         f"stage{grid.pp_rank}"))
     else:
         trainer = Trainer(cfg, ctx)
         if args.tp > 1:

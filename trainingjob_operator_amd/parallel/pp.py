@@ -119,14 +119,17 @@ class GPipeSchedule:
     (stages-1)/micro_batches. (1F1B is a round-2 refinement.)"""
 
     def __init__(self, stage: LlamaStage, stage_idx: int, n_stages: int,
-                 group=None, device=None):
+                 group=None, device=None,
+                 pp_ranks: Optional[List[int]] = None):
         self.stage = stage
         self.idx = stage_idx
         self.n = n_stages
         self.group = group
         self.device = device or "cpu"
-        ranks = (list(range(dist.get_world_size(group)))
-                 if dist.is_initialized() else [0])
+        # GLOBAL ranks of this pipeline, ordered by stage (p2p dst/src are
+        # global ranks even with a subgroup). Default: pure-PP world where
+        # rank == stage.
+        ranks = pp_ranks if pp_ranks is not None else list(range(n_stages))
         self.prev_rank = ranks[stage_idx - 1] if stage_idx > 0 else None
         self.next_rank = ranks[stage_idx + 1] \
             if stage_idx < n_stages - 1 else None
@@ -272,13 +275,21 @@ class OneFOneBSchedule(GPipeSchedule):
 
 
 class PPTrainer:
-    """Minimal pipeline trainer: one LlamaStage per rank, per-stage flat
-    param/grad store + fused AdamW, GPipe schedule per optimizer step.
+    """Pipeline trainer: one LlamaStage per rank, per-stage flat param/grad
+    store + fused AdamW, a 1F1B (or GPipe) schedule per optimizer step.
     Composes with the launcher's env bootstrap exactly like the DP Trainer
-    (training.py); DPxPP grids come from parallel/groups.py in round 2."""
+    (training.py).
 
-    def __init__(self, cfg, stage_idx: int, n_stages: int, device=None,
-                 act_dtype=None, schedule: str = "1f1b"):
+    Pure PP: pass stage_idx/n_stages (rank == stage, world == pp).
+    DP x PP: pass a GridTopology from parallel/groups.build_grid(pp_size=N)
+    — each dp replica runs its own pipeline over distinct data, and each
+    stage all-reduces its flat grad across its dp peers (one bucket: the
+    whole stage slice) before the optimizer step, exactly the DDP seam of
+    ddp.py collapsed to a single flat buffer."""
+
+    def __init__(self, cfg, stage_idx: Optional[int] = None,
+                 n_stages: Optional[int] = None, device=None,
+                 act_dtype=None, schedule: str = "1f1b", grid=None):
         from ..models.config import CONFIGS
         from ..parallel.flat import FlatParamStore
         from ..optim import FlatAdamW
@@ -287,6 +298,19 @@ class PPTrainer:
         self.cfg = cfg
         mcfg = CONFIGS[cfg.model]
         self.device = torch.device(device or "cpu")
+        if grid is not None:
+            assert grid.tp_size == 1, "PPxTP composition is roadmap"
+            stage_idx, n_stages = grid.pp_rank, grid.pp_size
+            pp_ranks = [(grid.dp_rank * grid.pp_size + p) * grid.tp_size
+                        + grid.tp_rank for p in range(grid.pp_size)]
+            self.dp_size, self.dp_rank = grid.dp_size, grid.dp_rank
+            self.dp_group = grid.dp_group
+        else:
+            assert stage_idx is not None and n_stages is not None
+            pp_ranks = None
+            self.dp_size, self.dp_rank, self.dp_group = 1, 0, None
+        self.grid = grid
+        # identical init on every dp replica of a stage (same seed)
         torch.manual_seed(cfg.seed)
         self.stage = LlamaStage.from_config(mcfg, stage_idx, n_stages,
                                             device=self.device)
@@ -296,20 +320,28 @@ class PPTrainer:
         sched_cls = {"gpipe": GPipeSchedule,
                      "1f1b": OneFOneBSchedule}[schedule]
         self.sched = sched_cls(self.stage, stage_idx, n_stages,
-                               device=self.device)
+                               device=self.device, pp_ranks=pp_ranks)
         self.store = FlatParamStore(self.stage, device=self.device)
         self.opt = FlatAdamW(self.store, lr=cfg.lr, betas=cfg.betas,
                              weight_decay=cfg.weight_decay,
                              clip_grad_norm=cfg.clip_grad_norm)
-        # every pp rank draws the same data; only the roles that need
-        # tokens/targets use them
-        self.data = synthetic_batches(cfg, self.device, rank=0)
+        # every pp rank of a replica draws the same stream; replicas draw
+        # DISTINCT streams (dp_rank-keyed), like the DP Trainer
+        self.data = synthetic_batches(cfg, self.device, rank=self.dp_rank)
         self.hidden_shape = (cfg.micro_batch, cfg.seq_len, mcfg.hidden_size)
         self.step_count = 0
 
     def train_step(self):
         micros = [next(self.data) for _ in range(self.cfg.grad_accum)]
         loss = self.sched.step(micros, self.hidden_shape, self.act_dtype)
+        if self.dp_size > 1:
+            # stage-peer gradient seam: SUM with 1/world pre-scale (gloo
+            # has no AVG; RCCL path matches ddp.py's convention)
+            self.store.flat_grad.mul_(1.0 / self.dp_size)
+            dist.all_reduce(self.store.flat_grad, group=self.dp_group)
+            if loss is not None:
+                loss = loss / self.dp_size
+                dist.all_reduce(loss, group=self.dp_group)
         self.opt.step()
         self.opt.zero_grad()
         self.step_count += 1
