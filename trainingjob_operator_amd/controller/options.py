@@ -30,6 +30,10 @@ class OperatorOptions:
     renew_deadline: float = 10.0
     retry_period: float = 3.0
     metrics_port: int = 0            # 0 == disabled
+    # EdlPolicy=Auto control loop (new semantics; the reference declared
+    # the fields but never read them — SURVEY.md §C15)
+    elastic_unschedulable_grace: float = 60.0
+    elastic_scaleup_interval: float = 300.0
 
     def creating_failure_policy(self) -> CreatingFailurePolicy:
         return CreatingFailurePolicy(
@@ -60,6 +64,15 @@ class OperatorOptions:
         ap.add_argument("--no-leader-elect", dest="leader_elect",
                         action="store_false")
         ap.add_argument("--metrics-port", type=int, default=d.metrics_port)
+        ap.add_argument("--elastic-unschedulable-grace", type=float,
+                        default=d.elastic_unschedulable_grace,
+                        help="seconds a pod may sit unschedulable before an "
+                             "EdlPolicy=Auto job scales down toward "
+                             "minReplicas")
+        ap.add_argument("--elastic-scaleup-interval", type=float,
+                        default=d.elastic_scaleup_interval,
+                        help="seconds between EdlPolicy=Auto +1 scale-up "
+                             "probes toward maxReplicas")
 
     @classmethod
     def from_args(cls, args: argparse.Namespace) -> "OperatorOptions":
@@ -72,4 +85,6 @@ class OperatorOptions:
             enable_creating_failed=args.enable_creating_failed,
             gc_period=args.gc_period, leader_elect=args.leader_elect,
             metrics_port=args.metrics_port,
+            elastic_unschedulable_grace=args.elastic_unschedulable_grace,
+            elastic_scaleup_interval=args.elastic_scaleup_interval,
         )

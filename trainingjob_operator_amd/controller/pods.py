@@ -23,6 +23,10 @@ log = logging.getLogger(__name__)
 
 EPOCH_ANNOTATION = f"{C.CRD_GROUP}/rendezvous-epoch"
 WORLD_SIZE_ANNOTATION = f"{C.CRD_GROUP}/world-size"
+# EdlPolicy=Auto state, per replica type (suffix "-<rt>"): the controller's
+# chosen effective world size and the last auto-resize timestamp
+TARGET_ANNOTATION = f"{C.CRD_GROUP}/target-replicas"
+LAST_RESIZE_ANNOTATION = f"{C.CRD_GROUP}/last-resize"
 
 
 def filter_pods_for_replica_type(pods: List[dict], rt: str) -> List[dict]:
@@ -176,6 +180,23 @@ class PodReconciler:
         job.status.replica_statuses[rtype] = engine.ReplicaStatus()
         job.status.restart_counts.setdefault(rtype, 0)
 
+        # --- EdlPolicy=Auto: controller-chosen world size in
+        # [minReplicas, maxReplicas] (new semantics).
+        # Scale-down when pods sit unschedulable past the grace period;
+        # +1 scale-up probes toward maxReplicas once everything runs and
+        # the probe interval elapsed. The chosen size is sticky via the
+        # target-replicas annotation; the world-size stale check below
+        # then drives the actual restart-at-new-size dance.
+        if spec.edl_policy == EdlPolicy.AUTO:
+            effective = self._auto_target(job, rtype, spec,
+                                          replica_pods, now)
+            if effective != replicas:
+                # in-memory only: spec is never persisted, but this sync's
+                # pod creation/env injection and the services reconciler
+                # must all see the effective size
+                spec.replicas = effective
+                replicas = effective
+
         # --- elastic resize detection (new semantics) ---
         if spec.edl_policy and spec.edl_policy != EdlPolicy.NEVER:
             stale = [p for p in replica_pods
@@ -294,6 +315,77 @@ class PodReconciler:
     def _count(self, job: AITrainingJob, rtype: str, pods: List[dict]):
         job.status.replica_statuses[rtype] = engine.count_replica_statuses(
             job.status.restart_counts.get(rtype, 0), pods)
+
+    def _auto_target(self, job: AITrainingJob, rtype: str,
+                     spec: ReplicaSpec, replica_pods: List[dict],
+                     now: float) -> int:
+        """EdlPolicy=Auto control loop: returns the effective replica
+        count for this sync and records it (plus the last-resize time) in
+        job annotations so it is sticky across syncs and restarts."""
+        from ..utils.k8stime import format_time, parse_time
+        rt = rtype.lower()
+        base = spec.replicas or 0
+        lo = spec.min_replicas if spec.min_replicas is not None else base
+        hi = spec.max_replicas if spec.max_replicas is not None else base
+        key = f"{TARGET_ANNOTATION}-{rt}"
+        tkey = f"{LAST_RESIZE_ANNOTATION}-{rt}"
+        try:
+            target = int(job.annotations[key])
+        except (KeyError, ValueError):
+            target = base
+        target = max(lo, min(hi, target))
+
+        def resize(new: int, why: str) -> int:
+            job.annotations[key] = str(new)
+            job.annotations[tkey] = format_time(now)
+            msg = (f"EdlPolicy=Auto: {rt} {target} -> {new} replicas "
+                   f"({why})")
+            log.info("%s: %s", job.key, msg)
+            if self.recorder:
+                self.recorder.event(job, "Normal", "ElasticScale", msg)
+            return new
+
+        grace = self.options.elastic_unschedulable_grace
+        stuck = [p for p in replica_pods
+                 if (s := self._unschedulable_since(p)) is not None
+                 and now - s >= grace]
+        if stuck and target > lo:
+            scheduled = len([p for p in replica_pods if ko.pod_node(p)])
+            new = max(lo, min(target - 1, scheduled))
+            return resize(new, f"{len(stuck)} pods unschedulable "
+                               f">{grace:.0f}s")
+
+        if not stuck and target < hi:
+            last = parse_time(job.annotations.get(tkey))
+            if last is None:
+                # start the probe clock on first sight, don't scale yet
+                job.annotations[tkey] = format_time(now)
+            else:
+                running = len([p for p in replica_pods
+                               if ko.pod_phase(p) == "Running"])
+                if (running >= target
+                        and now - last >= self.options
+                        .elastic_scaleup_interval):
+                    return resize(min(hi, target + 1), "capacity probe")
+
+        job.annotations[key] = str(target)
+        return target
+
+    @staticmethod
+    def _unschedulable_since(pod: dict):
+        """Epoch seconds since the scheduler marked the pod unschedulable
+        (PodScheduled=False while Pending and unbound), else None."""
+        if ko.pod_phase(pod) != "Pending" or ko.pod_node(pod):
+            return None
+        from ..utils.k8stime import parse_time
+        for cond in (pod.get("status") or {}).get("conditions") or []:
+            if cond.get("type") == "PodScheduled" and \
+                    cond.get("status") == "False":
+                return (parse_time(cond.get("lastTransitionTime"))
+                        or parse_time((pod.get("metadata") or {})
+                                      .get("creationTimestamp"))
+                        or 0.0)
+        return None
 
     @staticmethod
     def _scheduling_message(pod: dict) -> str:

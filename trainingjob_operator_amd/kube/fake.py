@@ -344,6 +344,25 @@ class FakeKubeApi(KubeApi):
             if ns == namespace:
                 self.set_pod_phase(ns, name, phase, **kw)
 
+    def set_pod_unschedulable(self, namespace: str, name: str,
+                              message: str = "0/8 nodes are available",
+                              since: Optional[float] = None):
+        """Scheduler-style Unschedulable: Pending, unbound, with a
+        PodScheduled=False condition stamped at `since` (default now)."""
+        with self._lock:
+            pod = self._pod(namespace, name)
+            pod["spec"].pop("nodeName", None)
+            status = pod.setdefault("status", {})
+            status["phase"] = "Pending"
+            conds = [c for c in status.get("conditions") or []
+                     if c.get("type") != "PodScheduled"]
+            conds.append({"type": "PodScheduled", "status": "False",
+                          "reason": "Unschedulable", "message": message,
+                          "lastTransitionTime": format_time(since)})
+            status["conditions"] = conds
+            pod["metadata"]["resourceVersion"] = str(next(self._rv))
+            self._emit("MODIFIED", "pod", pod)
+
     def bind_pod(self, namespace: str, name: str, node: str):
         with self._lock:
             pod = self._pod(namespace, name)
