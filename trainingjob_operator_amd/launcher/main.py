@@ -27,6 +27,8 @@ def _saves_ckpt(args, ctx, trainer) -> bool:
     (every tp shard, its own stream); PP -> every stage."""
     if args.pp > 1:
         return trainer.dp_rank == 0   # every stage, dp-replica 0 only
+    if args.ep:
+        return trainer.topo.edp_rank == 0  # every shard, edp-replica 0
     if args.tp > 1:
         return trainer.topo.dp_rank == 0
     return ctx.is_rank0
@@ -52,6 +54,10 @@ def main(argv=None) -> int:
     ap.add_argument("--pp", type=int, default=int(os.environ.get(
         "TRAININGJOB_PP_SIZE", "1")),
         help="pipeline-parallel stages (world must equal pp; rank = stage)")
+    ap.add_argument("--ep", type=int, default=int(os.environ.get(
+        "TRAININGJOB_EP_SIZE", "0")),
+        help="expert-parallel group size for MoE models (world = edp x ep; "
+             "ep ranks adjacent; 0 = off)")
     ap.add_argument("--pp-schedule", choices=("1f1b", "gpipe"),
                     default=os.environ.get("TRAININGJOB_PP_SCHEDULE", "1f1b"),
                     help="pipeline schedule: 1f1b bounds live microbatches "
@@ -95,6 +101,17 @@ def main(argv=None) -> int:
         # written by the dp_rank==0 replica — see _saves_ckpt)
         ckpt = Checkpointer(os.path.join(args.ckpt_dir,
                                          f"stage{grid.pp_rank}"))
+    elif args.ep:
+        assert args.tp == 1, "EPxTP composition is roadmap"
+        from ..parallel.ep import EPTrainer
+        import torch
+        dev = torch.device(f"cuda:{ctx.local_rank}"
+                           if torch.cuda.is_available() else "cpu")
+        trainer = EPTrainer(cfg, ep_size=args.ep, device=dev)
+        # each ep rank owns a distinct expert shard: per-ep-rank streams,
+        # written by the edp_rank==0 replica
+        ckpt = Checkpointer(os.path.join(
+            args.ckpt_dir, f"ep{trainer.topo.ep_rank}"))
     else:
         trainer = Trainer(cfg, ctx)
         if args.tp > 1:

@@ -80,3 +80,9 @@ class MoELlamaModel(nn.Module):
             logits.reshape(T, -1).contiguous(), targets.reshape(T))
         n_valid = (targets.reshape(T) != -100).sum().clamp(min=1)
         return per_tok.sum() / n_valid
+
+
+# launcher/EPTrainer look models up by name like the dense families
+from .config import CONFIGS  # noqa: E402
+
+CONFIGS[MOE_TINY.name] = MOE_TINY

@@ -14,7 +14,8 @@ gloo-verified against a single-process MoE in tests/test_ep_gloo.py.
 """
 from __future__ import annotations
 
-from typing import List
+from dataclasses import dataclass
+from typing import List, Optional
 
 import torch
 import torch.distributed as dist
@@ -46,6 +47,11 @@ def _a2a_raw(inp: torch.Tensor, in_splits: List[int], out_splits: List[int],
                                input_split_sizes=in_splits, group=group)
         return out
     rank = dist.get_rank(group)
+
+    def _glob(r: int) -> int:
+        # p2p dst/src are GLOBAL ranks even when a group is passed
+        return dist.get_global_rank(group, r) if group is not None else r
+
     in_chunks = list(inp.split(in_splits, dim=0))
     out_chunks = list(out.split(out_splits, dim=0))
     reqs = []
@@ -54,10 +60,11 @@ def _a2a_raw(inp: torch.Tensor, in_splits: List[int], out_splits: List[int],
             out_chunks[peer].copy_(in_chunks[peer])
             continue
         if in_splits[peer] > 0:
-            reqs.append(dist.isend(in_chunks[peer].contiguous(), dst=peer,
-                                   group=group))
+            reqs.append(dist.isend(in_chunks[peer].contiguous(),
+                                   dst=_glob(peer), group=group))
         if out_splits[peer] > 0:
-            reqs.append(dist.irecv(out_chunks[peer], src=peer, group=group))
+            reqs.append(dist.irecv(out_chunks[peer], src=_glob(peer),
+                                   group=group))
     for r in reqs:
         r.wait()
     return out
@@ -122,7 +129,9 @@ class MoEMLP(nn.Module):
         xt = x.reshape(-1, H)
         T = xt.shape[0]
 
-        logits = self.router(xt.float())
+        # routing in fp32 regardless of model dtype (gate numerics);
+        # grads flow back into the (possibly bf16) router weight
+        logits = nn.functional.linear(xt.float(), self.router.weight.float())
         probs = torch.softmax(logits, dim=-1)
         gates, idx = probs.topk(self.top_k, dim=-1)          # [T, k]
         gates = gates / gates.sum(dim=-1, keepdim=True)
@@ -216,3 +225,172 @@ class MoEMLP(nn.Module):
             ex.gate_proj.weight.copy_(src.gate_proj.weight)
             ex.up_proj.weight.copy_(src.up_proj.weight)
             ex.down_proj.weight.copy_(src.down_proj.weight)
+
+
+@dataclass
+class EPTopology:
+    """world = edp x ep. EP groups are CONTIGUOUS ranks (token all-to-all
+    stays on adjacent GPUs over direct xGMI links); expert-dp peers — the
+    ranks holding the same expert shard in different EP groups — are
+    strided by ep_size."""
+    world: int
+    rank: int
+    ep_size: int
+    ep_rank: int
+    ep_group: Optional[object]
+    edp_size: int
+    edp_rank: int
+    edp_group: Optional[object]
+
+
+def build_ep_topology(ep_size: int = 0) -> EPTopology:
+    """ep_size == 0 means 'the whole world is one EP group'. Every rank
+    must call with the same ep_size (new_group is collective)."""
+    if not dist.is_initialized():
+        return EPTopology(1, 0, max(ep_size, 1), 0, None, 1, 0, None)
+    world = dist.get_world_size()
+    rank = dist.get_rank()
+    ep = ep_size or world
+    if world % ep != 0:
+        raise ValueError(f"world {world} not divisible by ep={ep}")
+    edp = world // ep
+    ep_rank, edp_rank = rank % ep, rank // ep
+    ep_group = edp_group = None
+    if ep > 1 and edp > 1:
+        for g in range(edp):
+            grp = dist.new_group(list(range(g * ep, (g + 1) * ep)))
+            if g == edp_rank:
+                ep_group = grp
+    elif ep > 1:
+        ep_group = None  # whole world == default group
+    if edp > 1:
+        for t in range(ep):
+            grp = dist.new_group(list(range(t, world, ep)))
+            if t == ep_rank:
+                edp_group = grp
+    return EPTopology(world, rank, ep, ep_rank, ep_group,
+                      edp, edp_rank, edp_group)
+
+
+def diversify_experts(model, seed: int, ep_rank: int = 0) -> None:
+    """Re-draw each expert's weights from a generator keyed by its GLOBAL
+    expert id, so (a) shards differ across ep ranks, (b) edp peers of the
+    same shard match bit-for-bit, and (c) a single-process full model
+    (ep_rank=0 owning every expert) reproduces every shard exactly."""
+    with torch.no_grad():
+        for li, blk in enumerate(model.blocks):
+            moe = blk.moe
+            for le, ex in enumerate(moe.experts):
+                geid = ep_rank * moe.experts_per_rank + le
+                g = torch.Generator().manual_seed(
+                    seed * 1_000_003 + li * 1009 + geid)
+                for w in (ex.gate_proj.weight, ex.up_proj.weight,
+                          ex.down_proj.weight):
+                    # nn.Linear default init: kaiming_uniform(a=sqrt(5))
+                    # == U(-1/sqrt(fan_in), 1/sqrt(fan_in))
+                    bound = 1.0 / (w.shape[1] ** 0.5)
+                    w.copy_(torch.empty(
+                        w.shape, dtype=torch.float32).uniform_(
+                        -bound, bound, generator=g).to(w.dtype))
+
+
+class EPTrainer:
+    """DP x EP trainer for the MoE-Llama family: every rank is a data-
+    parallel worker over its own batch stream; each EP group shards the
+    experts and exchanges tokens by all-to-all; dense (non-expert)
+    parameters are replicated world-wide.
+
+    Gradient seams (loss = mean over the global batch):
+      * whole flat grad pre-scaled by 1/world,
+      * dense spans all-reduced over the WORLD,
+      * expert spans all-reduced over the expert-dp group (the edp peers
+        holding the same shard) — a no-op when ep == world.
+    Global grad-norm clip counts each expert once: dense normsq (identical
+    on every rank) + expert normsq all-reduced over the EP group.
+    """
+
+    def __init__(self, cfg, ep_size: int = 0, device=None):
+        from ..models.config import CONFIGS
+        from ..models.moe_llama import MoELlamaModel
+        from ..optim import FlatAdamW
+        from ..parallel.flat import FlatParamStore, _aligned
+        from ..training import synthetic_batches
+
+        self.cfg = cfg
+        mcfg = CONFIGS[cfg.model]
+        self.device = torch.device(device or "cpu")
+        self.topo = build_ep_topology(ep_size)
+        torch.manual_seed(cfg.seed)       # identical dense init everywhere
+        with torch.device(self.device):
+            model = MoELlamaModel(mcfg, ep_group=self.topo.ep_group)
+        diversify_experts(model, cfg.seed, self.topo.ep_rank)
+        self.model = model.to(torch.bfloat16)
+        from ..ops import make_inv_freq
+        self.model.inv_freq = make_inv_freq(mcfg.head_dim, mcfg.rope_theta,
+                                            device=self.device)
+        self.store = FlatParamStore(self.model, device=self.device)
+        # optimizer clips nothing; EPTrainer applies the EP-aware clip
+        self.opt = FlatAdamW(self.store, lr=cfg.lr, betas=cfg.betas,
+                             weight_decay=cfg.weight_decay,
+                             clip_grad_norm=0.0)
+        self._build_spans(_aligned)
+        self.data = synthetic_batches(cfg, self.device, rank=self.topo.rank)
+        self.step_count = 0
+
+    def _build_spans(self, _aligned) -> None:
+        """Split the flat buffer into dense vs expert spans, merging
+        adjacent same-class params (padding between them is always zero,
+        so reducing it is harmless)."""
+        items = sorted(
+            ((off, off + _aligned(numel), ".experts." in name)
+             for name, (off, numel) in self.store.offsets.items()),
+            key=lambda t: t[0])
+        self.dense_spans, self.expert_spans = [], []
+        for start, end, is_exp in items:
+            spans = self.expert_spans if is_exp else self.dense_spans
+            if spans and spans[-1][1] == start:
+                spans[-1] = (spans[-1][0], end)
+            else:
+                spans.append((start, end))
+
+    def _reduce_grads(self) -> None:
+        if self.topo.world == 1:
+            return
+        fg = self.store.flat_grad
+        fg.mul_(1.0 / self.topo.world)
+        for s, e in self.dense_spans:
+            dist.all_reduce(fg[s:e])                       # whole world
+        if self.topo.edp_size > 1:
+            for s, e in self.expert_spans:
+                dist.all_reduce(fg[s:e], group=self.topo.edp_group)
+
+    def _clip_grads(self) -> None:
+        clip = self.cfg.clip_grad_norm
+        if not clip or clip <= 0:
+            return
+        fg = self.store.flat_grad
+        expert_nsq = fg.new_zeros((), dtype=torch.float32)
+        for s, e in self.expert_spans:
+            expert_nsq += fg[s:e].float().pow(2).sum()
+        dense_nsq = fg.new_zeros((), dtype=torch.float32)
+        for s, e in self.dense_spans:
+            dense_nsq += fg[s:e].float().pow(2).sum()
+        if self.topo.ep_size > 1 and self.topo.world > 1:
+            dist.all_reduce(expert_nsq, group=self.topo.ep_group)
+        gnorm = (dense_nsq + expert_nsq).sqrt()
+        if float(gnorm) > clip:
+            fg.mul_(clip / float(gnorm))
+
+    def train_step(self):
+        cfg = self.cfg
+        loss = None
+        for _ in range(cfg.grad_accum):
+            tokens, targets = next(self.data)
+            loss = self.model(tokens, targets)
+            (loss / cfg.grad_accum).backward()
+        self._reduce_grads()
+        self._clip_grads()
+        self.opt.step(grad_pre_scale=1.0)
+        self.opt.zero_grad()
+        self.step_count += 1
+        return loss.detach()
