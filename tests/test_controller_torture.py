@@ -196,3 +196,69 @@ def test_torture_multirole(seed):
     api.set_node_ready("node-0", True)
     for _ in range(6):
         tc.sync_once(f"{NS}/torture")
+
+
+def make_auto_job(rng):
+    j = make_job(rng)
+    rs = j["spec"]["replicaSpecs"]["trainer"]
+    rs["edlPolicy"] = "Auto"
+    rs["minReplicas"] = 1
+    rs["maxReplicas"] = 6
+    rs["replicas"] = rng.randint(1, 4)
+    return j
+
+
+@pytest.mark.parametrize("seed", [3, 17, 404, 8080])
+def test_torture_auto_elastic(seed):
+    """EdlPolicy=Auto under a randomized event stream with a simulated
+    clock: unschedulable pods, faults, node flips. The controller's chosen
+    target must always stay inside [min, max] and never materialize more
+    pods than maxReplicas."""
+    from trainingjob_operator_amd.controller.pods import TARGET_ANNOTATION
+    rng = random.Random(seed)
+    api = FakeKubeApi()
+    tc = TrainingJobController(api, OperatorOptions(
+        elastic_unschedulable_grace=5.0, elastic_scaleup_interval=20.0))
+    api.create_job(NS, make_auto_job(rng))
+    t = 1_700_000_000.0
+    history = {}
+
+    for step in range(120):
+        t += rng.uniform(1.0, 10.0)
+        action = rng.random()
+        pods = api.pod_names(NS)
+        try:
+            if action < 0.35 and pods:
+                api.set_pod_phase(NS, rng.choice(pods), rng.choice(
+                    ["Running", "Running", "Running", "Succeeded"]))
+            elif action < 0.50 and pods:
+                api.set_pod_phase(NS, rng.choice(pods), "Failed",
+                                  exit_code=rng.choice([137, 128, 1]))
+            elif action < 0.62 and pods:
+                # scheduler pressure: a pod cannot land anywhere
+                api.set_pod_unschedulable(NS, rng.choice(pods),
+                                          since=t - rng.uniform(0, 10))
+            elif action < 0.70:
+                api.set_node_ready("node-0", rng.random() < 0.8)
+        except KeyError:
+            pass
+
+        tc.sync_once(f"{NS}/torture", now=t)
+        job = AITrainingJob.from_dict(api.get_job(NS, "torture"))
+        assert job.status.phase in ALL_PHASES
+        rc = job.status.restart_counts.get("trainer", 0)
+        assert rc >= history.get("rc", 0)
+        history["rc"] = rc
+        tgt = job.annotations.get(f"{TARGET_ANNOTATION}-trainer")
+        if tgt is not None:
+            assert 1 <= int(tgt) <= 6, f"target {tgt} out of [min,max]"
+        assert len(api.pod_names(NS)) <= 6, "more pods than maxReplicas"
+        if job.status.phase in ENDING_PHASES and not api.pod_names(NS):
+            break
+
+    # settle
+    api.set_node_ready("node-0", True)
+    for _ in range(10):
+        t += 5.0
+        tc.sync_once(f"{NS}/torture", now=t)
+        assert len(api.pod_names(NS)) <= 6
