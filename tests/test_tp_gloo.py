@@ -369,3 +369,42 @@ def test_grid_topology_8ranks(tmp_path):
                                                    f"g{r}.json"))))
               for r in range(8)}
     assert len(set(coords.values())) == 8  # bijective rank <-> coordinate
+
+
+def _tp_clip_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from trainingjob_operator_amd.parallel.dist_ctx import DistContext
+        from trainingjob_operator_amd.training import TrainConfig, Trainer
+        ctx = DistContext(rank=rank, world_size=world, backend="gloo")
+        # clip low enough to engage on every step
+        cfg = TrainConfig(model="llama-tiny", micro_batch=1, grad_accum=2,
+                          seq_len=32, lr=1e-2, clip_grad_norm=0.05,
+                          tp_size=world)
+        trainer = Trainer(cfg, ctx)
+        assert trainer._tp_spans is not None
+        sharded, replicated = trainer._tp_spans
+        assert sharded and replicated
+        for _ in range(3):
+            trainer.train_step()
+        # the clip factor must be IDENTICAL on every tp peer (global norm,
+        # not local) or the replicated params drift — require bit-equality
+        flat = trainer.store.flat_param
+        reps = torch.cat([flat[s:e] for s, e in replicated]).contiguous()
+        peers = [torch.empty_like(reps) for _ in range(world)]
+        dist.all_gather(peers, reps, group=trainer.topo.tp_group)
+        assert torch.equal(peers[0], peers[1]), \
+            "replicated params diverged across tp peers under clipping"
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_tp_clip_keeps_replicated_params_identical():
+    """Grad-norm clipping under TP uses the GLOBAL norm (sharded normsq
+    all-reduced over the tp group + replicated counted once); a local-norm
+    clip would scale peers differently and silently diverge the
+    replicated norms/embeddings."""
+    mp.spawn(_tp_clip_worker, args=(2, _free_port()), nprocs=2, join=True)

@@ -313,7 +313,7 @@ class EPTrainer:
         from ..models.config import CONFIGS
         from ..models.moe_llama import MoELlamaModel
         from ..optim import FlatAdamW
-        from ..parallel.flat import FlatParamStore, _aligned
+        from ..parallel.flat import FlatParamStore, classify_spans
         from ..training import synthetic_batches
 
         self.cfg = cfg
@@ -333,25 +333,10 @@ class EPTrainer:
         self.opt = FlatAdamW(self.store, lr=cfg.lr, betas=cfg.betas,
                              weight_decay=cfg.weight_decay,
                              clip_grad_norm=0.0)
-        self._build_spans(_aligned)
+        self.expert_spans, self.dense_spans = classify_spans(
+            self.store, lambda n: ".experts." in n)
         self.data = synthetic_batches(cfg, self.device, rank=self.topo.rank)
         self.step_count = 0
-
-    def _build_spans(self, _aligned) -> None:
-        """Split the flat buffer into dense vs expert spans, merging
-        adjacent same-class params (padding between them is always zero,
-        so reducing it is harmless)."""
-        items = sorted(
-            ((off, off + _aligned(numel), ".experts." in name)
-             for name, (off, numel) in self.store.offsets.items()),
-            key=lambda t: t[0])
-        self.dense_spans, self.expert_spans = [], []
-        for start, end, is_exp in items:
-            spans = self.expert_spans if is_exp else self.dense_spans
-            if spans and spans[-1][1] == start:
-                spans[-1] = (spans[-1][0], end)
-            else:
-                spans.append((start, end))
 
     def _reduce_grads(self) -> None:
         if self.topo.world == 1:

@@ -112,3 +112,23 @@ class FlatParamStore:
     def load_flat_param(self, flat: torch.Tensor) -> None:
         self.flat_param.copy_(flat.to(self.flat_param.device,
                                       self.flat_param.dtype))
+
+
+def classify_spans(store: FlatParamStore, predicate):
+    """Split the flat buffer into (matching, rest) span lists by
+    predicate(param_name). Adjacent same-class params are merged INCLUDING
+    the alignment padding between them — padding is zero-initialized and
+    only param slices are ever written, so collectives/norms over it are
+    harmless. Spans are [start, end) offsets into flat_param/flat_grad."""
+    items = sorted(
+        ((off, off + _aligned(numel), bool(predicate(name)))
+         for name, (off, numel) in store.offsets.items()),
+        key=lambda t: t[0])
+    match, rest = [], []
+    for start, end, is_match in items:
+        spans = match if is_match else rest
+        if spans and spans[-1][1] == start:
+            spans[-1] = (spans[-1][0], end)
+        else:
+            spans.append((start, end))
+    return match, rest

@@ -102,6 +102,7 @@ class ColumnParallelLinear(nn.Module):
         self.copy_input = copy_input
         self.weight = nn.Parameter(
             torch.empty(self.out_per_rank, in_features, dtype=dtype))
+        self.weight.tp_sharded = True  # distinct shard per tp rank
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if self.copy_input:
@@ -129,6 +130,7 @@ class RowParallelLinear(nn.Module):
         self.reduce_output = reduce_output
         self.weight = nn.Parameter(
             torch.empty(out_features, self.in_per_rank, dtype=dtype))
+        self.weight.tp_sharded = True  # distinct shard per tp rank
 
     def forward(self, x_shard: torch.Tensor) -> torch.Tensor:
         partial = nn.functional.linear(x_shard, self.weight)

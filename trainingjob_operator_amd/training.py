@@ -105,9 +105,22 @@ class Trainer:
         self.ddp = DDPEngine(self.store, process_group=self.topo.dp_group,
                              bucket_bytes=cfg.bucket_bytes,
                              world_size=self.topo.dp_size)
+        # Under TP the flat buffer mixes per-rank shards with replicated
+        # params, so the optimizer's LOCAL-norm clip would scale tp peers
+        # differently and silently diverge the replicated params. The
+        # trainer applies a TP-aware global clip instead (_tp_clip).
+        tp_aware_clip = self.topo.tp_size > 1 and cfg.clip_grad_norm > 0
         self.opt = FlatAdamW(self.store, lr=cfg.lr, betas=cfg.betas,
                              weight_decay=cfg.weight_decay,
-                             clip_grad_norm=cfg.clip_grad_norm)
+                             clip_grad_norm=(0.0 if tp_aware_clip
+                                             else cfg.clip_grad_norm))
+        self._tp_spans = None
+        if tp_aware_clip:
+            from .parallel.flat import classify_spans
+            named = dict(self.model.named_parameters())
+            self._tp_spans = classify_spans(
+                self.store,
+                lambda n: getattr(named[n], "tp_sharded", False))
         # tp peers train on the SAME data (they hold shards of one replica)
         self.data = synthetic_batches(cfg, device, self.topo.dp_rank)
         self.step_count = 0
@@ -126,10 +139,39 @@ class Trainer:
                 # scale so accumulated grads average over micro-batches
                 (loss / cfg.grad_accum).backward()
         self.ddp.finish_backward()
-        self.opt.step(grad_pre_scale=self.ddp.grad_pre_scale,
-                      in_graph_capture=in_graph)
+        if self._tp_spans is not None:
+            self._tp_clip()
+            self.opt.step(grad_pre_scale=1.0, in_graph_capture=in_graph)
+        else:
+            self.opt.step(grad_pre_scale=self.ddp.grad_pre_scale,
+                          in_graph_capture=in_graph)
         self.opt.zero_grad()
         return loss
+
+    def _tp_clip(self) -> None:
+        """Global grad-norm clip under TP: sharded-param normsq summed
+        over the tp group plus replicated normsq counted once. The DP
+        pre-scale is folded into the flat grad first so every tp peer
+        applies the IDENTICAL clip factor (bit-aligned replicated params).
+        Not hipGraph-capturable (host sync on the norm) — TP is not the
+        graphed flagship path."""
+        import torch.distributed as dist
+        fg = self.store.flat_grad
+        if self.ddp.grad_pre_scale != 1.0:
+            fg.mul_(self.ddp.grad_pre_scale)
+        sharded, replicated = self._tp_spans
+        nsq_sh = fg.new_zeros((), dtype=torch.float32)
+        for s, e in sharded:
+            nsq_sh += fg[s:e].float().pow(2).sum()
+        if self.topo.tp_group is not None:
+            dist.all_reduce(nsq_sh, group=self.topo.tp_group)
+        nsq_rep = fg.new_zeros((), dtype=torch.float32)
+        for s, e in replicated:
+            nsq_rep += fg[s:e].float().pow(2).sum()
+        gnorm = float((nsq_sh + nsq_rep).sqrt())
+        clip = self.cfg.clip_grad_norm
+        if gnorm > clip:
+            fg.mul_(clip / gnorm)
 
     def _capture_graph(self) -> None:
         """Warm up twice on a side stream, then record one full step."""
