@@ -310,6 +310,9 @@ class PPTrainer:
             pp_ranks = None
             self.dp_size, self.dp_rank, self.dp_group = 1, 0, None
         self.grid = grid
+        # all stages of MY pipeline (grad-norm seam); None == default
+        # group in the pure-PP world where rank == stage
+        self.pp_group = grid.pp_group if grid is not None else None
         # identical init on every dp replica of a stage (same seed)
         torch.manual_seed(cfg.seed)
         self.stage = LlamaStage.from_config(mcfg, stage_idx, n_stages,
@@ -322,14 +325,31 @@ class PPTrainer:
         self.sched = sched_cls(self.stage, stage_idx, n_stages,
                                device=self.device, pp_ranks=pp_ranks)
         self.store = FlatParamStore(self.stage, device=self.device)
+        # the optimizer's LOCAL-norm clip would scale each stage by its
+        # own norm; the trainer applies the cross-stage global clip
         self.opt = FlatAdamW(self.store, lr=cfg.lr, betas=cfg.betas,
                              weight_decay=cfg.weight_decay,
-                             clip_grad_norm=cfg.clip_grad_norm)
+                             clip_grad_norm=0.0)
         # every pp rank of a replica draws the same stream; replicas draw
         # DISTINCT streams (dp_rank-keyed), like the DP Trainer
         self.data = synthetic_batches(cfg, self.device, rank=self.dp_rank)
         self.hidden_shape = (cfg.micro_batch, cfg.seq_len, mcfg.hidden_size)
         self.step_count = 0
+
+    def _clip_grads(self) -> None:
+        """Global grad-norm clip across the pipeline: each stage's normsq
+        summed over the pp group so every stage applies the IDENTICAL
+        factor — matching the unsharded model's clip semantics."""
+        clip = self.cfg.clip_grad_norm
+        if not clip or clip <= 0:
+            return
+        fg = self.store.flat_grad
+        nsq = fg.float().pow(2).sum()
+        if dist.is_initialized() and self.sched.n > 1:
+            dist.all_reduce(nsq, group=self.pp_group)
+        gnorm = float(nsq.sqrt())
+        if gnorm > clip:
+            fg.mul_(clip / gnorm)
 
     def train_step(self):
         micros = [next(self.data) for _ in range(self.cfg.grad_accum)]
@@ -342,6 +362,7 @@ class PPTrainer:
             if loss is not None:
                 loss = loss / self.dp_size
                 dist.all_reduce(loss, group=self.dp_group)
+        self._clip_grads()
         self.opt.step()
         self.opt.zero_grad()
         self.step_count += 1
