@@ -90,17 +90,18 @@ def main(argv=None) -> int:
         checkpoint_activations=args.checkpoint_activations,
         tp_size=args.tp)
     if args.pp > 1:
-        assert args.tp == 1, "DPxTPxPP grids are roadmap; use tp or pp"
-        assert ctx.world_size % args.pp == 0, \
-            f"world {ctx.world_size} not divisible by pp={args.pp}"
+        assert ctx.world_size % (args.pp * max(args.tp, 1)) == 0, \
+            f"world {ctx.world_size} not divisible by " \
+            f"pp*tp={args.pp * max(args.tp, 1)}"
         from ..parallel.groups import build_grid
         from ..parallel.pp import PPTrainer
-        grid = build_grid(tp_size=1, pp_size=args.pp)
+        grid = build_grid(tp_size=max(args.tp, 1), pp_size=args.pp)
         trainer = PPTrainer(cfg, schedule=args.pp_schedule, grid=grid)
-        # per-stage checkpoint streams (each stage slice its own stream,
-        # written by the dp_rank==0 replica — see _saves_ckpt)
-        ckpt = Checkpointer(os.path.join(args.ckpt_dir,
-                                         f"stage{grid.pp_rank}"))
+        # per-(stage, tp-shard) checkpoint streams, written by the
+        # dp_rank==0 replica — see _saves_ckpt
+        sub = (f"stage{grid.pp_rank}_tp{grid.tp_rank}" if args.tp > 1
+               else f"stage{grid.pp_rank}")
+        ckpt = Checkpointer(os.path.join(args.ckpt_dir, sub))
     elif args.ep:
         assert args.tp == 1, "EPxTP composition is roadmap"
         from ..parallel.ep import EPTrainer
@@ -139,9 +140,11 @@ def main(argv=None) -> int:
     if args.metrics_port and ctx.is_rank0:
         from .worker_metrics import WorkerMetrics
         metrics = WorkerMetrics(args.metrics_port)
-    # tp ranks share one replica: whole-job tokens count DP replicas
+    # tp ranks and pp stages share one replica: whole-job tokens count DP
+    # replicas only (EP ranks are all data workers: divisor 1)
     tokens_per_step = (cfg.tokens_per_step_per_rank()
-                       * (ctx.world_size // max(args.tp, 1)))
+                       * (ctx.world_size
+                          // (max(args.tp, 1) * max(args.pp, 1))))
     t_last = time.time()
     while trainer.step_count < args.steps and not stop_requested["flag"]:
         loss = trainer.train_step()

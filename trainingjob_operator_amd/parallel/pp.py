@@ -299,7 +299,6 @@ class PPTrainer:
         mcfg = CONFIGS[cfg.model]
         self.device = torch.device(device or "cpu")
         if grid is not None:
-            assert grid.tp_size == 1, "PPxTP composition is roadmap"
             stage_idx, n_stages = grid.pp_rank, grid.pp_size
             pp_ranks = [(grid.dp_rank * grid.pp_size + p) * grid.tp_size
                         + grid.tp_rank for p in range(grid.pp_size)]
@@ -315,8 +314,24 @@ class PPTrainer:
         self.pp_group = grid.pp_group if grid is not None else None
         # identical init on every dp replica of a stage (same seed)
         torch.manual_seed(cfg.seed)
-        self.stage = LlamaStage.from_config(mcfg, stage_idx, n_stages,
-                                            device=self.device)
+        if grid is not None and grid.tp_size > 1:
+            # PP x TP: slice a tensor-parallel model into stages — the
+            # stage machinery is block-generic (TPBlock outputs the same
+            # full-size (x, residual) pair after its row-parallel
+            # all-reduce, so the P2P seam is unchanged)
+            from ..training import build_model
+            from .tp_llama import TPLlamaModel
+            full = build_model(mcfg, self.device,
+                               cfg.checkpoint_activations)
+            tpm = TPLlamaModel(mcfg, group=grid.tp_group).to(
+                full.embed.weight.dtype).to(self.device)
+            tpm.inv_freq = full.inv_freq
+            tpm.shard_from_full(full)
+            del full
+            self.stage = LlamaStage.from_model(tpm, stage_idx, n_stages)
+        else:
+            self.stage = LlamaStage.from_config(mcfg, stage_idx, n_stages,
+                                                device=self.device)
         # the flat store keeps parameters in bf16; boundary activations
         # travel in the same dtype
         self.act_dtype = act_dtype or torch.bfloat16
@@ -330,6 +345,15 @@ class PPTrainer:
         self.opt = FlatAdamW(self.store, lr=cfg.lr, betas=cfg.betas,
                              weight_decay=cfg.weight_decay,
                              clip_grad_norm=0.0)
+        self._tp_spans = None
+        self.tp_group = None
+        if grid is not None and grid.tp_size > 1:
+            from .flat import classify_spans
+            named = dict(self.stage.named_parameters())
+            self._tp_spans = classify_spans(
+                self.store,
+                lambda n: getattr(named[n], "tp_sharded", False))
+            self.tp_group = grid.tp_group
         # every pp rank of a replica draws the same stream; replicas draw
         # DISTINCT streams (dp_rank-keyed), like the DP Trainer
         self.data = synthetic_batches(cfg, self.device, rank=self.dp_rank)
@@ -344,7 +368,18 @@ class PPTrainer:
         if not clip or clip <= 0:
             return
         fg = self.store.flat_grad
-        nsq = fg.float().pow(2).sum()
+        if self._tp_spans is not None:
+            # count each tp shard once: sharded normsq summed over the tp
+            # group, replicated params counted locally
+            sharded, replicated = self._tp_spans
+            nsq = fg.new_zeros((), dtype=torch.float32)
+            for s_, e_ in sharded:
+                nsq += fg[s_:e_].float().pow(2).sum()
+            dist.all_reduce(nsq, group=self.tp_group)
+            for s_, e_ in replicated:
+                nsq += fg[s_:e_].float().pow(2).sum()
+        else:
+            nsq = fg.float().pow(2).sum()
         if dist.is_initialized() and self.sched.n > 1:
             dist.all_reduce(nsq, group=self.pp_group)
         gnorm = float(nsq.sqrt())
