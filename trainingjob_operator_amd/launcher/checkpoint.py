@@ -48,6 +48,11 @@ class Checkpointer:
                 "step": trainer.opt.step_count,
             },
             "train_config": trainer.cfg.__dict__.copy(),
+            # flat-layout metadata: lets reshard.py reinterpret the stream
+            # name-wise when converting between parallel topologies
+            "names": list(trainer.store.names),
+            "offsets": dict(trainer.store.offsets),
+            "shapes": dict(getattr(trainer.store, "shapes", {})),
             "time": time.time(),
         }
 

@@ -37,8 +37,10 @@ class FlatParamStore:
 
         offset = 0
         self.offsets: Dict[str, Tuple[int, int]] = {}
+        self.shapes: Dict[str, Tuple[int, ...]] = {}
         for n, p in named:
             self.offsets[n] = (offset, p.numel())
+            self.shapes[n] = tuple(p.shape)
             offset += _aligned(p.numel())
         self.total = _aligned(offset)
 
