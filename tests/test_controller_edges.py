@@ -90,3 +90,59 @@ def test_gc_spares_other_namespaces_pods():
     # cluster-wide GC does collect it
     gc_all = GarbageCollector(api)
     assert gc_all.clean_garbage_pods(time.time()) == 1
+
+
+def test_fault_tolerant_widens_node_fail_restart():
+    """spec.faultTolerant (dead in the reference) makes node loss
+    retryable under ANY restart policy except Never."""
+    for ft, expect_restart in ((True, True), (False, False)):
+        api = FakeKubeApi()
+        tc = TrainingJobController(api, OperatorOptions())
+        api.create_job(NS, make_job(faultTolerant=ft))  # OnFailure-like?
+        j = api.get_job(NS, "e")
+        j["spec"]["replicaSpecs"]["trainer"]["restartPolicy"] = "OnFailure"
+        api.update_job(NS, "e", j)
+        tc.sync_once(f"{NS}/e")
+        api.set_all_pods_phase(NS, "Running")
+        tc.sync_once(f"{NS}/e")
+        api.set_node_ready("node-0", False)
+        tc.sync_once(f"{NS}/e")
+        job = AITrainingJob.from_dict(api.get_job(NS, "e"))
+        if expect_restart:
+            assert job.status.restart_replica_name == "trainer"
+        else:
+            assert job.status.restart_replica_name == ""
+            assert Phase.NODE_FAIL in job.annotations
+            tc.sync_once(f"{NS}/e")  # deferred finalization
+            job = AITrainingJob.from_dict(api.get_job(NS, "e"))
+            assert job.status.phase == Phase.NODE_FAIL
+
+
+def test_never_policy_ignores_fault_tolerant():
+    api = FakeKubeApi()
+    tc = TrainingJobController(api, OperatorOptions())
+    job = make_job(faultTolerant=True)
+    job["spec"]["replicaSpecs"]["trainer"]["restartPolicy"] = "Never"
+    api.create_job(NS, job)
+    tc.sync_once(f"{NS}/e")
+    api.set_all_pods_phase(NS, "Running")
+    tc.sync_once(f"{NS}/e")
+    api.set_node_ready("node-0", False)
+    tc.sync_once(f"{NS}/e")
+    job = AITrainingJob.from_dict(api.get_job(NS, "e"))
+    assert job.status.restart_replica_name == ""
+    assert Phase.NODE_FAIL in job.annotations
+    tc.sync_once(f"{NS}/e")
+    assert AITrainingJob.from_dict(
+        api.get_job(NS, "e")).status.phase == Phase.NODE_FAIL
+
+
+def test_last_reconcile_time_stamped_on_persist():
+    from trainingjob_operator_amd.utils.k8stime import parse_time
+    api = FakeKubeApi()
+    tc = TrainingJobController(api, OperatorOptions())
+    api.create_job(NS, make_job())
+    tc.sync_once(f"{NS}/e")
+    job = AITrainingJob.from_dict(api.get_job(NS, "e"))
+    t = parse_time(job.status.last_reconcile_time)
+    assert t is not None and t > 0

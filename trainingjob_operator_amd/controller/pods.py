@@ -238,7 +238,8 @@ class PodReconciler:
             d = engine.container_decision(
                 pod, spec.restart_policy, job.spec.restarting_exit_code,
                 node_ready, creating_transition,
-                self.options.creating_failure_policy(), now)
+                self.options.creating_failure_policy(), now,
+                fault_tolerant=job.spec.fault_tolerant)
             if d.message and d.phase == Phase.FAILED:
                 failed_reason.append(d.message)
 

@@ -171,6 +171,11 @@ class StatusEngine:
         """Write status (+ annotations, which carry the pending-termination
         marker) back to the API with optimistic-conflict retry
         (reference: status.go:285-305, improved to re-read on conflict)."""
+        # real semantics for status.lastReconcileTime (declared but never
+        # set in the reference, SURVEY.md C15): the time of the last
+        # status write — stamped here, not per-sync, to avoid API churn
+        from ..utils.k8stime import format_time
+        job.status.last_reconcile_time = format_time()
         for attempt in range(retries):
             try:
                 current = self.api.get_job(job.namespace, job.name)

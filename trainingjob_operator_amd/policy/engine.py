@@ -84,6 +84,7 @@ def container_decision(
     creating_condition_transition: Optional[float],
     policy: CreatingFailurePolicy,
     now: float,
+    fault_tolerant: bool = False,
 ) -> ContainerDecision:
     """Map one pod's container statuses to (phase, is_restart, message).
 
@@ -94,6 +95,8 @@ def container_decision(
     pod.go:439-455 builds the map only from Ready nodes).
     creating_condition_transition: unix time of the job's Creating condition
     lastTransitionTime, or None if the job has no True Creating condition.
+    fault_tolerant: spec.faultTolerant — widens node-failure restarts to
+    every restart policy except Never (new semantics for a dead ref field).
     """
     status = pod.get("status") or {}
     pod_spec = pod.get("spec") or {}
@@ -159,6 +162,12 @@ def container_decision(
     if node_name and node_name not in node_ready:
         if restart_policy in (RestartPolicy.ON_NODE_FAIL_WITH_EXIT_CODE,
                               RestartPolicy.ON_NODE_FAIL, RestartPolicy.ALWAYS):
+            is_restart = True
+        elif fault_tolerant and restart_policy != RestartPolicy.NEVER:
+            # spec.faultTolerant (declared but dead in the reference,
+            # SURVEY.md C15): node loss is retryable under ANY restart
+            # policy except Never — the job survives node failures even
+            # when its policy only covers container exits
             is_restart = True
         return ContainerDecision(
             Phase.NODE_FAIL, is_restart,
