@@ -156,3 +156,40 @@ def test_resharded_ckpt_resumes_as_tp(tmp_path):
     resume with the TP trainer, losses match."""
     mp.spawn(_tp_resume_worker, args=(2, _free_port(), str(tmp_path)),
              nprocs=2, join=True)
+
+
+def test_full_ep_full_roundtrip(tmp_path):
+    """MoE: full <-> ep=2 expert-shard streams (experts renumbered,
+    dense replicated)."""
+    from trainingjob_operator_amd.parallel.ep import EPTrainer
+    cfg = TrainConfig(model="moe-tiny", micro_batch=1, grad_accum=2,
+                      seq_len=16, lr=1e-3)
+    tr = EPTrainer(cfg)        # no dist -> single-process full MoE
+    for _ in range(2):
+        tr.train_step()
+    full_dir = os.path.join(str(tmp_path), "full")
+    ep_dir = os.path.join(str(tmp_path), "ep")
+    back_dir = os.path.join(str(tmp_path), "back")
+    Checkpointer(full_dir).save_async(tr, blocking=True)
+    paths = reshard("moe-tiny", full_dir, ep_dir, "full", "ep=2")
+    assert len(paths) == 2
+    s0 = stream_to_named(load_stream(Checkpointer(
+        os.path.join(ep_dir, "ep0")).latest()))
+    s1 = stream_to_named(load_stream(Checkpointer(
+        os.path.join(ep_dir, "ep1")).latest()))
+    # shard 0 local expert 1 == full global expert 1; shard 1 local 0 == 2
+    full_named = stream_to_named(load_stream(Checkpointer(full_dir).latest()))
+    assert torch.equal(
+        s0["blocks.0.moe.experts.1.gate_proj.weight"]["param"],
+        full_named["blocks.0.moe.experts.1.gate_proj.weight"]["param"])
+    assert torch.equal(
+        s1["blocks.0.moe.experts.0.gate_proj.weight"]["param"],
+        full_named["blocks.0.moe.experts.2.gate_proj.weight"]["param"])
+    # router replicated on both shards
+    assert torch.equal(s0["blocks.0.moe.router.weight"]["param"],
+                       s1["blocks.0.moe.router.weight"]["param"])
+    reshard("moe-tiny", ep_dir, back_dir, "ep=2", "full")
+    orig = load_stream(Checkpointer(full_dir).latest())
+    back = load_stream(Checkpointer(back_dir).latest())
+    for a, b in zip(_flats(orig), _flats(back)):
+        assert torch.equal(a, b)

@@ -1,5 +1,6 @@
 """Checkpoint topology resharding: convert saved training streams between
-parallel layouts — full (pure DP) <-> PP stage slices <-> TP shards —
+parallel layouts — full (pure DP) <-> PP stage slices <-> TP shards <->
+EP expert shards —
 carrying parameters AND AdamW moments (p32/m/v transform identically to
 their parameters, since the optimizer flats mirror the param flat layout).
 
@@ -10,8 +11,7 @@ is replicated — checkpoint.py), PP/TP changes run through here.
 
 Streams are the exact files Checkpointer writes (ckpt_stepNNNNNNNN.pt with
 names/offsets/shapes metadata). Directory conventions match the launcher:
-full -> DIR/, PP -> DIR/stage{i}/, TP -> DIR/tp{r}/, PPxTP ->
-DIR/stage{i}_tp{r}/.
+full -> DIR/, PP -> DIR/stage{i}/, TP -> DIR/tp{r}/, EP -> DIR/ep{r}/.
 """
 from __future__ import annotations
 
@@ -55,14 +55,23 @@ def _store_order(model) -> List[str]:
     return names
 
 
+def _is_moe(cfg: LlamaConfig) -> bool:
+    from ..models.moe_llama import MoELlamaConfig
+    return isinstance(cfg, MoELlamaConfig)
+
+
 def _model_order(cfg: LlamaConfig, kind: str) -> List[str]:
     """Flat layout order for a target model, built on the meta device
     (no allocation)."""
     with torch.device("meta"):
-        if kind == "full":
+        if kind == "full" and _is_moe(cfg):
+            from ..models.moe_llama import MoELlamaModel
+            model = MoELlamaModel(cfg)
+        elif kind == "full":
             from ..models.llama import LlamaModel
             model = LlamaModel(cfg)
         elif kind == "tp":
+            assert not _is_moe(cfg), "TP of MoE models is not supported"
             from ..parallel.tp_llama import TPLlamaModel
             model = TPLlamaModel(cfg)
         else:
@@ -230,6 +239,53 @@ def split_tp(full_named: Dict, cfg: LlamaConfig, tp: int) -> List[Dict]:
     return out
 
 
+# -- EP: expert shards <-> full -------------------------------------------
+
+def _expert_gid(name: str) -> int:
+    return int(name.split(".experts.")[1].split(".")[0])
+
+
+def _rename_expert(name: str, new_id: int) -> str:
+    pre, rest = name.split(".experts.")
+    _, tail = rest.split(".", 1)
+    return f"{pre}.experts.{new_id}.{tail}"
+
+
+def merge_ep(shard_named: List[Dict], cfg) -> Dict:
+    """Per-ep-rank named dicts (rank order) -> full-model named dict:
+    local expert ids remap to global (rank r owns the contiguous block
+    [r*E/ep, (r+1)*E/ep) — parallel/ep.py MoEMLP); non-expert params are
+    replicated, taken from rank 0."""
+    ep = len(shard_named)
+    per = cfg.n_experts // ep
+    full = {}
+    for name, t in shard_named[0].items():
+        if ".experts." not in name:
+            full[name] = t
+    for r, named in enumerate(shard_named):
+        for name, t in named.items():
+            if ".experts." in name:
+                full[_rename_expert(name, r * per + _expert_gid(name))] = t
+    return full
+
+
+def split_ep(full_named: Dict, cfg, ep: int) -> List[Dict]:
+    """Inverse of merge_ep."""
+    per = cfg.n_experts // ep
+    out = []
+    for r in range(ep):
+        named = {}
+        for name, t in full_named.items():
+            if ".experts." not in name:
+                named[name] = t
+                continue
+            gid = _expert_gid(name)
+            if r * per <= gid < (r + 1) * per:
+                named[_rename_expert(name, gid - r * per)] = t
+        out.append(named)
+    return out
+
+
 # -- directory-level conversion -------------------------------------------
 
 def _latest_state(directory: str) -> dict:
@@ -287,6 +343,11 @@ def reshard(model: str, in_dir: str, out_dir: str,
                   for r in range(sn)]
         full = merge_tp([stream_to_named(s) for s in states], cfg)
         meta = states[0]
+    elif skind == "ep":
+        states = [_latest_state(os.path.join(in_dir, f"ep{r}"))
+                  for r in range(sn)]
+        full = merge_ep([stream_to_named(s) for s in states], cfg)
+        meta = states[0]
     else:
         raise ValueError(src)
 
@@ -310,6 +371,21 @@ def reshard(model: str, in_dir: str, out_dir: str,
         for r, named in enumerate(shards):
             written.append(_write_state(
                 os.path.join(out_dir, f"tp{r}"),
+                named_to_stream(named, order, meta)))
+    elif dkind == "ep":
+        full_order = _model_order(cfg, "full")
+        per = cfg.n_experts // dn
+        for r, named in enumerate(split_ep(full, cfg, dn)):
+            base = r * per
+            order = []
+            for n in full_order:
+                if ".experts." not in n:
+                    order.append(n)
+                elif base <= _expert_gid(n) < base + per:
+                    order.append(_rename_expert(n, _expert_gid(n) - base))
+            assert set(order) == set(named)
+            written.append(_write_state(
+                os.path.join(out_dir, f"ep{r}"),
                 named_to_stream(named, order, meta)))
     else:
         raise ValueError(dst)
