@@ -160,3 +160,21 @@ def test_manual_policy_ignores_auto_annotations():
     api.set_all_pods_phase(NS, "Running")
     tc.sync_once(f"{NS}/auto", now=t0 + 1000)
     assert len(api.pod_names(NS)) == 2
+
+
+def test_metrics_count_auto_resizes():
+    from trainingjob_operator_amd.controller.metrics import OperatorMetrics
+    m = OperatorMetrics(port=0)
+    api = FakeKubeApi()
+    tc = TrainingJobController(api, OperatorOptions(
+        elastic_unschedulable_grace=60.0), metrics=m)
+    api.create_job(NS, make_job(replicas=4, mn=2, mx=4))
+    t0 = time.time()
+    tc.sync_once(f"{NS}/auto", now=t0)
+    api.set_all_pods_phase(NS, "Running")
+    api.set_pod_unschedulable(NS, "auto-trainer-3", since=t0 - 120)
+    tc.sync_once(f"{NS}/auto", now=t0 + 1)
+    down = m.elastic_resizes_total.labels(direction="down")
+    assert down._value.get() == 1
+    # the restart the resize triggers is counted too
+    assert m.restarts_total.labels(scope="All")._value.get() == 1

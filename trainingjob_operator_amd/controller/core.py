@@ -127,6 +127,10 @@ class TrainingJobController:
                     engine.update_job_conditions(job, Phase.TERMINATING,
                                                  msg, now)
                     job.status.restart_replica_name = rtype
+                    if self.metrics:
+                        self.metrics.restarts_total.labels(
+                            scope=job.spec.replica_specs[rtype]
+                            .restart_scope or "All").inc()
                     break
                 self.service_reconciler.reconcile(job, services, rtype)
 
@@ -134,6 +138,18 @@ class TrainingJobController:
         # between Terminating and Restarting (controller.go:380, status.go:113)
         self.status_engine.update_status(job, pods, services, ending_phases,
                                          message, now)
+
+        if self.metrics:
+            from .pods import TARGET_ANNOTATION
+            for k, v in job.annotations.items():
+                prev = original_annotations.get(k)
+                if k.startswith(TARGET_ANNOTATION) and prev not in (None, v):
+                    try:
+                        direction = "up" if int(v) > int(prev) else "down"
+                    except ValueError:
+                        continue
+                    self.metrics.elastic_resizes_total.labels(
+                        direction=direction).inc()
 
         if job.status.to_dict() != original_status or \
                 dict(job.annotations) != original_annotations:

@@ -21,6 +21,9 @@ class OperatorMetrics:
             "aitj_restarts_total", "Pod restarts triggered", ["scope"])
         self.jobs_by_phase = Gauge(
             "aitj_jobs", "Jobs currently in phase", ["phase"])
+        self.elastic_resizes_total = Counter(
+            "aitj_elastic_resizes_total",
+            "EdlPolicy=Auto world resizes by direction", ["direction"])
         if port:
             start_http_server(port)
             log.info("metrics on :%d/metrics", port)
