@@ -16,7 +16,7 @@ full -> DIR/, PP -> DIR/stage{i}/, TP -> DIR/tp{r}/, EP -> DIR/ep{r}/.
 from __future__ import annotations
 
 import os
-from typing import Dict, List, Tuple
+from typing import Dict, List
 
 import torch
 
