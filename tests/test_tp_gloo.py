@@ -408,3 +408,46 @@ def test_tp_clip_keeps_replicated_params_identical():
     clip would scale peers differently and silently diverge the
     replicated norms/embeddings."""
     mp.spawn(_tp_clip_worker, args=(2, _free_port()), nprocs=2, join=True)
+
+
+def _sp_trainer_worker(rank, world, port, sp, outdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from trainingjob_operator_amd.parallel.dist_ctx import DistContext
+        from trainingjob_operator_amd.training import TrainConfig, Trainer
+        ctx = DistContext(rank=rank, world_size=world, backend="gloo")
+        cfg = TrainConfig(model="llama-tiny", micro_batch=1, grad_accum=2,
+                          seq_len=32, lr=1e-3, bucket_bytes=4096,
+                          tp_size=2 if world == 4 else 1,
+                          sequence_parallel=sp and world == 4)
+        trainer = Trainer(cfg, ctx)
+        losses = [trainer.train_step().item() for _ in range(3)]
+        import json
+        tag = "sp" if cfg.sequence_parallel else f"tp{cfg.tp_size}"
+        with open(os.path.join(outdir, f"{tag}_rank{rank}.json"), "w") as f:
+            json.dump(losses, f)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_trainer_dp2_sp2_matches_dp2(tmp_path):
+    """The sequence-parallel trainer (dp2 x tp2-SP via cfg.sequence_parallel)
+    must reproduce pure-dp2 losses on the same per-dp-rank data — SP is an
+    exact reformulation of TP up to bf16 seam rounding."""
+    import json
+    out = str(tmp_path)
+    mp.spawn(_sp_trainer_worker, args=(4, _free_port(), True, out),
+             nprocs=4, join=True)
+    mp.spawn(_sp_trainer_worker, args=(2, _free_port(), False, out),
+             nprocs=2, join=True)
+
+    def load(tag, rank):
+        return json.load(open(os.path.join(out, f"{tag}_rank{rank}.json")))
+
+    assert load("sp", 0) == pytest.approx(load("sp", 1), abs=1e-3)
+    assert load("sp", 2) == pytest.approx(load("sp", 3), abs=1e-3)
+    assert load("sp", 0) == pytest.approx(load("tp1", 0), abs=3e-2)
+    assert load("sp", 2) == pytest.approx(load("tp1", 1), abs=3e-2)

@@ -54,6 +54,10 @@ def main(argv=None) -> int:
     ap.add_argument("--pp", type=int, default=int(os.environ.get(
         "TRAININGJOB_PP_SIZE", "1")),
         help="pipeline-parallel stages (world must equal pp; rank = stage)")
+    ap.add_argument("--sp", action="store_true",
+                    default=os.environ.get("TRAININGJOB_SP", "") == "1",
+                    help="sequence parallelism on top of --tp (Megatron "
+                         "SP: seq-sharded norms/residual)")
     ap.add_argument("--ep", type=int, default=int(os.environ.get(
         "TRAININGJOB_EP_SIZE", "0")),
         help="expert-parallel group size for MoE models (world = edp x ep; "
@@ -84,11 +88,14 @@ def main(argv=None) -> int:
     t_start = time.time()
     dist_ctx.init_process_group(ctx)
 
+    if args.sp:
+        assert args.tp > 1, "--sp requires --tp > 1"
+        assert args.seq_len % args.tp == 0, "--sp needs seq_len % tp == 0"
     cfg = TrainConfig(
         model=args.model, micro_batch=args.micro_batch,
         grad_accum=args.grad_accum, seq_len=args.seq_len, lr=args.lr,
         checkpoint_activations=args.checkpoint_activations,
-        tp_size=args.tp)
+        tp_size=args.tp, sequence_parallel=args.sp)
     if args.pp > 1:
         assert ctx.world_size % (args.pp * max(args.tp, 1)) == 0, \
             f"world {ctx.world_size} not divisible by " \

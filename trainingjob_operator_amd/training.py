@@ -43,6 +43,10 @@ class TrainConfig:
     # contiguous -> adjacent GPUs over xGMI). 1 = pure DP (the flagship
     # 8B config — one replica fits a 288 GB MI355X).
     tp_size: int = 1
+    # Megatron-style sequence parallelism on top of TP (tp_size > 1):
+    # norms/residual run on seq shards with all-gather/reduce-scatter
+    # seams instead of the f/g all-reduces (parallel/sp.py)
+    sequence_parallel: bool = False
 
     @property
     def model_config(self) -> LlamaConfig:
@@ -89,11 +93,14 @@ class Trainer:
         from .parallel.groups import build_topology
         self.topo = build_topology(cfg.tp_size)
         if cfg.tp_size > 1:
-            from .parallel.tp_llama import TPLlamaModel
+            if cfg.sequence_parallel:
+                from .parallel.sp import SPLlamaModel as _TPModel
+            else:
+                from .parallel.tp_llama import TPLlamaModel as _TPModel
             full = build_model(cfg.model_config, device,
                                cfg.checkpoint_activations)
-            self.model = TPLlamaModel(cfg.model_config,
-                                      group=self.topo.tp_group).to(
+            self.model = _TPModel(cfg.model_config,
+                                  group=self.topo.tp_group).to(
                 full.embed.weight.dtype).to(device)
             self.model.inv_freq = full.inv_freq
             self.model.shard_from_full(full)
@@ -139,6 +146,10 @@ class Trainer:
                 # scale so accumulated grads average over micro-batches
                 (loss / cfg.grad_accum).backward()
         self.ddp.finish_backward()
+        if cfg.sequence_parallel and self.topo.tp_size > 1:
+            # seq-sharded params (norms/embed) saw only this rank's
+            # positions: sum their grads over the tp group BEFORE clip
+            self.model.allreduce_sp_grads()
         if self._tp_spans is not None:
             self._tp_clip()
             self.opt.step(grad_pre_scale=1.0, in_graph_capture=in_graph)
