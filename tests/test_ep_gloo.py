@@ -243,3 +243,41 @@ def test_launcher_ep_mode(tmp_path):
     for r in range(2):
         names = os.listdir(os.path.join(str(tmp_path), f"ep{r}"))
         assert any(n.startswith("ckpt_step") for n in names), names
+
+
+def test_moe_aux_loss_balances():
+    """Load-balance aux loss: ~1.0 for a (near-)uniform router, maximal
+    (~n_experts/top_k-ish) when everything routes to one expert; coef=0
+    reduces the model loss to plain CE."""
+    import dataclasses
+    from trainingjob_operator_amd.models.moe_llama import (
+        MOE_TINY, MoELlamaModel,
+    )
+    torch.manual_seed(5)
+    model = MoELlamaModel(MOE_TINY)
+    x = torch.randn(4, 8, MOE_TINY.hidden_size)
+    moe = model.blocks[0].moe
+    moe(x)
+    uniform_aux = float(moe.aux_loss.detach())
+    assert 0.8 < uniform_aux < 1.6, uniform_aux
+
+    # force single-expert routing: positive inputs + a large positive
+    # row-0 weight make expert 0 every token's top-1
+    with torch.no_grad():
+        moe.router.weight.zero_()
+        moe.router.weight[0].fill_(5.0)
+    moe(torch.rand(4, 8, MOE_TINY.hidden_size))
+    assert float(moe.aux_loss.detach()) > uniform_aux * 1.3
+
+    # coef=0: model loss equals plain CE (aux contributes nothing)
+    cfg0 = dataclasses.replace(MOE_TINY, aux_loss_coef=0.0)
+    torch.manual_seed(5)
+    m0 = MoELlamaModel(cfg0)
+    torch.manual_seed(5)
+    m1 = MoELlamaModel(MOE_TINY)
+    g = torch.Generator().manual_seed(6)
+    tokens = torch.randint(0, MOE_TINY.vocab_size, (2, 16), generator=g)
+    l0 = float(m0(tokens, tokens).detach())
+    l1 = float(m1(tokens, tokens).detach())
+    assert l1 > l0  # aux adds a positive term
+    assert abs((l1 - l0) - MOE_TINY.aux_loss_coef * 1.0) < 0.02

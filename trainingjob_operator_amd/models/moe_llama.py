@@ -22,6 +22,9 @@ class MoELlamaConfig(LlamaConfig):
     n_experts: int = 8
     top_k: int = 2
     expert_ff: Optional[int] = None  # default: intermediate_size // 2
+    # Switch-style load-balance aux loss coefficient (0 disables); the
+    # per-layer aux terms are averaged and added to the LM loss
+    aux_loss_coef: float = 0.01
 
 
 MOE_TINY = MoELlamaConfig(
@@ -79,7 +82,11 @@ class MoELlamaModel(nn.Module):
         per_tok = fused_cross_entropy(
             logits.reshape(T, -1).contiguous(), targets.reshape(T))
         n_valid = (targets.reshape(T) != -100).sum().clamp(min=1)
-        return per_tok.sum() / n_valid
+        loss = per_tok.sum() / n_valid
+        if self.cfg.aux_loss_coef:
+            aux = sum(blk.moe.aux_loss for blk in self.blocks)
+            loss = loss + self.cfg.aux_loss_coef * aux / len(self.blocks)
+        return loss
 
 
 # launcher/EPTrainer look models up by name like the dense families

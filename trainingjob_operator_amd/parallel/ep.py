@@ -147,6 +147,14 @@ class MoEMLP(nn.Module):
         counts = torch.bincount(flat_e, minlength=self.n_experts)
         in_splits = self._expert_owner_splits(counts)
 
+        # Switch-style load-balance auxiliary loss over THIS rank's
+        # tokens: E * sum_e f_e * P_e, where f_e is the routed fraction
+        # (non-differentiable counts) and P_e the mean router prob
+        # (differentiable) — minimized at 1.0 by a uniform router.
+        # Consumed by MoELlamaModel (aux_loss_coef); harmless elsewhere.
+        f = counts.float() / max(T * self.top_k, 1)
+        self.aux_loss = self.n_experts * (f.detach() * probs.mean(0)).sum()
+
         # exchange split sizes: every rank needs how much each peer sends it
         ep = _group_size(self.group)
         if ep > 1:
