@@ -34,13 +34,17 @@ MOE_TINY = MoELlamaConfig(
 
 
 class MoEBlock(nn.Module):
-    def __init__(self, cfg: MoELlamaConfig, ep_group=None):
+    def __init__(self, cfg: MoELlamaConfig, ep_group=None, tp_group=None):
         super().__init__()
         self.cfg = cfg
-        self.attn = Attention(cfg)
+        if tp_group is not None:
+            from ..parallel.tp_llama import TPAttention
+            self.attn = TPAttention(cfg, tp_group)
+        else:
+            self.attn = Attention(cfg)
         ff = cfg.expert_ff or cfg.intermediate_size // 2
         self.moe = MoEMLP(cfg.hidden_size, ff, cfg.n_experts, cfg.top_k,
-                          group=ep_group)
+                          group=ep_group, tp_group=tp_group)
         self.input_norm_weight = nn.Parameter(torch.ones(cfg.hidden_size))
         self.post_attn_norm_weight = nn.Parameter(
             torch.ones(cfg.hidden_size))
@@ -56,11 +60,17 @@ class MoEBlock(nn.Module):
 
 
 class MoELlamaModel(nn.Module):
-    def __init__(self, cfg: MoELlamaConfig, ep_group=None):
+    """ep_group shards experts; tp_group (EP x TP) additionally tensor-
+    shards attention and each expert's matrices. Initialize a sharded
+    instance from an unsharded one with shard_from_full."""
+
+    def __init__(self, cfg: MoELlamaConfig, ep_group=None, tp_group=None):
         super().__init__()
         self.cfg = cfg
+        self.ep_group = ep_group
+        self.tp_group = tp_group
         self.embed = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
-        self.blocks = nn.ModuleList(MoEBlock(cfg, ep_group)
+        self.blocks = nn.ModuleList(MoEBlock(cfg, ep_group, tp_group)
                                     for _ in range(cfg.num_layers))
         self.final_norm_weight = nn.Parameter(torch.ones(cfg.hidden_size))
         self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False)
@@ -87,6 +97,54 @@ class MoELlamaModel(nn.Module):
             aux = sum(blk.moe.aux_loss for blk in self.blocks)
             loss = loss + self.cfg.aux_loss_coef * aux / len(self.blocks)
         return loss
+
+    @torch.no_grad()
+    def shard_from_full(self, full: "MoELlamaModel") -> None:
+        """Load this rank's (ep, tp) shards from an unsharded MoE model:
+        attention slices exactly like TPLlamaModel.shard_from_full; each
+        owned expert (global id = ep_rank * experts_per_rank + local)
+        takes gate/up ROW blocks and down COLUMN blocks of the full
+        expert; norms/embed/head/router replicate."""
+        import torch.distributed as dist
+        from ..parallel.tp import _group_size, shard_from
+        g = self.tp_group
+        ep = self.ep_group
+        ep_rank = dist.get_rank(ep) if (
+            dist.is_initialized() and _group_size(ep) > 1) else 0
+        self.embed.weight.copy_(full.embed.weight)
+        self.lm_head.weight.copy_(full.lm_head.weight)
+        self.final_norm_weight.copy_(full.final_norm_weight)
+        q_size = self.cfg.num_heads * self.cfg.head_dim
+        kv_size = self.cfg.num_kv_heads * self.cfg.head_dim
+        for blk, fblk in zip(self.blocks, full.blocks):
+            blk.input_norm_weight.copy_(fblk.input_norm_weight)
+            blk.post_attn_norm_weight.copy_(fblk.post_attn_norm_weight)
+            if g is not None:
+                qkv = fblk.attn.qkv_proj.weight
+                wq, wk, wv = qkv.split([q_size, kv_size, kv_size], dim=0)
+                blk.attn.q_proj.weight.copy_(shard_from(wq, 0, g))
+                blk.attn.k_proj.weight.copy_(shard_from(wk, 0, g))
+                blk.attn.v_proj.weight.copy_(shard_from(wv, 0, g))
+                blk.attn.o_proj.weight.copy_(
+                    shard_from(fblk.attn.o_proj.weight, 1, g))
+            else:
+                blk.attn.qkv_proj.weight.copy_(fblk.attn.qkv_proj.weight)
+                blk.attn.o_proj.weight.copy_(fblk.attn.o_proj.weight)
+            blk.moe.router.weight.copy_(fblk.moe.router.weight)
+            per = blk.moe.experts_per_rank
+            for le, ex in enumerate(blk.moe.experts):
+                src = fblk.moe.experts[ep_rank * per + le]
+                if g is not None:
+                    ex.gate_proj.weight.copy_(
+                        shard_from(src.gate_proj.weight, 0, g))
+                    ex.up_proj.weight.copy_(
+                        shard_from(src.up_proj.weight, 0, g))
+                    ex.down_proj.weight.copy_(
+                        shard_from(src.down_proj.weight, 1, g))
+                else:
+                    ex.gate_proj.weight.copy_(src.gate_proj.weight)
+                    ex.up_proj.weight.copy_(src.up_proj.weight)
+                    ex.down_proj.weight.copy_(src.down_proj.weight)
 
 
 # launcher/EPTrainer look models up by name like the dense families

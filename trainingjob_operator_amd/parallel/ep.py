@@ -100,23 +100,54 @@ class Expert(nn.Module):
                                      self.up_proj(x).contiguous()))
 
 
+class TPExpert(nn.Module):
+    """One SwiGLU expert with its weights TENSOR-sharded over the TP group
+    (EP x TP composition: experts split across EP ranks, each expert's
+    matrices split across TP ranks). Input tokens arrive replicated on
+    every tp peer (each peer ran its own ep-plane all-to-all on identical
+    activations); copy_input=False because the single f-op at the MoE
+    entry already sums the partial input grads — a per-expert f would
+    double-count. The Row all-reduce re-assembles full outputs."""
+
+    def __init__(self, hidden: int, ff: int, tp_group=None):
+        super().__init__()
+        from .tp import ColumnParallelLinear, RowParallelLinear
+        self.gate_proj = ColumnParallelLinear(hidden, ff, tp_group,
+                                              copy_input=False)
+        self.up_proj = ColumnParallelLinear(hidden, ff, tp_group,
+                                            copy_input=False)
+        self.down_proj = RowParallelLinear(ff, hidden, tp_group,
+                                           reduce_output=True)
+
+    def forward(self, x):
+        return self.down_proj(swiglu(self.gate_proj(x).contiguous(),
+                                     self.up_proj(x).contiguous()))
+
+
 class MoEMLP(nn.Module):
     """Dropless top-k mixture-of-experts MLP, experts sharded over the EP
     group. With ep_size == 1 it is a plain (single-process) MoE — the
-    reference the EP tests compare against."""
+    reference the EP tests compare against. With a tp_group the experts
+    are additionally tensor-sharded (TPExpert)."""
 
     def __init__(self, hidden: int, ff: int, n_experts: int, top_k: int = 2,
-                 group=None):
+                 group=None, tp_group=None):
         super().__init__()
         self.group = group
+        self.tp_group = tp_group
         self.n_experts = n_experts
         self.top_k = top_k
         ep = _group_size(group)
         assert n_experts % ep == 0, (n_experts, ep)
         self.experts_per_rank = n_experts // ep
         self.router = nn.Linear(hidden, n_experts, bias=False)
-        self.experts = nn.ModuleList(
-            Expert(hidden, ff) for _ in range(self.experts_per_rank))
+        if tp_group is not None and _group_size(tp_group) > 1:
+            self.experts = nn.ModuleList(
+                TPExpert(hidden, ff, tp_group)
+                for _ in range(self.experts_per_rank))
+        else:
+            self.experts = nn.ModuleList(
+                Expert(hidden, ff) for _ in range(self.experts_per_rank))
 
     def _expert_owner_splits(self, counts: torch.Tensor) -> List[int]:
         """counts per expert [E] -> tokens destined per EP rank."""
@@ -127,6 +158,12 @@ class MoEMLP(nn.Module):
         orig_shape = x.shape
         H = orig_shape[-1]
         xt = x.reshape(-1, H)
+        if self.tp_group is not None and _group_size(self.tp_group) > 1:
+            # the ONE f-op for the whole MoE under EP x TP: forward
+            # identity, backward sums the tp peers' partial input grads
+            # (each peer only backprops its own expert shards)
+            from .tp import _CopyToTP
+            xt = _CopyToTP.apply(xt, self.tp_group)
         T = xt.shape[0]
 
         # routing in fp32 regardless of model dtype (gate numerics);
