@@ -378,3 +378,131 @@ def test_ep_x_tp_model_matches_single_process():
     the MoE model reproduce the unsharded loss and every gradient class
     (dense, attention shards, router, expert shards)."""
     mp.spawn(_epxtp_worker, args=(4, _free_port()), nprocs=4, join=True)
+
+
+def _epxtp_trainer_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from trainingjob_operator_amd.models.moe_llama import (
+            MOE_TINY, MoELlamaModel,
+        )
+        from trainingjob_operator_amd.ops import make_inv_freq
+        from trainingjob_operator_amd.parallel.ep import (
+            EPTrainer, diversify_experts, solo_group,
+        )
+        from trainingjob_operator_amd.parallel.tp import shard_from
+        from trainingjob_operator_amd.training import (
+            TrainConfig, synthetic_batches,
+        )
+        cfg = TrainConfig(model="moe-tiny", micro_batch=2, grad_accum=2,
+                          seq_len=16, lr=2e-3, clip_grad_norm=1.0)
+        tr = EPTrainer(cfg, ep_size=2, tp_size=2)     # world 4 = ep2 x tp2
+        topo = tr.topo
+        assert (topo.edp_size, topo.ep_size, topo.tp_size) == (1, 2, 2)
+        assert topo.data_replicas == 2
+
+        # single-process reference accumulating BOTH data replicas' micros
+        solo = solo_group()
+        torch.manual_seed(cfg.seed)
+        ref = MoELlamaModel(MOE_TINY, ep_group=solo)
+        diversify_experts(ref, cfg.seed, ep_rank=0)
+        ref = ref.to(torch.bfloat16)
+        ref.inv_freq = make_inv_freq(MOE_TINY.head_dim, MOE_TINY.rope_theta)
+        for r in range(topo.data_replicas):
+            data = synthetic_batches(cfg, torch.device("cpu"), rank=r)
+            for _ in range(cfg.grad_accum):
+                tokens, targets = next(data)
+                (ref(tokens, targets)
+                 / (cfg.grad_accum * topo.data_replicas)).backward()
+
+        # trainer's first step up to the gradient seam
+        for _ in range(cfg.grad_accum):
+            tokens, targets = next(tr.data)
+            (tr.model(tokens, targets) / cfg.grad_accum).backward()
+        tr._reduce_grads()
+
+        tol = dict(atol=3e-2, rtol=8e-2)   # bf16 + tp re-association
+        ref_named = dict(ref.named_parameters())
+        blk = tr.model.blocks[0]
+        fblk = ref.blocks[0]
+        # replicated: router + norm (dense-dp-averaged over both replicas)
+        assert torch.allclose(blk.moe.router.weight.grad.float(),
+                              ref_named["blocks.0.moe.router.weight"]
+                              .grad.float(), **tol)
+        assert torch.allclose(blk.input_norm_weight.grad.float(),
+                              fblk.input_norm_weight.grad.float(), **tol)
+        # attention tp shard
+        q_size = MOE_TINY.num_heads * MOE_TINY.head_dim
+        kv = MOE_TINY.num_kv_heads * MOE_TINY.head_dim
+        gq = fblk.attn.qkv_proj.weight.grad.split([q_size, kv, kv], 0)[0]
+        assert torch.allclose(blk.attn.q_proj.weight.grad.float(),
+                              shard_from(gq, 0, topo.tp_group).float(),
+                              **tol)
+        # expert (ep x tp) shard
+        per = blk.moe.experts_per_rank
+        for le, ex in enumerate(blk.moe.experts):
+            src = fblk.moe.experts[topo.ep_rank * per + le]
+            if src.gate_proj.weight.grad is None:
+                continue
+            assert torch.allclose(
+                ex.gate_proj.weight.grad.float(),
+                shard_from(src.gate_proj.weight.grad, 0,
+                           topo.tp_group).float(), **tol)
+
+        # finish the step and take two more full steps
+        tr._clip_grads()
+        tr.opt.step(grad_pre_scale=1.0)
+        tr.opt.zero_grad()
+        tr.step_count += 1
+        for _ in range(2):
+            loss = tr.train_step()
+        # tp peers agree on the loss bit-for-bit (identical data + model)
+        mx = loss.clone()
+        mn = loss.clone()
+        dist.all_reduce(mx, op=dist.ReduceOp.MAX, group=topo.tp_group)
+        dist.all_reduce(mn, op=dist.ReduceOp.MIN, group=topo.tp_group)
+        assert torch.equal(mx, mn)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_ep_x_tp_trainer_matches_single_process():
+    """EP2 x TP2 trainer on 4 gloo ranks: first-step gradients (every
+    shard class) match a single process over both data replicas' batches;
+    tp peers stay loss-identical across optimizer steps."""
+    mp.spawn(_epxtp_trainer_worker, args=(4, _free_port()), nprocs=4,
+             join=True)
+
+
+def _full_epxtp_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from trainingjob_operator_amd.parallel.ep import EPTrainer
+        from trainingjob_operator_amd.training import TrainConfig
+        cfg = TrainConfig(model="moe-tiny", micro_batch=1, grad_accum=2,
+                          seq_len=16, lr=2e-3, clip_grad_norm=1.0)
+        tr = EPTrainer(cfg, ep_size=2, tp_size=2)  # 8 = edp2 x ep2 x tp2
+        topo = tr.topo
+        assert (topo.edp_size, topo.ep_size, topo.tp_size) == (2, 2, 2)
+        for _ in range(2):
+            tr.train_step()
+        # edp peers (same (ep, tp) shard, different data) bit-identical
+        flat = tr.store.flat_param
+        peers = [torch.empty_like(flat) for _ in range(topo.edp_size)]
+        dist.all_gather(peers, flat, group=topo.edp_group)
+        assert torch.equal(peers[0], peers[1])
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_full_edp_ep_tp_grid_trains():
+    """The full edp2 x ep2 x tp2 3D MoE grid on 8 gloo ranks steps and
+    keeps expert-dp peers bit-identical."""
+    mp.spawn(_full_epxtp_worker, args=(8, _free_port()), nprocs=8,
+             join=True)

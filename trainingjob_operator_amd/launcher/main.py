@@ -91,6 +91,7 @@ def main(argv=None) -> int:
     if args.sp:
         assert args.tp > 1, "--sp requires --tp > 1"
         assert args.seq_len % args.tp == 0, "--sp needs seq_len % tp == 0"
+        assert not args.ep, "--sp with --ep is not supported"
     cfg = TrainConfig(
         model=args.model, micro_batch=args.micro_batch,
         grad_accum=args.grad_accum, seq_len=args.seq_len, lr=args.lr,
@@ -110,16 +111,17 @@ def main(argv=None) -> int:
                else f"stage{grid.pp_rank}")
         ckpt = Checkpointer(os.path.join(args.ckpt_dir, sub))
     elif args.ep:
-        assert args.tp == 1, "EPxTP composition is roadmap"
         from ..parallel.ep import EPTrainer
         import torch
         dev = torch.device(f"cuda:{ctx.local_rank}"
                            if torch.cuda.is_available() else "cpu")
-        trainer = EPTrainer(cfg, ep_size=args.ep, device=dev)
-        # each ep rank owns a distinct expert shard: per-ep-rank streams,
+        trainer = EPTrainer(cfg, ep_size=args.ep, device=dev,
+                            tp_size=args.tp)
+        # each (ep, tp) coordinate owns a distinct shard: its own stream,
         # written by the edp_rank==0 replica
-        ckpt = Checkpointer(os.path.join(
-            args.ckpt_dir, f"ep{trainer.topo.ep_rank}"))
+        sub = (f"ep{trainer.topo.ep_rank}_tp{trainer.topo.tp_rank}"
+               if args.tp > 1 else f"ep{trainer.topo.ep_rank}")
+        ckpt = Checkpointer(os.path.join(args.ckpt_dir, sub))
     else:
         trainer = Trainer(cfg, ctx)
         if args.tp > 1:

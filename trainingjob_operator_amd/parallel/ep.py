@@ -274,10 +274,16 @@ class MoEMLP(nn.Module):
 
 @dataclass
 class EPTopology:
-    """world = edp x ep. EP groups are CONTIGUOUS ranks (token all-to-all
-    stays on adjacent GPUs over direct xGMI links); expert-dp peers — the
-    ranks holding the same expert shard in different EP groups — are
-    strided by ep_size."""
+    """world = edp x ep x tp, rank = ((edp_r*ep + ep_r)*tp + tp_r) —
+    tp innermost (adjacent GPUs over direct xGMI), then EP planes (token
+    all-to-all among same-tp ranks), then expert-dp.
+
+    Groups:
+      ep_group      token-dispatch plane: same (edp_r, tp_r)
+      edp_group     same expert shard across edp replicas: same (ep_r, tp_r)
+      tp_group      tensor shards of one model instance: same (edp_r, ep_r)
+      dense_dp_group  all data replicas of a dense/attn shard: same tp_r
+    None means the default (whole-world) group."""
     world: int
     rank: int
     ep_size: int
@@ -286,35 +292,84 @@ class EPTopology:
     edp_size: int
     edp_rank: int
     edp_group: Optional[object]
+    tp_size: int = 1
+    tp_rank: int = 0
+    tp_group: Optional[object] = None
+    dense_dp_group: Optional[object] = None
+    data_rank: int = 0        # index among the edp*ep data replicas
+    data_replicas: int = 1
 
 
-def build_ep_topology(ep_size: int = 0) -> EPTopology:
-    """ep_size == 0 means 'the whole world is one EP group'. Every rank
-    must call with the same ep_size (new_group is collective)."""
+def build_ep_topology(ep_size: int = 0, tp_size: int = 1) -> EPTopology:
+    """ep_size == 0 means 'everything left after tp is one EP group'.
+    Every rank must call with the same sizes (new_group is collective)."""
     if not dist.is_initialized():
-        return EPTopology(1, 0, max(ep_size, 1), 0, None, 1, 0, None)
+        return EPTopology(1, 0, max(ep_size, 1), 0, None, 1, 0, None,
+                          tp_size=max(tp_size, 1))
     world = dist.get_world_size()
     rank = dist.get_rank()
-    ep = ep_size or world
-    if world % ep != 0:
-        raise ValueError(f"world {world} not divisible by ep={ep}")
-    edp = world // ep
-    ep_rank, edp_rank = rank % ep, rank // ep
-    ep_group = edp_group = None
-    if ep > 1 and edp > 1:
-        for g in range(edp):
-            grp = dist.new_group(list(range(g * ep, (g + 1) * ep)))
-            if g == edp_rank:
-                ep_group = grp
-    elif ep > 1:
-        ep_group = None  # whole world == default group
+    tp = max(tp_size, 1)
+    if world % tp != 0:
+        raise ValueError(f"world {world} not divisible by tp={tp}")
+    ep = ep_size or world // tp
+    if (world // tp) % ep != 0:
+        raise ValueError(f"world {world} not divisible by ep*tp={ep * tp}")
+    edp = world // (ep * tp)
+    tp_r = rank % tp
+    ep_r = (rank // tp) % ep
+    edp_r = rank // (ep * tp)
+
+    def _mk(groups_ranks, mine_key):
+        mine = None
+        for key, ranks in groups_ranks:
+            if len(ranks) == world:
+                return None if key == mine_key else mine
+            grp = dist.new_group(ranks)
+            if key == mine_key:
+                mine = grp
+        return mine
+
+    ep_group = edp_group = tp_group = dense_dp_group = None
+    if ep > 1:
+        ep_group = _mk([((d, t), [(d * ep + e) * tp + t
+                                  for e in range(ep)])
+                        for d in range(edp) for t in range(tp)],
+                       (edp_r, tp_r))
     if edp > 1:
-        for t in range(ep):
-            grp = dist.new_group(list(range(t, world, ep)))
-            if t == ep_rank:
-                edp_group = grp
-    return EPTopology(world, rank, ep, ep_rank, ep_group,
-                      edp, edp_rank, edp_group)
+        edp_group = _mk([((e, t), [(d * ep + e) * tp + t
+                                   for d in range(edp)])
+                         for e in range(ep) for t in range(tp)],
+                        (ep_r, tp_r))
+    if tp > 1:
+        tp_group = _mk([((d, e), [(d * ep + e) * tp + t
+                                  for t in range(tp)])
+                        for d in range(edp) for e in range(ep)],
+                       (edp_r, ep_r))
+    if edp * ep > 1:
+        dense_dp_group = _mk([(t, [(d * ep + e) * tp + t
+                                   for d in range(edp)
+                                   for e in range(ep)])
+                              for t in range(tp)], tp_r)
+    return EPTopology(world, rank, ep, ep_r, ep_group,
+                      edp, edp_r, edp_group,
+                      tp_size=tp, tp_rank=tp_r, tp_group=tp_group,
+                      dense_dp_group=dense_dp_group,
+                      data_rank=edp_r * ep + ep_r,
+                      data_replicas=edp * ep)
+
+
+def solo_group():
+    """A 1-member group per rank (collective: every rank creates every
+    solo group). Passing it as ep_group yields a truly UNSHARDED model
+    inside an initialized world, where group=None would mean world."""
+    if not dist.is_initialized():
+        return None
+    mine = None
+    for r in range(dist.get_world_size()):
+        g = dist.new_group([r])
+        if r == dist.get_rank():
+            mine = g
+    return mine
 
 
 def diversify_experts(model, seed: int, ep_rank: int = 0) -> None:
@@ -340,21 +395,28 @@ def diversify_experts(model, seed: int, ep_rank: int = 0) -> None:
 
 
 class EPTrainer:
-    """DP x EP trainer for the MoE-Llama family: every rank is a data-
-    parallel worker over its own batch stream; each EP group shards the
-    experts and exchanges tokens by all-to-all; dense (non-expert)
-    parameters are replicated world-wide.
+    """DP x EP (x TP) trainer for the MoE-Llama family: every (edp, ep)
+    coordinate is a data-parallel worker over its own batch stream (tp
+    peers share data); each EP plane shards the experts and exchanges
+    tokens by all-to-all; with tp_size > 1 attention and each expert's
+    matrices are additionally tensor-sharded (TPExpert).
 
-    Gradient seams (loss = mean over the global batch):
-      * whole flat grad pre-scaled by 1/world,
-      * dense spans all-reduced over the WORLD,
+    Gradient seams (loss = mean over the global batch of edp*ep data
+    replicas):
+      * whole flat grad pre-scaled by 1/(edp*ep),
       * expert spans all-reduced over the expert-dp group (the edp peers
-        holding the same shard) — a no-op when ep == world.
-    Global grad-norm clip counts each expert once: dense normsq (identical
-    on every rank) + expert normsq all-reduced over the EP group.
+        holding the same shard) — a no-op when edp == 1,
+      * every other span (dense replicated + attention tp-shards)
+        all-reduced over the dense-dp group (all data replicas at my
+        tp_rank; the whole world when tp == 1).
+    The global grad-norm clip counts every shard exactly once:
+      * expert normsq summed over the ep plane AND the tp group,
+      * attention-shard normsq summed over the tp group,
+      * replicated normsq counted locally (identical on every rank).
     """
 
-    def __init__(self, cfg, ep_size: int = 0, device=None):
+    def __init__(self, cfg, ep_size: int = 0, device=None,
+                 tp_size: int = 1):
         from ..models.config import CONFIGS
         from ..models.moe_llama import MoELlamaModel
         from ..optim import FlatAdamW
@@ -364,50 +426,90 @@ class EPTrainer:
         self.cfg = cfg
         mcfg = CONFIGS[cfg.model]
         self.device = torch.device(device or "cpu")
-        self.topo = build_ep_topology(ep_size)
+        self.topo = build_ep_topology(ep_size, tp_size)
         torch.manual_seed(cfg.seed)       # identical dense init everywhere
-        with torch.device(self.device):
-            model = MoELlamaModel(mcfg, ep_group=self.topo.ep_group)
-        diversify_experts(model, cfg.seed, self.topo.ep_rank)
+        if self.topo.tp_size > 1:
+            # build the unsharded model once and slice both tp and ep
+            # shards from it (exact shard_from_full semantics)
+            with torch.device(self.device):
+                full = MoELlamaModel(mcfg, ep_group=solo_group())
+            diversify_experts(full, cfg.seed, ep_rank=0)  # all experts
+            with torch.device(self.device):
+                model = MoELlamaModel(mcfg, ep_group=self.topo.ep_group,
+                                      tp_group=self.topo.tp_group)
+            model.shard_from_full(full)
+            del full
+        else:
+            with torch.device(self.device):
+                model = MoELlamaModel(mcfg, ep_group=self.topo.ep_group)
+            diversify_experts(model, cfg.seed, self.topo.ep_rank)
         self.model = model.to(torch.bfloat16)
         from ..ops import make_inv_freq
         self.model.inv_freq = make_inv_freq(mcfg.head_dim, mcfg.rope_theta,
                                             device=self.device)
         self.store = FlatParamStore(self.model, device=self.device)
-        # optimizer clips nothing; EPTrainer applies the EP-aware clip
+        # optimizer clips nothing; EPTrainer applies the topology-aware clip
         self.opt = FlatAdamW(self.store, lr=cfg.lr, betas=cfg.betas,
                              weight_decay=cfg.weight_decay,
                              clip_grad_norm=0.0)
-        self.expert_spans, self.dense_spans = classify_spans(
+        self.expert_spans, rest = classify_spans(
             self.store, lambda n: ".experts." in n)
-        self.data = synthetic_batches(cfg, self.device, rank=self.topo.rank)
+        if self.topo.tp_size > 1:
+            named = dict(self.model.named_parameters())
+            shard_rest = {n for n in named
+                          if ".experts." not in n
+                          and getattr(named[n], "tp_sharded", False)}
+            self.attn_shard_spans, _ = classify_spans(
+                self.store, lambda n: n in shard_rest)
+            _, self.replicated_spans = classify_spans(
+                self.store,
+                lambda n: n in shard_rest or ".experts." in n)
+        else:
+            self.attn_shard_spans = []
+            self.replicated_spans = rest
+        self.dense_spans = rest   # everything non-expert: one sync class
+        self.data = synthetic_batches(cfg, self.device,
+                                      rank=self.topo.data_rank)
         self.step_count = 0
 
     def _reduce_grads(self) -> None:
-        if self.topo.world == 1:
+        t = self.topo
+        if t.data_replicas == 1:
             return
         fg = self.store.flat_grad
-        fg.mul_(1.0 / self.topo.world)
+        fg.mul_(1.0 / t.data_replicas)
         for s, e in self.dense_spans:
-            dist.all_reduce(fg[s:e])                       # whole world
-        if self.topo.edp_size > 1:
+            dist.all_reduce(fg[s:e], group=t.dense_dp_group)
+        if t.edp_size > 1:
             for s, e in self.expert_spans:
-                dist.all_reduce(fg[s:e], group=self.topo.edp_group)
+                dist.all_reduce(fg[s:e], group=t.edp_group)
 
     def _clip_grads(self) -> None:
         clip = self.cfg.clip_grad_norm
         if not clip or clip <= 0:
             return
+        t = self.topo
         fg = self.store.flat_grad
         expert_nsq = fg.new_zeros((), dtype=torch.float32)
         for s, e in self.expert_spans:
             expert_nsq += fg[s:e].float().pow(2).sum()
-        dense_nsq = fg.new_zeros((), dtype=torch.float32)
-        for s, e in self.dense_spans:
-            dense_nsq += fg[s:e].float().pow(2).sum()
-        if self.topo.ep_size > 1 and self.topo.world > 1:
-            dist.all_reduce(expert_nsq, group=self.topo.ep_group)
-        gnorm = (dense_nsq + expert_nsq).sqrt()
+        if t.ep_size > 1 and t.world > 1:
+            dist.all_reduce(expert_nsq, group=t.ep_group)
+        if t.tp_size > 1:
+            dist.all_reduce(expert_nsq, group=t.tp_group)
+            shard_nsq = fg.new_zeros((), dtype=torch.float32)
+            for s, e in self.attn_shard_spans:
+                shard_nsq += fg[s:e].float().pow(2).sum()
+            dist.all_reduce(shard_nsq, group=t.tp_group)
+            rep_nsq = fg.new_zeros((), dtype=torch.float32)
+            for s, e in self.replicated_spans:
+                rep_nsq += fg[s:e].float().pow(2).sum()
+            gnorm = (expert_nsq + shard_nsq + rep_nsq).sqrt()
+        else:
+            dense_nsq = fg.new_zeros((), dtype=torch.float32)
+            for s, e in self.dense_spans:
+                dense_nsq += fg[s:e].float().pow(2).sum()
+            gnorm = (dense_nsq + expert_nsq).sqrt()
         if float(gnorm) > clip:
             fg.mul_(clip / float(gnorm))
 
