@@ -435,3 +435,92 @@ def test_full_3d_grid_trains():
     """DP2 x PP2 x TP2 on 8 gloo ranks: the full 3D composition steps and
     keeps dp peers bit-identical."""
     mp.spawn(_full3d_worker, args=(8, _free_port()), nprocs=8, join=True)
+
+
+def _ppxep_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from trainingjob_operator_amd.models.moe_llama import (
+            MOE_TINY, MoELlamaModel,
+        )
+        from trainingjob_operator_amd.ops import make_inv_freq
+        from trainingjob_operator_amd.parallel.ep import (
+            diversify_experts, solo_group,
+        )
+        from trainingjob_operator_amd.parallel.groups import build_grid
+        from trainingjob_operator_amd.parallel.pp import (
+            PPTrainer, partition_layers,
+        )
+        from trainingjob_operator_amd.training import (
+            TrainConfig, synthetic_batches,
+        )
+        cfg = TrainConfig(model="moe-tiny", micro_batch=2, grad_accum=2,
+                          seq_len=16, lr=2e-3, clip_grad_norm=1.0)
+        grid = build_grid(tp_size=1, pp_size=2)   # dp axis == EP plane
+        assert (grid.dp_size, grid.pp_size) == (2, 2)
+        tr = PPTrainer(cfg, grid=grid, schedule="gpipe")
+        assert tr._expert_spans, "MoE stage must classify expert spans"
+
+        # single-process reference over BOTH pipelines' data streams
+        solo = solo_group()
+        torch.manual_seed(cfg.seed)
+        ref = MoELlamaModel(MOE_TINY, ep_group=solo)
+        diversify_experts(ref, cfg.seed, ep_rank=0)
+        ref = ref.to(torch.bfloat16)
+        ref.inv_freq = make_inv_freq(MOE_TINY.head_dim, MOE_TINY.rope_theta)
+        for r in range(grid.dp_size):
+            data = synthetic_batches(cfg, torch.device("cpu"), rank=r)
+            for _ in range(cfg.grad_accum):
+                tokens, targets = next(data)
+                (ref(tokens, targets)
+                 / (cfg.grad_accum * grid.dp_size)).backward()
+
+        # one schedule pass + the PP x EP gradient seam (no optimizer)
+        micros = [next(tr.data) for _ in range(cfg.grad_accum)]
+        tr.sched.step(micros, tr.hidden_shape, tr.act_dtype)
+        fg = tr.store.flat_grad
+        fg.mul_(1.0 / grid.dp_size)
+        for s_, e_ in tr._moe_dense_spans:
+            dist.all_reduce(fg[s_:e_], group=grid.dp_group)
+
+        tol = dict(atol=3e-2, rtol=8e-2)
+        ref_named = dict(ref.named_parameters())
+        parts = [list(r_) for r_ in
+                 partition_layers(MOE_TINY.num_layers, grid.pp_size)]
+        per = tr.stage.blocks[0].moe.experts_per_rank
+        for name in tr.store.offsets:
+            rname = name
+            if name.startswith("blocks."):
+                loc, tail = name[len("blocks."):].split(".", 1)
+                glob = parts[grid.pp_rank][int(loc)]
+                if ".experts." in tail:
+                    pre, rest = tail.split(".experts.")
+                    le, t2 = rest.split(".", 1)
+                    tail = (f"{pre}.experts."
+                            f"{grid.dp_rank * per + int(le)}.{t2}")
+                rname = f"blocks.{glob}.{tail}"
+            mine = tr.store.grad_view(name).float()
+            want = ref_named[rname].grad.float().reshape(-1)
+            assert torch.allclose(mine, want, **tol), \
+                f"{name}: max err {(mine - want).abs().max()}"
+
+        # finish: clip + step, then two full steps through train_step
+        tr._clip_grads()
+        tr.opt.step()
+        tr.opt.zero_grad()
+        tr.step_count += 1
+        for _ in range(2):
+            tr.train_step()
+        assert tr.step_count == 3
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_pp_x_ep_grid_matches_single_process():
+    """PP2 x EP2 on 4 gloo ranks (MoE stages, experts sharded across the
+    stage plane): first-step gradients — including middle-stage router
+    aux grads — match a single process over both pipelines' batches."""
+    mp.spawn(_ppxep_worker, args=(4, _free_port()), nprocs=4, join=True)

@@ -98,6 +98,9 @@ def main(argv=None) -> int:
         checkpoint_activations=args.checkpoint_activations,
         tp_size=args.tp, sequence_parallel=args.sp)
     if args.pp > 1:
+        assert not args.ep, \
+            "--pp with a MoE model shards experts automatically: the " \
+            "grid's dp axis IS the EP plane (drop --ep)"
         assert ctx.world_size % (args.pp * max(args.tp, 1)) == 0, \
             f"world {ctx.world_size} not divisible by " \
             f"pp*tp={args.pp * max(args.tp, 1)}"

@@ -112,6 +112,48 @@ class LlamaStage(nn.Module):
         n_valid = (targets.reshape(T) != -100).sum().clamp(min=1)
         return per_tok.sum() / n_valid
 
+    def pop_aux(self):
+        """Auxiliary loss of the micro-batch that JUST ran forward, for
+        stages whose extra losses cannot ride the pipeline (MoEStage);
+        dense stages have none."""
+        return None
+
+
+class MoEStage(LlamaStage):
+    """Pipeline stage of a MoELlamaModel (PP x EP): blocks are MoEBlocks
+    whose MoEMLPs dispatch tokens over the stage's EP plane (the ranks
+    holding the other expert shards of the SAME stage). The Switch aux
+    term of this stage's routers backprops LOCALLY via pop_aux — only the
+    last stage owns the LM loss, so middle stages feed their aux gradient
+    into the schedule's backward alongside the received activation
+    grads."""
+
+    total_layers: int = 0
+    aux_coef: float = 0.0
+
+    @classmethod
+    def from_moe_model(cls, model, stage: int,
+                       n_stages: int) -> "MoEStage":
+        st = cls.from_model(model, stage, n_stages)
+        st.total_layers = model.cfg.num_layers
+        st.aux_coef = model.cfg.aux_loss_coef
+        return st
+
+    def _stage_aux(self):
+        aux = sum(blk.moe.aux_loss for blk in self.blocks)
+        return self.aux_coef * aux / max(self.total_layers, 1)
+
+    def forward(self, x, residual, targets=None):
+        out = super().forward(x, residual, targets)
+        if self.is_last and targets is not None and self.aux_coef:
+            out = out + self._stage_aux()   # fold into the stage loss
+        return out
+
+    def pop_aux(self):
+        if self.is_last or not self.aux_coef:
+            return None
+        return self._stage_aux()
+
 
 class GPipeSchedule:
     """Fill-drain (GPipe) schedule: all micro-batch forwards, then all
@@ -169,20 +211,28 @@ class GPipeSchedule:
                 loss = out / n_micro
                 loss_total = loss.detach() if loss_total is None \
                     else loss_total + loss.detach()
-                saved.append((x_in, res_in, loss, None))
+                saved.append((x_in, res_in, loss, None, None))
             else:
                 x_out, res_out = out
+                aux = self.stage.pop_aux()
                 self._send(x_out, self.next_rank)
                 self._send(res_out, self.next_rank)
-                saved.append((x_in, res_in, x_out, res_out))
+                saved.append((x_in, res_in, x_out, res_out,
+                              None if aux is None else aux / n_micro))
 
-        for x_in, res_in, a, b in reversed(saved):
+        for x_in, res_in, a, b, aux in reversed(saved):
             if self.is_last_stage:
                 a.backward()  # loss already scaled by 1/n_micro
             else:
                 dx = self._recv_grad(a)
                 dres = self._recv_grad(b)
-                torch.autograd.backward((a, b), (dx, dres))
+                if aux is None:
+                    torch.autograd.backward((a, b), (dx, dres))
+                else:
+                    # the stage-local aux loss (MoE router balance) rides
+                    # the same backward pass as the pipeline grads
+                    torch.autograd.backward(
+                        (a, b, aux), (dx, dres, torch.ones_like(aux)))
             if not self.is_first_stage:
                 self._send(x_in.grad, self.prev_rank)
                 self._send(res_in.grad, self.prev_rank)
@@ -239,17 +289,21 @@ class OneFOneBSchedule(GPipeSchedule):
                 out = self.stage(x_in, res_in,
                                  targets if self.is_last_stage else None)
             if self.is_last_stage:
-                outstanding.append((x_in, res_in, out / n_micro, None))
+                outstanding.append((x_in, res_in, out / n_micro, None,
+                                    None))
             else:
                 x_out, res_out = out
+                aux = self.stage.pop_aux()
                 self._send(x_out, self.next_rank)
                 self._send(res_out, self.next_rank)
-                outstanding.append((x_in, res_in, x_out, res_out))
+                outstanding.append((x_in, res_in, x_out, res_out,
+                                    None if aux is None
+                                    else aux / n_micro))
             self.peak_live = max(self.peak_live, len(outstanding))
 
         def backward():
             nonlocal loss_total
-            x_in, res_in, a, b = outstanding.pop(0)
+            x_in, res_in, a, b, aux = outstanding.pop(0)
             if self.is_last_stage:
                 loss_total = a.detach() if loss_total is None \
                     else loss_total + a.detach()
@@ -257,7 +311,11 @@ class OneFOneBSchedule(GPipeSchedule):
             else:
                 dx = self._recv_grad(a)
                 dres = self._recv_grad(b)
-                torch.autograd.backward((a, b), (dx, dres))
+                if aux is None:
+                    torch.autograd.backward((a, b), (dx, dres))
+                else:
+                    torch.autograd.backward(
+                        (a, b, aux), (dx, dres, torch.ones_like(aux)))
             if not self.is_first_stage:
                 self._send(x_in.grad, self.prev_rank)
                 self._send(res_in.grad, self.prev_rank)
@@ -312,9 +370,28 @@ class PPTrainer:
         # all stages of MY pipeline (grad-norm seam); None == default
         # group in the pure-PP world where rank == stage
         self.pp_group = grid.pp_group if grid is not None else None
+        from ..models.moe_llama import MoELlamaConfig
+        self._is_moe = isinstance(mcfg, MoELlamaConfig)
         # identical init on every dp replica of a stage (same seed)
         torch.manual_seed(cfg.seed)
-        if grid is not None and grid.tp_size > 1:
+        if self._is_moe:
+            # PP x EP: the grid's dp axis doubles as the EP plane — the
+            # ranks holding the SAME stage in the other pipelines hold
+            # the other expert shards of that stage (groups.py: "EP == DP
+            # for MoE"). Pure PP (no grid / dp 1) keeps all experts local.
+            assert grid is None or grid.tp_size == 1, \
+                "PP x EP x TP is roadmap"
+            from ..models.moe_llama import MoELlamaModel
+            from .ep import diversify_experts, solo_group
+            if grid is not None and grid.dp_size > 1:
+                ep_plane, plane_rank = grid.dp_group, grid.dp_rank
+            else:
+                ep_plane, plane_rank = solo_group(), 0
+            with torch.device(self.device):
+                moem = MoELlamaModel(mcfg, ep_group=ep_plane)
+            diversify_experts(moem, cfg.seed, ep_rank=plane_rank)
+            self.stage = MoEStage.from_moe_model(moem, stage_idx, n_stages)
+        elif grid is not None and grid.tp_size > 1:
             # PP x TP: slice a tensor-parallel model into stages — the
             # stage machinery is block-generic (TPBlock outputs the same
             # full-size (x, residual) pair after its row-parallel
@@ -354,6 +431,11 @@ class PPTrainer:
                 self.store,
                 lambda n: getattr(named[n], "tp_sharded", False))
             self.tp_group = grid.tp_group
+        self._expert_spans = self._moe_dense_spans = None
+        if self._is_moe:
+            from .flat import classify_spans
+            self._expert_spans, self._moe_dense_spans = classify_spans(
+                self.store, lambda n: ".experts." in n)
         # every pp rank of a replica draws the same stream; replicas draw
         # DISTINCT streams (dp_rank-keyed), like the DP Trainer
         self.data = synthetic_batches(cfg, self.device, rank=self.dp_rank)
@@ -378,6 +460,16 @@ class PPTrainer:
             dist.all_reduce(nsq, group=self.tp_group)
             for s_, e_ in replicated:
                 nsq += fg[s_:e_].float().pow(2).sum()
+        elif self._expert_spans is not None and self.dp_size > 1:
+            # PP x EP: each plane member holds distinct experts — sum
+            # their normsq over the plane; dense counted once (identical
+            # across the plane after the all-reduce above)
+            nsq = fg.new_zeros((), dtype=torch.float32)
+            for s_, e_ in self._expert_spans:
+                nsq += fg[s_:e_].float().pow(2).sum()
+            dist.all_reduce(nsq, group=self.dp_group)
+            for s_, e_ in self._moe_dense_spans:
+                nsq += fg[s_:e_].float().pow(2).sum()
         else:
             nsq = fg.float().pow(2).sum()
         if dist.is_initialized() and self.sched.n > 1:
@@ -392,8 +484,17 @@ class PPTrainer:
         if self.dp_size > 1:
             # stage-peer gradient seam: SUM with 1/world pre-scale (gloo
             # has no AVG; RCCL path matches ddp.py's convention)
-            self.store.flat_grad.mul_(1.0 / self.dp_size)
-            dist.all_reduce(self.store.flat_grad, group=self.dp_group)
+            fg = self.store.flat_grad
+            fg.mul_(1.0 / self.dp_size)
+            if self._expert_spans is None:
+                dist.all_reduce(fg, group=self.dp_group)
+            else:
+                # PP x EP: the "dp" peers hold DIFFERENT experts at the
+                # same flat offsets; expert grads are already complete
+                # (the backward all-to-all summed every pipeline's
+                # tokens) and only need the 1/replicas scale
+                for s_, e_ in self._moe_dense_spans:
+                    dist.all_reduce(fg[s_:e_], group=self.dp_group)
             if loss is not None:
                 loss = loss / self.dp_size
                 dist.all_reduce(loss, group=self.dp_group)
