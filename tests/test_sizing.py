@@ -79,3 +79,18 @@ def test_declared_params_precedence():
     assert sizing.declared_params(j) == 123  # count wins over name
     assert sizing.declared_params(_job(model_ann="nope")) is None
     assert sizing.declared_params(_job()) is None
+
+
+def test_sharded_model_admission():
+    """model-shards annotation: a pp x tp sharded 70B fits 8 x 1-GPU pods;
+    under-provisioned shard counts are rejected."""
+    j = _job(model_ann="llama3-70b", gpus=1)
+    j.annotations[sizing.MODEL_SHARDS_ANNOTATION] = "8"
+    j.spec.replica_specs["trainer"].replicas = 8
+    assert validate(j) == []
+    # unsharded 70B on 1-GPU pods is rejected
+    assert validate(_job(model_ann="llama3-70b", gpus=1)) != []
+    # declaring more shards than provided GPUs is rejected
+    j.spec.replica_specs["trainer"].replicas = 4
+    errs = validate(j)
+    assert len(errs) == 1 and "model-shards=8" in errs[0]

@@ -23,6 +23,8 @@ from .constants import CRD_GROUP, MI355X_HBM_BYTES
 
 MODEL_PARAMS_ANNOTATION = f"{CRD_GROUP}/model-params"
 MODEL_NAME_ANNOTATION = f"{CRD_GROUP}/model"
+# model sharded over N devices (pp x tp): per-GPU state = total / N
+MODEL_SHARDS_ANNOTATION = f"{CRD_GROUP}/model-shards"
 
 # ~8% HBM headroom: allocator fragmentation, RCCL buffers, HIP runtime.
 _HEADROOM = 0.92
@@ -95,6 +97,7 @@ def min_gpus_for(estimate: MemoryEstimate,
 # (models/config.py); annotation "elasticdeeplearning.ai/model" selects one.
 KNOWN_MODELS = {
     "llama3-8b": 8_030_000_000,
+    "llama3-70b": 70_600_000_000,
     "llama-1b": 1_100_000_000,
 }
 
@@ -111,3 +114,12 @@ def declared_params(job) -> Optional[int]:
     if name:
         return KNOWN_MODELS.get(name)
     return None
+
+
+def declared_shards(job) -> int:
+    """How many ways the model is sharded across devices (the launcher's
+    pp x tp degree), from the model-shards annotation; 1 = unsharded/DP."""
+    try:
+        return max(1, int(job.annotations.get(MODEL_SHARDS_ANNOTATION, 1)))
+    except (TypeError, ValueError):
+        return 1

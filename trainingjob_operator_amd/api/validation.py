@@ -145,18 +145,26 @@ def validate_hbm_sizing(job: AITrainingJob) -> List[str]:
     n_params = sizing.declared_params(job)
     if n_params is None:
         return errors
-    est = sizing.estimate_training_bytes(n_params)
+    # a pp x tp sharded model (model-shards annotation) spreads its state
+    # over N devices: admission checks the per-shard slice
+    shards = sizing.declared_shards(job)
+    est = sizing.estimate_training_bytes(n_params // shards)
     for rtype, rs in job.spec.replica_specs.items():
         gpus = gpus_requested(rs)
         if gpus <= 0:
+            continue
+        if shards > 1 and (rs.replicas or 0) * gpus < shards:
+            errors.append(
+                f"spec.replicaSpecs[{rtype}]: model-shards={shards} but "
+                f"the job only provides {(rs.replicas or 0) * gpus} GPUs")
             continue
         if not sizing.fits_per_gpu(est, gpus):
             need = sizing.min_gpus_for(est)
             errors.append(
                 f"spec.replicaSpecs[{rtype}]: model of {n_params:,} params "
-                f"needs ~{est.total_gb:.0f} GB of HBM per replica but the "
-                f"pod requests {gpus} x 288 GB GPU(s); request at least "
-                f"{need} amd.com/gpu (or shard the model)")
+                f"(/{shards} shards) needs ~{est.total_gb:.0f} GB of HBM "
+                f"per replica but the pod requests {gpus} x 288 GB GPU(s); "
+                f"request at least {need} amd.com/gpu (or shard further)")
     return errors
 
 

@@ -55,4 +55,13 @@ LLAMA_SMOKE = LlamaConfig(
     intermediate_size=8192, num_layers=2, num_heads=16, num_kv_heads=8,
     head_dim=128, rope_theta=500000.0)
 
-CONFIGS = {c.name: c for c in (LLAMA3_8B, LLAMA_TINY, LLAMA_1B, LLAMA_SMOKE)}
+# Llama-3-70B: the canonical multi-GPU target — its ~1.12 TB of training
+# state (16 B/param) outgrows one 288 GB MI355X, so it runs PP (and/or TP)
+# across the node; sizing.py admission computes the same bound.
+LLAMA3_70B = LlamaConfig(
+    name="llama3-70b", vocab_size=128256, hidden_size=8192,
+    intermediate_size=28672, num_layers=80, num_heads=64, num_kv_heads=8,
+    head_dim=128, rope_theta=500000.0)
+
+CONFIGS = {c.name: c for c in (LLAMA3_8B, LLAMA_TINY, LLAMA_1B,
+                               LLAMA_SMOKE, LLAMA3_70B)}
