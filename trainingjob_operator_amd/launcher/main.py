@@ -25,6 +25,8 @@ log = logging.getLogger("launcher")
 def _saves_ckpt(args, ctx, trainer) -> bool:
     """Who writes checkpoints: DP -> rank 0; TP -> the dp_rank==0 replica
     (every tp shard, its own stream); PP -> every stage."""
+    if args.zero1:
+        return True                   # every rank owns a moment shard
     if args.pp > 1:
         return trainer.dp_rank == 0   # every stage, dp-replica 0 only
     if args.ep:
@@ -54,6 +56,10 @@ def main(argv=None) -> int:
     ap.add_argument("--pp", type=int, default=int(os.environ.get(
         "TRAININGJOB_PP_SIZE", "1")),
         help="pipeline-parallel stages (world must equal pp; rank = stage)")
+    ap.add_argument("--zero1", action="store_true",
+                    default=os.environ.get("TRAININGJOB_ZERO1", "") == "1",
+                    help="ZeRO-1: shard optimizer state across the dp "
+                         "group (pure-DP jobs)")
     ap.add_argument("--sp", action="store_true",
                     default=os.environ.get("TRAININGJOB_SP", "") == "1",
                     help="sequence parallelism on top of --tp (Megatron "
@@ -88,6 +94,9 @@ def main(argv=None) -> int:
     t_start = time.time()
     dist_ctx.init_process_group(ctx)
 
+    if args.zero1:
+        assert args.tp == 1 and args.pp == 1 and not args.ep, \
+            "--zero1 launcher support is pure-DP (composed grids: roadmap)"
     if args.sp:
         assert args.tp > 1, "--sp requires --tp > 1"
         assert args.seq_len % args.tp == 0, "--sp needs seq_len % tp == 0"
@@ -96,7 +105,7 @@ def main(argv=None) -> int:
         model=args.model, micro_batch=args.micro_batch,
         grad_accum=args.grad_accum, seq_len=args.seq_len, lr=args.lr,
         checkpoint_activations=args.checkpoint_activations,
-        tp_size=args.tp, sequence_parallel=args.sp)
+        tp_size=args.tp, sequence_parallel=args.sp, zero1=args.zero1)
     if args.pp > 1:
         assert not args.ep, \
             "--pp with a MoE model shards experts automatically: the " \
@@ -127,7 +136,13 @@ def main(argv=None) -> int:
         ckpt = Checkpointer(os.path.join(args.ckpt_dir, sub))
     else:
         trainer = Trainer(cfg, ctx)
-        if args.tp > 1:
+        if args.zero1 and ctx.world_size > 1:
+            # every rank owns a distinct optimizer-state shard: its own
+            # checkpoint stream (the full bf16 params ride each stream,
+            # so params still resume anywhere — only moments are sharded)
+            ckpt = Checkpointer(os.path.join(args.ckpt_dir,
+                                             f"zero{ctx.rank}"))
+        elif args.tp > 1:
             # each tp rank owns a distinct shard: per-tp-rank checkpoint
             # streams, written by the dp_rank==0 replica
             ckpt = Checkpointer(os.path.join(

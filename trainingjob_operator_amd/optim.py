@@ -8,7 +8,7 @@ step). CPU path uses the equivalent reference implementation.
 """
 from __future__ import annotations
 
-from typing import Optional
+from typing import Optional, Tuple
 
 import torch
 
@@ -21,7 +21,12 @@ _N_PARTIALS = 2048
 class FlatAdamW:
     def __init__(self, store: FlatParamStore, lr: float = 3e-4,
                  betas=(0.9, 0.95), eps: float = 1e-8,
-                 weight_decay: float = 0.1, clip_grad_norm: float = 0.0):
+                 weight_decay: float = 0.1, clip_grad_norm: float = 0.0,
+                 shard: Optional[Tuple[int, int]] = None):
+        """shard=(start, end): ZeRO-1 — master weights and moments cover
+        only that slice of the flat buffer; step() updates only the slice
+        (the grad-norm clip still reads the FULL gradient). The caller
+        re-assembles flat_param across the dp group after the step."""
         self.store = store
         self.lr = lr
         self.beta1, self.beta2 = betas
@@ -29,8 +34,11 @@ class FlatAdamW:
         self.weight_decay = weight_decay
         self.clip = clip_grad_norm
         self.step_count = 0
+        self.shard = shard
+        s0, s1 = shard if shard is not None else (0, store.total)
+        self._s0, self._s1 = s0, s1
         dev = store.flat_param.device
-        self.p32 = store.flat_param.to(torch.float32)
+        self.p32 = store.flat_param[s0:s1].to(torch.float32)
         self.m = torch.zeros_like(self.p32)
         self.v = torch.zeros_like(self.p32)
         if dev.type == "cuda":
@@ -68,7 +76,12 @@ class FlatAdamW:
         if not in_graph_capture:
             self.bump_step()
         st = self.store
-        n = st.total
+        s0, s1 = self._s0, self._s1
+        n = s1 - s0
+        if n <= 0:
+            return
+        grad = st.flat_grad[s0:s1]
+        param = st.flat_param[s0:s1]
         bc1 = 1.0 - self.beta1 ** max(self.step_count, 1)
         bc2 = 1.0 - self.beta2 ** max(self.step_count, 1)
         if st.flat_param.is_cuda:
@@ -78,13 +91,15 @@ class FlatAdamW:
             sp = native.stream_ptr()
             normsq_ptr = None
             if self.clip > 0.0:
-                lib.l2normsq(sp, st.flat_grad.data_ptr(), n,
+                # the clip norm is GLOBAL: always over the full gradient,
+                # even when the update covers only this rank's shard
+                lib.l2normsq(sp, st.flat_grad.data_ptr(), st.total,
                              self._partials.data_ptr(), _N_PARTIALS,
                              self._normsq.data_ptr())
                 normsq_ptr = self._normsq.data_ptr()
             lib.adamw_step(sp, self.p32.data_ptr(), self.m.data_ptr(),
-                           self.v.data_ptr(), st.flat_grad.data_ptr(),
-                           st.flat_param.data_ptr(), normsq_ptr, n, self.lr,
+                           self.v.data_ptr(), grad.data_ptr(),
+                           param.data_ptr(), normsq_ptr, n, self.lr,
                            self.beta1, self.beta2, self.eps,
                            self.weight_decay, bc1, bc2, self.clip,
                            grad_pre_scale, self._bc.data_ptr())
@@ -92,8 +107,8 @@ class FlatAdamW:
             normsq = None
             if self.clip > 0.0:
                 normsq = st.flat_grad.float().pow(2).sum()
-            reference.adamw_step(self.p32, self.m, self.v, st.flat_grad,
-                                 st.flat_param, self.lr, self.beta1,
+            reference.adamw_step(self.p32, self.m, self.v, grad,
+                                 param, self.lr, self.beta1,
                                  self.beta2, self.eps, self.weight_decay,
                                  self.step_count, self.clip, normsq,
                                  grad_pre_scale)
@@ -112,8 +127,15 @@ class FlatAdamW:
                 "step": self.step_count}
 
     def load_state_dict(self, sd: dict) -> None:
+        if sd["p32"].numel() != self.p32.numel():
+            raise ValueError(
+                f"optimizer state length {sd['p32'].numel()} does not "
+                f"match this rank's {self.p32.numel()} — a ZeRO-1 "
+                f"checkpoint must resume at the same dp size (or be "
+                f"resharded)")
         self.p32.copy_(sd["p32"])
         self.m.copy_(sd["m"])
         self.v.copy_(sd["v"])
         self.step_count = int(sd["step"])
-        self.store.flat_param.copy_(self.p32.to(torch.bfloat16))
+        self.store.flat_param[self._s0:self._s1].copy_(
+            self.p32.to(torch.bfloat16))

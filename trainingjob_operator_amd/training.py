@@ -47,6 +47,10 @@ class TrainConfig:
     # norms/residual run on seq shards with all-gather/reduce-scatter
     # seams instead of the f/g all-reduces (parallel/sp.py)
     sequence_parallel: bool = False
+    # ZeRO-1: shard the fp32 master weights + AdamW moments across the dp
+    # group (12 bytes/param -> 12/dp); each rank updates its slice, then
+    # the updated bf16 params are re-assembled across the group
+    zero1: bool = False
 
     @property
     def model_config(self) -> LlamaConfig:
@@ -117,10 +121,22 @@ class Trainer:
         # differently and silently diverge the replicated params. The
         # trainer applies a TP-aware global clip instead (_tp_clip).
         tp_aware_clip = self.topo.tp_size > 1 and cfg.clip_grad_norm > 0
+        self._zero_shards = None
+        shard = None
+        if cfg.zero1 and self.topo.dp_size > 1:
+            from .parallel.flat import _aligned
+            dp, r = self.topo.dp_size, self.topo.dp_rank
+            chunk = _aligned(-(-self.store.total // dp))
+            self._zero_shards = [
+                (min(i * chunk, self.store.total),
+                 min((i + 1) * chunk, self.store.total))
+                for i in range(dp)]
+            shard = self._zero_shards[r]
         self.opt = FlatAdamW(self.store, lr=cfg.lr, betas=cfg.betas,
                              weight_decay=cfg.weight_decay,
                              clip_grad_norm=(0.0 if tp_aware_clip
-                                             else cfg.clip_grad_norm))
+                                             else cfg.clip_grad_norm),
+                             shard=shard)
         self._tp_spans = None
         if tp_aware_clip:
             from .parallel.flat import classify_spans
@@ -156,8 +172,23 @@ class Trainer:
         else:
             self.opt.step(grad_pre_scale=self.ddp.grad_pre_scale,
                           in_graph_capture=in_graph)
+        if self._zero_shards is not None:
+            self._zero_allgather_params()
         self.opt.zero_grad()
         return loss
+
+    def _zero_allgather_params(self) -> None:
+        """ZeRO-1 re-assembly: each dp rank broadcasts the flat_param
+        slice it just updated. (Per-slice broadcast keeps uneven tail
+        shards trivial; an all_gather_into_tensor fast path over RCCL is
+        a round-2 optimization.)"""
+        import torch.distributed as dist
+        g = self.topo.dp_group
+        for r, (s, e) in enumerate(self._zero_shards):
+            if e <= s:
+                continue
+            src = dist.get_global_rank(g, r) if g is not None else r
+            dist.broadcast(self.store.flat_param[s:e], src=src, group=g)
 
     def _tp_clip(self) -> None:
         """Global grad-norm clip under TP: sharded-param normsq summed
