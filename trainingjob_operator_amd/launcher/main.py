@@ -50,6 +50,13 @@ def main(argv=None) -> int:
     ap.add_argument("--ckpt-every", type=int, default=50)
     ap.add_argument("--log-every", type=int, default=10)
     ap.add_argument("--checkpoint-activations", action="store_true")
+    ap.add_argument("--data-path", default=os.environ.get(
+        "TRAININGJOB_DATA_PATH", ""),
+        help="flat token-id binary to train on (uint16/uint32; "
+             "memory-mapped); empty = synthetic data")
+    ap.add_argument("--data-dtype", choices=("uint16", "uint32"),
+                    default=os.environ.get("TRAININGJOB_DATA_DTYPE",
+                                           "uint16"))
     ap.add_argument("--tp", type=int, default=int(os.environ.get(
         "TRAININGJOB_TP_SIZE", "1")),
         help="tensor-parallel degree (world = dp x tp; tp ranks adjacent)")
@@ -105,7 +112,8 @@ def main(argv=None) -> int:
         model=args.model, micro_batch=args.micro_batch,
         grad_accum=args.grad_accum, seq_len=args.seq_len, lr=args.lr,
         checkpoint_activations=args.checkpoint_activations,
-        tp_size=args.tp, sequence_parallel=args.sp, zero1=args.zero1)
+        tp_size=args.tp, sequence_parallel=args.sp, zero1=args.zero1,
+        data_path=args.data_path, data_dtype=args.data_dtype)
     if args.pp > 1:
         assert not args.ep, \
             "--pp with a MoE model shards experts automatically: the " \

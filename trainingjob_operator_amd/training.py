@@ -51,6 +51,10 @@ class TrainConfig:
     # group (12 bytes/param -> 12/dp); each rank updates its slice, then
     # the updated bf16 params are re-assembled across the group
     zero1: bool = False
+    # real data: path to a flat token-id binary (launcher/data.py);
+    # "" = deterministic synthetic stream
+    data_path: str = ""
+    data_dtype: str = "uint16"
 
     @property
     def model_config(self) -> LlamaConfig:
@@ -145,7 +149,8 @@ class Trainer:
                 self.store,
                 lambda n: getattr(named[n], "tp_sharded", False))
         # tp peers train on the SAME data (they hold shards of one replica)
-        self.data = synthetic_batches(cfg, device, self.topo.dp_rank)
+        from .launcher.data import make_batches
+        self.data = make_batches(cfg, device, self.topo.dp_rank)
         self.step_count = 0
         self._graph = None
         self._static_batches = None
