@@ -1,0 +1,154 @@
+"""KV-cached autoregressive generation for the Llama family.
+
+Training is the flagship workload, but a framework its users can switch
+to needs an eval/inference path: greedy or temperature/top-k sampling
+with a per-layer KV cache (prefill once, then one-token decode steps).
+
+The decode path reuses the training modules' weights and the fused
+rmsnorm/SwiGLU ops; RoPE uses an explicit-position rotation here (the
+training kernel derives positions from flat indices, which is wrong for
+a 1-token step at position p) — matching ops/reference.py rope_rotate
+semantics (Neox half-rotation). Decode-side kernel fusion (paged KV,
+fused decode attention) is round-2 serving work — see ROADMAP.md.
+"""
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+
+from ..ops import fused_rmsnorm
+from .config import LlamaConfig
+from .llama import LlamaModel, _sdpa
+
+
+class KVCache:
+    """Per-layer [B, n_kv, max_len, D] key/value buffers + fill length."""
+
+    def __init__(self, cfg: LlamaConfig, batch: int, max_len: int,
+                 device, dtype):
+        shape = (batch, cfg.num_kv_heads, max_len, cfg.head_dim)
+        self.k: List[torch.Tensor] = [
+            torch.zeros(shape, device=device, dtype=dtype)
+            for _ in range(cfg.num_layers)]
+        self.v: List[torch.Tensor] = [
+            torch.zeros(shape, device=device, dtype=dtype)
+            for _ in range(cfg.num_layers)]
+        self.max_len = max_len
+        self.len = 0
+
+
+def _rope_at(x: torch.Tensor, inv_freq: torch.Tensor,
+             pos0: int) -> torch.Tensor:
+    """Neox half-rotation with explicit positions pos0..pos0+S-1.
+    x: [B, S, n_heads, D]."""
+    B, S, nh, D = x.shape
+    half = D // 2
+    pos = (torch.arange(S, device=x.device) + pos0).float()
+    ang = pos[:, None] * inv_freq[None, :].float()       # [S, half]
+    cos = ang.cos()[None, :, None, :]
+    sin = ang.sin()[None, :, None, :]
+    xf = x.float()
+    x1, x2 = xf[..., :half], xf[..., half:]
+    return torch.cat([x1 * cos - x2 * sin,
+                      x1 * sin + x2 * cos], dim=-1).to(x.dtype)
+
+
+def _attn_cached(attn, x: torch.Tensor, inv_freq: torch.Tensor,
+                 ck: torch.Tensor, cv: torch.Tensor,
+                 pos0: int) -> torch.Tensor:
+    cfg = attn.cfg
+    B, S, _ = x.shape
+    qkv = attn.qkv_proj(x)
+    q, k, v = qkv.split([attn.q_size, attn.kv_size, attn.kv_size], dim=-1)
+    q = q.reshape(B, S, cfg.num_heads, cfg.head_dim)
+    k = k.reshape(B, S, cfg.num_kv_heads, cfg.head_dim)
+    v = v.reshape(B, S, cfg.num_kv_heads, cfg.head_dim)
+    q = _rope_at(q, inv_freq, pos0).transpose(1, 2)      # [B, nh, S, D]
+    k = _rope_at(k, inv_freq, pos0).transpose(1, 2)
+    v = v.transpose(1, 2)
+    ck[:, :, pos0:pos0 + S] = k
+    cv[:, :, pos0:pos0 + S] = v
+    kk = ck[:, :, :pos0 + S]
+    vv = cv[:, :, :pos0 + S]
+    gqa = cfg.num_heads != cfg.num_kv_heads
+    # prefill (S>1, empty cache) is causal among the new tokens; a decode
+    # step (S==1) attends to everything cached
+    o = _sdpa(q, kk, vv, enable_gqa=gqa) if S > 1 else \
+        _sdpa(q, kk, vv, enable_gqa=gqa, is_causal=False)
+    o = o.transpose(1, 2).reshape(B, S, cfg.num_heads * cfg.head_dim)
+    return attn.o_proj(o)
+
+
+@torch.no_grad()
+def _forward_cached(model: LlamaModel, tokens: torch.Tensor,
+                    cache: KVCache) -> torch.Tensor:
+    """Run tokens through the model appending to the cache; returns
+    logits [B, S, V]."""
+    cfg = model.cfg
+    pos0 = cache.len
+    x = model.embed(tokens)
+    residual = None
+    for li, blk in enumerate(model.blocks):
+        normed, residual = fused_rmsnorm(x, blk.input_norm_weight,
+                                         residual, cfg.norm_eps)
+        attn_out = _attn_cached(blk.attn, normed, model.inv_freq,
+                                cache.k[li], cache.v[li], pos0)
+        normed, residual = fused_rmsnorm(attn_out,
+                                         blk.post_attn_norm_weight,
+                                         residual, cfg.norm_eps)
+        x = blk.mlp(normed)
+    cache.len = pos0 + tokens.shape[1]
+    normed, _ = fused_rmsnorm(x, model.final_norm_weight, residual,
+                              cfg.norm_eps)
+    return model.lm_head(normed)
+
+
+def _sample(logits: torch.Tensor, temperature: float, top_k: int,
+            generator: Optional[torch.Generator]) -> torch.Tensor:
+    """logits [B, V] -> next tokens [B]."""
+    if temperature <= 0.0:
+        return logits.argmax(dim=-1)
+    logits = logits.float() / temperature
+    if top_k > 0 and top_k < logits.shape[-1]:
+        kth = logits.topk(top_k, dim=-1).values[:, -1:]
+        logits = logits.masked_fill(logits < kth, float("-inf"))
+    probs = torch.softmax(logits, dim=-1)
+    return torch.multinomial(probs, 1, generator=generator).squeeze(-1)
+
+
+@torch.no_grad()
+def generate(model: LlamaModel, tokens: torch.Tensor,
+             max_new_tokens: int, temperature: float = 0.0,
+             top_k: int = 0, eos_token: Optional[int] = None,
+             seed: Optional[int] = None) -> torch.Tensor:
+    """Autoregressive generation: prefill the prompt once, then KV-cached
+    one-token decode steps. temperature 0 = greedy. Returns
+    [B, prompt + generated]."""
+    was_training = model.training
+    model.eval()
+    try:
+        B, S0 = tokens.shape
+        cache = KVCache(model.cfg, B, S0 + max_new_tokens,
+                        tokens.device, model.embed.weight.dtype)
+        gen = None
+        if seed is not None:
+            gen = torch.Generator(device="cpu").manual_seed(seed)
+        logits = _forward_cached(model, tokens, cache)
+        out = tokens
+        done = torch.zeros(B, dtype=torch.bool, device=tokens.device)
+        for _ in range(max_new_tokens):
+            nxt = _sample(logits[:, -1].cpu(), temperature, top_k,
+                          gen).to(tokens.device)
+            if eos_token is not None:
+                nxt = torch.where(done, torch.full_like(nxt, eos_token),
+                                  nxt)
+                done = done | (nxt == eos_token)
+            out = torch.cat([out, nxt[:, None]], dim=1)
+            if eos_token is not None and bool(done.all()):
+                break
+            logits = _forward_cached(model, nxt[:, None], cache)
+        return out
+    finally:
+        if was_training:
+            model.train()

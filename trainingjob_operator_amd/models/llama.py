@@ -27,7 +27,7 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 
-def _sdpa(q, k, v, enable_gqa=False):
+def _sdpa(q, k, v, enable_gqa=False, is_causal=True):
     """SDPA with a backend override knob (AITJ_SDPA_BACKEND =
     flash|efficient|math|native|auto) — pins a backend for debugging or
     selects the hand-written CDNA4 forward (ops/attention.py)."""
@@ -41,14 +41,14 @@ def _sdpa(q, k, v, enable_gqa=False):
         return flash_attention(q.contiguous() if q.stride(-1) != 1 else q,
                                k, v)
     if backend == "auto" or not q.is_cuda:
-        return F.scaled_dot_product_attention(q, k, v, is_causal=True,
+        return F.scaled_dot_product_attention(q, k, v, is_causal=is_causal,
                                               enable_gqa=enable_gqa)
     from torch.nn.attention import SDPBackend, sdpa_kernel
     mapping = {"flash": SDPBackend.FLASH_ATTENTION,
                "efficient": SDPBackend.EFFICIENT_ATTENTION,
                "math": SDPBackend.MATH}
     with sdpa_kernel([mapping[backend]]):
-        return F.scaled_dot_product_attention(q, k, v, is_causal=True,
+        return F.scaled_dot_product_attention(q, k, v, is_causal=is_causal,
                                               enable_gqa=enable_gqa)
 
 from ..ops import (apply_rope, fused_cross_entropy, fused_rmsnorm,
