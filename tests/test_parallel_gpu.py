@@ -57,3 +57,18 @@ def test_reshard_roundtrip_gpu(tmp_path):
     tr2 = Trainer(cfg, device=torch.device(DEV))
     assert Checkpointer(back_dir).load_latest(tr2) == tr.step_count
     assert torch.equal(tr2.store.flat_param, tr.store.flat_param)
+
+
+def test_generate_gpu():
+    """KV-cached decode path on the MI355X (fused rmsnorm/SwiGLU run in
+    the cached forward too)."""
+    import torch
+    from trainingjob_operator_amd.models.config import CONFIGS
+    from trainingjob_operator_amd.models.generate import generate
+    from trainingjob_operator_amd.training import build_model
+    model = build_model(CONFIGS["llama-smoke"], torch.device(DEV))
+    g = torch.Generator().manual_seed(3)
+    prompt = torch.randint(0, 32000, (2, 16), generator=g).to(DEV)
+    out = generate(model, prompt, max_new_tokens=8)
+    assert out.shape == (2, 24)
+    assert int(out.max()) < 32000 and int(out.min()) >= 0
