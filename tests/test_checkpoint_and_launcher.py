@@ -127,3 +127,25 @@ def test_worker_metrics_endpoint(tmp_path):
     finally:
         proc.kill()
         proc.wait(timeout=30)
+
+
+def test_launcher_eval_loop(tmp_path, monkeypatch):
+    """--eval-every runs a held-out eval (distinct seed stream) and traces
+    it; single process, no dist."""
+    import json
+    import os
+    os.environ.pop("RANK", None)
+    os.environ.pop("WORLD_SIZE", None)
+    trace_path = os.path.join(str(tmp_path), "trace.jsonl")
+    monkeypatch.setenv("AITJ_TRACE", trace_path)
+    from trainingjob_operator_amd.launcher.main import main
+    rc = main(["--model", "llama-tiny", "--steps", "4", "--seq-len", "32",
+               "--grad-accum", "1", "--micro-batch", "1",
+               "--ckpt-every", "100", "--log-every", "1",
+               "--eval-every", "2", "--eval-batches", "2",
+               "--ckpt-dir", str(tmp_path)])
+    assert rc == 0
+    events = [json.loads(l) for l in open(trace_path)]
+    evals = [e for e in events if e["kind"] == "eval"]
+    assert len(evals) == 2                  # steps 2 and 4
+    assert all("eval_loss" in e and e["eval_loss"] > 0 for e in evals)

@@ -22,6 +22,41 @@ import time
 log = logging.getLogger("launcher")
 
 
+def _evaluate(trainer, cfg, n_batches: int):
+    """Mean loss over a HELD-OUT stream (distinct seed from training) —
+    dp-averaged so every rank logs the same number. None for trainers
+    without a local loss (pipeline stages)."""
+    import dataclasses
+
+    import torch
+    import torch.distributed as dist
+
+    model = getattr(trainer, "model", None)
+    if model is None:
+        return None
+    from .data import make_batches
+    eval_cfg = dataclasses.replace(cfg, seed=cfg.seed + 7777)
+    rank = getattr(getattr(trainer, "topo", None), "dp_rank",
+                   getattr(getattr(trainer, "topo", None), "data_rank", 0))         if hasattr(trainer, "topo") else 0
+    data = make_batches(eval_cfg, trainer.device
+                        if hasattr(trainer, "device") else None, rank or 0)
+    was_training = model.training
+    model.eval()
+    total = 0.0
+    with torch.no_grad():
+        for _ in range(n_batches):
+            tokens, targets = next(data)
+            total += float(model(tokens, targets))
+    if was_training:
+        model.train()
+    loss = total / max(n_batches, 1)
+    if dist.is_initialized() and dist.get_world_size() > 1:
+        t = torch.tensor(loss / dist.get_world_size())
+        dist.all_reduce(t)
+        loss = float(t)
+    return loss
+
+
 def _saves_ckpt(args, ctx, trainer) -> bool:
     """Who writes checkpoints: DP -> rank 0; TP -> the dp_rank==0 replica
     (every tp shard, its own stream); PP -> every stage."""
@@ -79,6 +114,11 @@ def main(argv=None) -> int:
                     default=os.environ.get("TRAININGJOB_PP_SCHEDULE", "1f1b"),
                     help="pipeline schedule: 1f1b bounds live microbatches "
                          "per stage; gpipe holds all of them")
+    ap.add_argument("--eval-every", type=int, default=int(os.environ.get(
+        "TRAININGJOB_EVAL_EVERY", "0")),
+        help="run a held-out eval every N steps (0 = off)")
+    ap.add_argument("--eval-batches", type=int, default=8,
+                    help="micro-batches per evaluation")
     ap.add_argument("--metrics-port", type=int, default=int(os.environ.get(
         "TRAININGJOB_METRICS_PORT", "0")),
         help="expose Prometheus worker metrics (tokens/s, step time, loss)")
@@ -199,6 +239,13 @@ def main(argv=None) -> int:
                         world_size=ctx.world_size)
             if metrics:
                 metrics.observe(step, loss.item(), tps)
+        if args.eval_every and step % args.eval_every == 0:
+            ev = _evaluate(trainer, cfg, args.eval_batches)
+            if ev is not None and ctx.is_rank0:
+                log.info("step %d eval_loss %.4f", step, ev)
+                trace.event("eval", step=step, eval_loss=round(ev, 4))
+                if metrics:
+                    metrics.observe_eval(ev)
         if step % args.ckpt_every == 0 and _saves_ckpt(args, ctx, trainer):
             ckpt.save_async(trainer)
             trace.event("checkpoint", step=step)

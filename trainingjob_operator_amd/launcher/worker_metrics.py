@@ -16,6 +16,8 @@ class WorkerMetrics:
             "Whole-job training throughput (tokens/s, all ranks)")
         self.step = Gauge("aitj_worker_step", "Current optimizer step")
         self.loss = Gauge("aitj_worker_loss", "Last training loss")
+        self.eval_loss = Gauge("aitj_worker_eval_loss",
+                               "Last held-out evaluation loss")
         start_http_server(port)
         log.info("worker metrics on :%d/metrics", port)
 
@@ -23,3 +25,6 @@ class WorkerMetrics:
         self.step.set(step)
         self.loss.set(loss)
         self.tokens_per_sec.set(tokens_per_sec)
+
+    def observe_eval(self, loss: float) -> None:
+        self.eval_loss.set(loss)
