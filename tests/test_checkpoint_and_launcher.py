@@ -129,22 +129,25 @@ def test_worker_metrics_endpoint(tmp_path):
         proc.wait(timeout=30)
 
 
-def test_launcher_eval_loop(tmp_path, monkeypatch):
+def test_launcher_eval_loop(tmp_path):
     """--eval-every runs a held-out eval (distinct seed stream) and traces
-    it; single process, no dist."""
+    it (subprocess: the tracer is a process-lifetime singleton)."""
     import json
-    import os
-    os.environ.pop("RANK", None)
-    os.environ.pop("WORLD_SIZE", None)
+    import subprocess
+    import sys
     trace_path = os.path.join(str(tmp_path), "trace.jsonl")
-    monkeypatch.setenv("AITJ_TRACE", trace_path)
-    from trainingjob_operator_amd.launcher.main import main
-    rc = main(["--model", "llama-tiny", "--steps", "4", "--seq-len", "32",
-               "--grad-accum", "1", "--micro-batch", "1",
-               "--ckpt-every", "100", "--log-every", "1",
-               "--eval-every", "2", "--eval-batches", "2",
-               "--ckpt-dir", str(tmp_path)])
-    assert rc == 0
+    env = {k: v for k, v in os.environ.items()
+           if k not in ("RANK", "WORLD_SIZE")}
+    env["AITJ_TRACE"] = trace_path
+    r = subprocess.run(
+        [sys.executable, "-m", "trainingjob_operator_amd.launcher.main",
+         "--model", "llama-tiny", "--steps", "4", "--seq-len", "32",
+         "--grad-accum", "1", "--micro-batch", "1",
+         "--ckpt-every", "100", "--log-every", "1",
+         "--eval-every", "2", "--eval-batches", "2",
+         "--ckpt-dir", str(tmp_path)],
+        env=env, capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-2000:]
     events = [json.loads(l) for l in open(trace_path)]
     evals = [e for e in events if e["kind"] == "eval"]
     assert len(evals) == 2                  # steps 2 and 4
