@@ -1,4 +1,5 @@
 """End-to-end training engine tests on CPU (tiny Llama, reference op path)."""
+import pytest
 import torch
 
 from trainingjob_operator_amd.models.config import LLAMA_TINY
@@ -71,3 +72,32 @@ def test_optimizer_updates_params():
     # fp32 master tracks the bf16 copy
     assert torch.allclose(trainer.opt.p32.to(torch.bfloat16),
                           trainer.store.flat_param)
+
+
+def test_lr_schedule_values():
+    from trainingjob_operator_amd.optim import lr_at
+    # warmup ramps linearly to base
+    assert lr_at(0, 1.0, warmup_steps=10) == pytest.approx(0.1)
+    assert lr_at(9, 1.0, warmup_steps=10) == pytest.approx(1.0)
+    # constant after warmup without decay
+    assert lr_at(500, 1.0, warmup_steps=10) == 1.0
+    # cosine midpoint and floor
+    assert lr_at(10 + 50, 1.0, 10, 100, 0.1) == pytest.approx(0.55)
+    assert lr_at(10 + 100, 1.0, 10, 100, 0.1) == pytest.approx(0.1)
+    assert lr_at(10_000, 1.0, 10, 100, 0.1) == pytest.approx(0.1)
+
+
+def test_trainer_applies_schedule():
+    from trainingjob_operator_amd.training import TrainConfig, Trainer
+    cfg = TrainConfig(model="llama-tiny", micro_batch=1, grad_accum=1,
+                      seq_len=16, lr=1e-2, warmup_steps=3,
+                      lr_decay_steps=10, min_lr=1e-3)
+    tr = Trainer(cfg)
+    lrs = []
+    for _ in range(5):
+        tr.train_step()
+        lrs.append(tr.opt.lr)
+    # ramp through warmup (updates 1-3), hold at the boundary, then decay
+    assert lrs[0] < lrs[1] < lrs[2] == pytest.approx(1e-2)
+    assert lrs[3] == pytest.approx(1e-2)
+    assert lrs[4] < lrs[3]

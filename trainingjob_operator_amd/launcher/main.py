@@ -80,6 +80,11 @@ def main(argv=None) -> int:
     ap.add_argument("--grad-accum", type=int, default=4)
     ap.add_argument("--seq-len", type=int, default=4096)
     ap.add_argument("--lr", type=float, default=3e-4)
+    ap.add_argument("--warmup-steps", type=int, default=0,
+                    help="linear LR warmup steps")
+    ap.add_argument("--lr-decay-steps", type=int, default=0,
+                    help="cosine-decay horizon after warmup (0 = constant)")
+    ap.add_argument("--min-lr", type=float, default=0.0)
     ap.add_argument("--ckpt-dir", default=os.environ.get(
         "TRAININGJOB_CKPT_DIR", "/tmp/aitj-ckpt"))
     ap.add_argument("--ckpt-every", type=int, default=50)
@@ -153,7 +158,9 @@ def main(argv=None) -> int:
         grad_accum=args.grad_accum, seq_len=args.seq_len, lr=args.lr,
         checkpoint_activations=args.checkpoint_activations,
         tp_size=args.tp, sequence_parallel=args.sp, zero1=args.zero1,
-        data_path=args.data_path, data_dtype=args.data_dtype)
+        data_path=args.data_path, data_dtype=args.data_dtype,
+        warmup_steps=args.warmup_steps, lr_decay_steps=args.lr_decay_steps,
+        min_lr=args.min_lr)
     if args.pp > 1:
         assert not args.ep, \
             "--pp with a MoE model shards experts automatically: the " \

@@ -18,6 +18,21 @@ from .parallel.flat import FlatParamStore
 _N_PARTIALS = 2048
 
 
+def lr_at(step: int, base_lr: float, warmup_steps: int = 0,
+          decay_steps: int = 0, min_lr: float = 0.0) -> float:
+    """Warmup + cosine-decay schedule (the standard LLM pretraining
+    shape): linear 0 -> base over warmup_steps, cosine base -> min_lr
+    over decay_steps, then flat min_lr. decay_steps == 0 disables decay
+    (constant base after warmup)."""
+    import math
+    if warmup_steps > 0 and step < warmup_steps:
+        return base_lr * (step + 1) / warmup_steps
+    if decay_steps <= 0:
+        return base_lr
+    t = min(max(step - warmup_steps, 0), decay_steps) / decay_steps
+    return min_lr + 0.5 * (base_lr - min_lr) * (1.0 + math.cos(math.pi * t))
+
+
 class FlatAdamW:
     def __init__(self, store: FlatParamStore, lr: float = 3e-4,
                  betas=(0.9, 0.95), eps: float = 1e-8,

@@ -515,6 +515,11 @@ class EPTrainer:
 
     def train_step(self):
         cfg = self.cfg
+        if cfg.warmup_steps or cfg.lr_decay_steps:
+            from ..optim import lr_at
+            self.opt.lr = lr_at(self.opt.step_count, cfg.lr,
+                                cfg.warmup_steps, cfg.lr_decay_steps,
+                                cfg.min_lr)
         loss = None
         for _ in range(cfg.grad_accum):
             tokens, targets = next(self.data)

@@ -480,6 +480,11 @@ class PPTrainer:
 
     def train_step(self):
         micros = [next(self.data) for _ in range(self.cfg.grad_accum)]
+        if self.cfg.warmup_steps or self.cfg.lr_decay_steps:
+            from ..optim import lr_at
+            self.opt.lr = lr_at(self.opt.step_count, self.cfg.lr,
+                                self.cfg.warmup_steps,
+                                self.cfg.lr_decay_steps, self.cfg.min_lr)
         loss = self.sched.step(micros, self.hidden_shape, self.act_dtype)
         if self.dp_size > 1:
             # stage-peer gradient seam: SUM with 1/world pre-scale (gloo

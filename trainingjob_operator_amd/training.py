@@ -55,6 +55,12 @@ class TrainConfig:
     # "" = deterministic synthetic stream
     data_path: str = ""
     data_dtype: str = "uint16"
+    # LR schedule: linear warmup to lr, then cosine decay to min_lr over
+    # lr_decay_steps (0 = constant lr after warmup). NOTE: not applied
+    # inside a captured hipGraph (the kernel arg is baked at capture).
+    warmup_steps: int = 0
+    lr_decay_steps: int = 0
+    min_lr: float = 0.0
 
     @property
     def model_config(self) -> LlamaConfig:
@@ -167,6 +173,11 @@ class Trainer:
                 # scale so accumulated grads average over micro-batches
                 (loss / cfg.grad_accum).backward()
         self.ddp.finish_backward()
+        if not in_graph and (cfg.warmup_steps or cfg.lr_decay_steps):
+            from .optim import lr_at
+            self.opt.lr = lr_at(self.opt.step_count, cfg.lr,
+                                cfg.warmup_steps, cfg.lr_decay_steps,
+                                cfg.min_lr)
         if cfg.sequence_parallel and self.topo.tp_size > 1:
             # seq-sharded params (norms/embed) saw only this rank's
             # positions: sum their grads over the tp group BEFORE clip
