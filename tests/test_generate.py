@@ -61,3 +61,28 @@ def test_eos_stops_early(model):
     eos = int(greedy[0, 4])       # the first token it would emit
     out = generate(model, prompt, 8, eos_token=eos)
     assert out.shape[1] == 5      # stopped right after emitting EOS
+
+
+def test_generate_cli(tmp_path):
+    """scripts/generate.py resumes a trained checkpoint and decodes."""
+    import os
+    import subprocess
+    import sys
+    from trainingjob_operator_amd.launcher.checkpoint import Checkpointer
+    from trainingjob_operator_amd.training import TrainConfig, Trainer
+    cfg = TrainConfig(model="llama-tiny", micro_batch=1, grad_accum=1,
+                      seq_len=16)
+    tr = Trainer(cfg)
+    tr.train_step()
+    ck = os.path.join(str(tmp_path), "ck")
+    Checkpointer(ck).save_async(tr, blocking=True)
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, os.path.join(root, "scripts", "generate.py"),
+         "--model", "llama-tiny", "--ckpt-dir", ck,
+         "--prompt-tokens", "1,2,3", "--max-new-tokens", "4"],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-2000:]
+    ids = [int(t) for t in r.stdout.strip().split(",")]
+    assert len(ids) == 4 and all(0 <= t < 512 for t in ids)
+    assert "loaded step 1" in r.stderr
