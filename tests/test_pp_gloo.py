@@ -524,3 +524,39 @@ def test_pp_x_ep_grid_matches_single_process():
     stage plane): first-step gradients — including middle-stage router
     aux grads — match a single process over both pipelines' batches."""
     mp.spawn(_ppxep_worker, args=(4, _free_port()), nprocs=4, join=True)
+
+
+def _pp_ckptact_worker(rank, world, port, outdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        import json
+        from trainingjob_operator_amd.parallel.pp import PPTrainer
+        from trainingjob_operator_amd.training import TrainConfig
+        losses = {}
+        for ca in (False, True):
+            cfg = TrainConfig(model="llama-tiny", micro_batch=1,
+                              grad_accum=2, seq_len=32, lr=1e-3,
+                              checkpoint_activations=ca)
+            tr = PPTrainer(cfg, stage_idx=rank, n_stages=world)
+            assert tr.stage.checkpoint_activations == ca
+            ls = [tr.train_step() for _ in range(2)]
+            losses[ca] = [None if l is None else float(l) for l in ls]
+        if rank == world - 1:
+            with open(os.path.join(outdir, f"ca_{rank}.json"), "w") as f:
+                json.dump(losses[False] + losses[True], f)
+            # recompute must reproduce the exact same training trajectory
+            assert losses[False] == losses[True]
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_pp_activation_checkpointing_matches(tmp_path):
+    """Stages with activation checkpointing reproduce the exact
+    non-checkpointed losses (recompute is deterministic)."""
+    mp.spawn(_pp_ckptact_worker, args=(2, _free_port(), str(tmp_path)),
+             nprocs=2, join=True)
+    import json
+    assert os.path.exists(os.path.join(str(tmp_path), "ca_1.json"))

@@ -51,6 +51,7 @@ class LlamaStage(nn.Module):
                  lm_head: Optional[nn.Linear], inv_freq: torch.Tensor):
         super().__init__()
         self.cfg = cfg
+        self.checkpoint_activations = False
         self.embed = embed
         self.blocks = nn.ModuleList(layers)
         self.lm_head = lm_head
@@ -98,7 +99,13 @@ class LlamaStage(nn.Module):
             x = self.embed(x)
             residual = None
         for blk in self.blocks:
-            x, residual = blk(x, residual, self.inv_freq)
+            if self.checkpoint_activations and self.training:
+                if residual is None:
+                    residual = torch.zeros_like(x)
+                x, residual = torch.utils.checkpoint.checkpoint(
+                    blk, x, residual, self.inv_freq, use_reentrant=False)
+            else:
+                x, residual = blk(x, residual, self.inv_freq)
         if not self.is_last:
             return x, residual
         normed, _ = fused_rmsnorm(x, self.final_norm_weight, residual,
@@ -391,6 +398,9 @@ class PPTrainer:
                 moem = MoELlamaModel(mcfg, ep_group=ep_plane)
             diversify_experts(moem, cfg.seed, ep_rank=plane_rank)
             self.stage = MoEStage.from_moe_model(moem, stage_idx, n_stages)
+            assert not cfg.checkpoint_activations, \
+                "activation checkpointing re-runs the MoE all-to-all at " \
+                "backward time — unsupported for MoE stages (roadmap)"
         elif grid is not None and grid.tp_size > 1:
             # PP x TP: slice a tensor-parallel model into stages — the
             # stage machinery is block-generic (TPBlock outputs the same
@@ -411,6 +421,7 @@ class PPTrainer:
                                                 device=self.device)
         # the flat store keeps parameters in bf16; boundary activations
         # travel in the same dtype
+        self.stage.checkpoint_activations = cfg.checkpoint_activations
         self.act_dtype = act_dtype or torch.bfloat16
         sched_cls = {"gpipe": GPipeSchedule,
                      "1f1b": OneFOneBSchedule}[schedule]
