@@ -57,6 +57,42 @@ def _evaluate(trainer, cfg, n_batches: int):
     return loss
 
 
+def _run_ps_mode(args) -> int:
+    """Asynchronous parameter-server training: ps ranks [0, N) + worker
+    ranks after, one torch.distributed world. Role/index come from the
+    operator's injected env (TRAININGJOB_REPLICA_NAME/_INDEX)."""
+    import torch
+    import torch.distributed as dist
+
+    from ..parallel.ps import run_role
+    from ..training import TrainConfig
+
+    n_ps = args.ps_servers
+    n_workers = args.ps_workers or int(os.environ.get(
+        "TRAINER_INSTANCES_NUM", "1"))
+    index = int(os.environ.get("TRAININGJOB_REPLICA_INDEX",
+                               os.environ.get("RANK", "0")))
+    role = "pserver" if args.role.lower().startswith("ps") else "trainer"
+    rank = index if role == "pserver" else n_ps + index
+    world = n_ps + n_workers
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "23456")
+    backend = "nccl" if torch.cuda.is_available() else "gloo"
+    dist.init_process_group(backend, rank=rank, world_size=world)
+    log.info("ps mode: role=%s rank=%d/%d (%d ps + %d workers)",
+             role, rank, world, n_ps, n_workers)
+    cfg = TrainConfig(model=args.model, micro_batch=args.micro_batch,
+                      grad_accum=args.grad_accum, seq_len=args.seq_len,
+                      lr=args.lr, clip_grad_norm=0.0,
+                      data_path=args.data_path, data_dtype=args.data_dtype)
+    try:
+        run_role(cfg, role, index, n_ps, n_workers, args.steps)
+    finally:
+        dist.destroy_process_group()
+    log.info("ps mode done (role=%s)", role)
+    return 0
+
+
 def _saves_ckpt(args, ctx, trainer) -> bool:
     """Who writes checkpoints: DP -> rank 0; TP -> the dp_rank==0 replica
     (every tp shard, its own stream); PP -> every stage."""
@@ -115,6 +151,17 @@ def main(argv=None) -> int:
         "TRAININGJOB_EP_SIZE", "0")),
         help="expert-parallel group size for MoE models (world = edp x ep; "
              "ep ranks adjacent; 0 = off)")
+    ap.add_argument("--ps-servers", type=int, default=int(os.environ.get(
+        "TRAININGJOB_PS_SERVERS", "0")),
+        help="asynchronous parameter-server mode: N pserver ranks + the "
+             "rest workers (role from TRAININGJOB_REPLICA_NAME; the "
+             "reference's pserver/trainer job shape)")
+    ap.add_argument("--ps-workers", type=int, default=0,
+        help="worker count in ps mode (default: TRAINER_INSTANCES_NUM "
+             "from the operator's env contract)")
+    ap.add_argument("--role", default=os.environ.get(
+        "TRAININGJOB_REPLICA_NAME", "trainer"),
+        help="this pod's replica role (injected by the operator)")
     ap.add_argument("--pp-schedule", choices=("1f1b", "gpipe"),
                     default=os.environ.get("TRAININGJOB_PP_SCHEDULE", "1f1b"),
                     help="pipeline schedule: 1f1b bounds live microbatches "
@@ -136,6 +183,9 @@ def main(argv=None) -> int:
     from ..parallel import dist_ctx
     from ..training import TrainConfig, Trainer
     from .checkpoint import Checkpointer
+
+    if args.ps_servers:
+        return _run_ps_mode(args)
 
     ctx = dist_ctx.from_env()
     epoch = os.environ.get("TRAININGJOB_RENDEZVOUS_EPOCH", "0")
