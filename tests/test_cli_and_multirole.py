@@ -60,3 +60,28 @@ def test_multirole_env_contract():
     assert env["RANK"] == "1"
     assert env["MASTER_ADDR"] == "ps-worker-0.ml"
     assert env["MASTER_PORT"] == "2223"      # the role's own aitj port
+
+
+def test_operator_yaml_config(tmp_path):
+    import argparse
+    import os
+    from trainingjob_operator_amd.controller.options import OperatorOptions
+    cfgfile = os.path.join(str(tmp_path), "op.yaml")
+    with open(cfgfile, "w") as f:
+        f.write("namespace: prod\nresync_period: 30.0\n"
+                "elastic_scaleup_interval: 120.0\n")
+    ap = argparse.ArgumentParser()
+    OperatorOptions.add_flags(ap)
+    # file values land; explicit flags override
+    o = OperatorOptions.from_args(ap.parse_args(["--config", cfgfile]))
+    assert o.namespace == "prod" and o.resync_period == 30.0
+    assert o.elastic_scaleup_interval == 120.0
+    o2 = OperatorOptions.from_args(ap.parse_args(
+        ["--config", cfgfile, "--namespace", "dev"]))
+    assert o2.namespace == "dev" and o2.resync_period == 30.0
+    # unknown keys rejected
+    with open(cfgfile, "w") as f:
+        f.write("not_an_option: 1\n")
+    import pytest
+    with pytest.raises(ValueError):
+        OperatorOptions.from_yaml(cfgfile)

@@ -43,8 +43,24 @@ class OperatorOptions:
         )
 
     @classmethod
+    def from_yaml(cls, path: str) -> "OperatorOptions":
+        """Optional YAML config (SURVEY.md §5 'Config / flag system'):
+        keys are the dataclass field names; unknown keys are rejected."""
+        import yaml
+        with open(path) as f:
+            data = yaml.safe_load(f) or {}
+        fields = {f.name for f in cls.__dataclass_fields__.values()}
+        unknown = set(data) - fields
+        if unknown:
+            raise ValueError(f"{path}: unknown option(s) {sorted(unknown)}")
+        return cls(**data)
+
+    @classmethod
     def add_flags(cls, ap: argparse.ArgumentParser) -> None:
         d = cls()
+        ap.add_argument("--config", default="",
+                        help="YAML config file (field-named keys); "
+                             "explicitly-passed flags override its values")
         ap.add_argument("--master", default=d.master,
                         help="API server URL (default: in-cluster or proxy)")
         ap.add_argument("--kubeconfig", default=d.kubeconfig)
@@ -76,6 +92,17 @@ class OperatorOptions:
 
     @classmethod
     def from_args(cls, args: argparse.Namespace) -> "OperatorOptions":
+        if getattr(args, "config", ""):
+            base = cls.from_yaml(args.config)
+            # explicit CLI values override the file's (argparse gives us
+            # the default when a flag was not passed; detect overrides by
+            # comparing to the dataclass defaults)
+            d = cls()
+            for f in cls.__dataclass_fields__.values():
+                cli = getattr(args, f.name, None)
+                if cli is not None and cli != getattr(d, f.name):
+                    setattr(base, f.name, cli)
+            return base
         return cls(
             master=args.master, kubeconfig=args.kubeconfig,
             run_in_cluster=args.run_in_cluster, thread_num=args.thread_num,
