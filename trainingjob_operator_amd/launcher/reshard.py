@@ -286,6 +286,48 @@ def split_ep(full_named: Dict, cfg, ep: int) -> List[Dict]:
     return out
 
 
+# -- ZeRO-1: per-rank moment shards <-> full --------------------------------
+
+def zero_chunk_bounds(total: int, dp: int) -> List:
+    """The exact shard math training.py uses (aligned ceil-div chunks)."""
+    chunk = _aligned(-(-total // dp))
+    return [(min(i * chunk, total), min((i + 1) * chunk, total))
+            for i in range(dp)]
+
+
+def merge_zero(states: List[dict]) -> dict:
+    """Per-rank ZeRO-1 streams (rank order) -> one full-layout state.
+    Every stream carries the FULL bf16 flat_param; p32/m/v are that
+    rank's chunk and concatenate back to full."""
+    full = dict(states[0])
+    total = states[0]["flat_param"].numel()
+    bounds = zero_chunk_bounds(total, len(states))
+    opt = {"step": states[0]["opt"]["step"]}
+    for key in ("p32", "m", "v"):
+        buf = torch.zeros(total, dtype=torch.float32)
+        for st, (s, e) in zip(states, bounds):
+            shard = st["opt"][key]
+            assert shard.numel() == e - s,                 f"zero shard {key} length {shard.numel()} != {e - s}"
+            buf[s:e].copy_(shard)
+        opt[key] = buf
+    full["opt"] = opt
+    return full
+
+
+def split_zero(state: dict, dp: int) -> List[dict]:
+    """Inverse of merge_zero: slice the full moments into dp chunks."""
+    total = state["flat_param"].numel()
+    out = []
+    for s, e in zero_chunk_bounds(total, dp):
+        st = dict(state)
+        st["opt"] = {"step": state["opt"]["step"],
+                     "p32": state["opt"]["p32"][s:e].clone(),
+                     "m": state["opt"]["m"][s:e].clone(),
+                     "v": state["opt"]["v"][s:e].clone()}
+        out.append(st)
+    return out
+
+
 # -- directory-level conversion -------------------------------------------
 
 def _latest_state(directory: str) -> dict:
@@ -316,8 +358,8 @@ def _tp_order(cfg: LlamaConfig, shard_named: Dict) -> List[str]:
 def reshard(model: str, in_dir: str, out_dir: str,
             src: str, dst: str) -> List[str]:
     """Convert the latest checkpoint in `in_dir` (layout `src`) to layout
-    `dst` under `out_dir`. Layouts: "full", "pp=N", "tp=N". Returns the
-    written paths."""
+    `dst` under `out_dir`. Layouts: "full", "pp=N", "tp=N", "ep=N",
+    "zero=N" (ZeRO-1 moment shards). Returns the written paths."""
     cfg = CONFIGS[model]
 
     def parse(lay):
@@ -329,8 +371,24 @@ def reshard(model: str, in_dir: str, out_dir: str,
     skind, sn = parse(src)
     dkind, dn = parse(dst)
 
-    # read -> one full named dict
-    if skind == "full":
+    # ZeRO-1 conversions operate on raw streams (the flat layout is the
+    # full model's already; only the optimizer moments are sharded)
+    if skind == "zero":
+        states = [_latest_state(os.path.join(in_dir, f"zero{r}"))
+                  for r in range(sn)]
+        merged = merge_zero(states)
+        if dkind == "full":
+            return [_write_state(out_dir, merged)]
+        if dkind == "zero":
+            return [_write_state(os.path.join(out_dir, f"zero{r}"), st)
+                    for r, st in enumerate(split_zero(merged, dn))]
+        st = merged
+        full, meta = stream_to_named(st), st
+    elif skind == "full" and dkind == "zero":
+        st = _latest_state(in_dir)
+        return [_write_state(os.path.join(out_dir, f"zero{r}"), sh)
+                for r, sh in enumerate(split_zero(st, dn))]
+    elif skind == "full":
         st = _latest_state(in_dir)
         full, meta = stream_to_named(st), st
     elif skind == "pp":
