@@ -77,3 +77,22 @@ def test_fault_then_resize_sequence(tmp_path):
     assert codes == [0]
     assert sup.report.final_step >= 14
     assert sup._epoch == 1
+
+
+@pytest.mark.timeout(600)
+def test_elastic_resize_up(tmp_path):
+    """Scale 1 -> 2 mid-run: the flat DP checkpoint resumes at the larger
+    world (BASELINE config 3 upward direction)."""
+    sup = LocalSupervisor(world_size=1, ckpt_dir=str(tmp_path),
+                          total_steps=10, master_port=29877)
+    sup.start_world()
+    deadline = time.monotonic() + 120
+    while sup.report.final_step < 3 and time.monotonic() < deadline:
+        time.sleep(0.2)
+    assert sup.report.final_step >= 3
+
+    sup.restart_world(time.monotonic(), new_world_size=2)  # scale 1 -> 2
+    codes = sup.wait(timeout=300)
+    assert codes == [0, 0]
+    assert sup.report.final_step >= 10
+    assert sup._epoch == 1
