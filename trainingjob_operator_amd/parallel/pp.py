@@ -165,7 +165,9 @@ class MoEStage(LlamaStage):
 class GPipeSchedule:
     """Fill-drain (GPipe) schedule: all micro-batch forwards, then all
     backwards in reverse — simple, correct, and bubble-bounded by
-    (stages-1)/micro_batches. (1F1B is a round-2 refinement.)"""
+    (stages-1)/micro_batches. OneFOneBSchedule below bounds the live
+    activations and is the PPTrainer default; GPipe remains as the
+    simpler reference schedule (--pp-schedule gpipe)."""
 
     def __init__(self, stage: LlamaStage, stage_idx: int, n_stages: int,
                  group=None, device=None,

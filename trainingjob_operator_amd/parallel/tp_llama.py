@@ -6,8 +6,9 @@ after the MLP (RowParallel down_proj) per direction — the minimal
 collective count for xGMI. Heads are sharded across TP ranks (num_heads
 and num_kv_heads must divide by tp), so RoPE/SDPA run on local heads with
 no communication. Embedding and lm_head stay replicated (at 8B the head is
-1 GB bf16 — replication is cheap against 288 GB HBM; vocab-parallel CE is
-a round-2 refinement).
+1 GB bf16 — replication is cheap against 288 GB HBM; a vocab-parallel
+LM head + sharded CE exists separately in parallel/vocab_parallel.py for
+memory-bound configs).
 
 `shard_from_full` loads a rank's shards from an unsharded LlamaModel, which
 is also how the gloo tests prove exact fwd/bwd parity
