@@ -1,12 +1,14 @@
-"""Pipeline parallelism: Llama stage partitioning + a GPipe schedule over
-torch.distributed P2P (RCCL send/recv on GPU, gloo in CPU tests).
+"""Pipeline parallelism: Llama stage partitioning + GPipe/1F1B schedules
+over torch.distributed P2P (RCCL send/recv on GPU, gloo in CPU tests).
 
-Completes the DP/TP/PP triple the launcher can form from the operator's
-injected env (BASELINE.json north star). The flagship 8B bench stays pure
-DP (one replica fits a 288 GB MI355X); PP is for models that outgrow one
-GPU. On one 8xMI355X node every stage boundary is a direct xGMI link, so
-the P2P activations ride point-to-point bandwidth (~153 GB/s/link) without
-touching collectives.
+The launcher forms pipelines from the operator's injected env
+(BASELINE.json north star); PPTrainer composes with DP (replicated
+pipelines), TP (TP-sharded stages) and EP (MoE stages whose experts
+shard across the stage plane) — docs/PARALLELISM.md. The flagship 8B
+bench stays pure DP (one replica fits a 288 GB MI355X); PP is for models
+that outgrow one GPU, e.g. llama3-70b. On one 8xMI355X node every stage
+boundary is a direct xGMI link, so the P2P activations ride
+point-to-point bandwidth (~153 GB/s/link) without touching collectives.
 
 Pipeline boundary payload: the pre-norm Llama block carries TWO tensors
 (branch output x and the running residual stream — models/llama.py Block),
