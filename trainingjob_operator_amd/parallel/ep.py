@@ -421,7 +421,7 @@ class EPTrainer:
         from ..models.moe_llama import MoELlamaModel
         from ..optim import FlatAdamW
         from ..parallel.flat import FlatParamStore, classify_spans
-        from ..training import synthetic_batches
+        from ..launcher.data import make_batches
 
         self.cfg = cfg
         mcfg = CONFIGS[cfg.model]
@@ -468,8 +468,8 @@ class EPTrainer:
             self.attn_shard_spans = []
             self.replicated_spans = rest
         self.dense_spans = rest   # everything non-expert: one sync class
-        self.data = synthetic_batches(cfg, self.device,
-                                      rank=self.topo.data_rank)
+        self.data = make_batches(cfg, self.device,
+                                rank=self.topo.data_rank)
         self.step_count = 0
 
     def _reduce_grads(self) -> None:

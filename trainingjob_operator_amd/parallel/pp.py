@@ -362,7 +362,7 @@ class PPTrainer:
         from ..models.config import CONFIGS
         from ..parallel.flat import FlatParamStore
         from ..optim import FlatAdamW
-        from ..training import synthetic_batches
+        from ..launcher.data import make_batches
 
         self.cfg = cfg
         mcfg = CONFIGS[cfg.model]
@@ -453,7 +453,7 @@ class PPTrainer:
                 self.store, lambda n: ".experts." in n)
         # every pp rank of a replica draws the same stream; replicas draw
         # DISTINCT streams (dp_rank-keyed), like the DP Trainer
-        self.data = synthetic_batches(cfg, self.device, rank=self.dp_rank)
+        self.data = make_batches(cfg, self.device, rank=self.dp_rank)
         self.hidden_shape = (cfg.micro_batch, cfg.seq_len, mcfg.hidden_size)
         self.step_count = 0
 

@@ -99,7 +99,8 @@ class PSWorker:
     """One worker rank: local fwd/bwd, per-chunk exchange with every ps."""
 
     def __init__(self, cfg, worker_index: int, n_ps: int, device=None):
-        from ..training import build_model, synthetic_batches
+        from ..launcher.data import make_batches
+        from ..training import build_model
         self.cfg = cfg
         self.device = torch.device(device or "cpu")
         torch.manual_seed(cfg.seed)
@@ -107,7 +108,7 @@ class PSWorker:
         self.store = FlatParamStore(self.model, device=self.device)
         self.bounds = chunk_bounds(self.store.total, n_ps)
         self.n_ps = n_ps
-        self.data = synthetic_batches(cfg, self.device, rank=worker_index)
+        self.data = make_batches(cfg, self.device, rank=worker_index)
         self.step_count = 0
 
     def train_step(self):
