@@ -76,3 +76,21 @@ def test_delete(cluster, capsys):
     api, _ = cluster
     assert cli.main(["delete", "j", "-n", NS], api=api) == 0
     assert api.list_jobs(NS) == []
+
+
+def test_logs(cluster, capsys):
+    api, _ = cluster
+    api.set_pod_log(NS, "j-trainer-0", "step 1 loss 6.2\nstep 2 loss 6.1")
+    api.set_pod_log(NS, "j-trainer-1", "hello from rank 1")
+    assert cli.main(["logs", "j", "-n", NS], api=api) == 0
+    out = capsys.readouterr().out
+    assert "==> j-trainer-0 <==" in out and "step 2 loss 6.1" in out
+    assert "hello from rank 1" in out
+    # single replica + tail
+    assert cli.main(["logs", "j", "--replica", "trainer-0", "--tail", "1",
+                     "-n", NS], api=api) == 0
+    out = capsys.readouterr().out
+    assert "step 2 loss 6.1" in out and "step 1" not in out
+    assert "==>" not in out
+    assert cli.main(["logs", "j", "--replica", "nope-9", "-n", NS],
+                    api=api) == 1

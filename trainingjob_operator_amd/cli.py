@@ -126,6 +126,28 @@ def cmd_resize(api, namespace: str, name: str, replicas: int,
     return 0
 
 
+def cmd_logs(api, namespace: str, name: str, replica: str,
+             tail: int, out=None) -> int:
+    out = out or sys.stdout
+    from .kube import objects as ko
+    pods = api.list_pods(namespace, ko.job_selector(name))
+    if replica:
+        pods = [p for p in pods
+                if ko.name_of(p) == f"{name}-{replica}"]
+    if not pods:
+        print(f"error: no pods for job {name!r}"
+              + (f" replica {replica!r}" if replica else ""),
+              file=sys.stderr)
+        return 1
+    for p in sorted(pods, key=ko.name_of):
+        pname = ko.name_of(p)
+        if len(pods) > 1:
+            print(f"==> {pname} <==", file=out)
+        print(api.read_pod_log(namespace, pname, tail_lines=tail or None),
+              file=out)
+    return 0
+
+
 def cmd_delete(api, namespace: str, name: str, out=None) -> int:
     out = out or sys.stdout
     api.delete_job(namespace, name)
@@ -150,6 +172,11 @@ def main(argv=None, api=None) -> int:
     r.add_argument("--role", default="trainer")
     x = sub.add_parser("delete", parents=[common])
     x.add_argument("name")
+    lg = sub.add_parser("logs", parents=[common])
+    lg.add_argument("name")
+    lg.add_argument("--replica", default="",
+                    help="single replica, e.g. trainer-0 (default: all)")
+    lg.add_argument("--tail", type=int, default=0)
     args = ap.parse_args(argv)
 
     if api is None:
@@ -164,6 +191,9 @@ def main(argv=None, api=None) -> int:
                           args.role)
     if args.cmd == "delete":
         return cmd_delete(api, args.namespace, args.name)
+    if args.cmd == "logs":
+        return cmd_logs(api, args.namespace, args.name, args.replica,
+                        args.tail)
     return 2
 
 

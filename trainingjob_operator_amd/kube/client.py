@@ -51,6 +51,8 @@ class KubeApi:
                   selector: Optional[Dict[str, str]] = None) -> List[dict]: ...
     def delete_pod(self, namespace: str, name: str,
                    grace_period: Optional[int] = None) -> None: ...
+    def read_pod_log(self, namespace: str, name: str,
+                     tail_lines: Optional[int] = None) -> str: ...
 
     # services -----------------------------------------------------------
     def create_service(self, namespace: str, svc: dict) -> dict: ...
@@ -171,6 +173,17 @@ class RealKubeApi(KubeApi):
             body["gracePeriodSeconds"] = grace_period
         self._req("DELETE", self._ns_path("pods", namespace) + "/" + name,
                   body or None)
+
+    def read_pod_log(self, namespace, name, tail_lines=None):
+        params = {}
+        if tail_lines is not None:
+            params["tailLines"] = str(tail_lines)
+        url = (self.base_url + self._ns_path("pods", namespace)
+               + f"/{name}/log")
+        r = self.session.get(url, params=params, timeout=30.0)
+        if r.status_code >= 400:
+            raise ApiError(r.status_code, r.text[:500])
+        return r.text
 
     # -- services ---------------------------------------------------------
     def create_service(self, namespace, svc):

@@ -29,6 +29,7 @@ class FakeKubeApi(KubeApi):
         self.jobs: Dict[Tuple[str, str], dict] = {}
         self.leases: Dict[Tuple[str, str], dict] = {}
         self.events: List[dict] = []
+        self.pod_logs: dict = {}
         self.crds: Dict[str, dict] = {}
         self.actions: List[tuple] = []
         self._uid = itertools.count(1)
@@ -343,6 +344,19 @@ class FakeKubeApi(KubeApi):
         for (ns, name) in list(self.pods):
             if ns == namespace:
                 self.set_pod_phase(ns, name, phase, **kw)
+
+    def set_pod_log(self, namespace: str, name: str, text: str):
+        with self._lock:
+            self._pod(namespace, name)  # existence check
+            self.pod_logs[(namespace, name)] = text
+
+    def read_pod_log(self, namespace, name, tail_lines=None):
+        with self._lock:
+            self._pod(namespace, name)
+            text = self.pod_logs.get((namespace, name), "")
+        if tail_lines is not None:
+            return "\n".join(text.splitlines()[-tail_lines:])
+        return text
 
     def set_pod_unschedulable(self, namespace: str, name: str,
                               message: str = "0/8 nodes are available",
