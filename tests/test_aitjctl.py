@@ -94,3 +94,20 @@ def test_logs(cluster, capsys):
     assert "==>" not in out
     assert cli.main(["logs", "j", "--replica", "nope-9", "-n", NS],
                     api=api) == 1
+
+
+def test_nodes(cluster, capsys, monkeypatch):
+    import json
+    from trainingjob_operator_amd.agent import gpu_health
+    from trainingjob_operator_amd.agent.node_agent import NodeAgent
+    api, _ = cluster
+    api.add_node("gpu-node", ready=True)
+    monkeypatch.setenv(gpu_health.FAKE_ENV, json.dumps(
+        {"gpus": [{"index": 0, "temp_c": 60},
+                  {"index": 1, "temp_c": 115, "message": "hot"}]}))
+    NodeAgent(api, "gpu-node", expected_gpus=2).probe_and_publish()
+    assert cli.main(["nodes"], api=api) == 0
+    out = capsys.readouterr().out
+    assert "NODE" in out and "gpu-node" in out
+    assert "False" in out and "gpu1: hot" in out
+    assert "node-0" in out   # plain node without the agent shows "-"

@@ -155,6 +155,35 @@ def cmd_delete(api, namespace: str, name: str, out=None) -> int:
     return 0
 
 
+def cmd_nodes(api, out=None) -> int:
+    """Per-node GPU health as published by the node agent (EDLGPUHealthy
+    condition + the gpu-health annotation detail)."""
+    import json
+
+    from .agent.node_agent import GPU_HEALTH_ANNOTATION, GPU_HEALTH_CONDITION
+    out = out or sys.stdout
+    rows = [["NODE", "READY", "GPU-HEALTH", "GPUS", "DETAIL"]]
+    for node in api.list_nodes():
+        name = node.get("metadata", {}).get("name", "?")
+        conds = {c.get("type"): c.get("status")
+                 for c in node.get("status", {}).get("conditions") or []}
+        ready = conds.get("Ready", "?")
+        gh = conds.get(GPU_HEALTH_CONDITION, "-")
+        ngpus, detail = "-", ""
+        ann = (node.get("metadata", {}).get("annotations") or {}).get(
+            GPU_HEALTH_ANNOTATION)
+        if ann:
+            try:
+                rep = json.loads(ann)
+                ngpus = str(len(rep.get("gpus", [])))
+                detail = rep.get("summary", "")[:50]
+            except ValueError:
+                detail = "unparseable annotation"
+        rows.append([name, ready, gh, ngpus, detail])
+    print(_fmt_table(rows), file=out)
+    return 0
+
+
 def main(argv=None, api=None) -> int:
     common = argparse.ArgumentParser(add_help=False)
     common.add_argument("-n", "--namespace", default="default")
@@ -172,6 +201,7 @@ def main(argv=None, api=None) -> int:
     r.add_argument("--role", default="trainer")
     x = sub.add_parser("delete", parents=[common])
     x.add_argument("name")
+    sub.add_parser("nodes", parents=[common])
     lg = sub.add_parser("logs", parents=[common])
     lg.add_argument("name")
     lg.add_argument("--replica", default="",
@@ -194,6 +224,8 @@ def main(argv=None, api=None) -> int:
     if args.cmd == "logs":
         return cmd_logs(api, args.namespace, args.name, args.replica,
                         args.tail)
+    if args.cmd == "nodes":
+        return cmd_nodes(api)
     return 2
 
 
