@@ -101,3 +101,29 @@ def test_trainer_applies_schedule():
     assert lrs[0] < lrs[1] < lrs[2] == pytest.approx(1e-2)
     assert lrs[3] == pytest.approx(1e-2)
     assert lrs[4] < lrs[3]
+
+
+def test_tied_embeddings():
+    """tie_embeddings shares one matrix between embed and lm_head (the
+    small-model convention) and the param count reflects it."""
+    import dataclasses
+    cfg = dataclasses.replace(LLAMA_TINY, tie_embeddings=True)
+    m = LlamaModel(cfg)
+    assert m.lm_head.weight is m.embed.weight
+    untied = LlamaModel(LLAMA_TINY)
+    v_h = LLAMA_TINY.vocab_size * LLAMA_TINY.hidden_size
+    n_tied = sum(p.numel() for p in m.parameters())
+    n_untied = sum(p.numel() for p in untied.parameters())
+    assert n_untied - n_tied == v_h
+    assert cfg.n_params == n_tied and LLAMA_TINY.n_params == n_untied
+    # flat store keeps ONE copy and training still steps
+    from trainingjob_operator_amd.training import TrainConfig, Trainer
+    import trainingjob_operator_amd.models.config as mc
+    mc.CONFIGS["llama-tiny-tied"] = dataclasses.replace(
+        cfg, name="llama-tiny-tied")
+    try:
+        tr = Trainer(TrainConfig(model="llama-tiny-tied", micro_batch=1,
+                                 grad_accum=1, seq_len=16))
+        assert float(tr.train_step()) > 0
+    finally:
+        mc.CONFIGS.pop("llama-tiny-tied", None)
