@@ -102,15 +102,28 @@ class TPBlock(nn.Module):
 
 
 class TPLlamaModel(nn.Module):
-    def __init__(self, cfg: LlamaConfig, group=None):
+    """vocab_parallel_head=True swaps the replicated lm_head (1 GB bf16 at
+    Llama-3's 128k vocab, plus full-vocab logits) for a row-sharded head
+    with the fused sharded CE (parallel/vocab_parallel.py): weight AND
+    logits shrink to 1/tp per rank."""
+
+    def __init__(self, cfg: LlamaConfig, group=None,
+                 vocab_parallel_head: bool = False):
         super().__init__()
         self.cfg = cfg
         self.group = group
+        self.vocab_parallel_head = vocab_parallel_head
         self.embed = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
         self.blocks = nn.ModuleList(TPBlock(cfg, group)
                                     for _ in range(cfg.num_layers))
         self.final_norm_weight = nn.Parameter(torch.ones(cfg.hidden_size))
-        self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size, bias=False)
+        if vocab_parallel_head:
+            from .vocab_parallel import VocabParallelLMHead
+            self.lm_head = VocabParallelLMHead(cfg.hidden_size,
+                                               cfg.vocab_size, group)
+        else:
+            self.lm_head = nn.Linear(cfg.hidden_size, cfg.vocab_size,
+                                     bias=False)
         self.register_buffer("inv_freq",
                              make_inv_freq(cfg.head_dim, cfg.rope_theta),
                              persistent=False)
@@ -122,6 +135,16 @@ class TPLlamaModel(nn.Module):
             x, residual = blk(x, residual, self.inv_freq)
         normed, _ = fused_rmsnorm(x, self.final_norm_weight, residual,
                                   self.cfg.norm_eps)
+        if self.vocab_parallel_head:
+            B, S, H = normed.shape
+            if targets is None:
+                from .tp import _GatherFromTP
+                return _GatherFromTP.apply(
+                    self.lm_head.proj(normed), self.group)
+            per_tok = self.lm_head.loss(
+                normed.reshape(B * S, H), targets.reshape(B * S))
+            n_valid = (targets.reshape(-1) != -100).sum().clamp(min=1)
+            return per_tok.sum() / n_valid
         logits = self.lm_head(normed)
         if targets is None:
             return logits
@@ -138,7 +161,11 @@ class TPLlamaModel(nn.Module):
         columns; norms/embed/head replicate."""
         g = self.group
         self.embed.weight.copy_(full.embed.weight)
-        self.lm_head.weight.copy_(full.lm_head.weight)
+        if getattr(self, "vocab_parallel_head", False):  # SPLlama: absent
+            self.lm_head.proj.weight.copy_(
+                shard_from(full.lm_head.weight, 0, g))
+        else:
+            self.lm_head.weight.copy_(full.lm_head.weight)
         self.final_norm_weight.copy_(full.final_norm_weight)
         q_size = self.cfg.num_heads * self.cfg.head_dim
         kv_size = self.cfg.num_kv_heads * self.cfg.head_dim

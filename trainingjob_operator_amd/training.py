@@ -43,6 +43,9 @@ class TrainConfig:
     # contiguous -> adjacent GPUs over xGMI). 1 = pure DP (the flagship
     # 8B config — one replica fits a 288 GB MI355X).
     tp_size: int = 1
+    # vocab-parallel LM head + sharded CE under TP: head weight and
+    # logits shrink to 1/tp per rank (parallel/vocab_parallel.py)
+    vocab_parallel: bool = False
     # Megatron-style sequence parallelism on top of TP (tp_size > 1):
     # norms/residual run on seq shards with all-gather/reduce-scatter
     # seams instead of the f/g all-reduces (parallel/sp.py)
@@ -107,14 +110,18 @@ class Trainer:
         from .parallel.groups import build_topology
         self.topo = build_topology(cfg.tp_size)
         if cfg.tp_size > 1:
+            kw = {}
             if cfg.sequence_parallel:
+                assert not cfg.vocab_parallel, \
+                    "vocab_parallel with sequence_parallel is roadmap"
                 from .parallel.sp import SPLlamaModel as _TPModel
             else:
                 from .parallel.tp_llama import TPLlamaModel as _TPModel
+                kw["vocab_parallel_head"] = cfg.vocab_parallel
             full = build_model(cfg.model_config, device,
                                cfg.checkpoint_activations)
             self.model = _TPModel(cfg.model_config,
-                                  group=self.topo.tp_group).to(
+                                  group=self.topo.tp_group, **kw).to(
                 full.embed.weight.dtype).to(device)
             self.model.inv_freq = full.inv_freq
             self.model.shard_from_full(full)
