@@ -143,6 +143,11 @@ def main(argv=None) -> int:
                     default=os.environ.get("TRAININGJOB_ZERO1", "") == "1",
                     help="ZeRO-1: shard optimizer state across the dp "
                          "group (pure-DP jobs)")
+    ap.add_argument("--vocab-parallel", action="store_true",
+                    default=os.environ.get("TRAININGJOB_VOCAB_PARALLEL",
+                                           "") == "1",
+                    help="shard the LM head + CE over the tp group "
+                         "(requires --tp > 1)")
     ap.add_argument("--sp", action="store_true",
                     default=os.environ.get("TRAININGJOB_SP", "") == "1",
                     help="sequence parallelism on top of --tp (Megatron "
@@ -199,6 +204,10 @@ def main(argv=None) -> int:
     if args.zero1:
         assert args.tp == 1 and args.pp == 1 and not args.ep, \
             "--zero1 launcher support is pure-DP (composed grids: roadmap)"
+    if args.vocab_parallel:
+        assert args.tp > 1 and args.pp == 1 and not args.ep, \
+            "--vocab-parallel is a pure-TP option"
+        assert not args.sp, "vocab-parallel with --sp is roadmap"
     if args.sp:
         assert args.tp > 1, "--sp requires --tp > 1"
         assert args.seq_len % args.tp == 0, "--sp needs seq_len % tp == 0"
@@ -208,6 +217,7 @@ def main(argv=None) -> int:
         grad_accum=args.grad_accum, seq_len=args.seq_len, lr=args.lr,
         checkpoint_activations=args.checkpoint_activations,
         tp_size=args.tp, sequence_parallel=args.sp, zero1=args.zero1,
+        vocab_parallel=args.vocab_parallel,
         data_path=args.data_path, data_dtype=args.data_dtype,
         warmup_steps=args.warmup_steps, lr_decay_steps=args.lr_decay_steps,
         min_lr=args.min_lr)
