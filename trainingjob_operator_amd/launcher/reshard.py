@@ -9,6 +9,9 @@ compute code); this is the MI355X stack's answer to "the elastic policy
 changed the topology, resume anyway": DP resizes need nothing (flat state
 is replicated — checkpoint.py), PP/TP changes run through here.
 
+Vocab-parallel-head TP streams (lm_head.proj.weight shards) are not yet
+convertible — the name-set assertions fail loudly rather than guessing.
+
 Streams are the exact files Checkpointer writes (ckpt_stepNNNNNNNN.pt with
 names/offsets/shapes metadata). Directory conventions match the launcher:
 full -> DIR/, PP -> DIR/stage{i}/, TP -> DIR/tp{r}/, EP -> DIR/ep{r}/.
