@@ -86,3 +86,21 @@ def test_generate_cli(tmp_path):
     ids = [int(t) for t in r.stdout.strip().split(",")]
     assert len(ids) == 4 and all(0 <= t < 512 for t in ids)
     assert "loaded step 1" in r.stderr
+
+
+def test_moe_generate_matches_uncached():
+    from trainingjob_operator_amd.models.moe_llama import (
+        MOE_TINY, MoELlamaModel,
+    )
+    torch.manual_seed(21)
+    moe = MoELlamaModel(MOE_TINY).eval()
+    g = torch.Generator().manual_seed(22)
+    prompt = torch.randint(0, MOE_TINY.vocab_size, (1, 6), generator=g)
+    out = generate(moe, prompt, max_new_tokens=5)
+    assert out.shape == (1, 11)
+    ref = prompt.clone()
+    with torch.no_grad():
+        for _ in range(5):
+            nxt = moe(ref)[:, -1].argmax(dim=-1)
+            ref = torch.cat([ref, nxt[:, None]], dim=1)
+    assert torch.equal(out, ref)

@@ -1,4 +1,5 @@
-"""KV-cached autoregressive generation for the Llama family.
+"""KV-cached autoregressive generation for the Llama family (dense and
+MoE — MoE blocks route each decoded token through their experts).
 
 Training is the flagship workload, but a framework its users can switch
 to needs an eval/inference path: greedy or temperature/top-k sampling
@@ -97,7 +98,9 @@ def _forward_cached(model: LlamaModel, tokens: torch.Tensor,
         normed, residual = fused_rmsnorm(attn_out,
                                          blk.post_attn_norm_weight,
                                          residual, cfg.norm_eps)
-        x = blk.mlp(normed)
+        # dense blocks carry .mlp; MoE blocks carry .moe (routing works
+        # per token, so single-token decode steps route normally)
+        x = blk.mlp(normed) if hasattr(blk, "mlp") else blk.moe(normed)
     cache.len = pos0 + tokens.shape[1]
     normed, _ = fused_rmsnorm(x, model.final_norm_weight, residual,
                               cfg.norm_eps)
