@@ -152,3 +152,27 @@ def test_launcher_eval_loop(tmp_path):
     evals = [e for e in events if e["kind"] == "eval"]
     assert len(evals) == 2                  # steps 2 and 4
     assert all("eval_loss" in e and e["eval_loss"] > 0 for e in evals)
+
+
+def test_bench_contract_two_ranks(tmp_path):
+    """The driver's exact multi-rank invocation shape: torchrun-style env,
+    2 gloo ranks, rank 0 prints ONE JSON line with the whole-job value."""
+    import json
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29733", "--no-python", "--",
+         sys.executable, os.path.join(root, "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1",
+         "--seq-len", "32", "--micro-batch", "1", "--grad-accum", "1"],
+        capture_output=True, text=True, timeout=420, cwd=root)
+    assert r.returncode == 0, r.stderr[-2000:]
+    lines = [l for l in r.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, r.stdout[-1500:]      # exactly one JSON line
+    rec = json.loads(lines[0])
+    assert rec["n_gpus"] == 2 and rec["steps"] == 2
+    assert rec["scaling"] == "weak" and rec["dtype"] == "bf16"
+    assert rec["value"] > 0 and rec["config"]["parallelism"] == "dp2"

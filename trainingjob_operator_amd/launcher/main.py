@@ -94,8 +94,9 @@ def _run_ps_mode(args) -> int:
 
 
 def _saves_ckpt(args, ctx, trainer) -> bool:
-    """Who writes checkpoints: DP -> rank 0; TP -> the dp_rank==0 replica
-    (every tp shard, its own stream); PP -> every stage."""
+    """Who writes checkpoints: DP -> rank 0; ZeRO-1 -> every rank (each
+    owns a moment shard); PP -> every stage's dp-replica 0; EP -> every
+    expert shard's edp-replica 0; TP -> every shard's dp-replica 0."""
     if args.zero1:
         return True                   # every rank owns a moment shard
     if args.pp > 1:
