@@ -560,3 +560,95 @@ def test_pp_activation_checkpointing_matches(tmp_path):
              nprocs=2, join=True)
     import json
     assert os.path.exists(os.path.join(str(tmp_path), "ca_1.json"))
+
+
+def _ppxepxtp_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from trainingjob_operator_amd.models.moe_llama import (
+            MOE_TINY, MoELlamaModel,
+        )
+        from trainingjob_operator_amd.ops import make_inv_freq
+        from trainingjob_operator_amd.parallel.ep import (
+            diversify_experts, solo_group,
+        )
+        from trainingjob_operator_amd.parallel.groups import build_grid
+        from trainingjob_operator_amd.parallel.pp import (
+            PPTrainer, partition_layers,
+        )
+        from trainingjob_operator_amd.parallel.tp import shard_from
+        from trainingjob_operator_amd.training import (
+            TrainConfig, synthetic_batches,
+        )
+        cfg = TrainConfig(model="moe-tiny", micro_batch=2, grad_accum=2,
+                          seq_len=16, lr=2e-3, clip_grad_norm=1.0)
+        grid = build_grid(tp_size=2, pp_size=2)  # 8 = plane2 x pp2 x tp2
+        assert (grid.dp_size, grid.pp_size, grid.tp_size) == (2, 2, 2)
+        tr = PPTrainer(cfg, grid=grid, schedule="gpipe")
+        assert tr._attn_shard_spans, "3-way clip classes must exist"
+
+        # single-process reference over BOTH planes' data streams
+        solo = solo_group()
+        torch.manual_seed(cfg.seed)
+        ref = MoELlamaModel(MOE_TINY, ep_group=solo)
+        diversify_experts(ref, cfg.seed, ep_rank=0)
+        ref = ref.to(torch.bfloat16)
+        ref.inv_freq = make_inv_freq(MOE_TINY.head_dim, MOE_TINY.rope_theta)
+        for r in range(grid.dp_size):
+            data = synthetic_batches(cfg, torch.device("cpu"), rank=r)
+            for _ in range(cfg.grad_accum):
+                tokens, targets = next(data)
+                (ref(tokens, targets)
+                 / (cfg.grad_accum * grid.dp_size)).backward()
+
+        micros = [next(tr.data) for _ in range(cfg.grad_accum)]
+        tr.sched.step(micros, tr.hidden_shape, tr.act_dtype)
+        fg = tr.store.flat_grad
+        fg.mul_(1.0 / grid.dp_size)
+        for s_, e_ in tr._moe_dense_spans:
+            dist.all_reduce(fg[s_:e_], group=grid.dp_group)
+
+        tol = dict(atol=3e-2, rtol=8e-2)
+        fblk = ref.blocks[partition_layers(
+            MOE_TINY.num_layers, grid.pp_size)[grid.pp_rank][0]]
+        blk = tr.stage.blocks[0]
+        # replicated (router), tp attn shard, and (plane x tp) expert shard
+        assert torch.allclose(blk.moe.router.weight.grad.float(),
+                              fblk.moe.router.weight.grad.float(), **tol)
+        q_size = MOE_TINY.num_heads * MOE_TINY.head_dim
+        kv = MOE_TINY.num_kv_heads * MOE_TINY.head_dim
+        gq = fblk.attn.qkv_proj.weight.grad.split([q_size, kv, kv], 0)[0]
+        assert torch.allclose(blk.attn.q_proj.weight.grad.float(),
+                              shard_from(gq, 0, grid.tp_group).float(),
+                              **tol)
+        per = blk.moe.experts_per_rank
+        src = fblk.moe.experts[grid.dp_rank * per + 0]
+        if src.gate_proj.weight.grad is not None:
+            assert torch.allclose(
+                blk.moe.experts[0].gate_proj.weight.grad.float(),
+                shard_from(src.gate_proj.weight.grad, 0,
+                           grid.tp_group).float(), **tol)
+
+        tr._clip_grads()
+        tr.opt.step()
+        tr.opt.zero_grad()
+        tr.step_count += 1
+        for _ in range(2):
+            loss = tr.train_step()
+        if loss is not None:     # last-stage ranks: tp peers bit-agree
+            mx, mn = loss.clone(), loss.clone()
+            dist.all_reduce(mx, op=dist.ReduceOp.MAX, group=grid.tp_group)
+            dist.all_reduce(mn, op=dist.ReduceOp.MIN, group=grid.tp_group)
+            assert torch.equal(mx, mn)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_pp_x_ep_x_tp_grid_matches_single_process():
+    """The full MoE 3D grid (expert-plane 2 x pp 2 x tp 2 on 8 ranks):
+    first-step gradients of every shard class match a single process over
+    both planes' batches; tp peers stay loss-identical across steps."""
+    mp.spawn(_ppxepxtp_worker, args=(8, _free_port()), nprocs=8, join=True)

@@ -386,21 +386,35 @@ class PPTrainer:
         # identical init on every dp replica of a stage (same seed)
         torch.manual_seed(cfg.seed)
         if self._is_moe:
-            # PP x EP: the grid's dp axis doubles as the EP plane — the
-            # ranks holding the SAME stage in the other pipelines hold
+            # PP x EP (x TP): the grid's dp axis doubles as the EP plane —
+            # the ranks holding the SAME stage in the other pipelines hold
             # the other expert shards of that stage (groups.py: "EP == DP
-            # for MoE"). Pure PP (no grid / dp 1) keeps all experts local.
-            assert grid is None or grid.tp_size == 1, \
-                "PP x EP x TP is roadmap"
+            # for MoE"); tp_size > 1 additionally tensor-shards attention
+            # and the experts (TPExpert). Pure PP (no grid / dp 1) keeps
+            # all experts local.
             from ..models.moe_llama import MoELlamaModel
             from .ep import diversify_experts, solo_group
             if grid is not None and grid.dp_size > 1:
                 ep_plane, plane_rank = grid.dp_group, grid.dp_rank
             else:
                 ep_plane, plane_rank = solo_group(), 0
-            with torch.device(self.device):
-                moem = MoELlamaModel(mcfg, ep_group=ep_plane)
-            diversify_experts(moem, cfg.seed, ep_rank=plane_rank)
+            tp_grp = grid.tp_group if grid is not None and \
+                grid.tp_size > 1 else None
+            if tp_grp is not None:
+                # build unsharded, diversify ALL experts, then slice both
+                # the tp and ep shards (exact shard_from_full semantics)
+                with torch.device(self.device):
+                    full = MoELlamaModel(mcfg, ep_group=solo_group())
+                diversify_experts(full, cfg.seed, ep_rank=0)
+                with torch.device(self.device):
+                    moem = MoELlamaModel(mcfg, ep_group=ep_plane,
+                                         tp_group=tp_grp)
+                moem.shard_from_full(full)
+                del full
+            else:
+                with torch.device(self.device):
+                    moem = MoELlamaModel(mcfg, ep_group=ep_plane)
+                diversify_experts(moem, cfg.seed, ep_rank=plane_rank)
             self.stage = MoEStage.from_moe_model(moem, stage_idx, n_stages)
             assert not cfg.checkpoint_activations, \
                 "activation checkpointing re-runs the MoE all-to-all at " \
@@ -447,10 +461,24 @@ class PPTrainer:
                 lambda n: getattr(named[n], "tp_sharded", False))
             self.tp_group = grid.tp_group
         self._expert_spans = self._moe_dense_spans = None
+        self._attn_shard_spans = self._moe_rep_spans = None
         if self._is_moe:
             from .flat import classify_spans
             self._expert_spans, self._moe_dense_spans = classify_spans(
                 self.store, lambda n: ".experts." in n)
+            if self.tp_group is not None:
+                # PP x EP x TP: three clip classes (experts sharded over
+                # plane x tp; attention over tp; the rest replicated)
+                named = dict(self.stage.named_parameters())
+                self._attn_shard_spans, _ = classify_spans(
+                    self.store,
+                    lambda n: getattr(named[n], "tp_sharded", False)
+                    and ".experts." not in n)
+                _, self._moe_rep_spans = classify_spans(
+                    self.store,
+                    lambda n: getattr(named[n], "tp_sharded", False)
+                    or ".experts." in n)
+                self._tp_spans = None   # MoE branch owns the clip
         # every pp rank of a replica draws the same stream; replicas draw
         # DISTINCT streams (dp_rank-keyed), like the DP Trainer
         self.data = make_batches(cfg, self.device, rank=self.dp_rank)
@@ -465,7 +493,22 @@ class PPTrainer:
         if not clip or clip <= 0:
             return
         fg = self.store.flat_grad
-        if self._tp_spans is not None:
+        if self._attn_shard_spans is not None:
+            # PP x EP x TP: experts are (plane x tp)-sharded — sum their
+            # normsq over BOTH; attention shards over tp; replicated once
+            nsq_e = fg.new_zeros((), dtype=torch.float32)
+            for s_, e_ in self._expert_spans:
+                nsq_e += fg[s_:e_].float().pow(2).sum()
+            if self.dp_size > 1:
+                dist.all_reduce(nsq_e, group=self.dp_group)
+            nsq_a = fg.new_zeros((), dtype=torch.float32)
+            for s_, e_ in self._attn_shard_spans:
+                nsq_a += fg[s_:e_].float().pow(2).sum()
+            nsq = nsq_e + nsq_a
+            dist.all_reduce(nsq, group=self.tp_group)
+            for s_, e_ in self._moe_rep_spans:
+                nsq += fg[s_:e_].float().pow(2).sum()
+        elif self._tp_spans is not None:
             # count each tp shard once: sharded normsq summed over the tp
             # group, replicated params counted locally
             sharded, replicated = self._tp_spans
