@@ -164,6 +164,65 @@ class MoEStage(LlamaStage):
         return self._stage_aux()
 
 
+class SPStage(LlamaStage):
+    """Pipeline stage of an SPLlamaModel (PP x TP with sequence
+    parallelism): blocks run on seq shards, so the P2P boundary payload
+    shrinks to S/tp per hop — each tp peer's pipeline ships only its own
+    shard. First stage scatters the embedding output; the last stage
+    gathers through the replicated-seam op before the (replicated) head
+    so its backward splits grads correctly."""
+
+    group = None   # tp group, set by PPTrainer after slicing
+
+    def _shard(self, x):
+        from .tp import _group_size
+        n = _group_size(self.group)
+        if n == 1:
+            return x
+        import torch.distributed as d
+        return x.chunk(n, dim=1)[d.get_rank(self.group)].contiguous()
+
+    def forward(self, x, residual, targets=None):
+        from ..ops import fused_cross_entropy as _ce
+        from .sp import _GatherSeqReplicated
+        if self.is_first:
+            x = self._shard(self.embed(x))
+            residual = None
+        for blk in self.blocks:
+            x, residual = blk(x, residual, self.inv_freq)
+        if not self.is_last:
+            return x, residual
+        normed, _ = fused_rmsnorm(x, self.final_norm_weight, residual,
+                                  self.cfg.norm_eps)
+        normed = _GatherSeqReplicated.apply(normed, self.group)
+        logits = self.lm_head(normed)
+        if targets is None:
+            return logits
+        T = logits.shape[0] * logits.shape[1]
+        per_tok = _ce(logits.reshape(T, -1).contiguous(),
+                      targets.reshape(T))
+        n_valid = (targets.reshape(T) != -100).sum().clamp(min=1)
+        return per_tok.sum() / n_valid
+
+    @torch.no_grad()
+    def allreduce_sp_grads(self) -> None:
+        """Seq-sharded params OF THIS STAGE (norms, embed if owned) saw
+        only this rank's positions: sum grads over the tp group (mirrors
+        SPLlamaModel.allreduce_sp_grads)."""
+        from .tp import _group_size
+        if _group_size(self.group) == 1:
+            return
+        params = ([b.input_norm_weight for b in self.blocks]
+                  + [b.post_attn_norm_weight for b in self.blocks])
+        if self.embed is not None:
+            params.append(self.embed.weight)
+        if self.final_norm_weight is not None:
+            params.append(self.final_norm_weight)
+        for p in params:
+            if p.grad is not None:
+                dist.all_reduce(p.grad, group=self.group)
+
+
 class GPipeSchedule:
     """Fill-drain (GPipe) schedule: all micro-batch forwards, then all
     backwards in reverse — simple, correct, and bubble-bounded by
@@ -392,6 +451,8 @@ class PPTrainer:
             # for MoE"); tp_size > 1 additionally tensor-shards attention
             # and the experts (TPExpert). Pure PP (no grid / dp 1) keeps
             # all experts local.
+            assert not cfg.sequence_parallel, \
+                "sequence parallelism with MoE stages is roadmap"
             from ..models.moe_llama import MoELlamaModel
             from .ep import diversify_experts, solo_group
             if grid is not None and grid.dp_size > 1:
@@ -423,17 +484,27 @@ class PPTrainer:
             # PP x TP: slice a tensor-parallel model into stages — the
             # stage machinery is block-generic (TPBlock outputs the same
             # full-size (x, residual) pair after its row-parallel
-            # all-reduce, so the P2P seam is unchanged)
+            # all-reduce, so the P2P seam is unchanged). With
+            # cfg.sequence_parallel the blocks are SPBlocks and the
+            # boundary payload is seq-SHARDED (S/tp per hop).
             from ..training import build_model
-            from .tp_llama import TPLlamaModel
+            if cfg.sequence_parallel:
+                from .sp import SPLlamaModel as _M
+            else:
+                from .tp_llama import TPLlamaModel as _M
             full = build_model(mcfg, self.device,
                                cfg.checkpoint_activations)
-            tpm = TPLlamaModel(mcfg, group=grid.tp_group).to(
+            tpm = _M(mcfg, group=grid.tp_group).to(
                 full.embed.weight.dtype).to(self.device)
             tpm.inv_freq = full.inv_freq
             tpm.shard_from_full(full)
             del full
-            self.stage = LlamaStage.from_model(tpm, stage_idx, n_stages)
+            if cfg.sequence_parallel:
+                self.stage = SPStage.from_model(tpm, stage_idx, n_stages)
+                self.stage.group = grid.tp_group
+            else:
+                self.stage = LlamaStage.from_model(tpm, stage_idx,
+                                                   n_stages)
         else:
             self.stage = LlamaStage.from_config(mcfg, stage_idx, n_stages,
                                                 device=self.device)
@@ -482,7 +553,11 @@ class PPTrainer:
         # every pp rank of a replica draws the same stream; replicas draw
         # DISTINCT streams (dp_rank-keyed), like the DP Trainer
         self.data = make_batches(cfg, self.device, rank=self.dp_rank)
-        self.hidden_shape = (cfg.micro_batch, cfg.seq_len, mcfg.hidden_size)
+        seq = cfg.seq_len
+        if grid is not None and cfg.sequence_parallel:
+            assert seq % grid.tp_size == 0
+            seq //= grid.tp_size        # seq-sharded boundary payload
+        self.hidden_shape = (cfg.micro_batch, seq, mcfg.hidden_size)
         self.step_count = 0
 
     def _clip_grads(self) -> None:
@@ -544,6 +619,8 @@ class PPTrainer:
                                 self.cfg.warmup_steps,
                                 self.cfg.lr_decay_steps, self.cfg.min_lr)
         loss = self.sched.step(micros, self.hidden_shape, self.act_dtype)
+        if self.cfg.sequence_parallel and isinstance(self.stage, SPStage):
+            self.stage.allreduce_sp_grads()
         if self.dp_size > 1:
             # stage-peer gradient seam: SUM with 1/world pre-scale (gloo
             # has no AVG; RCCL path matches ddp.py's convention)
