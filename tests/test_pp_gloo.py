@@ -708,3 +708,29 @@ def test_pp_x_sp_matches_pure_pp(tmp_path):
     assert all(x is not None for x in l2 + l3 + ref)
     assert l2 == pytest.approx(l3, abs=1e-3)     # tp peers agree
     assert l2 == pytest.approx(ref, abs=3e-2)    # matches pure PP
+
+
+def _ppsp_launcher_worker(rank, world, port, ckdir):
+    os.environ.update({
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+        "RANK": str(rank), "WORLD_SIZE": str(world),
+        "LOCAL_RANK": str(rank),
+    })
+    from trainingjob_operator_amd.launcher.main import main
+    rc = main(["--model", "llama-tiny", "--steps", "3", "--seq-len", "32",
+               "--grad-accum", "2", "--micro-batch", "1",
+               "--ckpt-every", "2", "--log-every", "1",
+               "--ckpt-dir", ckdir, "--pp", "2", "--tp", "2", "--sp"])
+    assert rc == 0
+
+
+@pytest.mark.timeout(600)
+def test_launcher_pp_tp_sp_mode(tmp_path):
+    """--pp 2 --tp 2 --sp end-to-end through the launcher."""
+    mp.spawn(_ppsp_launcher_worker, args=(4, _free_port(), str(tmp_path)),
+             nprocs=4, join=True)
+    for s in range(2):
+        for t in range(2):
+            names = os.listdir(os.path.join(str(tmp_path),
+                                            f"stage{s}_tp{t}"))
+            assert any(n.startswith("ckpt_step") for n in names), names
