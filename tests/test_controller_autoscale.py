@@ -178,3 +178,49 @@ def test_metrics_count_auto_resizes():
     assert down._value.get() == 1
     # the restart the resize triggers is counted too
     assert m.restarts_total.labels(scope="All")._value.get() == 1
+
+
+def test_auto_role_beside_fixed_role():
+    """Multi-role job: the Auto trainer role scales down while the fixed
+    pserver role is untouched."""
+    api = FakeKubeApi()
+    tc = TrainingJobController(api, OperatorOptions(
+        elastic_unschedulable_grace=60.0))
+    api.create_job(NS, {
+        "apiVersion": C.API_VERSION, "kind": C.CRD_KIND,
+        "metadata": {"name": "auto", "namespace": NS},
+        "spec": {"replicaSpecs": {
+            "pserver": {
+                "replicas": 1, "restartPolicy": "OnFailure",
+                "restartScope": "Replica",
+                "template": {"spec": {"containers": [{
+                    "name": "aitj-ps",
+                    "ports": [{"name": "aitj-p",
+                               "containerPort": 6000}]}]}},
+            },
+            "trainer": {
+                "replicas": 3, "minReplicas": 2, "maxReplicas": 4,
+                "edlPolicy": "Auto",
+                "restartPolicy": "OnFailure", "restartScope": "Replica",
+                "template": {"spec": {"containers": [{
+                    "name": "aitj-main",
+                    "ports": [{"name": "aitj-p",
+                               "containerPort": 5000}]}]}},
+            },
+        }},
+    })
+    t0 = time.time()
+    tc.sync_once(f"{NS}/auto", now=t0)
+    assert len(api.pod_names(NS)) == 4
+    api.set_all_pods_phase(NS, "Running")
+    api.set_pod_unschedulable(NS, "auto-trainer-2", since=t0 - 120)
+    tc.sync_once(f"{NS}/auto", now=t0 + 1)
+    j = job_of(api)
+    assert j.annotations[f"{TARGET_ANNOTATION}-trainer"] == "2"
+    assert f"{TARGET_ANNOTATION}-pserver" not in j.annotations
+    # restart scope Replica: only the trainer role restarts
+    tc.sync_once(f"{NS}/auto", now=t0 + 2)
+    tc.sync_once(f"{NS}/auto", now=t0 + 3)
+    names = api.pod_names(NS)
+    assert "auto-pserver-0" in names
+    assert sum(1 for n in names if "trainer" in n) == 2
