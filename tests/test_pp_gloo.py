@@ -734,3 +734,38 @@ def test_launcher_pp_tp_sp_mode(tmp_path):
             names = os.listdir(os.path.join(str(tmp_path),
                                             f"stage{s}_tp{t}"))
             assert any(n.startswith("ckpt_step") for n in names), names
+
+
+def _deep_1f1b_worker(rank, world, port, outdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        import json
+        from trainingjob_operator_amd.parallel.pp import PPTrainer
+        from trainingjob_operator_amd.training import TrainConfig
+        cfg = TrainConfig(model="llama-tiny", micro_batch=1, grad_accum=6,
+                          seq_len=32, lr=1e-3)
+        tr = PPTrainer(cfg, stage_idx=rank, n_stages=world)  # 1f1b default
+        for _ in range(2):
+            tr.train_step()
+        # the 1F1B memory bound: at most (stages - stage - 1) + 1 live
+        # micro-batches per stage, NOT grad_accum (=6) like GPipe
+        with open(os.path.join(outdir, f"peak{rank}.json"), "w") as f:
+            json.dump(tr.sched.peak_live, f)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(600)
+def test_deep_1f1b_memory_bound(tmp_path):
+    """4-stage 1F1B with 6 micro-batches: every stage's peak live
+    activations equal its warmup depth + 1 — the whole point of 1F1B."""
+    import json
+    world = 4
+    mp.spawn(_deep_1f1b_worker, args=(world, _free_port(), str(tmp_path)),
+             nprocs=world, join=True)
+    for r in range(world):
+        peak = json.load(open(os.path.join(str(tmp_path),
+                                           f"peak{r}.json")))
+        assert peak == (world - r - 1) + 1, (r, peak)
