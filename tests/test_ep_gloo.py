@@ -506,3 +506,40 @@ def test_full_edp_ep_tp_grid_trains():
     keeps expert-dp peers bit-identical."""
     mp.spawn(_full_epxtp_worker, args=(8, _free_port()), nprocs=8,
              join=True)
+
+
+def _starved_expert_worker(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from trainingjob_operator_amd.parallel.ep import MoEMLP
+        torch.manual_seed(17)
+        moe = MoEMLP(hidden=16, ff=32, n_experts=4, top_k=2, group=None)
+        # force every token's top-2 onto rank 0's experts (0 and 1):
+        # rank 1 receives ZERO tokens and must still survive fwd+bwd
+        with torch.no_grad():
+            moe.router.weight.zero_()
+            moe.router.weight[0].fill_(8.0)
+            moe.router.weight[1].fill_(7.0)
+        x = torch.rand(2, 6, 16).requires_grad_()   # positive -> logits>0
+        y = moe(x)
+        y.sum().backward()
+        assert x.grad is not None
+        if rank == 1:
+            # starved experts get no gradient, and that's fine
+            for ex in moe.experts:
+                g = ex.gate_proj.weight.grad
+                assert g is None or float(g.abs().sum()) == 0.0
+        else:
+            assert moe.experts[0].gate_proj.weight.grad is not None
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_starved_expert_rank_survives():
+    """Dropless routing edge: a rank whose experts receive zero tokens
+    must pass forward/backward (empty all-to-all splits)."""
+    mp.spawn(_starved_expert_worker, args=(2, _free_port()), nprocs=2,
+             join=True)
