@@ -428,20 +428,35 @@ class PPTrainer:
         self.device = torch.device(device or "cpu")
         if grid is not None:
             stage_idx, n_stages = grid.pp_rank, grid.pp_size
-            pp_ranks = [(grid.dp_rank * grid.pp_size + p) * grid.tp_size
-                        + grid.tp_rank for p in range(grid.pp_size)]
+            pp_ranks = getattr(grid, "pp_global_ranks", None) or [
+                (grid.dp_rank * grid.pp_size + p) * grid.tp_size
+                + grid.tp_rank for p in range(grid.pp_size)]
             self.dp_size, self.dp_rank = grid.dp_size, grid.dp_rank
             self.dp_group = grid.dp_group
+            # 4-axis MoE grids (groups.build_moe_grid) add true data
+            # parallelism on top of the expert plane
+            self.edp_size = getattr(grid, "edp_size", 1)
+            self.edp_group = getattr(grid, "edp_group", None)
+            self.data_replicas = getattr(grid, "data_replicas",
+                                         grid.dp_size)
+            self.data_rank = getattr(grid, "data_rank", grid.dp_rank)
+            self.dense_dp_group = getattr(grid, "dense_dp_group", None) \
+                or grid.dp_group
         else:
             assert stage_idx is not None and n_stages is not None
             pp_ranks = None
             self.dp_size, self.dp_rank, self.dp_group = 1, 0, None
+            self.edp_size, self.edp_group = 1, None
+            self.data_replicas, self.data_rank = 1, 0
+            self.dense_dp_group = None
         self.grid = grid
         # all stages of MY pipeline (grad-norm seam); None == default
         # group in the pure-PP world where rank == stage
         self.pp_group = grid.pp_group if grid is not None else None
         from ..models.moe_llama import MoELlamaConfig
         self._is_moe = isinstance(mcfg, MoELlamaConfig)
+        assert self.edp_size == 1 or self._is_moe, \
+            "4-axis grids (edp) are for MoE pipelines"
         # identical init on every dp replica of a stage (same seed)
         torch.manual_seed(cfg.seed)
         if self._is_moe:
@@ -552,7 +567,7 @@ class PPTrainer:
                 self._tp_spans = None   # MoE branch owns the clip
         # every pp rank of a replica draws the same stream; replicas draw
         # DISTINCT streams (dp_rank-keyed), like the DP Trainer
-        self.data = make_batches(cfg, self.device, rank=self.dp_rank)
+        self.data = make_batches(cfg, self.device, rank=self.data_rank)
         seq = cfg.seq_len
         if grid is not None and cfg.sequence_parallel:
             assert seq % grid.tp_size == 0
@@ -621,23 +636,28 @@ class PPTrainer:
         loss = self.sched.step(micros, self.hidden_shape, self.act_dtype)
         if self.cfg.sequence_parallel and isinstance(self.stage, SPStage):
             self.stage.allreduce_sp_grads()
-        if self.dp_size > 1:
-            # stage-peer gradient seam: SUM with 1/world pre-scale (gloo
-            # has no AVG; RCCL path matches ddp.py's convention)
+        if self.data_replicas > 1:
+            # stage-peer gradient seam: SUM with 1/replicas pre-scale
+            # (gloo has no AVG; RCCL path matches ddp.py's convention)
             fg = self.store.flat_grad
-            fg.mul_(1.0 / self.dp_size)
+            fg.mul_(1.0 / self.data_replicas)
             if self._expert_spans is None:
                 dist.all_reduce(fg, group=self.dp_group)
             else:
-                # PP x EP: the "dp" peers hold DIFFERENT experts at the
+                # PP x EP: the plane peers hold DIFFERENT experts at the
                 # same flat offsets; expert grads are already complete
-                # (the backward all-to-all summed every pipeline's
-                # tokens) and only need the 1/replicas scale
+                # within the plane (the backward all-to-all summed every
+                # pipeline's tokens) — they sync only across edp replicas
                 for s_, e_ in self._moe_dense_spans:
-                    dist.all_reduce(fg[s_:e_], group=self.dp_group)
+                    dist.all_reduce(fg[s_:e_], group=self.dense_dp_group)
+                if self.edp_size > 1:
+                    for s_, e_ in self._expert_spans:
+                        dist.all_reduce(fg[s_:e_], group=self.edp_group)
             if loss is not None:
-                loss = loss / self.dp_size
-                dist.all_reduce(loss, group=self.dp_group)
+                loss = loss / self.data_replicas
+                dist.all_reduce(loss, group=self.dense_dp_group
+                                if self._expert_spans is not None
+                                else self.dp_group)
         self._clip_grads()
         self.opt.step()
         self.opt.zero_grad()
