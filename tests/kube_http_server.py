@@ -78,6 +78,17 @@ class _Handler(BaseHTTPRequestHandler):
         q = parse_qs(u.query)
         watching = q.get("watch", ["false"])[0] == "true"
         try:
+            if u.path.endswith("/log") and "/pods/" in u.path:
+                parts = u.path.split("/")
+                ns, name = parts[4], parts[6]
+                tail = q.get("tailLines", [None])[0]
+                text = self.fake.read_pod_log(
+                    ns, name, int(tail) if tail else None)
+                self.send_response(200)
+                self.send_header("Content-Type", "text/plain")
+                self.end_headers()
+                self.wfile.write(text.encode())
+                return None
             if (m := POD_RE.match(u.path)):
                 ns, name = m.groups()
                 if watching:

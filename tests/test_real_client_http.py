@@ -127,3 +127,12 @@ def test_full_controller_over_http(cluster):
     tc.sync_once(f"{NS}/e2e")
     job = AITrainingJob.from_dict(fake.get_job(NS, "e2e"))
     assert job.status.phase == Phase.SUCCEEDED
+
+
+def test_pod_log_over_http(cluster):
+    fake, api = cluster
+    fake.create_pod("d", {"metadata": {"name": "p1", "namespace": "d"},
+                          "spec": {"containers": []}})
+    fake.set_pod_log("d", "p1", "line1\nline2\nline3")
+    assert api.read_pod_log("d", "p1") == "line1\nline2\nline3"
+    assert api.read_pod_log("d", "p1", tail_lines=1) == "line3"
