@@ -136,3 +136,15 @@ def test_pod_log_over_http(cluster):
     fake.set_pod_log("d", "p1", "line1\nline2\nline3")
     assert api.read_pod_log("d", "p1") == "line1\nline2\nline3"
     assert api.read_pod_log("d", "p1", tail_lines=1) == "line3"
+
+
+def test_aitjctl_over_http(cluster, capsys):
+    """The ops CLI works against the real REST client end-to-end."""
+    fake, api = cluster
+    fake.create_job("d", job_manifest("cli-job"))
+    from trainingjob_operator_amd import cli
+    assert cli.main(["get", "-n", "d"], api=api) == 0
+    out = capsys.readouterr().out
+    assert "cli-job" in out
+    assert cli.main(["describe", "cli-job", "-n", "d"], api=api) == 0
+    assert "Name:      cli-job" in capsys.readouterr().out
