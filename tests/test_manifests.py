@@ -47,3 +47,31 @@ def test_crd_has_status_subresource_and_shortname():
     v = crd["spec"]["versions"][0]
     assert v["subresources"] == {"status": {}}
     assert crd["spec"]["names"]["shortNames"] == ["aitj"]
+
+
+def test_validate_job_cli():
+    import glob
+    import subprocess
+    import sys
+    paths = sorted(glob.glob(os.path.join(
+        ROOT, "manifests", "examples", "*.yaml")))
+    r = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "scripts", "validate_job.py")]
+        + paths, capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert r.stdout.count("OK") == len(paths)
+    # an invalid spec is rejected with a non-zero exit
+    import tempfile
+    with tempfile.NamedTemporaryFile("w", suffix=".yaml",
+                                     delete=False) as f:
+        f.write("apiVersion: elasticdeeplearning.ai/v1\n"
+                "kind: AITrainingJob\n"
+                "metadata: {name: Bad_Name}\n"
+                "spec: {replicaSpecs: {t: {template: {spec: "
+                "{containers: [{name: main}]}}}}}\n")
+        bad = f.name
+    r = subprocess.run(
+        [sys.executable, os.path.join(ROOT, "scripts", "validate_job.py"),
+         bad], capture_output=True, text=True, timeout=300)
+    os.unlink(bad)
+    assert r.returncode != 0
