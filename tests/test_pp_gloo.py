@@ -806,3 +806,30 @@ def test_moe_4axis_grid_trains():
     the MoE pipeline grid — expert grads all-reduce across edp and the
     replicas stay bit-identical."""
     mp.spawn(_moe4d_worker, args=(8, _free_port()), nprocs=8, join=True)
+
+
+def _moe4d_launcher_worker(rank, world, port, ckdir):
+    os.environ.update({
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+        "RANK": str(rank), "WORLD_SIZE": str(world),
+        "LOCAL_RANK": str(rank),
+    })
+    from trainingjob_operator_amd.launcher.main import main
+    rc = main(["--model", "moe-tiny", "--steps", "3", "--seq-len", "16",
+               "--grad-accum", "2", "--micro-batch", "1",
+               "--ckpt-every", "2", "--log-every", "1",
+               "--ckpt-dir", ckdir, "--pp", "2", "--moe-plane", "2"])
+    assert rc == 0
+
+
+@pytest.mark.timeout(600)
+def test_launcher_moe_4axis(tmp_path):
+    """--pp 2 --moe-plane 2 on 8 ranks = edp2 x plane2 x pp2 through the
+    launcher; the edp-replica-0 half writes per-(stage, plane) streams."""
+    mp.spawn(_moe4d_launcher_worker, args=(8, _free_port(), str(tmp_path)),
+             nprocs=8, join=True)
+    for s in range(2):
+        for pl in range(2):
+            names = os.listdir(os.path.join(str(tmp_path),
+                                            f"stage{s}_pl{pl}"))
+            assert any(n.startswith("ckpt_step") for n in names), names

@@ -99,6 +99,9 @@ def _saves_ckpt(args, ctx, trainer) -> bool:
     expert shard's edp-replica 0; TP -> every shard's dp-replica 0."""
     if args.zero1:
         return True                   # every rank owns a moment shard
+    if args.pp > 1 and getattr(args, "moe_plane", 0):
+        # 4-axis: every (stage, plane, tp) shard, edp-replica 0 only
+        return trainer.edp_rank == 0
     if args.pp > 1:
         return trainer.dp_rank == 0   # every stage, dp-replica 0 only
     if args.ep:
@@ -168,6 +171,11 @@ def main(argv=None) -> int:
     ap.add_argument("--role", default=os.environ.get(
         "TRAININGJOB_REPLICA_NAME", "trainer"),
         help="this pod's replica role (injected by the operator)")
+    ap.add_argument("--moe-plane", type=int, default=int(os.environ.get(
+        "TRAININGJOB_MOE_PLANE", "0")),
+        help="expert-plane size for 4-axis MoE pipeline grids "
+             "(world = edp x plane x pp x tp; requires --pp and a MoE "
+             "model; 0 = plane inferred as world/(pp*tp))")
     ap.add_argument("--pp-schedule", choices=("1f1b", "gpipe"),
                     default=os.environ.get("TRAININGJOB_PP_SCHEDULE", "1f1b"),
                     help="pipeline schedule: 1f1b bounds live microbatches "
@@ -229,14 +237,25 @@ def main(argv=None) -> int:
         assert ctx.world_size % (args.pp * max(args.tp, 1)) == 0, \
             f"world {ctx.world_size} not divisible by " \
             f"pp*tp={args.pp * max(args.tp, 1)}"
-        from ..parallel.groups import build_grid
         from ..parallel.pp import PPTrainer
-        grid = build_grid(tp_size=max(args.tp, 1), pp_size=args.pp)
+        if args.moe_plane:
+            # 4-axis MoE grid: true data parallelism (edp) on top of the
+            # expert plane x pp x tp
+            from ..parallel.groups import build_moe_grid
+            grid = build_moe_grid(plane_size=args.moe_plane,
+                                  pp_size=args.pp,
+                                  tp_size=max(args.tp, 1))
+        else:
+            from ..parallel.groups import build_grid
+            grid = build_grid(tp_size=max(args.tp, 1), pp_size=args.pp)
         trainer = PPTrainer(cfg, schedule=args.pp_schedule, grid=grid)
-        # per-(stage, tp-shard) checkpoint streams, written by the
-        # dp_rank==0 replica — see _saves_ckpt
-        sub = (f"stage{grid.pp_rank}_tp{grid.tp_rank}" if args.tp > 1
-               else f"stage{grid.pp_rank}")
+        # per-(stage, tp-shard[, plane-shard]) checkpoint streams,
+        # written by the dp_rank==0 (and edp_rank==0) replica
+        sub = f"stage{grid.pp_rank}"
+        if args.moe_plane:
+            sub += f"_pl{grid.dp_rank}"
+        if args.tp > 1:
+            sub += f"_tp{grid.tp_rank}"
         ckpt = Checkpointer(os.path.join(args.ckpt_dir, sub))
     elif args.ep:
         from ..parallel.ep import EPTrainer

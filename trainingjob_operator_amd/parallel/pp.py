@@ -436,6 +436,7 @@ class PPTrainer:
             # 4-axis MoE grids (groups.build_moe_grid) add true data
             # parallelism on top of the expert plane
             self.edp_size = getattr(grid, "edp_size", 1)
+            self.edp_rank = getattr(grid, "edp_rank", 0)
             self.edp_group = getattr(grid, "edp_group", None)
             self.data_replicas = getattr(grid, "data_replicas",
                                          grid.dp_size)
@@ -446,7 +447,7 @@ class PPTrainer:
             assert stage_idx is not None and n_stages is not None
             pp_ranks = None
             self.dp_size, self.dp_rank, self.dp_group = 1, 0, None
-            self.edp_size, self.edp_group = 1, None
+            self.edp_size, self.edp_rank, self.edp_group = 1, 0, None
             self.data_replicas, self.data_rank = 1, 0
             self.dense_dp_group = None
         self.grid = grid
