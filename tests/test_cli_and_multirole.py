@@ -85,3 +85,30 @@ def test_operator_yaml_config(tmp_path):
     import pytest
     with pytest.raises(ValueError):
         OperatorOptions.from_yaml(cfgfile)
+
+
+def test_launcher_mode_flag_validation(tmp_path, monkeypatch):
+    """Invalid mode combinations fail fast with clear messages."""
+    import os
+    import subprocess
+    import sys
+    env = {k: v for k, v in os.environ.items()
+           if k not in ("RANK", "WORLD_SIZE")}
+    env.update({"MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29741"})
+
+    def run(*flags):
+        return subprocess.run(
+            [sys.executable, "-m",
+             "trainingjob_operator_amd.launcher.main",
+             "--model", "llama-tiny", "--steps", "1", "--seq-len", "32",
+             "--ckpt-dir", str(tmp_path)] + list(flags),
+            env=env, capture_output=True, text=True, timeout=300)
+
+    r = run("--sp")                       # sp without tp
+    assert r.returncode != 0 and "--sp requires --tp" in r.stderr
+    r = run("--zero1", "--tp", "2")       # zero1 with tp
+    assert r.returncode != 0 and "pure-DP" in r.stderr
+    r = run("--vocab-parallel")           # vp without tp
+    assert r.returncode != 0 and "pure-TP" in r.stderr
+    r = run("--pp", "2", "--ep", "2")     # pp + ep flag clash
+    assert r.returncode != 0 and "drop --ep" in r.stderr
