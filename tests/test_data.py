@@ -66,3 +66,37 @@ def test_make_batches_dispatch(token_file):
     cfg.data_path = "/nonexistent/corpus.bin"
     with pytest.raises(FileNotFoundError):
         make_batches(cfg, None)
+
+
+def _pp_data_worker(rank, world, port, data_path, ckdir):
+    import torch.distributed  # noqa: F401
+    os.environ.update({
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+        "RANK": str(rank), "WORLD_SIZE": str(world),
+        "LOCAL_RANK": str(rank),
+    })
+    from trainingjob_operator_amd.launcher.main import main
+    rc = main(["--model", "llama-tiny", "--steps", "2", "--seq-len", "32",
+               "--grad-accum", "1", "--micro-batch", "1",
+               "--ckpt-every", "10", "--log-every", "1",
+               "--ckpt-dir", ckdir, "--pp", str(world),
+               "--data-path", data_path])
+    assert rc == 0
+
+
+def test_pp_launcher_trains_on_token_file(token_file, tmp_path):
+    """--pp with --data-path: pipeline stages draw the same real-data
+    stream (stage 0 consumes tokens, the last consumes targets)."""
+    import torch.multiprocessing as mp
+
+    def _free_port():
+        import socket
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+        s.close()
+        return port
+
+    mp.spawn(_pp_data_worker,
+             args=(2, _free_port(), token_file, str(tmp_path)),
+             nprocs=2, join=True)
