@@ -115,6 +115,12 @@ class _Handler(BaseHTTPRequestHandler):
                 if name:
                     return self._send(200, self.fake.get_job(ns, name))
                 return self._send(200, {"items": self.fake.list_jobs(ns)})
+            if "/events" in u.path and u.path.startswith("/api/v1/"):
+                ns = u.path.split("/")[4]
+                fs = q.get("fieldSelector", [""])[0]
+                name = fs.split("=", 1)[1] if "=" in fs else None
+                return self._send(200, {"items": self.fake.list_events(
+                    ns, involved_name=name)})
             if (m := LEASE_RE.match(u.path)):
                 ns, name = m.groups()
                 return self._send(200, self.fake.get_lease(ns, name))
@@ -154,6 +160,12 @@ class _Handler(BaseHTTPRequestHandler):
             if (m := JOB_RE.match(u.path)):
                 ns, name, _status = m.groups()
                 return self._send(200, self.fake.update_job(ns, name, body))
+            if "/events" in u.path and u.path.startswith("/api/v1/"):
+                ns = u.path.split("/")[4]
+                fs = q.get("fieldSelector", [""])[0]
+                name = fs.split("=", 1)[1] if "=" in fs else None
+                return self._send(200, {"items": self.fake.list_events(
+                    ns, involved_name=name)})
             if (m := LEASE_RE.match(u.path)):
                 return self._send(200, self.fake.update_lease(
                     m.group(1), m.group(2), body))

@@ -111,3 +111,11 @@ def test_nodes(cluster, capsys, monkeypatch):
     assert "NODE" in out and "gpu-node" in out
     assert "False" in out and "gpu1: hot" in out
     assert "node-0" in out   # plain node without the agent shows "-"
+
+
+def test_describe_shows_events(cluster, capsys):
+    api, _ = cluster
+    assert cli.main(["describe", "j", "-n", NS], api=api) == 0
+    out = capsys.readouterr().out
+    assert "Events:" in out
+    assert "SuccessfulCreatePod" in out

@@ -148,3 +148,15 @@ def test_aitjctl_over_http(cluster, capsys):
     assert "cli-job" in out
     assert cli.main(["describe", "cli-job", "-n", "d"], api=api) == 0
     assert "Name:      cli-job" in capsys.readouterr().out
+
+
+def test_list_events_over_http(cluster):
+    fake, api = cluster
+    fake.create_job("d", job_manifest("evjob"))
+    fake.events.append({
+        "metadata": {"name": "evjob.1", "namespace": "d"},
+        "involvedObject": {"name": "evjob"},
+        "type": "Normal", "reason": "Test", "message": "hello"})
+    evs = api.list_events("d", involved_name="evjob")
+    assert len(evs) == 1 and evs[0]["reason"] == "Test"
+    assert api.list_events("d", involved_name="other") == []

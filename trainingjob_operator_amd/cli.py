@@ -100,6 +100,20 @@ def cmd_describe(api, namespace: str, name: str, out=None) -> int:
             rows.append(["  " + c.type, c.status, c.reason or "",
                          (c.message or "")[:60]])
         print(_fmt_table(rows), file=out)
+    try:
+        events = api.list_events(namespace, involved_name=name)
+    except Exception:                      # older servers: best-effort
+        events = []
+    if events:
+        print("Events:", file=out)
+        rows = [["  AGE", "TYPE", "REASON", "MESSAGE"]]
+        for e in events[-10:]:
+            rows.append([
+                "  " + _age(e.get("lastTimestamp")
+                            or e.get("firstTimestamp")),
+                e.get("type", ""), e.get("reason", ""),
+                (e.get("message") or "")[:60]])
+        print(_fmt_table(rows), file=out)
     return 0
 
 

@@ -76,6 +76,8 @@ class KubeApi:
 
     # events / leases ----------------------------------------------------
     def create_event(self, namespace: str, event: dict) -> None: ...
+    def list_events(self, namespace: str,
+                    involved_name: Optional[str] = None) -> List[dict]: ...
     def get_lease(self, namespace: str, name: str) -> dict: ...
     def create_lease(self, namespace: str, lease: dict) -> dict: ...
     def update_lease(self, namespace: str, name: str, lease: dict) -> dict: ...
@@ -244,6 +246,14 @@ class RealKubeApi(KubeApi):
             self._req("POST", self._ns_path("events", namespace), event)
         except ApiError:
             pass  # events are best-effort
+
+    def list_events(self, namespace, involved_name=None):
+        params = {}
+        if involved_name:
+            params["fieldSelector"] = \
+                f"involvedObject.name={involved_name}"
+        return self._req("GET", self._ns_path("events", namespace),
+                         params=params)["items"]
 
     def _lease_path(self, namespace):
         return (f"/apis/coordination.k8s.io/v1/namespaces/{namespace}/leases")

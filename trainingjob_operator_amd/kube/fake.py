@@ -232,6 +232,19 @@ class FakeKubeApi(KubeApi):
             self._emit("DELETED", "job", j)
 
     # -- events / leases --------------------------------------------------
+    def list_events(self, namespace, involved_name=None):
+        with self._lock:
+            out = []
+            for e in self.events:
+                if namespace and \
+                        e.get("metadata", {}).get("namespace") != namespace:
+                    continue
+                if involved_name and e.get("involvedObject", {}) \
+                        .get("name") != involved_name:
+                    continue
+                out.append(copy.deepcopy(e))
+            return out
+
     def create_event(self, namespace, event):
         with self._lock:
             self.events.append(event)
