@@ -176,3 +176,37 @@ def test_bench_contract_two_ranks(tmp_path):
     assert rec["n_gpus"] == 2 and rec["steps"] == 2
     assert rec["scaling"] == "weak" and rec["dtype"] == "bf16"
     assert rec["value"] > 0 and rec["config"]["parallelism"] == "dp2"
+
+
+def test_sigusr1_checkpoints_on_demand(tmp_path):
+    """SIGUSR1 forces a checkpoint at the next step boundary (planned
+    maintenance hook), independent of --ckpt-every."""
+    import signal
+    import subprocess
+    import sys
+    import time
+    env = {k: v for k, v in os.environ.items()
+           if k not in ("RANK", "WORLD_SIZE")}
+    env.update({"MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29747"})
+    p = subprocess.Popen(
+        [sys.executable, "-m", "trainingjob_operator_amd.launcher.main",
+         "--model", "llama-tiny", "--steps", "400", "--seq-len", "32",
+         "--grad-accum", "1", "--micro-batch", "1",
+         "--ckpt-every", "100000", "--log-every", "50",
+         "--ckpt-dir", str(tmp_path)],
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        text=True)
+    try:
+        time.sleep(4)                     # let it make a few steps
+        p.send_signal(signal.SIGUSR1)
+        deadline = time.monotonic() + 60
+        while time.monotonic() < deadline:
+            if any(n.startswith("ckpt_step")
+                   for n in os.listdir(str(tmp_path))):
+                break
+            time.sleep(0.3)
+        names = os.listdir(str(tmp_path))
+        assert any(n.startswith("ckpt_step") for n in names), names
+    finally:
+        p.terminate()
+        p.wait(timeout=60)

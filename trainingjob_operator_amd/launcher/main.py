@@ -291,10 +291,17 @@ def main(argv=None) -> int:
              time.time() - t_start)
 
     stop_requested = {"flag": False}
+    ckpt_requested = {"flag": False}
 
     def on_term(signum, frame):
         stop_requested["flag"] = True
     signal.signal(signal.SIGTERM, on_term)
+
+    def on_usr1(signum, frame):
+        # on-demand checkpoint (e.g. before planned node maintenance):
+        # kubectl exec ... -- kill -USR1 1
+        ckpt_requested["flag"] = True
+    signal.signal(signal.SIGUSR1, on_usr1)
 
     from ..utils.tracing import tracer
     trace = tracer("worker")
@@ -333,9 +340,12 @@ def main(argv=None) -> int:
                 trace.event("eval", step=step, eval_loss=round(ev, 4))
                 if metrics:
                     metrics.observe_eval(ev)
-        if step % args.ckpt_every == 0 and _saves_ckpt(args, ctx, trainer):
+        want_ckpt = step % args.ckpt_every == 0 or ckpt_requested["flag"]
+        if want_ckpt and _saves_ckpt(args, ctx, trainer):
             ckpt.save_async(trainer)
-            trace.event("checkpoint", step=step)
+            trace.event("checkpoint", step=step,
+                        on_demand=ckpt_requested["flag"])
+        ckpt_requested["flag"] = False
 
     if _saves_ckpt(args, ctx, trainer):
         ckpt.save_async(trainer, blocking=True)
