@@ -119,3 +119,15 @@ def test_describe_shows_events(cluster, capsys):
     out = capsys.readouterr().out
     assert "Events:" in out
     assert "SuccessfulCreatePod" in out
+
+
+def test_get_all_namespaces(cluster, capsys):
+    api, _ = cluster
+    api.create_job("other", {
+        "apiVersion": C.API_VERSION, "kind": C.CRD_KIND,
+        "metadata": {"name": "elsewhere", "namespace": "other"},
+        "spec": {"replicaSpecs": {"t": {"template": {"spec": {
+            "containers": [{"name": "aitj-x"}]}}}}}})
+    assert cli.main(["get", "-A"], api=api) == 0
+    out = capsys.readouterr().out
+    assert "j" in out and "elsewhere" in out

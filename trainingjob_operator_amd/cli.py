@@ -201,6 +201,7 @@ def cmd_nodes(api, out=None) -> int:
 def main(argv=None, api=None) -> int:
     common = argparse.ArgumentParser(add_help=False)
     common.add_argument("-n", "--namespace", default="default")
+    common.add_argument("-A", "--all-namespaces", action="store_true")
     common.add_argument("--master", default="",
                         help="API server URL (default: kube proxy / "
                              "in-cluster)")
@@ -227,7 +228,7 @@ def main(argv=None, api=None) -> int:
         from .kube.client import RealKubeApi
         api = RealKubeApi(base_url=args.master or None)
     if args.cmd == "get":
-        return cmd_get(api, args.namespace)
+        return cmd_get(api, "" if args.all_namespaces else args.namespace)
     if args.cmd == "describe":
         return cmd_describe(api, args.namespace, args.name)
     if args.cmd == "resize":
