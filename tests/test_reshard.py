@@ -245,3 +245,13 @@ def test_zero_streams_reshard_to_full_and_back(tmp_path):
     z0 = load_stream(Checkpointer(os.path.join(z3_dir, "zero0")).latest())
     m = z0["opt"]["p32"].numel()
     assert torch.equal(z0["opt"]["p32"], full["opt"]["p32"][:m])
+
+
+def test_reshard_rejects_wrong_model(tmp_path):
+    """Converting with the wrong --model fails loudly (name/shape
+    mismatches) instead of writing a corrupt stream."""
+    _, _, full_dir = _trained_full_ckpt(tmp_path)
+    out = os.path.join(str(tmp_path), "out")
+    with pytest.raises(Exception):
+        reshard("llama-1b", full_dir, out, "full", "tp=2")
+    assert not os.path.exists(os.path.join(out, "tp1"))
