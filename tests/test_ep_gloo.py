@@ -543,3 +543,17 @@ def test_starved_expert_rank_survives():
     must pass forward/backward (empty all-to-all splits)."""
     mp.spawn(_starved_expert_worker, args=(2, _free_port()), nprocs=2,
              join=True)
+
+
+def test_moe_param_formula_and_8x7b():
+    from trainingjob_operator_amd.models.moe_llama import (
+        MOE_8X7B, MOE_TINY, MoELlamaModel,
+    )
+    m = MoELlamaModel(MOE_TINY)
+    assert MOE_TINY.n_params == sum(p.numel() for p in m.parameters())
+    assert MOE_TINY.active_params < MOE_TINY.n_params
+    # Mixtral-class: ~47B total / ~13B active
+    assert 45e9 < MOE_8X7B.n_params < 49e9
+    assert 12e9 < MOE_8X7B.active_params < 14e9
+    from trainingjob_operator_amd.models.config import CONFIGS
+    assert CONFIGS["moe-8x7b"] is MOE_8X7B

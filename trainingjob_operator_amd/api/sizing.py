@@ -98,6 +98,7 @@ def min_gpus_for(estimate: MemoryEstimate,
 KNOWN_MODELS = {
     "llama3-8b": 8_030_000_000,
     "llama3-70b": 70_600_000_000,
+    "moe-8x7b": 46_700_000_000,
     "llama-1b": 1_100_000_000,
 }
 

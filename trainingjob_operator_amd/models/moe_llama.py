@@ -26,6 +26,28 @@ class MoELlamaConfig(LlamaConfig):
     # per-layer aux terms are averaged and added to the LM loss
     aux_loss_coef: float = 0.01
 
+    @property
+    def n_params(self) -> int:
+        """Total parameters (the dense formula's MLP term replaced by
+        router + n_experts SwiGLU experts)."""
+        H, V, L = self.hidden_size, self.vocab_size, self.num_layers
+        kv = self.num_kv_heads * self.head_dim
+        q = self.num_heads * self.head_dim
+        ff = self.expert_ff or self.intermediate_size // 2
+        attn = H * q + 2 * H * kv + q * H
+        moe = H * self.n_experts + self.n_experts * 3 * H * ff
+        per_layer = attn + moe + 2 * H
+        embed = V * H * (1 if self.tie_embeddings else 2)
+        return L * per_layer + embed + H
+
+    @property
+    def active_params(self) -> int:
+        """Parameters touched per token (top-k experts instead of all)."""
+        ff = self.expert_ff or self.intermediate_size // 2
+        inactive = (self.n_experts - self.top_k) * 3 * \
+            self.hidden_size * ff * self.num_layers
+        return self.n_params - inactive
+
 
 MOE_TINY = MoELlamaConfig(
     name="moe-tiny", vocab_size=512, hidden_size=64, intermediate_size=128,
@@ -147,7 +169,18 @@ class MoELlamaModel(nn.Module):
                     ex.down_proj.weight.copy_(src.down_proj.weight)
 
 
+# Mixtral-class MoE: 8 experts x top-2, ~47B total / ~13B active params.
+# The EP target config: one expert shard per MI355X on an 8-GPU node
+# (launcher --ep 8), or EP x TP for headroom.
+MOE_8X7B = MoELlamaConfig(
+    name="moe-8x7b", vocab_size=32000, hidden_size=4096,
+    intermediate_size=28672, num_layers=32, num_heads=32, num_kv_heads=8,
+    head_dim=128, rope_theta=1_000_000.0, n_experts=8, top_k=2,
+    expert_ff=14336)
+
+
 # launcher/EPTrainer look models up by name like the dense families
 from .config import CONFIGS  # noqa: E402
 
 CONFIGS[MOE_TINY.name] = MOE_TINY
+CONFIGS[MOE_8X7B.name] = MOE_8X7B
