@@ -119,7 +119,10 @@ def declared_params(job) -> Optional[int]:
 
 def declared_shards(job) -> int:
     """How many ways the model is sharded across devices (the launcher's
-    pp x tp degree), from the model-shards annotation; 1 = unsharded/DP."""
+    pp x tp degree, or ep for MoE), from the model-shards annotation;
+    1 = unsharded/DP. Admission divides the state uniformly — a slight
+    underestimate for EP (dense params replicate across the plane), which
+    the sizing headroom absorbs."""
     try:
         return max(1, int(job.annotations.get(MODEL_SHARDS_ANNOTATION, 1)))
     except (TypeError, ValueError):
