@@ -72,3 +72,22 @@ def test_generate_gpu():
     out = generate(model, prompt, max_new_tokens=8)
     assert out.shape == (2, 24)
     assert int(out.max()) < 32000 and int(out.min()) >= 0
+
+
+def test_moe_generate_gpu():
+    import torch
+    from trainingjob_operator_amd.models.generate import generate
+    from trainingjob_operator_amd.models.moe_llama import (
+        MOE_TINY, MoELlamaModel,
+    )
+    torch.manual_seed(5)
+    m = MoELlamaModel(MOE_TINY).to(torch.bfloat16).to(DEV)
+    from trainingjob_operator_amd.ops import make_inv_freq
+    m.inv_freq = make_inv_freq(MOE_TINY.head_dim, MOE_TINY.rope_theta,
+                               device=DEV)
+    g = torch.Generator().manual_seed(6)
+    prompt = torch.randint(0, MOE_TINY.vocab_size, (2, 8),
+                           generator=g).to(DEV)
+    out = generate(m, prompt, max_new_tokens=6)
+    assert out.shape == (2, 14)
+    assert int(out.max()) < MOE_TINY.vocab_size
