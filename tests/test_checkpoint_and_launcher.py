@@ -210,3 +210,28 @@ def test_sigusr1_checkpoints_on_demand(tmp_path):
     finally:
         p.terminate()
         p.wait(timeout=60)
+
+
+def test_fault_injection_knob_and_resume(tmp_path):
+    """AITJ_FAULT_STEP kills the worker with the retryable code 137;
+    a relaunch resumes from its checkpoint and completes."""
+    import subprocess
+    import sys
+    env = {k: v for k, v in os.environ.items()
+           if k not in ("RANK", "WORLD_SIZE")}
+    env.update({"MASTER_ADDR": "127.0.0.1", "MASTER_PORT": "29751",
+                "AITJ_FAULT_STEP": "3"})
+    cmd = [sys.executable, "-m",
+           "trainingjob_operator_amd.launcher.main",
+           "--model", "llama-tiny", "--steps", "6", "--seq-len", "32",
+           "--grad-accum", "1", "--micro-batch", "1",
+           "--ckpt-every", "1", "--log-every", "1",
+           "--ckpt-dir", str(tmp_path)]
+    r = subprocess.run(cmd, env=env, capture_output=True, text=True,
+                       timeout=300)
+    assert r.returncode == 137, (r.returncode, r.stderr[-500:])
+    env.pop("AITJ_FAULT_STEP")
+    r = subprocess.run(cmd, env=env, capture_output=True, text=True,
+                       timeout=300)
+    assert r.returncode == 0, r.stderr[-1000:]
+    assert "resumed from step" in r.stderr + r.stdout

@@ -314,10 +314,18 @@ def main(argv=None) -> int:
     tokens_per_step = (cfg.tokens_per_step_per_rank()
                        * (ctx.world_size
                           // (max(args.tp, 1) * max(args.pp, 1))))
+    # chaos knob: die with a retryable code at step N — exercises the
+    # operator's ExitCode restart + checkpoint-resume path in-cluster
+    fault_step = int(os.environ.get("AITJ_FAULT_STEP", "0"))
     t_last = time.time()
     while trainer.step_count < args.steps and not stop_requested["flag"]:
         loss = trainer.train_step()
         step = trainer.step_count
+        if fault_step and step == fault_step:
+            ckpt.wait()
+            log.error("AITJ_FAULT_STEP=%d: injected fault, exiting 137",
+                      fault_step)
+            os._exit(137)
         if loss is None:  # non-last pipeline stages produce no loss
             if step % args.ckpt_every == 0:
                 ckpt.save_async(trainer)
