@@ -127,3 +127,9 @@ def test_tied_embeddings():
         assert float(tr.train_step()) > 0
     finally:
         mc.CONFIGS.pop("llama-tiny-tied", None)
+
+
+def test_dense_trainer_rejects_moe_configs():
+    import trainingjob_operator_amd.models.moe_llama  # register  # noqa
+    with pytest.raises(ValueError, match="MoE config"):
+        Trainer(TrainConfig(model="moe-tiny"))

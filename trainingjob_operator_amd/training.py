@@ -99,6 +99,12 @@ def synthetic_batches(cfg: TrainConfig, device: torch.device,
 class Trainer:
     def __init__(self, cfg: TrainConfig, ctx: Optional[DistContext] = None,
                  device: Optional[torch.device] = None):
+        from .models.moe_llama import MoELlamaConfig
+        if isinstance(cfg.model_config, MoELlamaConfig):
+            raise ValueError(
+                f"{cfg.model!r} is a MoE config: train it with EPTrainer "
+                "(launcher --ep) or PPTrainer (--pp); the dense Trainer "
+                "would silently build a dense model from it")
         self.cfg = cfg
         self.ctx = ctx or DistContext()
         if device is None:
