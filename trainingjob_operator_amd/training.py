@@ -75,6 +75,11 @@ class TrainConfig:
 
 def build_model(cfg: LlamaConfig, device: torch.device,
                 checkpoint_activations: bool = False) -> LlamaModel:
+    from .models.moe_llama import MoELlamaConfig
+    if isinstance(cfg, MoELlamaConfig):
+        raise ValueError(
+            f"{cfg.name!r} is a MoE config; build MoELlamaModel directly "
+            "(EPTrainer / PPTrainer do)")
     with torch.device(device):
         model = LlamaModel(cfg, checkpoint_activations=checkpoint_activations)
     model = model.to(torch.bfloat16)
