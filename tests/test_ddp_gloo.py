@@ -124,3 +124,17 @@ def test_ddp_grads_match_single_process_average():
     expected = acc / world
     # bf16 all-reduce rounding: tolerance scaled to grad magnitude
     assert torch.allclose(dist_grads, expected, atol=5e-3, rtol=5e-2)
+
+
+@pytest.mark.timeout(600)
+def test_ddp_eight_ranks_stay_in_sync():
+    """The driver's 8-GPU shape: bucketed all-reduce keeps all 8 data
+    ranks bit-identical (small buckets force many ordered launches)."""
+    port = _free_port()
+    world = 8
+    with mp.Manager() as mgr:
+        results = mgr.dict()
+        mp.spawn(_worker, args=(world, port, results), nprocs=world,
+                 join=True)
+        assert len(results) == world
+        assert len({results[r] for r in range(world)}) == 1
