@@ -224,3 +224,33 @@ def test_auto_role_beside_fixed_role():
     names = api.pod_names(NS)
     assert "auto-pserver-0" in names
     assert sum(1 for n in names if "trainer" in n) == 2
+
+
+def test_manual_patch_resets_auto_target():
+    """Patching spec.replicas on an Auto role resets the control loop's
+    target: explicit user intent outranks the last automatic choice."""
+    api = FakeKubeApi()
+    tc = TrainingJobController(api, OperatorOptions(
+        elastic_unschedulable_grace=60.0))
+    api.create_job(NS, make_job(replicas=4, mn=2, mx=4))
+    t0 = time.time()
+    tc.sync_once(f"{NS}/auto", now=t0)
+    api.set_all_pods_phase(NS, "Running")
+    api.set_pod_unschedulable(NS, "auto-trainer-3", since=t0 - 120)
+    tc.sync_once(f"{NS}/auto", now=t0 + 1)       # auto-down to 3
+    assert job_of(api).annotations[
+        f"{TARGET_ANNOTATION}-trainer"] == "3"
+    tc.sync_once(f"{NS}/auto", now=t0 + 2)
+    tc.sync_once(f"{NS}/auto", now=t0 + 3)
+    assert len(api.pod_names(NS)) == 3
+
+    # the user patches spec.replicas to 2 (e.g. via aitjctl resize)
+    j = api.get_job(NS, "auto")
+    j["spec"]["replicaSpecs"]["trainer"]["replicas"] = 2
+    api.update_job(NS, "auto", j)
+    tc.sync_once(f"{NS}/auto", now=t0 + 4)
+    assert job_of(api).annotations[
+        f"{TARGET_ANNOTATION}-trainer"] == "2"
+    tc.sync_once(f"{NS}/auto", now=t0 + 5)
+    tc.sync_once(f"{NS}/auto", now=t0 + 6)
+    assert len(api.pod_names(NS)) == 2

@@ -346,6 +346,16 @@ class PodReconciler:
                 self.recorder.event(job, "Normal", "ElasticScale", msg)
             return new
 
+        # a LIVE spec.replicas patch on an Auto role resets the target:
+        # explicit user intent outranks the control loop's last choice
+        bkey = f"{TARGET_ANNOTATION}-base-{rt}"
+        seen_base = job.annotations.get(bkey)
+        if seen_base != str(base):
+            job.annotations[bkey] = str(base)
+            if seen_base is not None:
+                return resize(max(lo, min(hi, base)),
+                              "manual spec.replicas change")
+
         grace = self.options.elastic_unschedulable_grace
         stuck = [p for p in replica_pods
                  if (s := self._unschedulable_since(p)) is not None
