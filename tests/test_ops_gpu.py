@@ -185,3 +185,48 @@ def test_swiglu_packed_matches_reference():
                           rtol=5e-2)
     assert torch.allclose(gu.grad[:, F:].float().cpu(), du_ref, atol=5e-2,
                           rtol=5e-2)
+
+
+def test_rmsnorm_generic_small_h():
+    """Round-1 SIGFPE regression: H outside the templated 2048*k set must
+    run the generic kernel (not crash, not return garbage). Covers the
+    moe-tiny/llama-tiny H=64 shape that killed GPUTEST_r01."""
+    for H in (64, 256, 1024, 2560):
+        T = 33
+        x = _mk((T, H)).requires_grad_()
+        r = _mk((T, H)).requires_grad_()
+        w = _mk((H,), scale=0.5).requires_grad_()
+        y, res = fused_rmsnorm(x, w, r, 1e-5)
+        dy = _mk((T, H))
+        (y.float() * dy.float()).sum().backward()
+
+        xa = x.detach().float().cpu().requires_grad_()
+        ra = r.detach().float().cpu().requires_grad_()
+        wa = w.detach().float().cpu().requires_grad_()
+        resa = xa + ra
+        ya = resa * torch.rsqrt(resa.pow(2).mean(-1, keepdim=True) + 1e-5) * wa
+        (ya * dy.float().cpu()).sum().backward()
+        assert torch.allclose(y.float().cpu(), ya.detach(), atol=BF16_ATOL,
+                              rtol=1e-2), f"H={H}"
+        assert torch.allclose(x.grad.float().cpu(), xa.grad, atol=5e-2,
+                              rtol=5e-2), f"H={H}"
+        assert torch.allclose(w.grad.float().cpu(), wa.grad,
+                              atol=wa.grad.abs().max() * 0.03 + 0.05,
+                              rtol=5e-2), f"H={H}"
+
+
+def test_launchers_reject_unsupported_shapes():
+    """Launchers return nonzero for shapes they cannot run and Python
+    raises — never a silent no-op with uninitialized outputs."""
+    x = _mk((4, 60))  # H % 8 != 0
+    w = _mk((60,))
+    with pytest.raises(RuntimeError, match="unsupported shape"):
+        fused_rmsnorm(x, w, None, 1e-5)
+    lib = native.load()
+    # direct launcher probes (no tensors touched on failure)
+    assert lib.rmsnorm_fwd(None, None, None, None, None, None, None,
+                           4, 60, 1e-5) != 0
+    assert lib.ce_fwd(None, None, None, None, None, 4, 100, -100) != 0
+    assert lib.rope(None, None, None, None, 4, 8, 16, 24, 1.0) != 0
+    assert lib.attn_fwd(None, None, None, None, None, None,
+                        0, 0, 0, 0, 0, 0, 0, 0, 0, 1, 4, 100, 1.0) != 0

@@ -30,13 +30,14 @@ class _NativeFlashAttention(torch.autograd.Function):
         lib = native.load(require=True)
         out = torch.empty(B, H, S, D, dtype=torch.bfloat16, device=q.device)
         lse = torch.empty(B, H, S, dtype=torch.float32, device=q.device)
-        lib.attn_fwd(
+        rc = lib.attn_fwd(
             native.stream_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
             out.data_ptr(), lse.data_ptr(),
             q.stride(0), q.stride(1), q.stride(2),
             k.stride(0), k.stride(1), k.stride(2),
             v.stride(0), v.stride(1), v.stride(2),
             B, H, S, scale)
+        native.check_rc(rc, "attn_fwd", f"B={B} H={H} S={S}")
         ctx.save_for_backward(q, k, v, out, lse)
         ctx.scale = scale
         return out
@@ -74,11 +75,12 @@ def flash_attention_fwd_only(q, k, v, scale=None):
     lib = native.load(require=True)
     out = torch.empty(B, H, S, D, dtype=torch.bfloat16, device=q.device)
     lse = torch.empty(B, H, S, dtype=torch.float32, device=q.device)
-    lib.attn_fwd(
+    rc = lib.attn_fwd(
         native.stream_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
         out.data_ptr(), lse.data_ptr(),
         q.stride(0), q.stride(1), q.stride(2),
         k.stride(0), k.stride(1), k.stride(2),
         v.stride(0), v.stride(1), v.stride(2),
         B, H, S, scale)
+    native.check_rc(rc, "attn_fwd", f"B={B} H={H} S={S}")
     return out, lse

@@ -37,9 +37,10 @@ class _FusedRMSNorm(torch.autograd.Function):
             y = torch.empty_like(x)
             res_out = torch.empty_like(x) if residual is not None else None
             rrms = torch.empty(rows, dtype=torch.float32, device=x.device)
-            lib.rmsnorm_fwd(native.stream_ptr(), _ptr(x), _ptr(residual),
-                            _ptr(weight), _ptr(y), _ptr(res_out), _ptr(rrms),
-                            rows, H, eps)
+            rc = lib.rmsnorm_fwd(native.stream_ptr(), _ptr(x),
+                                 _ptr(residual), _ptr(weight), _ptr(y),
+                                 _ptr(res_out), _ptr(rrms), rows, H, eps)
+            native.check_rc(rc, "rmsnorm_fwd", f"H={H}")
         else:
             y, res_out, rrms = reference.rmsnorm_fwd(x, residual, weight, eps)
         saved_res = res_out if res_out is not None else x
@@ -58,16 +59,19 @@ class _FusedRMSNorm(torch.autograd.Function):
             if dres is not None:
                 dres = dres.contiguous()
             dx = torch.empty_like(res_out)
-            P = lib.rmsnorm_bwd_partials(rows)
+            P = lib.rmsnorm_bwd_partials(rows, H)
             dw_partial = torch.empty(P, H, dtype=torch.float32,
                                      device=dy.device)
             # dres (grad via the residual stream) is fused into dx in-kernel
-            lib.rmsnorm_bwd(native.stream_ptr(), _ptr(dy), _ptr(res_out),
-                            _ptr(weight), _ptr(rrms), _ptr(dres), _ptr(dx),
-                            _ptr(dw_partial), rows, H)
+            rc = lib.rmsnorm_bwd(native.stream_ptr(), _ptr(dy),
+                                 _ptr(res_out), _ptr(weight), _ptr(rrms),
+                                 _ptr(dres), _ptr(dx), _ptr(dw_partial),
+                                 rows, H)
+            native.check_rc(rc, "rmsnorm_bwd", f"H={H}")
             dw32 = torch.zeros(H, dtype=torch.float32, device=dy.device)
-            lib.rmsnorm_dw_reduce(native.stream_ptr(), _ptr(dw_partial), P,
-                                  _ptr(dw32), H)
+            rc = lib.rmsnorm_dw_reduce(native.stream_ptr(),
+                                       _ptr(dw_partial), P, _ptr(dw32), H)
+            native.check_rc(rc, "rmsnorm_dw_reduce", f"H={H} P={P}")
             dw = dw32.to(weight.dtype)
         else:
             dx, dw32 = reference.rmsnorm_bwd(dy, res_out, weight, rrms)
@@ -121,8 +125,9 @@ def _rope_run(x: torch.Tensor, inv_freq: torch.Tensor, seq_len: int,
         assert x.dtype == torch.bfloat16 and xt.is_contiguous()
         lib = _hip()
         out = torch.empty_like(xt)
-        lib.rope(native.stream_ptr(), _ptr(xt), _ptr(out), _ptr(inv_freq),
-                 xt.shape[0], n_heads, seq_len, D, sign)
+        rc = lib.rope(native.stream_ptr(), _ptr(xt), _ptr(out),
+                      _ptr(inv_freq), xt.shape[0], n_heads, seq_len, D, sign)
+        native.check_rc(rc, "rope", f"D={D}")
     else:
         out = reference.rope_rotate(xt, inv_freq, seq_len, sign)
     return out.reshape(orig_shape)
@@ -154,8 +159,9 @@ class _SwiGLU(torch.autograd.Function):
             assert g.dtype == torch.bfloat16 and g.is_contiguous()
             lib = _hip()
             out = torch.empty_like(g)
-            lib.swiglu_fwd(native.stream_ptr(), _ptr(g), _ptr(u), _ptr(out),
-                           g.numel())
+            rc = lib.swiglu_fwd(native.stream_ptr(), _ptr(g), _ptr(u),
+                                _ptr(out), g.numel())
+            native.check_rc(rc, "swiglu_fwd", f"n={g.numel()}")
             return out
         return reference.swiglu_fwd(g, u)
 
@@ -167,8 +173,9 @@ class _SwiGLU(torch.autograd.Function):
             dout = dout.contiguous()
             dg = torch.empty_like(g)
             du = torch.empty_like(u)
-            lib.swiglu_bwd(native.stream_ptr(), _ptr(dout), _ptr(g), _ptr(u),
-                           _ptr(dg), _ptr(du), g.numel())
+            rc = lib.swiglu_bwd(native.stream_ptr(), _ptr(dout), _ptr(g),
+                                _ptr(u), _ptr(dg), _ptr(du), g.numel())
+            native.check_rc(rc, "swiglu_bwd", f"n={g.numel()}")
             return dg, du
         return reference.swiglu_bwd(dout, g, u)
 
@@ -194,8 +201,9 @@ class _SwiGLUPacked(torch.autograd.Function):
             lib = _hip()
             out = torch.empty(*gu.shape[:-1], F, dtype=gu.dtype,
                               device=gu.device)
-            lib.swiglu_packed_fwd(native.stream_ptr(), _ptr(gu), _ptr(out),
-                                  rows, F)
+            rc = lib.swiglu_packed_fwd(native.stream_ptr(), _ptr(gu),
+                                       _ptr(out), rows, F)
+            native.check_rc(rc, "swiglu_packed_fwd", f"F={F}")
             return out
         g, u = gu.split([F, F], dim=-1)
         return reference.swiglu_fwd(g.contiguous(), u.contiguous())
@@ -209,8 +217,9 @@ class _SwiGLUPacked(torch.autograd.Function):
             lib = _hip()
             dout = dout.contiguous()
             dgu = torch.empty_like(gu)
-            lib.swiglu_packed_bwd(native.stream_ptr(), _ptr(dout), _ptr(gu),
-                                  _ptr(dgu), rows, F)
+            rc = lib.swiglu_packed_bwd(native.stream_ptr(), _ptr(dout),
+                                       _ptr(gu), _ptr(dgu), rows, F)
+            native.check_rc(rc, "swiglu_packed_bwd", f"F={F}")
             return dgu
         g, u = gu.split([F, F], dim=-1)
         dg, du = reference.swiglu_bwd(dout, g.contiguous(), u.contiguous())
@@ -238,8 +247,9 @@ class _FusedCE(torch.autograd.Function):
             t32 = targets.to(torch.int32).contiguous()
             lse = torch.empty(T, dtype=torch.float32, device=logits.device)
             loss = torch.empty(T, dtype=torch.float32, device=logits.device)
-            lib.ce_fwd(native.stream_ptr(), _ptr(logits), _ptr(t32), _ptr(lse),
-                       _ptr(loss), T, V, ignore_index)
+            rc = lib.ce_fwd(native.stream_ptr(), _ptr(logits), _ptr(t32),
+                            _ptr(lse), _ptr(loss), T, V, ignore_index)
+            native.check_rc(rc, "ce_fwd", f"V={V}")
         else:
             loss, lse = reference.ce_fwd(logits, targets, ignore_index)
             t32 = targets.to(torch.int32)
@@ -255,8 +265,10 @@ class _FusedCE(torch.autograd.Function):
         if logits.is_cuda:
             lib = _hip()
             dlogits = torch.empty_like(logits)
-            lib.ce_bwd(native.stream_ptr(), _ptr(logits), _ptr(t32), _ptr(lse),
-                       _ptr(gscale), _ptr(dlogits), T, V, ctx.ignore_index)
+            rc = lib.ce_bwd(native.stream_ptr(), _ptr(logits), _ptr(t32),
+                            _ptr(lse), _ptr(gscale), _ptr(dlogits), T, V,
+                            ctx.ignore_index)
+            native.check_rc(rc, "ce_bwd", f"V={V}")
         else:
             dlogits = reference.ce_bwd(logits, t32, lse, gscale,
                                        ctx.ignore_index)

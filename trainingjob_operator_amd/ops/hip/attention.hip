@@ -231,16 +231,18 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   }
 }
 
-extern "C" void attn_fwd(void* stream, const void* q, const void* k,
-                         const void* v, void* out, void* lse,
-                         long q_sb, long q_sh, long q_ss,
-                         long k_sb, long k_sh, long k_ss,
-                         long v_sb, long v_sh, long v_ss,
-                         int batch, int n_heads, int S, float scale) {
+extern "C" int attn_fwd(void* stream, const void* q, const void* k,
+                        const void* v, void* out, void* lse,
+                        long q_sb, long q_sh, long q_ss,
+                        long k_sb, long k_sh, long k_ss,
+                        long v_sb, long v_sh, long v_ss,
+                        int batch, int n_heads, int S, float scale) {
+  if (S <= 0 || S % ATTN_BM != 0 || n_heads <= 0 || batch <= 0) return -1;
   dim3 grid(S / ATTN_BM, n_heads, batch), block(256);
   hipLaunchKernelGGL(attn_fwd_kernel, grid, block, 0,
                      reinterpret_cast<hipStream_t>(stream),
                      (const u16*)q, (const u16*)k, (const u16*)v, (u16*)out,
                      (float*)lse, q_sb, q_sh, q_ss, k_sb, k_sh, k_ss,
                      v_sb, v_sh, v_ss, n_heads, S, scale);
+  return 0;
 }

@@ -216,6 +216,111 @@ __global__ void rmsnorm_bwd_kernel(
   }
 }
 
+// Generic-H variants (any H % 8 == 0): used for shapes the register-cached
+// templated kernels cannot cover (H/8 not a multiple of 256, or H > 16384).
+// Two streaming passes instead of a register row cache; one row per block.
+// These keep small models (tests use H=64) on the native path instead of a
+// silent fallback — the launchers return a status code and Python raises on
+// anything unsupported (round-1 GPUTEST SIGFPE fix).
+
+__global__ void rmsnorm_fwd_generic_kernel(
+    const uint4* __restrict__ x, const uint4* __restrict__ res_in,
+    const uint4* __restrict__ w, uint4* __restrict__ y,
+    uint4* __restrict__ res_out, float* __restrict__ rrms_out,
+    int H8, float eps) {
+  __shared__ float lds[WAVES_PER_BLOCK];
+  const long row = blockIdx.x;
+  const uint4* xr = x + row * H8;
+  const uint4* rr = res_in ? res_in + row * H8 : nullptr;
+  uint4* yr = y + row * H8;
+  uint4* ror = res_out ? res_out + row * H8 : nullptr;
+
+  float sumsq = 0.f;
+  for (int idx = threadIdx.x; idx < H8; idx += BLOCK) {
+    BF8 a; a.v = xr[idx];
+    float vals[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) vals[j] = bf2f(a.h[j]);
+    if (rr) {
+      BF8 b; b.v = rr[idx];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vals[j] += bf2f(b.h[j]);
+      BF8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) o.h[j] = f2bf(vals[j]);
+      ror[idx] = o.v;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vals[j] = bf2f(o.h[j]);
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) sumsq += vals[j] * vals[j];
+  }
+  const float total = block_reduce_sum(sumsq, lds);
+  const float rrms = rsqrtf(total / (float)(H8 * 8) + eps);
+  if (threadIdx.x == 0 && rrms_out) rrms_out[row] = rrms;
+
+  // pass 2: each thread re-reads exactly the indices it wrote (same stride
+  // walk), so reading the rounded residual back needs no extra sync.
+  const uint4* src = rr ? (const uint4*)ror : xr;
+  for (int idx = threadIdx.x; idx < H8; idx += BLOCK) {
+    BF8 a; a.v = src[idx];
+    BF8 wv; wv.v = w[idx];
+    BF8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o.h[j] = f2bf(bf2f(a.h[j]) * rrms * bf2f(wv.h[j]));
+    yr[idx] = o.v;
+  }
+}
+
+__global__ void rmsnorm_bwd_generic_kernel(
+    const uint4* __restrict__ dy, const uint4* __restrict__ res_out,
+    const uint4* __restrict__ w, const float* __restrict__ rrms_in,
+    const uint4* __restrict__ dres, uint4* __restrict__ dx,
+    float* __restrict__ dw_partial, int H8) {
+  __shared__ float lds[WAVES_PER_BLOCK];
+  const long row = blockIdx.x;
+  const int H = H8 * 8;
+  const uint4* dyr = dy + row * H8;
+  const uint4* xr = res_out + row * H8;
+  uint4* dxr = dx + row * H8;
+  const float rr = rrms_in[row];
+
+  float dot = 0.f;
+  for (int idx = threadIdx.x; idx < H8; idx += BLOCK) {
+    BF8 a; a.v = xr[idx];
+    BF8 d; d.v = dyr[idx];
+    BF8 wv; wv.v = w[idx];
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      dot += bf2f(d.h[j]) * bf2f(wv.h[j]) * bf2f(a.h[j]);
+  }
+  const float m = block_reduce_sum(dot, lds) * rr * rr / (float)H;
+  float* dwp = dw_partial + row * (long)H;
+  for (int idx = threadIdx.x; idx < H8; idx += BLOCK) {
+    BF8 a; a.v = xr[idx];
+    BF8 d; d.v = dyr[idx];
+    BF8 wv; wv.v = w[idx];
+    BF8 o;
+    if (dres) {
+      BF8 dr; dr.v = dres[row * H8 + idx];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o.h[j] = f2bf(rr * (bf2f(d.h[j]) * bf2f(wv.h[j])
+                            - bf2f(a.h[j]) * m) + bf2f(dr.h[j]));
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        o.h[j] = f2bf(rr * (bf2f(d.h[j]) * bf2f(wv.h[j])
+                            - bf2f(a.h[j]) * m));
+    }
+    dxr[idx] = o.v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      dwp[idx * 8 + j] = bf2f(d.h[j]) * bf2f(a.h[j]) * rr;
+  }
+}
+
 __global__ void rmsnorm_dw_reduce_kernel(const float* __restrict__ dw_partial,
                                          int n_partials,
                                          float* __restrict__ dw_out, int H,
@@ -629,123 +734,175 @@ void tr_probe(void* stream, void* out, int base_elems, int mode) {
                      (u16*)out, base_elems, mode);
 }
 
-void rmsnorm_fwd(void* stream, const void* x, const void* res_in, const void* w,
-                 void* y, void* res_out, void* rrms, long n_rows, int H,
-                 float eps) {
+// Every launcher returns 0 on success, nonzero on an unsupported shape;
+// the Python side (ops/functional.py) raises on nonzero — no silent
+// fallback, no uninitialized output buffers (round-1 SIGFPE postmortem).
+
+static inline bool rmsnorm_templated(int H) {
   const int H8 = H / 8;
-  const int niter = H8 / BLOCK;
+  return H % 8 == 0 && H8 % BLOCK == 0 && H8 / BLOCK >= 1 && H8 / BLOCK <= 8;
+}
+
+int rmsnorm_fwd(void* stream, const void* x, const void* res_in, const void* w,
+                void* y, void* res_out, void* rrms, long n_rows, int H,
+                float eps) {
+  if (H % 8 != 0 || H <= 0) return -1;
+  if (n_rows <= 0) return 0;
+  const int H8 = H / 8;
   dim3 grid((unsigned)n_rows), block(BLOCK);
+  if (rmsnorm_templated(H)) {
+    const int niter = H8 / BLOCK;
 #define CASE(N) \
   case N: hipLaunchKernelGGL((rmsnorm_fwd_kernel<N>), grid, block, 0, STREAM, \
       (const uint4*)x, (const uint4*)res_in, (const uint4*)w, (uint4*)y, \
-      (uint4*)res_out, (float*)rrms, H8, eps); break;
-  switch (niter) {
-    CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
-    default:
-      fprintf(stderr, "rmsnorm_fwd: unsupported H=%d\n", H);
-  }
+      (uint4*)res_out, (float*)rrms, H8, eps); return 0;
+    switch (niter) {
+      CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
+    }
 #undef CASE
+  }
+  hipLaunchKernelGGL(rmsnorm_fwd_generic_kernel, grid, block, 0, STREAM,
+                     (const uint4*)x, (const uint4*)res_in, (const uint4*)w,
+                     (uint4*)y, (uint4*)res_out, (float*)rrms, H8, eps);
+  return 0;
 }
 
-long rmsnorm_bwd_partials(long n_rows) {
-  return (n_rows + DW_ROWS - 1) / DW_ROWS;
+long rmsnorm_bwd_partials(long n_rows, int H) {
+  // templated path: DW_ROWS rows share one partial; generic: one per row
+  if (rmsnorm_templated(H)) return (n_rows + DW_ROWS - 1) / DW_ROWS;
+  return n_rows;
 }
 
-void rmsnorm_bwd(void* stream, const void* dy, const void* res_out,
-                 const void* w, const void* rrms, const void* dres, void* dx,
-                 void* dw_partial, long n_rows, int H) {
+int rmsnorm_bwd(void* stream, const void* dy, const void* res_out,
+                const void* w, const void* rrms, const void* dres, void* dx,
+                void* dw_partial, long n_rows, int H) {
+  if (H % 8 != 0 || H <= 0) return -1;
+  if (n_rows <= 0) return 0;
   const int H8 = H / 8;
-  const int niter = H8 / BLOCK;
-  dim3 grid((unsigned)rmsnorm_bwd_partials(n_rows)), block(BLOCK);
+  dim3 block(BLOCK);
+  if (rmsnorm_templated(H)) {
+    const int niter = H8 / BLOCK;
+    dim3 grid((unsigned)((n_rows + DW_ROWS - 1) / DW_ROWS));
 #define CASE(N) \
   case N: hipLaunchKernelGGL((rmsnorm_bwd_kernel<N>), grid, block, 0, STREAM, \
       (const uint4*)dy, (const uint4*)res_out, (const uint4*)w, \
       (const float*)rrms, (const uint4*)dres, (uint4*)dx, (float*)dw_partial, \
-      n_rows, H8); break;
-  switch (niter) {
-    CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
-    default:
-      fprintf(stderr, "rmsnorm_bwd: unsupported H=%d\n", H);
-  }
+      n_rows, H8); return 0;
+    switch (niter) {
+      CASE(1) CASE(2) CASE(3) CASE(4) CASE(5) CASE(6) CASE(7) CASE(8)
+    }
 #undef CASE
+  }
+  dim3 grid((unsigned)n_rows);
+  hipLaunchKernelGGL(rmsnorm_bwd_generic_kernel, grid, block, 0, STREAM,
+                     (const uint4*)dy, (const uint4*)res_out, (const uint4*)w,
+                     (const float*)rrms, (const uint4*)dres, (uint4*)dx,
+                     (float*)dw_partial, H8);
+  return 0;
 }
 
-void rmsnorm_dw_reduce(void* stream, const void* dw_partial, long n_partials,
-                       void* dw_f32, int H) {
-  // aim for ~1024 blocks to fill 256 CUs
-  int ny = (int)((1024 + (H / BLOCK) - 1) / (H / BLOCK));
+int rmsnorm_dw_reduce(void* stream, const void* dw_partial, long n_partials,
+                      void* dw_f32, int H) {
+  if (H <= 0 || n_partials <= 0) return -1;
+  // aim for ~1024 blocks total to fill 256 CUs; bx >= 1 always (the round-1
+  // form divided by H/BLOCK, which is 0 for H < 256 -> host SIGFPE)
+  const int bx = (H + BLOCK - 1) / BLOCK;
+  int ny = (1024 + bx - 1) / bx;
   if (ny < 1) ny = 1;
   if (ny > n_partials) ny = (int)n_partials;
   const int rows_per_block = (int)((n_partials + ny - 1) / ny);
-  dim3 grid((H + BLOCK - 1) / BLOCK, ny), block(BLOCK);
+  dim3 grid(bx, ny), block(BLOCK);
   hipLaunchKernelGGL(rmsnorm_dw_reduce_kernel, grid, block, 0, STREAM,
                      (const float*)dw_partial, (int)n_partials, (float*)dw_f32,
                      H, rows_per_block);
+  return 0;
 }
 
-void rope(void* stream, const void* x, void* out, const void* inv_freq,
-          long n_tokens, int n_heads, int S, int D, float sign) {
+int rope(void* stream, const void* x, void* out, const void* inv_freq,
+         long n_tokens, int n_heads, int S, int D, float sign) {
+  if (D <= 0 || D % 16 != 0 || n_heads <= 0 || S <= 0) return -1;
+  if (n_tokens <= 0) return 0;
   const int vec_per_half = (D / 2) / 8;
   const long total_vec = n_tokens * (long)n_heads * vec_per_half;
   dim3 grid(elementwise_grid(total_vec)), block(BLOCK);
   hipLaunchKernelGGL(rope_kernel, grid, block, 0, STREAM, (const uint4*)x,
                      (uint4*)out, (const float*)inv_freq, total_vec,
                      vec_per_half, n_heads, S, D, sign);
+  return 0;
 }
 
-void swiglu_fwd(void* stream, const void* g, const void* u, void* out, long n) {
+int swiglu_fwd(void* stream, const void* g, const void* u, void* out, long n) {
+  if (n % 8 != 0) return -1;
+  if (n == 0) return 0;
   const long n8 = n / 8;
   dim3 grid(elementwise_grid(n8)), block(BLOCK);
   hipLaunchKernelGGL(swiglu_fwd_kernel, grid, block, 0, STREAM,
                      (const uint4*)g, (const uint4*)u, (uint4*)out, n8);
+  return 0;
 }
 
-void swiglu_packed_fwd(void* stream, const void* gu, void* out, long n_rows,
-                       int F) {
+int swiglu_packed_fwd(void* stream, const void* gu, void* out, long n_rows,
+                      int F) {
+  if (F <= 0 || F % 8 != 0) return -1;
+  if (n_rows <= 0) return 0;
   const int F8 = F / 8;
   dim3 grid(elementwise_grid(n_rows * (long)F8)), block(BLOCK);
   hipLaunchKernelGGL(swiglu_packed_fwd_kernel, grid, block, 0, STREAM,
                      (const uint4*)gu, (uint4*)out, n_rows, F8);
+  return 0;
 }
 
-void swiglu_packed_bwd(void* stream, const void* dout, const void* gu,
-                       void* dgu, long n_rows, int F) {
+int swiglu_packed_bwd(void* stream, const void* dout, const void* gu,
+                      void* dgu, long n_rows, int F) {
+  if (F <= 0 || F % 8 != 0) return -1;
+  if (n_rows <= 0) return 0;
   const int F8 = F / 8;
   dim3 grid(elementwise_grid(n_rows * (long)F8)), block(BLOCK);
   hipLaunchKernelGGL(swiglu_packed_bwd_kernel, grid, block, 0, STREAM,
                      (const uint4*)dout, (const uint4*)gu, (uint4*)dgu,
                      n_rows, F8);
+  return 0;
 }
 
-void swiglu_bwd(void* stream, const void* dout, const void* g, const void* u,
-                void* dg, void* du, long n) {
+int swiglu_bwd(void* stream, const void* dout, const void* g, const void* u,
+               void* dg, void* du, long n) {
+  if (n % 8 != 0) return -1;
+  if (n == 0) return 0;
   const long n8 = n / 8;
   dim3 grid(elementwise_grid(n8)), block(BLOCK);
   hipLaunchKernelGGL(swiglu_bwd_kernel, grid, block, 0, STREAM,
                      (const uint4*)dout, (const uint4*)g, (const uint4*)u,
                      (uint4*)dg, (uint4*)du, n8);
+  return 0;
 }
 
-void ce_fwd(void* stream, const void* logits, const void* targets, void* lse,
-            void* loss, long n_rows, int V, int ignore_index) {
+int ce_fwd(void* stream, const void* logits, const void* targets, void* lse,
+           void* loss, long n_rows, int V, int ignore_index) {
+  if (V <= 0 || V % 8 != 0) return -1;
+  if (n_rows <= 0) return 0;
   dim3 grid((unsigned)n_rows), block(BLOCK);
   hipLaunchKernelGGL(ce_fwd_kernel, grid, block, 0, STREAM,
                      (const uint4*)logits, (const int*)targets, (float*)lse,
                      (float*)loss, V, ignore_index);
+  return 0;
 }
 
-void ce_bwd(void* stream, const void* logits, const void* targets,
-            const void* lse, const void* gscale, void* dlogits, long n_rows,
-            int V, int ignore_index) {
+int ce_bwd(void* stream, const void* logits, const void* targets,
+           const void* lse, const void* gscale, void* dlogits, long n_rows,
+           int V, int ignore_index) {
+  if (V <= 0 || V % 8 != 0) return -1;
+  if (n_rows <= 0) return 0;
   dim3 grid((unsigned)n_rows), block(BLOCK);
   hipLaunchKernelGGL(ce_bwd_kernel, grid, block, 0, STREAM,
                      (const uint4*)logits, (const int*)targets,
                      (const float*)lse, (const float*)gscale, (uint4*)dlogits,
                      V, ignore_index);
+  return 0;
 }
 
-void l2normsq(void* stream, const void* grad, long n, void* partials,
-              int n_partials, void* out) {
+int l2normsq(void* stream, const void* grad, long n, void* partials,
+             int n_partials, void* out) {
+  if (n % 8 != 0 || n <= 0 || n_partials <= 0) return -1;
   const long n8 = n / 8;
   int grid = elementwise_grid(n8);
   if (grid > n_partials) grid = n_partials;
@@ -753,12 +910,15 @@ void l2normsq(void* stream, const void* grad, long n, void* partials,
                      STREAM, (const uint4*)grad, n8, (float*)partials);
   hipLaunchKernelGGL(reduce_partials_kernel, dim3(1), dim3(BLOCK), 0, STREAM,
                      (const float*)partials, grid, (float*)out);
+  return 0;
 }
 
-void adamw_step(void* stream, void* p32, void* m, void* v, const void* grad,
-                void* p_bf16, const void* normsq, long n, float lr, float beta1,
-                float beta2, float eps, float weight_decay, float bc1,
-                float bc2, float clip, float pre_scale, const void* bc_dev) {
+int adamw_step(void* stream, void* p32, void* m, void* v, const void* grad,
+               void* p_bf16, const void* normsq, long n, float lr, float beta1,
+               float beta2, float eps, float weight_decay, float bc1,
+               float bc2, float clip, float pre_scale, const void* bc_dev) {
+  if (n % 4 != 0) return -1;
+  if (n == 0) return 0;
   const long n4 = n / 4;
   dim3 grid(elementwise_grid(n4)), block(BLOCK);
   hipLaunchKernelGGL(adamw_kernel, grid, block, 0, STREAM, (float*)p32,
@@ -766,6 +926,7 @@ void adamw_step(void* stream, void* p32, void* m, void* v, const void* grad,
                      (const float*)normsq, n4, lr, beta1, beta2, eps,
                      weight_decay, bc1, bc2, clip, pre_scale,
                      (const float*)bc_dev);
+  return 0;
 }
 
 }  // extern "C"

@@ -56,27 +56,37 @@ def load(require: bool = True) -> Optional[ctypes.CDLL]:
         return None
     lib = ctypes.CDLL(LIB_PATH)
     vp, l, i, f = ctypes.c_void_p, ctypes.c_long, ctypes.c_int, ctypes.c_float
+    # every launcher returns int: 0 = launched, nonzero = unsupported shape
+    # (callers raise via check_rc — no silent no-op launches)
     _sig(lib.hipops_arch_check, [], i)
-    _sig(lib.rmsnorm_fwd, [vp, vp, vp, vp, vp, vp, vp, l, i, f])
-    _sig(lib.rmsnorm_bwd, [vp, vp, vp, vp, vp, vp, vp, vp, l, i])
-    _sig(lib.rmsnorm_bwd_partials, [l], l)
-    _sig(lib.rmsnorm_dw_reduce, [vp, vp, l, vp, i])
-    _sig(lib.rope, [vp, vp, vp, vp, l, i, i, i, f])
-    _sig(lib.swiglu_fwd, [vp, vp, vp, vp, l])
-    _sig(lib.swiglu_bwd, [vp, vp, vp, vp, vp, vp, l])
-    _sig(lib.swiglu_packed_fwd, [vp, vp, vp, l, i])
-    _sig(lib.swiglu_packed_bwd, [vp, vp, vp, vp, l, i])
-    _sig(lib.ce_fwd, [vp, vp, vp, vp, vp, l, i, i])
-    _sig(lib.ce_bwd, [vp, vp, vp, vp, vp, vp, l, i, i])
-    _sig(lib.l2normsq, [vp, vp, l, vp, i, vp])
+    _sig(lib.rmsnorm_fwd, [vp, vp, vp, vp, vp, vp, vp, l, i, f], i)
+    _sig(lib.rmsnorm_bwd, [vp, vp, vp, vp, vp, vp, vp, vp, l, i], i)
+    _sig(lib.rmsnorm_bwd_partials, [l, i], l)
+    _sig(lib.rmsnorm_dw_reduce, [vp, vp, l, vp, i], i)
+    _sig(lib.rope, [vp, vp, vp, vp, l, i, i, i, f], i)
+    _sig(lib.swiglu_fwd, [vp, vp, vp, vp, l], i)
+    _sig(lib.swiglu_bwd, [vp, vp, vp, vp, vp, vp, l], i)
+    _sig(lib.swiglu_packed_fwd, [vp, vp, vp, l, i], i)
+    _sig(lib.swiglu_packed_bwd, [vp, vp, vp, vp, l, i], i)
+    _sig(lib.ce_fwd, [vp, vp, vp, vp, vp, l, i, i], i)
+    _sig(lib.ce_bwd, [vp, vp, vp, vp, vp, vp, l, i, i], i)
+    _sig(lib.l2normsq, [vp, vp, l, vp, i, vp], i)
     _sig(lib.adamw_step, [vp, vp, vp, vp, vp, vp, vp, l,
-                          f, f, f, f, f, f, f, f, f, vp])
+                          f, f, f, f, f, f, f, f, f, vp], i)
     _sig(lib.mfma_probe, [vp, vp, vp, vp])
     _sig(lib.attn_fwd, [vp, vp, vp, vp, vp, vp,
-                        l, l, l, l, l, l, l, l, l, i, i, i, f])
+                        l, l, l, l, l, l, l, l, l, i, i, i, f], i)
     assert lib.hipops_arch_check() == 950
     _lib = lib
     return lib
+
+
+def check_rc(rc: int, op: str, detail: str = "") -> None:
+    """Raise on a nonzero launcher status (unsupported shape)."""
+    if rc != 0:
+        raise RuntimeError(
+            f"hipops {op}: unsupported shape ({detail}) — the kernel "
+            f"launcher refused to launch (rc={rc})")
 
 
 def available() -> bool:

@@ -108,16 +108,18 @@ class FlatAdamW:
             if self.clip > 0.0:
                 # the clip norm is GLOBAL: always over the full gradient,
                 # even when the update covers only this rank's shard
-                lib.l2normsq(sp, st.flat_grad.data_ptr(), st.total,
-                             self._partials.data_ptr(), _N_PARTIALS,
-                             self._normsq.data_ptr())
+                rc = lib.l2normsq(sp, st.flat_grad.data_ptr(), st.total,
+                                  self._partials.data_ptr(), _N_PARTIALS,
+                                  self._normsq.data_ptr())
+                native.check_rc(rc, "l2normsq", f"n={st.total}")
                 normsq_ptr = self._normsq.data_ptr()
-            lib.adamw_step(sp, self.p32.data_ptr(), self.m.data_ptr(),
-                           self.v.data_ptr(), grad.data_ptr(),
-                           param.data_ptr(), normsq_ptr, n, self.lr,
-                           self.beta1, self.beta2, self.eps,
-                           self.weight_decay, bc1, bc2, self.clip,
-                           grad_pre_scale, self._bc.data_ptr())
+            rc = lib.adamw_step(sp, self.p32.data_ptr(), self.m.data_ptr(),
+                                self.v.data_ptr(), grad.data_ptr(),
+                                param.data_ptr(), normsq_ptr, n, self.lr,
+                                self.beta1, self.beta2, self.eps,
+                                self.weight_decay, bc1, bc2, self.clip,
+                                grad_pre_scale, self._bc.data_ptr())
+            native.check_rc(rc, "adamw_step", f"n={n}")
         else:
             normsq = None
             if self.clip > 0.0:
