@@ -1,6 +1,7 @@
 """Bisect the PP llama-tiny SIGABRT: dense tiny trainer vs PP trainer."""
-import faulthandler, sys
+import faulthandler, os, sys
 faulthandler.enable()
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 
 which = sys.argv[1]
