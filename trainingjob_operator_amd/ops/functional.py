@@ -123,6 +123,11 @@ def _rope_run(x: torch.Tensor, inv_freq: torch.Tensor, seq_len: int,
     xt = x.reshape(-1, n_heads, D)
     if x.is_cuda:
         assert x.dtype == torch.bfloat16 and xt.is_contiguous()
+        assert inv_freq.device == x.device and \
+            inv_freq.dtype == torch.float32, (
+            f"inv_freq must be fp32 on {x.device}, got "
+            f"{inv_freq.dtype}@{inv_freq.device} (a host pointer would "
+            f"memory-fault the GPU)")
         lib = _hip()
         out = torch.empty_like(xt)
         rc = lib.rope(native.stream_ptr(), _ptr(xt), _ptr(out),
@@ -141,7 +146,10 @@ def apply_rope(x: torch.Tensor, inv_freq: torch.Tensor,
 
 
 def make_inv_freq(head_dim: int, theta: float = 500000.0,
-                  device="cpu") -> torch.Tensor:
+                  device=None) -> torch.Tensor:
+    # device=None respects an ambient `with torch.device(...)` context —
+    # an explicit "cpu" default left standalone PP stages with a host
+    # inv_freq under a CUDA model (GPU memory fault in the rope kernel).
     return 1.0 / (theta ** (torch.arange(0, head_dim, 2,
                                          dtype=torch.float32,
                                          device=device) / head_dim))
