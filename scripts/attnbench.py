@@ -9,7 +9,8 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 import torch.nn.functional as F
 
-from trainingjob_operator_amd.ops.attention import flash_attention_fwd_only
+from trainingjob_operator_amd.ops.attention import (
+    flash_attention_fwd_only, flash_attention_fwd_v5)
 
 
 def main():
@@ -24,6 +25,8 @@ def main():
     def run():
         if which == "native":
             flash_attention_fwd_only(q, k, v)
+        elif which == "v5":
+            flash_attention_fwd_v5(q, k, v)
         else:
             F.scaled_dot_product_attention(q, k, v, is_causal=True)
 

@@ -35,7 +35,8 @@ def _sdpa(q, k, v, enable_gqa=False, is_causal=True):
     if backend == "native" and q.is_cuda:
         from ..ops.attention import flash_attention
         groups = q.shape[1] // k.shape[1]
-        if groups > 1:  # the native kernel wants equal head counts
+        if groups > 1 and q.shape[2] % 256 != 0:
+            # only the v6 kernel (S%256==0) maps GQA heads natively
             k = k.repeat_interleave(groups, dim=1)
             v = v.repeat_interleave(groups, dim=1)
         return flash_attention(q.contiguous() if q.stride(-1) != 1 else q,

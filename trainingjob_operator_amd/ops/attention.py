@@ -1,10 +1,18 @@
 """Native MFMA flash-attention forward (hip/attention.hip) with the aten
 flash backward for training.
 
+Two hand-written CDNA4 forwards:
+  * v6 (S % 256 == 0): 8-wave swapped-operand kernel - 32x32x16 MFMA,
+    in-register softmax (q row lane-local on both products), defer-rescale,
+    double-buffered K/V staging, native GQA. The production path.
+  * v5 (S % 64 == 0 fallback): 4-wave 16x16x32 kernel with the LDS P
+    round-trip; requires equal head counts (GQA expanded by the caller).
+
 The forward emits exactly what aten::_scaled_dot_product_flash_attention_
-backward consumes (O + logsumexp at matching scale semantics), so the
-hand-written CDNA4 forward drops into autograd with the library backward.
-Enable via AITJ_SDPA_BACKEND=native (models/llama.py _sdpa).
+backward consumes (O + logsumexp at natural-log scale), so the hand-written
+forward drops into autograd with the library backward until the native
+backward lands. GQA head counts pass straight through (the aten backward
+reduces dk/dv to the kv heads).
 """
 from __future__ import annotations
 
@@ -18,26 +26,34 @@ from . import native
 
 def _supported(q: torch.Tensor, k: torch.Tensor) -> bool:
     B, H, S, D = q.shape
-    return (q.dtype == torch.bfloat16 and D == 128 and S % 64 == 0
+    HKV = k.shape[1]
+    return (q.dtype == torch.bfloat16 and D == 128
+            and (S % 256 == 0 or (S % 64 == 0 and HKV == H))
             and q.stride(-1) == 1 and k.stride(-1) == 1
-            and k.shape[1] == H)
+            and H % HKV == 0)
+
+
+def _run_fwd(q, k, v, scale):
+    B, H, S, D = q.shape
+    HKV = k.shape[1]
+    lib = native.load(require=True)
+    out = torch.empty(B, H, S, D, dtype=torch.bfloat16, device=q.device)
+    lse = torch.empty(B, H, S, dtype=torch.float32, device=q.device)
+    rc = lib.attn_fwd(
+        native.stream_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+        out.data_ptr(), lse.data_ptr(),
+        q.stride(0), q.stride(1), q.stride(2),
+        k.stride(0), k.stride(1), k.stride(2),
+        v.stride(0), v.stride(1), v.stride(2),
+        B, H, HKV, S, scale)
+    native.check_rc(rc, "attn_fwd", f"B={B} H={H} HKV={HKV} S={S}")
+    return out, lse
 
 
 class _NativeFlashAttention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, scale):
-        B, H, S, D = q.shape
-        lib = native.load(require=True)
-        out = torch.empty(B, H, S, D, dtype=torch.bfloat16, device=q.device)
-        lse = torch.empty(B, H, S, dtype=torch.float32, device=q.device)
-        rc = lib.attn_fwd(
-            native.stream_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
-            out.data_ptr(), lse.data_ptr(),
-            q.stride(0), q.stride(1), q.stride(2),
-            k.stride(0), k.stride(1), k.stride(2),
-            v.stride(0), v.stride(1), v.stride(2),
-            B, H, S, scale)
-        native.check_rc(rc, "attn_fwd", f"B={B} H={H} S={S}")
+        out, lse = _run_fwd(q, k, v, scale)
         ctx.save_for_backward(q, k, v, out, lse)
         ctx.scale = scale
         return out
@@ -58,12 +74,14 @@ class _NativeFlashAttention(torch.autograd.Function):
 
 def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
                     scale: Optional[float] = None) -> torch.Tensor:
-    """Causal flash attention over [B, H, S, D=128] bf16 (S % 64 == 0)."""
+    """Causal flash attention over [B, H, S, D=128] bf16; k/v may carry
+    fewer (GQA) heads when S % 256 == 0 (the v6 kernel maps them)."""
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
     assert _supported(q, k), (
-        f"native flash attention needs bf16, D=128, S%64==0, d-contiguous "
-        f"strides; got {q.shape} {q.dtype}")
+        f"native flash attention needs bf16, D=128, S%256==0 (or S%64==0 "
+        f"with equal head counts), d-contiguous strides; got {q.shape} "
+        f"{q.dtype} kv={k.shape}")
     return _NativeFlashAttention.apply(q, k, v, scale)
 
 
@@ -71,16 +89,23 @@ def flash_attention_fwd_only(q, k, v, scale=None):
     """Forward-only entry returning (out, lse) for tests/benchmarks."""
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
+    return _run_fwd(q, k, v, scale)
+
+
+def flash_attention_fwd_v5(q, k, v, scale=None):
+    """The 4-wave v5 forward, kept callable for A/B benchmarking."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
     B, H, S, D = q.shape
     lib = native.load(require=True)
     out = torch.empty(B, H, S, D, dtype=torch.bfloat16, device=q.device)
     lse = torch.empty(B, H, S, dtype=torch.float32, device=q.device)
-    rc = lib.attn_fwd(
+    rc = lib.attn_fwd_v5(
         native.stream_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
         out.data_ptr(), lse.data_ptr(),
         q.stride(0), q.stride(1), q.stride(2),
         k.stride(0), k.stride(1), k.stride(2),
         v.stride(0), v.stride(1), v.stride(2),
         B, H, S, scale)
-    native.check_rc(rc, "attn_fwd", f"B={B} H={H} S={S}")
+    native.check_rc(rc, "attn_fwd_v5", f"B={B} H={H} S={S}")
     return out, lse

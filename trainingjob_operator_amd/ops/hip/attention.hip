@@ -231,12 +231,310 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   }
 }
 
+// ===========================================================================
+// v6: 8-wave swapped-operand forward (guide §B "8-warp 32×32 ladder").
+//
+// Structure: one 512-thread workgroup owns 256 q rows (32 per wave); K/V
+// tiles of 64 keys double-buffered in LDS; Q rows live in registers.
+// BOTH MFMA products are operand-swapped so the q index stays lane-local
+// end to end:
+//   S^T = K @ Q^T   via mfma_32x32x16(A=K-frag, B=Q-frag)  -> C[kv][q=lane&31]
+//   O^T = V^T @ P   via mfma_32x32x16(A=Vt-frag, B=P-frag) -> C[d][q=lane&31]
+// so the online-softmax state (m, l) is one scalar pair per lane (its q row,
+// 16 kv per half-wave), the row reduce is an in-register tree plus ONE
+// __shfl_xor(32) half-merge, and P never round-trips through LDS
+// (v5 paid a pack + ds_write + barrier + ds_read per tile for that).
+// Softmax runs in exp2 space (scale2 = scale*log2e folded into the S scale).
+// Techniques: T12 (cvt_pk_bf16_f32 + permlane32_swap P repack), T13
+// (defer-rescale THR=8), T14 (issue next tile's global loads before QK^T,
+// LDS write before PV), causal tile skip + diagonal-only masking, reversed
+// qb launch order (longest blocks first), native GQA (kv head = h / group).
+// ===========================================================================
+
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+#define V6_BM 256
+#define V6_QBLK 32
+#define V6_BN 64
+// per-buffer LDS halves (u16 counts): K [64][128] swizzled rows, then the
+// transposed V image (pitch 144 B + 256 B XOR tail)
+#define V6_K_U16 (V6_BN * ATTN_D)                        // 8192
+#define V6_V_U16 (ATTN_D * (VT_PITCH_B / 2) + 128)       // 9344
+#define V6_BUF_U16 (V6_K_U16 + V6_V_U16)
+
+__device__ __forceinline__ unsigned cvt_pk_bf16(float lo, float hi) {
+  unsigned r;
+  asm("v_cvt_pk_bf16_f32 %0, %1, %2" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+
+__global__ __launch_bounds__(512) void attn_fwd_v6_kernel(
+    const u16* __restrict__ q, const u16* __restrict__ k,
+    const u16* __restrict__ v, u16* __restrict__ out,
+    float* __restrict__ lse,
+    long q_sb, long q_sh, long q_ss,
+    long k_sb, long k_sh, long k_ss,
+    long v_sb, long v_sh, long v_ss,
+    int n_heads, int gqa_group, int S, float scale2) {
+  const int qb = gridDim.x - 1 - blockIdx.x;   // longest-running blocks first
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int hkv = h / gqa_group;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int lane = tid & 63;
+  const int low = lane & 31;
+  const int hi = lane >> 5;
+
+  __shared__ __attribute__((aligned(16))) u16 lds[2][V6_BUF_U16];
+
+  const int q0w = qb * V6_BM + wid * V6_QBLK;
+  const int qrow = q0w + low;                  // this lane's q row
+
+  // Q row in registers: 8 k-chunks, lane holds Q[qrow][kc*16 + hi*8 + j]
+  union F8 { bf16x8 v; uint4 u; u16 h[8]; };
+  const u16* qptr = q + (long)b * q_sb + (long)h * q_sh + (long)qrow * q_ss;
+  F8 qf[8];
+#pragma unroll
+  for (int kc = 0; kc < 8; ++kc)
+    qf[kc].u = *reinterpret_cast<const uint4*>(qptr + kc * 16 + hi * 8);
+
+  // O^T accumulators: 4 d-subtiles; lane holds O[d=crow(r,hi)+32*ds][qrow]
+  f32x16 oacc[4];
+#pragma unroll
+  for (int ds = 0; ds < 4; ++ds)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) oacc[ds][r] = 0.f;
+  float m_run = -INFINITY, l_run = 0.f;
+
+  const u16* kbase = k + (long)b * k_sb + (long)hkv * k_sh;
+  const u16* vbase = v + (long)b * v_sb + (long)hkv * v_sh;
+  const int n_tiles = (qb * V6_BM + V6_BM) / V6_BN;   // causal upper bound
+
+  char* lds0 = reinterpret_cast<char*>(&lds[0][0]);
+  char* lds1 = reinterpret_cast<char*>(&lds[1][0]);
+  const int VOFF = V6_K_U16 * 2;               // V image byte offset in a buf
+
+  // staging: 512 threads, 2 uint4 of K + 2 of V each (1 KiB rows of 16 B)
+  uint4 kst[2], vst[2];
+#define V6_LOAD_TILE(T)                                                     \
+  _Pragma("unroll")                                                         \
+  for (int i = 0; i < 2; ++i) {                                             \
+    const int idx = i * 512 + tid;                                          \
+    const int r_ = idx >> 4;                                                \
+    const int c8_ = (idx & 15) * 8;                                         \
+    kst[i] = *reinterpret_cast<const uint4*>(                               \
+        kbase + (long)((T) * V6_BN + r_) * k_ss + c8_);                     \
+    vst[i] = *reinterpret_cast<const uint4*>(                               \
+        vbase + (long)((T) * V6_BN + r_) * v_ss + c8_);                     \
+  }
+#define V6_WRITE_TILE(BUF)                                                  \
+  _Pragma("unroll")                                                         \
+  for (int i = 0; i < 2; ++i) {                                             \
+    const int idx = i * 512 + tid;                                          \
+    const int r_ = idx >> 4;                                                \
+    const int c8_ = (idx & 15) * 8;                                         \
+    *reinterpret_cast<uint4*>(&(BUF)[k_byte(r_, c8_)]) = kst[i];            \
+    union { uint4 u; u16 h[8]; } vv_;                                       \
+    vv_.u = vst[i];                                                         \
+    _Pragma("unroll")                                                       \
+    for (int j = 0; j < 8; ++j)                                             \
+      *reinterpret_cast<u16*>(&(BUF)[VOFF + vt_byte(c8_ + j, r_)]) = vv_.h[j]; \
+  }
+
+  V6_LOAD_TILE(0)
+  V6_WRITE_TILE(lds0)
+  __syncthreads();
+
+  int cur = 0;
+  for (int t = 0; t < n_tiles; ++t) {
+    const int kv0 = t * V6_BN;
+    const bool have_next = (t + 1) < n_tiles;
+    if (have_next) V6_LOAD_TILE(t + 1)        // issue early (T14)
+
+    char* kb = cur ? lds1 : lds0;
+    char* vbuf = kb + VOFF;
+    const bool active = kv0 <= q0w + V6_QBLK - 1;
+    const bool need_mask = kv0 + V6_BN > q0w;
+
+    f32x16 s0, s1;
+    union PF { bf16x8 v; unsigned u[4]; } pa[4];
+    if (active) {
+      // ---- S^T = K @ Q^T over 2 kv-subtiles of 32 ----
+#pragma unroll
+      for (int r = 0; r < 16; ++r) { s0[r] = 0.f; s1[r] = 0.f; }
+#pragma unroll
+      for (int kc = 0; kc < 8; ++kc) {
+        bf16x8 a0 = *reinterpret_cast<const bf16x8*>(
+            &kb[k_byte(low, kc * 16 + hi * 8)]);
+        s0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, qf[kc].v, s0, 0, 0, 0);
+        bf16x8 a1 = *reinterpret_cast<const bf16x8*>(
+            &kb[k_byte(32 + low, kc * 16 + hi * 8)]);
+        s1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, qf[kc].v, s1, 0, 0, 0);
+      }
+      // ---- scale into exp2 space + causal mask + tile max ----
+      float pm = -INFINITY;
+      if (need_mask) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kvr = kv0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float x0 = s0[r] * scale2;
+          if (kvr > qrow) x0 = -1e30f;
+          s0[r] = x0;
+          float x1 = s1[r] * scale2;
+          if (kvr + 32 > qrow) x1 = -1e30f;
+          s1[r] = x1;
+          pm = fmaxf(pm, fmaxf(x0, x1));
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          s0[r] *= scale2;
+          s1[r] *= scale2;
+          pm = fmaxf(pm, fmaxf(s0[r], s1[r]));
+        }
+      }
+      pm = fmaxf(pm, __shfl_xor(pm, 32, 64));   // merge q-row halves
+
+      // ---- defer-rescale (T13, THR=8 in exp2 space) ----
+      if (!__all(pm - m_run <= 8.0f)) {
+        const float mn = fmaxf(m_run, pm);
+        const float alpha = (m_run == -INFINITY) ? 0.f : exp2f(m_run - mn);
+        m_run = mn;
+        l_run *= alpha;
+#pragma unroll
+        for (int ds = 0; ds < 4; ++ds)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) oacc[ds][r] *= alpha;
+      }
+      // ---- P = exp2(s - m), row sum ----
+      float psum = 0.f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const float p0 = exp2f(s0[r] - m_run);
+        const float p1 = exp2f(s1[r] - m_run);
+        s0[r] = p0;
+        s1[r] = p1;
+        psum += p0 + p1;
+      }
+      psum += __shfl_xor(psum, 32, 64);
+      l_run += psum;
+
+      // ---- T12 repack: P f32 regs -> bf16 MFMA fragments, in-register ----
+      // chunk c of 16 kv: lane holds P[kv = c*16 + hi*8 + j][qrow]
+#pragma unroll
+      for (int sub = 0; sub < 2; ++sub) {
+#pragma unroll
+        for (int cc = 0; cc < 2; ++cc) {
+          const int pb = cc * 8;
+          unsigned a0, b0, a1, b1;
+          if (sub == 0) {
+            a0 = cvt_pk_bf16(s0[pb + 0], s0[pb + 1]);
+            b0 = cvt_pk_bf16(s0[pb + 4], s0[pb + 5]);
+            a1 = cvt_pk_bf16(s0[pb + 2], s0[pb + 3]);
+            b1 = cvt_pk_bf16(s0[pb + 6], s0[pb + 7]);
+          } else {
+            a0 = cvt_pk_bf16(s1[pb + 0], s1[pb + 1]);
+            b0 = cvt_pk_bf16(s1[pb + 4], s1[pb + 5]);
+            a1 = cvt_pk_bf16(s1[pb + 2], s1[pb + 3]);
+            b1 = cvt_pk_bf16(s1[pb + 6], s1[pb + 7]);
+          }
+          auto r02 = __builtin_amdgcn_permlane32_swap(a0, b0, false, false);
+          auto r13 = __builtin_amdgcn_permlane32_swap(a1, b1, false, false);
+          PF f;
+          f.u[0] = r02[0];
+          f.u[1] = r13[0];
+          f.u[2] = r02[1];
+          f.u[3] = r13[1];
+          pa[sub * 2 + cc] = f;
+        }
+      }
+    }
+
+    // ---- stage tile t+1 into the other buffer (write lands under PV) ----
+    if (have_next) {
+      char* nb = cur ? lds0 : lds1;
+      V6_WRITE_TILE(nb)
+    }
+
+    if (active) {
+      // ---- O^T += V^T @ P over 4 kv-chunks x 4 d-subtiles ----
+#pragma unroll
+      for (int c = 0; c < 4; ++c) {
+#pragma unroll
+        for (int ds = 0; ds < 4; ++ds) {
+          bf16x8 vf = *reinterpret_cast<const bf16x8*>(
+              &vbuf[vt_byte(ds * 32 + low, c * 16 + hi * 8)]);
+          oacc[ds] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              vf, pa[c].v, oacc[ds], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // ---- epilogue: O /= l, store bf16; LSE in natural-log space ----
+  const float inv_l = 1.0f / l_run;
+  u16* orow = out + (((long)b * n_heads + h) * S + qrow) * ATTN_D;
+#pragma unroll
+  for (int ds = 0; ds < 4; ++ds) {
+#pragma unroll
+    for (int g = 0; g < 4; ++g) {
+      const int d0 = 8 * g + 4 * hi + 32 * ds;
+      const unsigned w0 = cvt_pk_bf16(oacc[ds][4 * g + 0] * inv_l,
+                                      oacc[ds][4 * g + 1] * inv_l);
+      const unsigned w1 = cvt_pk_bf16(oacc[ds][4 * g + 2] * inv_l,
+                                      oacc[ds][4 * g + 3] * inv_l);
+      uint2 wv;
+      wv.x = w0;
+      wv.y = w1;
+      *reinterpret_cast<uint2*>(orow + d0) = wv;
+    }
+  }
+  if (hi == 0)
+    lse[((long)b * n_heads + h) * S + qrow] =
+        m_run * 0.6931471805599453f + __logf(l_run);
+}
+
 extern "C" int attn_fwd(void* stream, const void* q, const void* k,
                         const void* v, void* out, void* lse,
                         long q_sb, long q_sh, long q_ss,
                         long k_sb, long k_sh, long k_ss,
                         long v_sb, long v_sh, long v_ss,
-                        int batch, int n_heads, int S, float scale) {
+                        int batch, int n_heads, int n_kv_heads, int S,
+                        float scale) {
+  if (S <= 0 || n_heads <= 0 || batch <= 0 || n_kv_heads <= 0 ||
+      n_heads % n_kv_heads != 0)
+    return -1;
+  if (S % V6_BM == 0) {
+    dim3 grid(S / V6_BM, n_heads, batch), block(512);
+    const float scale2 = scale * 1.4426950408889634f;   // fold log2(e)
+    hipLaunchKernelGGL(attn_fwd_v6_kernel, grid, block, 0,
+                       reinterpret_cast<hipStream_t>(stream),
+                       (const u16*)q, (const u16*)k, (const u16*)v, (u16*)out,
+                       (float*)lse, q_sb, q_sh, q_ss, k_sb, k_sh, k_ss,
+                       v_sb, v_sh, v_ss, n_heads, n_heads / n_kv_heads, S,
+                       scale2);
+    return 0;
+  }
+  if (S % ATTN_BM != 0 || n_kv_heads != n_heads) return -1;  // v5: no GQA
+  dim3 grid(S / ATTN_BM, n_heads, batch), block(256);
+  hipLaunchKernelGGL(attn_fwd_kernel, grid, block, 0,
+                     reinterpret_cast<hipStream_t>(stream),
+                     (const u16*)q, (const u16*)k, (const u16*)v, (u16*)out,
+                     (float*)lse, q_sb, q_sh, q_ss, k_sb, k_sh, k_ss,
+                     v_sb, v_sh, v_ss, n_heads, S, scale);
+  return 0;
+}
+
+// v5 kept callable for A/B benchmarking (scripts/attnbench.py v5)
+extern "C" int attn_fwd_v5(void* stream, const void* q, const void* k,
+                           const void* v, void* out, void* lse,
+                           long q_sb, long q_sh, long q_ss,
+                           long k_sb, long k_sh, long k_ss,
+                           long v_sb, long v_sh, long v_ss,
+                           int batch, int n_heads, int S, float scale) {
   if (S <= 0 || S % ATTN_BM != 0 || n_heads <= 0 || batch <= 0) return -1;
   dim3 grid(S / ATTN_BM, n_heads, batch), block(256);
   hipLaunchKernelGGL(attn_fwd_kernel, grid, block, 0,

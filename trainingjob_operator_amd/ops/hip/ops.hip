@@ -677,6 +677,38 @@ __global__ void mfma_probe_kernel(const u16* __restrict__ A,
 }
 
 // ===========================================================================
+// 32x32x16 MFMA layout probe (test-only): one v_mfma_f32_32x32x16_bf16 under
+// the layout assumptions the v6 attention kernel builds on:
+//   A[32x16]: lane l holds A[l & 31][(l >> 5) * 8 + j], j = 0..7
+//   B[16x32]: lane l holds B[(l >> 5) * 8 + j][l & 31]
+//   C[32x32]: lane l reg r holds C[(r & 3) + 8 * (r >> 2) + 4 * (l >> 5)][l & 31]
+// Verified on silicon with asymmetric inputs (guide G9).
+// ===========================================================================
+
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+__global__ void mfma_probe32_kernel(const u16* __restrict__ A,
+                                    const u16* __restrict__ B,
+                                    float* __restrict__ C) {
+  const int lane = threadIdx.x & 63;
+  const int low = lane & 31;
+  const int hi = lane >> 5;
+  union { bf16x8 v; u16 h[8]; } a, b;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    a.h[j] = A[low * 16 + hi * 8 + j];        // A[m=low][k]
+    b.h[j] = B[(hi * 8 + j) * 32 + low];      // B[k][n=low]
+  }
+  f32x16 c;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) c[r] = 0.f;
+  c = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a.v, b.v, c, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 16; ++r)
+    C[((r & 3) + 8 * (r >> 2) + 4 * hi) * 32 + low] = c[r];
+}
+
+// ===========================================================================
 // ds_read_b64_tr_b16 semantics probe (test-only): LDS holds raw u16 == its
 // element index; each lane issues one transpose-read at a mode-dependent
 // address and dumps its 4 raw results, revealing the exact lane/elem ->
@@ -726,6 +758,11 @@ int hipops_arch_check() {
 
 void mfma_probe(void* stream, const void* A, const void* B, void* C) {
   hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, STREAM,
+                     (const u16*)A, (const u16*)B, (float*)C);
+}
+
+void mfma_probe32(void* stream, const void* A, const void* B, void* C) {
+  hipLaunchKernelGGL(mfma_probe32_kernel, dim3(1), dim3(64), 0, STREAM,
                      (const u16*)A, (const u16*)B, (float*)C);
 }
 

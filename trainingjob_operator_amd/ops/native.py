@@ -75,7 +75,10 @@ def load(require: bool = True) -> Optional[ctypes.CDLL]:
                           f, f, f, f, f, f, f, f, f, vp], i)
     _sig(lib.mfma_probe, [vp, vp, vp, vp])
     _sig(lib.attn_fwd, [vp, vp, vp, vp, vp, vp,
-                        l, l, l, l, l, l, l, l, l, i, i, i, f], i)
+                        l, l, l, l, l, l, l, l, l, i, i, i, i, f], i)
+    _sig(lib.attn_fwd_v5, [vp, vp, vp, vp, vp, vp,
+                           l, l, l, l, l, l, l, l, l, i, i, i, f], i)
+    _sig(lib.mfma_probe32, [vp, vp, vp, vp])
     assert lib.hipops_arch_check() == 950
     _lib = lib
     return lib
