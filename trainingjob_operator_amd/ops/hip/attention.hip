@@ -58,8 +58,18 @@ __device__ __forceinline__ int k_byte(int row, int col_elem) {
 // the XOR (<=240 B) may cross row ends, so the buffer carries 256 B of
 // tail padding.
 #define VT_PITCH_B 144
+// slot-XOR key k(m), m = d>>3: (m&7) | (((m>>1)&1)<<3).
+// Constraints (brute-force verified over the full image):
+//  * byte-map injective: rows 8m-1/8m share a 256-B block at odd m, so
+//    adjacent key pairs must flip slot-bit-3 together (the 9m key violated
+//    this and corrupted the image);
+//  * v6 b128 reads (32-d spans, lane groups {0-3,12-15,20-27}): <=2-way
+//    (the v5 key (d>>3)&15 was 3-way there);
+//  * v5 reads <=2-way, write scatter <=4-way (k mod 8 = m mod 8 spread).
 __device__ __forceinline__ int vt_byte(int d, int kv) {
-  return (d * VT_PITCH_B + kv * 2) ^ (((d >> 3) & 15) << 4);
+  const int m = d >> 3;
+  return (d * VT_PITCH_B + kv * 2) ^
+         ((((m & 7) | (((m >> 1) & 1) << 3))) << 4);
 }
 
 // P strip swizzle (row length 128 B = 8 slots): XOR with row & 7.
@@ -268,6 +278,15 @@ __device__ __forceinline__ unsigned cvt_pk_bf16(float lo, float hi) {
   return r;
 }
 
+// raw v_exp_f32 (2^x): libm exp2f lowers to a ~5-instruction guarded
+// sequence (ldexp + cndmask pairs); softmax arguments are <= 0 and
+// underflow-to-0 is exactly what we want.
+__device__ __forceinline__ float exp2_raw(float x) {
+  float r;
+  asm("v_exp_f32 %0, %1" : "=v"(r) : "v"(x));
+  return r;
+}
+
 __global__ __launch_bounds__(512) void attn_fwd_v6_kernel(
     const u16* __restrict__ q, const u16* __restrict__ k,
     const u16* __restrict__ v, u16* __restrict__ out,
@@ -315,42 +334,69 @@ __global__ __launch_bounds__(512) void attn_fwd_v6_kernel(
   char* lds1 = reinterpret_cast<char*>(&lds[1][0]);
   const int VOFF = V6_K_U16 * 2;               // V image byte offset in a buf
 
-  // staging: 512 threads, 2 uint4 of K + 2 of V each (1 KiB rows of 16 B)
-  uint4 kst[2], vst[2];
-#define V6_LOAD_TILE(T)                                                     \
-  _Pragma("unroll")                                                         \
-  for (int i = 0; i < 2; ++i) {                                             \
-    const int idx = i * 512 + tid;                                          \
-    const int r_ = idx >> 4;                                                \
-    const int c8_ = (idx & 15) * 8;                                         \
-    kst[i] = *reinterpret_cast<const uint4*>(                               \
-        kbase + (long)((T) * V6_BN + r_) * k_ss + c8_);                     \
-    vst[i] = *reinterpret_cast<const uint4*>(                               \
-        vbase + (long)((T) * V6_BN + r_) * v_ss + c8_);                     \
+  // K staging goes by global->LDS DMA (glds): zero in-flight registers and
+  // zero ds_write instructions; the T2 swizzle is inverted onto the SOURCE
+  // address (rule 21: glds writes lane-linear), which keeps the LDS image
+  // byte-identical to the ds_write version and costs nothing in coalescing
+  // (the XOR permutes 16-B slots within one 256-B row = the same two 128-B
+  // requests per 16 lanes). V cannot go by glds (transposed image) and
+  // keeps register staging: 2 uint4 per thread, written after QK^T.
+  const u16* kgp[2];
+  const u16* vgp[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int idx = i * 512 + tid;
+    const int L = idx * 16;                       // lane-linear LDS byte
+    const int kr = L >> 8;
+    const int kin = (((L & 255) ^ ((kr & 15) << 4)) >> 1);
+    kgp[i] = kbase + (long)kr * k_ss + kin;
+    // V pairing: thread owns kv rows (2rp, 2rp+1) at column chunk c8 so the
+    // scatter writes whole u32 kv-pairs (8 ds_write_b32, not 16 ds_write_b16)
+    vgp[i] = vbase + (long)((tid >> 4) * 2 + i) * v_ss + (tid & 15) * 8;
   }
-#define V6_WRITE_TILE(BUF)                                                  \
+  const long kstep = (long)V6_BN * k_ss;
+  const long vstep = (long)V6_BN * v_ss;
+  const int wu64 = __builtin_amdgcn_readfirstlane(tid >> 6);
+
+  uint4 vst[2];
+#define V6_GLDS_K(BUFI)                                                     \
   _Pragma("unroll")                                                         \
   for (int i = 0; i < 2; ++i) {                                             \
-    const int idx = i * 512 + tid;                                          \
-    const int r_ = idx >> 4;                                                \
-    const int c8_ = (idx & 15) * 8;                                         \
-    *reinterpret_cast<uint4*>(&(BUF)[k_byte(r_, c8_)]) = kst[i];            \
-    union { uint4 u; u16 h[8]; } vv_;                                       \
-    vv_.u = vst[i];                                                         \
+    __builtin_amdgcn_global_load_lds(                                       \
+        (const u32*)kgp[i], (u32*)&lds[BUFI][(i * 512 + wu64 * 64) * 8],    \
+        16, 0, 0);                                                          \
+    kgp[i] += kstep;                                                        \
+  }
+#define V6_LOAD_V()                                                         \
+  _Pragma("unroll")                                                         \
+  for (int i = 0; i < 2; ++i) {                                             \
+    vst[i] = *reinterpret_cast<const uint4*>(vgp[i]);                       \
+    vgp[i] += vstep;                                                        \
+  }
+#define V6_WRITE_V(BUF)                                                     \
+  {                                                                         \
+    const int rp2_ = (tid >> 4) * 2;                                        \
+    const int c8_ = (tid & 15) * 8;                                         \
+    union { uint4 u; u16 h[8]; } va_, vb_;                                  \
+    va_.u = vst[0];                                                         \
+    vb_.u = vst[1];                                                         \
     _Pragma("unroll")                                                       \
-    for (int j = 0; j < 8; ++j)                                             \
-      *reinterpret_cast<u16*>(&(BUF)[VOFF + vt_byte(c8_ + j, r_)]) = vv_.h[j]; \
+    for (int j = 0; j < 8; ++j) {                                           \
+      const u32 pair_ = (u32)va_.h[j] | ((u32)vb_.h[j] << 16);              \
+      *reinterpret_cast<u32*>(&(BUF)[VOFF + vt_byte(c8_ + j, rp2_)]) = pair_; \
+    }                                                                       \
   }
 
-  V6_LOAD_TILE(0)
-  V6_WRITE_TILE(lds0)
+  V6_GLDS_K(0)
+  V6_LOAD_V()
+  V6_WRITE_V(lds0)
   __syncthreads();
 
   int cur = 0;
   for (int t = 0; t < n_tiles; ++t) {
     const int kv0 = t * V6_BN;
     const bool have_next = (t + 1) < n_tiles;
-    if (have_next) V6_LOAD_TILE(t + 1)        // issue early (T14)
+    if (have_next) { V6_GLDS_K(cur ^ 1) V6_LOAD_V() }   // issue early (T14)
 
     char* kb = cur ? lds1 : lds0;
     char* vbuf = kb + VOFF;
@@ -359,6 +405,7 @@ __global__ __launch_bounds__(512) void attn_fwd_v6_kernel(
 
     f32x16 s0, s1;
     union PF { bf16x8 v; unsigned u[4]; } pa[4];
+    float pm = -INFINITY;
     if (active) {
       // ---- S^T = K @ Q^T over 2 kv-subtiles of 32 ----
 #pragma unroll
@@ -373,7 +420,6 @@ __global__ __launch_bounds__(512) void attn_fwd_v6_kernel(
         s1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, qf[kc].v, s1, 0, 0, 0);
       }
       // ---- scale into exp2 space + causal mask + tile max ----
-      float pm = -INFINITY;
       if (need_mask) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
@@ -395,11 +441,19 @@ __global__ __launch_bounds__(512) void attn_fwd_v6_kernel(
         }
       }
       pm = fmaxf(pm, __shfl_xor(pm, 32, 64));   // merge q-row halves
-
+    }
+    // stage tile t+1 now: QK^T covered the load latency, and writing here
+    // (instead of after the P repack) ends kst/vst's live range before the
+    // softmax -- the 252-VGPR scratch parking came from carrying them there
+    if (have_next) {
+      char* nb = cur ? lds0 : lds1;
+      V6_WRITE_V(nb)
+    }
+    if (active) {
       // ---- defer-rescale (T13, THR=8 in exp2 space) ----
       if (!__all(pm - m_run <= 8.0f)) {
         const float mn = fmaxf(m_run, pm);
-        const float alpha = (m_run == -INFINITY) ? 0.f : exp2f(m_run - mn);
+        const float alpha = (m_run == -INFINITY) ? 0.f : exp2_raw(m_run - mn);
         m_run = mn;
         l_run *= alpha;
 #pragma unroll
@@ -411,8 +465,8 @@ __global__ __launch_bounds__(512) void attn_fwd_v6_kernel(
       float psum = 0.f;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        const float p0 = exp2f(s0[r] - m_run);
-        const float p1 = exp2f(s1[r] - m_run);
+        const float p0 = exp2_raw(s0[r] - m_run);
+        const float p1 = exp2_raw(s1[r] - m_run);
         s0[r] = p0;
         s1[r] = p1;
         psum += p0 + p1;
@@ -449,12 +503,6 @@ __global__ __launch_bounds__(512) void attn_fwd_v6_kernel(
           pa[sub * 2 + cc] = f;
         }
       }
-    }
-
-    // ---- stage tile t+1 into the other buffer (write lands under PV) ----
-    if (have_next) {
-      char* nb = cur ? lds0 : lds1;
-      V6_WRITE_TILE(nb)
     }
 
     if (active) {
