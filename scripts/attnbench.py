@@ -10,7 +10,7 @@ import torch
 import torch.nn.functional as F
 
 from trainingjob_operator_amd.ops.attention import (
-    flash_attention_fwd_only, flash_attention_fwd_v5)
+    flash_attention_fwd_only, flash_attention_fwd_nw8, flash_attention_fwd_v5)
 
 
 def main():
@@ -27,6 +27,8 @@ def main():
             flash_attention_fwd_only(q, k, v)
         elif which == "v5":
             flash_attention_fwd_v5(q, k, v)
+        elif which == "nw8":
+            flash_attention_fwd_nw8(q, k, v)
         else:
             F.scaled_dot_product_attention(q, k, v, is_causal=True)
 

@@ -287,7 +287,8 @@ __device__ __forceinline__ float exp2_raw(float x) {
   return r;
 }
 
-__global__ __launch_bounds__(512) void attn_fwd_v6_kernel(
+template <int NW>   // waves per workgroup: 4 (BM=128, 2 WGs/CU) or 8 (BM=256)
+__global__ __launch_bounds__(NW * 64) void attn_fwd_v6_kernel(
     const u16* __restrict__ q, const u16* __restrict__ k,
     const u16* __restrict__ v, u16* __restrict__ out,
     float* __restrict__ lse,
@@ -295,6 +296,10 @@ __global__ __launch_bounds__(512) void attn_fwd_v6_kernel(
     long k_sb, long k_sh, long k_ss,
     long v_sb, long v_sh, long v_ss,
     int n_heads, int gqa_group, int S, float scale2) {
+  constexpr int NT = NW * 64;          // threads
+  constexpr int BM = NW * 32;          // q rows per workgroup
+  constexpr int KP = 1024 / NT;        // K glds pieces per thread (16 B each)
+  constexpr int VU = 512 / NT;         // V kv-pair units per thread
   const int qb = gridDim.x - 1 - blockIdx.x;   // longest-running blocks first
   const int h = blockIdx.y;
   const int b = blockIdx.z;
@@ -307,7 +312,7 @@ __global__ __launch_bounds__(512) void attn_fwd_v6_kernel(
 
   __shared__ __attribute__((aligned(16))) u16 lds[2][V6_BUF_U16];
 
-  const int q0w = qb * V6_BM + wid * V6_QBLK;
+  const int q0w = qb * BM + wid * V6_QBLK;
   const int qrow = q0w + low;                  // this lane's q row
 
   // Q row in registers: 8 k-chunks, lane holds Q[qrow][kc*16 + hi*8 + j]
@@ -328,7 +333,7 @@ __global__ __launch_bounds__(512) void attn_fwd_v6_kernel(
 
   const u16* kbase = k + (long)b * k_sb + (long)hkv * k_sh;
   const u16* vbase = v + (long)b * v_sb + (long)hkv * v_sh;
-  const int n_tiles = (qb * V6_BM + V6_BM) / V6_BN;   // causal upper bound
+  const int n_tiles = (qb * BM + BM) / V6_BN;   // causal upper bound
 
   char* lds0 = reinterpret_cast<char*>(&lds[0][0]);
   char* lds1 = reinterpret_cast<char*>(&lds[1][0]);
@@ -340,46 +345,54 @@ __global__ __launch_bounds__(512) void attn_fwd_v6_kernel(
   // byte-identical to the ds_write version and costs nothing in coalescing
   // (the XOR permutes 16-B slots within one 256-B row = the same two 128-B
   // requests per 16 lanes). V cannot go by glds (transposed image) and
-  // keeps register staging: 2 uint4 per thread, written after QK^T.
-  const u16* kgp[2];
-  const u16* vgp[2];
+  // keeps register staging: kv-row pairs written as u32 after QK^T.
+  const u16* kgp[KP];
+  const u16* vgp[2 * VU];
 #pragma unroll
-  for (int i = 0; i < 2; ++i) {
-    const int idx = i * 512 + tid;
+  for (int i = 0; i < KP; ++i) {
+    const int idx = i * NT + tid;
     const int L = idx * 16;                       // lane-linear LDS byte
     const int kr = L >> 8;
     const int kin = (((L & 255) ^ ((kr & 15) << 4)) >> 1);
     kgp[i] = kbase + (long)kr * k_ss + kin;
-    // V pairing: thread owns kv rows (2rp, 2rp+1) at column chunk c8 so the
-    // scatter writes whole u32 kv-pairs (8 ds_write_b32, not 16 ds_write_b16)
-    vgp[i] = vbase + (long)((tid >> 4) * 2 + i) * v_ss + (tid & 15) * 8;
+  }
+#pragma unroll
+  for (int u = 0; u < VU; ++u) {
+    const int unit = u * NT + tid;                // (rp, c8) unit
+    const int rp2 = (unit >> 4) * 2;
+    const int c8 = (unit & 15) * 8;
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+      vgp[u * 2 + i] = vbase + (long)(rp2 + i) * v_ss + c8;
   }
   const long kstep = (long)V6_BN * k_ss;
   const long vstep = (long)V6_BN * v_ss;
   const int wu64 = __builtin_amdgcn_readfirstlane(tid >> 6);
 
-  uint4 vst[2];
+  uint4 vst[2 * VU];
 #define V6_GLDS_K(BUFI)                                                     \
   _Pragma("unroll")                                                         \
-  for (int i = 0; i < 2; ++i) {                                             \
+  for (int i = 0; i < KP; ++i) {                                            \
     __builtin_amdgcn_global_load_lds(                                       \
-        (const u32*)kgp[i], (u32*)&lds[BUFI][(i * 512 + wu64 * 64) * 8],    \
+        (const u32*)kgp[i], (u32*)&lds[BUFI][(i * NT + wu64 * 64) * 8],     \
         16, 0, 0);                                                          \
     kgp[i] += kstep;                                                        \
   }
 #define V6_LOAD_V()                                                         \
   _Pragma("unroll")                                                         \
-  for (int i = 0; i < 2; ++i) {                                             \
+  for (int i = 0; i < 2 * VU; ++i) {                                        \
     vst[i] = *reinterpret_cast<const uint4*>(vgp[i]);                       \
     vgp[i] += vstep;                                                        \
   }
 #define V6_WRITE_V(BUF)                                                     \
-  {                                                                         \
-    const int rp2_ = (tid >> 4) * 2;                                        \
-    const int c8_ = (tid & 15) * 8;                                         \
-    union { uint4 u; u16 h[8]; } va_, vb_;                                  \
-    va_.u = vst[0];                                                         \
-    vb_.u = vst[1];                                                         \
+  _Pragma("unroll")                                                         \
+  for (int u = 0; u < VU; ++u) {                                            \
+    const int unit_ = u * NT + tid;                                         \
+    const int rp2_ = (unit_ >> 4) * 2;                                      \
+    const int c8_ = (unit_ & 15) * 8;                                       \
+    union { uint4 u4; u16 h[8]; } va_, vb_;                                 \
+    va_.u4 = vst[u * 2];                                                    \
+    vb_.u4 = vst[u * 2 + 1];                                                \
     _Pragma("unroll")                                                       \
     for (int j = 0; j < 8; ++j) {                                           \
       const u32 pair_ = (u32)va_.h[j] | ((u32)vb_.h[j] << 16);              \
@@ -443,8 +456,8 @@ __global__ __launch_bounds__(512) void attn_fwd_v6_kernel(
       pm = fmaxf(pm, __shfl_xor(pm, 32, 64));   // merge q-row halves
     }
     // stage tile t+1 now: QK^T covered the load latency, and writing here
-    // (instead of after the P repack) ends kst/vst's live range before the
-    // softmax -- the 252-VGPR scratch parking came from carrying them there
+    // (instead of after the P repack) ends vst's live range before the
+    // softmax (the 252-VGPR scratch parking came from carrying it there)
     if (have_next) {
       char* nb = cur ? lds0 : lds1;
       V6_WRITE_V(nb)
@@ -503,9 +516,7 @@ __global__ __launch_bounds__(512) void attn_fwd_v6_kernel(
           pa[sub * 2 + cc] = f;
         }
       }
-    }
 
-    if (active) {
       // ---- O^T += V^T @ P over 4 kv-chunks x 4 d-subtiles ----
 #pragma unroll
       for (int c = 0; c < 4; ++c) {
@@ -555,10 +566,23 @@ extern "C" int attn_fwd(void* stream, const void* q, const void* k,
   if (S <= 0 || n_heads <= 0 || batch <= 0 || n_kv_heads <= 0 ||
       n_heads % n_kv_heads != 0)
     return -1;
-  if (S % V6_BM == 0) {
-    dim3 grid(S / V6_BM, n_heads, batch), block(512);
+  if (S % 128 == 0) {
+    // BM=256 preferred: halves the K/V staging traffic vs BM=128 (each kv
+    // tile is read by half as many workgroups); measured 344 vs 587 us at
+    // B1H32S4096. BM=128 covers the S%128 shapes.
     const float scale2 = scale * 1.4426950408889634f;   // fold log2(e)
-    hipLaunchKernelGGL(attn_fwd_v6_kernel, grid, block, 0,
+    if (S % 256 == 0) {
+      dim3 grid(S / 256, n_heads, batch), block(512);
+      hipLaunchKernelGGL((attn_fwd_v6_kernel<8>), grid, block, 0,
+                         reinterpret_cast<hipStream_t>(stream),
+                         (const u16*)q, (const u16*)k, (const u16*)v,
+                         (u16*)out, (float*)lse, q_sb, q_sh, q_ss,
+                         k_sb, k_sh, k_ss, v_sb, v_sh, v_ss, n_heads,
+                         n_heads / n_kv_heads, S, scale2);
+      return 0;
+    }
+    dim3 grid(S / 128, n_heads, batch), block(256);
+    hipLaunchKernelGGL((attn_fwd_v6_kernel<4>), grid, block, 0,
                        reinterpret_cast<hipStream_t>(stream),
                        (const u16*)q, (const u16*)k, (const u16*)v, (u16*)out,
                        (float*)lse, q_sb, q_sh, q_ss, k_sb, k_sh, k_ss,
@@ -573,6 +597,26 @@ extern "C" int attn_fwd(void* stream, const void* q, const void* k,
                      (const u16*)q, (const u16*)k, (const u16*)v, (u16*)out,
                      (float*)lse, q_sb, q_sh, q_ss, k_sb, k_sh, k_ss,
                      v_sb, v_sh, v_ss, n_heads, S, scale);
+  return 0;
+}
+
+// 8-wave instantiation kept callable for A/B benchmarking
+extern "C" int attn_fwd_nw8(void* stream, const void* q, const void* k,
+                            const void* v, void* out, void* lse,
+                            long q_sb, long q_sh, long q_ss,
+                            long k_sb, long k_sh, long k_ss,
+                            long v_sb, long v_sh, long v_ss,
+                            int batch, int n_heads, int n_kv_heads, int S,
+                            float scale) {
+  if (S <= 0 || S % 256 != 0 || n_heads % n_kv_heads != 0) return -1;
+  dim3 grid(S / 256, n_heads, batch), block(512);
+  const float scale2 = scale * 1.4426950408889634f;
+  hipLaunchKernelGGL((attn_fwd_v6_kernel<8>), grid, block, 0,
+                     reinterpret_cast<hipStream_t>(stream),
+                     (const u16*)q, (const u16*)k, (const u16*)v, (u16*)out,
+                     (float*)lse, q_sb, q_sh, q_ss, k_sb, k_sh, k_ss,
+                     v_sb, v_sh, v_ss, n_heads, n_heads / n_kv_heads, S,
+                     scale2);
   return 0;
 }
 
