@@ -17,6 +17,7 @@ reduces dk/dv to the kv heads).
 from __future__ import annotations
 
 import math
+import os
 from typing import Optional
 
 import torch
@@ -61,7 +62,10 @@ class _NativeFlashAttention(torch.autograd.Function):
     @staticmethod
     def backward(ctx, gout):
         q, k, v, out, lse = ctx.saved_tensors
-        S = q.shape[2]
+        B, H, S, D = q.shape
+        if S % 256 == 0 and os.environ.get("AITJ_ATTN_BWD") != "aten":
+            return (*_native_backward(q, k, v, out, lse, gout, ctx.scale),
+                    None)
         # None -> undefined Tensor: selects the dense (non-varlen) path in
         # the aten flash backward (empty tensors select varlen and fail).
         philox = torch.empty(0, dtype=torch.int64, device=q.device)
@@ -70,6 +74,28 @@ class _NativeFlashAttention(torch.autograd.Function):
             None, None, S, S, 0.0, True,
             philox, philox, scale=ctx.scale)
         return dq, dk, dv, None
+
+
+def _native_backward(q, k, v, out, lse, gout, scale):
+    """Hand-written CDNA4 flash backward (delta + dq + dv + dk kernels)."""
+    B, H, S, D = q.shape
+    HKV = k.shape[1]
+    lib = native.load(require=True)
+    dout = gout.contiguous()
+    delta = torch.empty(B, H, S, dtype=torch.float32, device=q.device)
+    dq = torch.empty(B, H, S, D, dtype=torch.bfloat16, device=q.device)
+    dk = torch.empty(B, HKV, S, D, dtype=torch.bfloat16, device=q.device)
+    dv = torch.empty(B, HKV, S, D, dtype=torch.bfloat16, device=q.device)
+    rc = lib.attn_bwd(
+        native.stream_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+        out.data_ptr(), dout.data_ptr(), lse.data_ptr(), delta.data_ptr(),
+        dq.data_ptr(), dk.data_ptr(), dv.data_ptr(),
+        q.stride(0), q.stride(1), q.stride(2),
+        k.stride(0), k.stride(1), k.stride(2),
+        v.stride(0), v.stride(1), v.stride(2),
+        B, H, HKV, S, scale)
+    native.check_rc(rc, "attn_bwd", f"B={B} H={H} HKV={HKV} S={S}")
+    return dq, dk, dv
 
 
 def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,

@@ -44,6 +44,12 @@ __device__ __forceinline__ u16 attn_f2bf(float f) {
   return r.x;
 }
 
+__device__ __forceinline__ float bf2f_(u16 b) {
+  union { u32 u; float f; } c;
+  c.u = ((u32)b) << 16;
+  return c.f;
+}
+
 // T2 swizzle for the K tile: 16-byte slot index XORed with row & 15.
 __device__ __forceinline__ int k_byte(int row, int col_elem) {
   return (row * ATTN_D * 2 + col_elem * 2) ^ ((row & 15) << 4);
@@ -634,5 +640,783 @@ extern "C" int attn_fwd_v5(void* stream, const void* q, const void* k,
                      (const u16*)q, (const u16*)k, (const u16*)v, (u16*)out,
                      (float*)lse, q_sb, q_sh, q_ss, k_sb, k_sh, k_ss,
                      v_sb, v_sh, v_ss, n_heads, S, scale);
+  return 0;
+}
+
+// ===========================================================================
+// v6 flash-attention BACKWARD (causal, bf16, D=128, GQA).
+//
+// Standard flash backward split into two walks plus a delta precompute:
+//   delta[q]   = rowsum(dO[q] * O[q])                       (attn_delta)
+//   P          = exp2(S*scale*log2e - lse*log2e)            (recomputed)
+//   dV[kv]    += P^T @ dO        dK[kv] += scale * dS^T @ Q (attn_bwd_dkdv)
+//   dS         = P o (dP - delta),  dP = dO @ V^T
+//   dQ[q]     += scale * dS @ K                             (attn_bwd_dq)
+//
+// No online softmax state (lse is known), so the backward is simpler per
+// tile than the forward. Both kernels reuse the v6 layout machinery:
+// operand-swapped 32x32x16 MFMAs keep the OWNED axis lane-local (q rows in
+// dq, kv rows in dkdv), P/dS repack stays in registers via
+// cvt_pk_bf16_f32 + permlane32_swap, row-major images use the k_byte
+// swizzle, transposed images the vt_byte image. GQA: dq walks q heads;
+// dkdv owns a kv head and loops the group's q heads, accumulating dK/dV
+// in registers across the loop.
+// ===========================================================================
+
+__global__ __launch_bounds__(256) void attn_delta_kernel(
+    const u16* __restrict__ dout, const u16* __restrict__ o,
+    float* __restrict__ delta, long n_rows) {
+  // one 16-lane group per row; 8 elems per thread
+  const long row = (long)blockIdx.x * 16 + (threadIdx.x >> 4);
+  if (row >= n_rows) return;
+  const int e8 = (threadIdx.x & 15) * 8;
+  union { uint4 u; u16 h[8]; } a, b;
+  a.u = *reinterpret_cast<const uint4*>(dout + row * ATTN_D + e8);
+  b.u = *reinterpret_cast<const uint4*>(o + row * ATTN_D + e8);
+  float acc = 0.f;
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    acc += bf2f_(a.h[j]) * bf2f_(b.h[j]);
+#pragma unroll
+  for (int off = 1; off < 16; off <<= 1) acc += __shfl_xor(acc, off, 64);
+  if ((threadIdx.x & 15) == 0) delta[row] = acc;
+}
+
+// ---------------------------------------------------------------------------
+// dQ kernel: one 512-thread WG owns 256 q rows (8 waves x 32, q lane-local).
+// Per 64-kv tile: S^T = K@Q^T (16 MFMA), dP^T = V@dO^T (16), dS^T packed,
+// dQ^T += K^T @ dS (16). LDS per buffer: K rows (k_byte) + K^T (vt image)
+// + V rows (k_byte) = 50.4 KB, double-buffered.
+// ---------------------------------------------------------------------------
+
+#define BQ_K_U16 (V6_BN * ATTN_D)                      // K row image
+#define BQ_KT_U16 (ATTN_D * (VT_PITCH_B / 2) + 128)    // K^T image
+#define BQ_V_U16 (V6_BN * ATTN_D)                      // V row image
+#define BQ_BUF_U16 (BQ_K_U16 + BQ_KT_U16 + BQ_V_U16)
+
+__global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
+    const u16* __restrict__ q, const u16* __restrict__ k,
+    const u16* __restrict__ v, const u16* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    u16* __restrict__ dq,
+    long q_sb, long q_sh, long q_ss,
+    long k_sb, long k_sh, long k_ss,
+    long v_sb, long v_sh, long v_ss,
+    int n_heads, int gqa_group, int S, float scale) {
+  const int qb = gridDim.x - 1 - blockIdx.x;
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int hkv = h / gqa_group;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int lane = tid & 63;
+  const int low = lane & 31;
+  const int hi = lane >> 5;
+  const float scale2 = scale * 1.4426950408889634f;
+
+  __shared__ __attribute__((aligned(16))) u16 lds[2][BQ_BUF_U16];
+
+  const int q0w = qb * 256 + wid * 32;
+  const int qrow = q0w + low;
+
+  union F8 { bf16x8 v; uint4 u; u16 h[8]; };
+  const u16* qptr = q + (long)b * q_sb + (long)h * q_sh + (long)qrow * q_ss;
+  // dO is produced contiguous [B,H,S,D] by autograd (we .contiguous() it)
+  const u16* doptr = dout + (((long)b * n_heads + h) * S + qrow) * ATTN_D;
+  F8 qf[8], dof[8];
+#pragma unroll
+  for (int kc = 0; kc < 8; ++kc) {
+    qf[kc].u = *reinterpret_cast<const uint4*>(qptr + kc * 16 + hi * 8);
+    dof[kc].u = *reinterpret_cast<const uint4*>(doptr + kc * 16 + hi * 8);
+  }
+  const long lrow = ((long)b * n_heads + h) * S + qrow;
+  const float lse2 = lse[lrow] * 1.4426950408889634f;
+  const float dlt = delta[lrow];
+
+  f32x16 dqacc[4];
+#pragma unroll
+  for (int ds = 0; ds < 4; ++ds)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dqacc[ds][r] = 0.f;
+
+  const u16* kbase = k + (long)b * k_sb + (long)hkv * k_sh;
+  const u16* vbase = v + (long)b * v_sb + (long)hkv * v_sh;
+  const int n_tiles = (qb + 1) * 4;           // (qb*256+256)/64
+
+  char* lds0 = reinterpret_cast<char*>(&lds[0][0]);
+  char* lds1 = reinterpret_cast<char*>(&lds[1][0]);
+  const int KTOFF = BQ_K_U16 * 2;
+  const int VROFF = (BQ_K_U16 + BQ_KT_U16) * 2;
+
+  // staging: K rows + V rows by glds (swizzle-inverted source); K^T by
+  // register scatter (kv-pair u32 writes into the vt image)
+  const u16* kgp[2];
+  const u16* vgp[2];
+  const u16* ktp[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int idx = i * 512 + tid;
+    const int L = idx * 16;
+    const int kr = L >> 8;
+    const int kin = (((L & 255) ^ ((kr & 15) << 4)) >> 1);
+    kgp[i] = kbase + (long)kr * k_ss + kin;
+    vgp[i] = vbase + (long)kr * v_ss + kin;
+  }
+  {
+    const int rp2 = (tid >> 4) * 2;
+    const int c8 = (tid & 15) * 8;
+    ktp[0] = kbase + (long)rp2 * k_ss + c8;
+    ktp[1] = kbase + (long)(rp2 + 1) * k_ss + c8;
+  }
+  const long kstep = (long)V6_BN * k_ss;
+  const long vstep = (long)V6_BN * v_ss;
+  const int wu64 = __builtin_amdgcn_readfirstlane(tid >> 6);
+  uint4 ktst[2];
+
+#define BQ_GLDS(BUFI)                                                       \
+  _Pragma("unroll")                                                         \
+  for (int i = 0; i < 2; ++i) {                                             \
+    __builtin_amdgcn_global_load_lds(                                       \
+        (const u32*)kgp[i], (u32*)&lds[BUFI][(i * 512 + wu64 * 64) * 8],    \
+        16, 0, 0);                                                          \
+    kgp[i] += kstep;                                                        \
+    __builtin_amdgcn_global_load_lds(                                       \
+        (const u32*)vgp[i],                                                 \
+        (u32*)&lds[BUFI][VROFF / 2 + (i * 512 + wu64 * 64) * 8],            \
+        16, 0, 0);                                                          \
+    vgp[i] += vstep;                                                        \
+  }
+#define BQ_LOAD_KT()                                                        \
+  {                                                                         \
+    ktst[0] = *reinterpret_cast<const uint4*>(ktp[0]);                      \
+    ktst[1] = *reinterpret_cast<const uint4*>(ktp[1]);                      \
+    ktp[0] += kstep;                                                        \
+    ktp[1] += kstep;                                                        \
+  }
+#define BQ_WRITE_KT(BUF)                                                    \
+  {                                                                         \
+    const int rp2_ = (tid >> 4) * 2;                                        \
+    const int c8_ = (tid & 15) * 8;                                         \
+    union { uint4 u4; u16 h[8]; } va_, vb_;                                 \
+    va_.u4 = ktst[0];                                                       \
+    vb_.u4 = ktst[1];                                                       \
+    _Pragma("unroll")                                                       \
+    for (int j = 0; j < 8; ++j) {                                           \
+      const u32 pair_ = (u32)va_.h[j] | ((u32)vb_.h[j] << 16);              \
+      *reinterpret_cast<u32*>(&(BUF)[KTOFF + vt_byte(c8_ + j, rp2_)]) = pair_; \
+    }                                                                       \
+  }
+
+  BQ_GLDS(0)
+  BQ_LOAD_KT()
+  BQ_WRITE_KT(lds0)
+  __syncthreads();
+
+  int cur = 0;
+  for (int t = 0; t < n_tiles; ++t) {
+    const int kv0 = t * V6_BN;
+    const bool have_next = (t + 1) < n_tiles;
+    if (have_next) { BQ_GLDS(cur ^ 1) BQ_LOAD_KT() }
+
+    char* kb = cur ? lds1 : lds0;
+    char* ktb = kb + KTOFF;
+    char* vb = kb + VROFF;
+    const bool active = kv0 <= q0w + 31;
+    const bool need_mask = kv0 + V6_BN > q0w;
+
+    union PF { bf16x8 v; unsigned u[4]; } pds[4];
+    if (active) {
+#pragma unroll
+      for (int sub = 0; sub < 2; ++sub) {
+        // ---- S^T subtile: C[kv32][q32] ----
+        f32x16 s;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) s[r] = 0.f;
+#pragma unroll
+        for (int kc = 0; kc < 8; ++kc) {
+          bf16x8 a = *reinterpret_cast<const bf16x8*>(
+              &kb[k_byte(sub * 32 + low, kc * 16 + hi * 8)]);
+          s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qf[kc].v, s, 0, 0, 0);
+        }
+        // ---- dP^T subtile: C[kv32][q32] ----
+        f32x16 dp;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) dp[r] = 0.f;
+#pragma unroll
+        for (int kc = 0; kc < 8; ++kc) {
+          bf16x8 a = *reinterpret_cast<const bf16x8*>(
+              &vb[k_byte(sub * 32 + low, kc * 16 + hi * 8)]);
+          dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, dof[kc].v, dp,
+                                                       0, 0, 0);
+        }
+        // ---- dS = P o (dP - delta) * scale,  P = 2^(s*scale2 - lse2) ----
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          float x = s[r] * scale2 - lse2;
+          if (need_mask) {
+            const int kvr = kv0 + sub * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+            if (kvr > qrow) x = -1e30f;
+          }
+          const float p = exp2_raw(x);
+          s[r] = p * (dp[r] - dlt) * scale;
+        }
+        // ---- repack dS -> bf16 fragments ----
+#pragma unroll
+        for (int cc = 0; cc < 2; ++cc) {
+          const int pb = cc * 8;
+          unsigned a0 = cvt_pk_bf16(s[pb + 0], s[pb + 1]);
+          unsigned b0 = cvt_pk_bf16(s[pb + 4], s[pb + 5]);
+          unsigned a1 = cvt_pk_bf16(s[pb + 2], s[pb + 3]);
+          unsigned b1 = cvt_pk_bf16(s[pb + 6], s[pb + 7]);
+          auto r02 = __builtin_amdgcn_permlane32_swap(a0, b0, false, false);
+          auto r13 = __builtin_amdgcn_permlane32_swap(a1, b1, false, false);
+          PF f;
+          f.u[0] = r02[0];
+          f.u[1] = r13[0];
+          f.u[2] = r02[1];
+          f.u[3] = r13[1];
+          pds[sub * 2 + cc] = f;
+        }
+      }
+    }
+    if (have_next) {
+      char* nb = cur ? lds0 : lds1;
+      BQ_WRITE_KT(nb)
+    }
+    if (active) {
+      // ---- dQ^T += K^T @ dS over 4 kv chunks x 4 d subtiles ----
+#pragma unroll
+      for (int c = 0; c < 4; ++c) {
+#pragma unroll
+        for (int ds = 0; ds < 4; ++ds) {
+          bf16x8 kf = *reinterpret_cast<const bf16x8*>(
+              &ktb[vt_byte(ds * 32 + low, c * 16 + hi * 8)]);
+          dqacc[ds] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              kf, pds[c].v, dqacc[ds], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // ---- epilogue: dq [B,H,S,D] contiguous ----
+  u16* dqrow = dq + (((long)b * n_heads + h) * S + qrow) * ATTN_D;
+#pragma unroll
+  for (int ds = 0; ds < 4; ++ds) {
+#pragma unroll
+    for (int g = 0; g < 4; ++g) {
+      const int d0 = 8 * g + 4 * hi + 32 * ds;
+      const unsigned w0 = cvt_pk_bf16(dqacc[ds][4 * g + 0],
+                                      dqacc[ds][4 * g + 1]);
+      const unsigned w1 = cvt_pk_bf16(dqacc[ds][4 * g + 2],
+                                      dqacc[ds][4 * g + 3]);
+      uint2 wv;
+      wv.x = w0;
+      wv.y = w1;
+      *reinterpret_cast<uint2*>(dqrow + d0) = wv;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// dV kernel: one 512-thread WG owns 256 kv rows (8 waves x 32, kv
+// lane-local); loops the GQA group's q heads and their q tiles.
+//   S^T tile: C[q32][kv32] = Q @ K^T   (K fragments persistent in registers)
+//   P = 2^(s*scale2 - lse2[q]) (masked on the diagonal band)
+//   dV^T += dO^T @ P: C[d][kv] via A = dO^T image, B = P-fragment repack
+// LDS per buffer: Q row image + dO^T image + lse tile.
+// ---------------------------------------------------------------------------
+
+#define BV_Q_U16 (V6_BN * ATTN_D)
+#define BV_DOT_U16 (ATTN_D * (VT_PITCH_B / 2) + 128)
+#define BV_LSE_U16 128                      // 64 f32
+#define BV_BUF_U16 (BV_Q_U16 + BV_DOT_U16 + BV_LSE_U16)
+
+__global__ __launch_bounds__(512) void attn_bwd_dv_kernel(
+    const u16* __restrict__ q, const u16* __restrict__ k,
+    const u16* __restrict__ dout, const float* __restrict__ lse,
+    u16* __restrict__ dv,
+    long q_sb, long q_sh, long q_ss,
+    long k_sb, long k_sh, long k_ss,
+    int n_heads, int n_kv_heads, int S, float scale) {
+  const int kvb = blockIdx.x;                // kv block of 256 (asc = heavy 1st)
+  const int hkv = blockIdx.y;
+  const int b = blockIdx.z;
+  const int gqa_group = n_heads / n_kv_heads;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int lane = tid & 63;
+  const int low = lane & 31;
+  const int hi = lane >> 5;
+  const float scale2 = scale * 1.4426950408889634f;
+
+  __shared__ __attribute__((aligned(16))) u16 lds[2][BV_BUF_U16];
+  char* lds0 = reinterpret_cast<char*>(&lds[0][0]);
+  char* lds1 = reinterpret_cast<char*>(&lds[1][0]);
+  const int DOTOFF = BV_Q_U16 * 2;
+  const int LSEOFF = (BV_Q_U16 + BV_DOT_U16) * 2;
+
+  const int kv0w = kvb * 256 + wid * 32;
+  const int kvrow = kv0w + low;              // this lane's kv row
+
+  // K row in registers (persistent across the whole WG lifetime)
+  union F8 { bf16x8 v; uint4 u; u16 h[8]; };
+  const u16* kptr = k + (long)b * k_sb + (long)hkv * k_sh + (long)kvrow * k_ss;
+  F8 kf[8];
+#pragma unroll
+  for (int kc = 0; kc < 8; ++kc)
+    kf[kc].u = *reinterpret_cast<const uint4*>(kptr + kc * 16 + hi * 8);
+
+  f32x16 dvacc[4];
+#pragma unroll
+  for (int ds = 0; ds < 4; ++ds)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dvacc[ds][r] = 0.f;
+
+  const int qt0 = kvb * 4;                   // first causal q tile
+  const int qtn = S / 64;
+  const int wu64 = __builtin_amdgcn_readfirstlane(tid >> 6);
+  uint4 dtst[2];
+
+  for (int g = 0; g < gqa_group; ++g) {
+    const int h = hkv * gqa_group + g;
+    const u16* qbase = q + (long)b * q_sb + (long)h * q_sh;
+    const u16* dobase = dout + (((long)b * n_heads + h) * S) * ATTN_D;
+    const float* lbase = lse + ((long)b * n_heads + h) * S;
+
+    // per-thread staging pointers for this head
+    const u16* qgp[2];
+    const u16* dtp[2];
+    {
+      const int idx0 = tid, idx1 = 512 + tid;
+      const int L0 = idx0 * 16, L1 = idx1 * 16;
+      const int r0 = L0 >> 8, r1 = L1 >> 8;
+      const int in0 = (((L0 & 255) ^ ((r0 & 15) << 4)) >> 1);
+      const int in1 = (((L1 & 255) ^ ((r1 & 15) << 4)) >> 1);
+      qgp[0] = qbase + (long)(qt0 * 64 + r0) * q_ss + in0;
+      qgp[1] = qbase + (long)(qt0 * 64 + r1) * q_ss + in1;
+      const int rp2 = (tid >> 4) * 2;
+      const int c8 = (tid & 15) * 8;
+      dtp[0] = dobase + (long)(qt0 * 64 + rp2) * ATTN_D + c8;
+      dtp[1] = dobase + (long)(qt0 * 64 + rp2 + 1) * ATTN_D + c8;
+    }
+    const long qstep = (long)64 * q_ss;
+    const long dostep = (long)64 * ATTN_D;
+
+#define BV_STAGE_IN(BUFI, QT)                                               \
+    {                                                                       \
+      _Pragma("unroll")                                                     \
+      for (int i = 0; i < 2; ++i) {                                         \
+        __builtin_amdgcn_global_load_lds(                                   \
+            (const u32*)qgp[i], (u32*)&lds[BUFI][(i * 512 + wu64 * 64) * 8],\
+            16, 0, 0);                                                      \
+        qgp[i] += qstep;                                                    \
+      }                                                                     \
+      dtst[0] = *reinterpret_cast<const uint4*>(dtp[0]);                    \
+      dtst[1] = *reinterpret_cast<const uint4*>(dtp[1]);                    \
+      dtp[0] += dostep;                                                     \
+      dtp[1] += dostep;                                                     \
+      if (tid < 64)                                                         \
+        *reinterpret_cast<float*>(                                          \
+            &lds[BUFI][LSEOFF / 2 + tid * 2]) =                             \
+            lbase[(QT) * 64 + tid] * 1.4426950408889634f;                   \
+    }
+#define BV_WRITE_DOT(BUF)                                                   \
+    {                                                                       \
+      const int rp2_ = (tid >> 4) * 2;                                      \
+      const int c8_ = (tid & 15) * 8;                                       \
+      union { uint4 u4; u16 h[8]; } va_, vb_;                               \
+      va_.u4 = dtst[0];                                                     \
+      vb_.u4 = dtst[1];                                                     \
+      _Pragma("unroll")                                                     \
+      for (int j = 0; j < 8; ++j) {                                         \
+        const u32 pair_ = (u32)va_.h[j] | ((u32)vb_.h[j] << 16);            \
+        *reinterpret_cast<u32*>(&(BUF)[DOTOFF + vt_byte(c8_ + j, rp2_)]) =  \
+            pair_;                                                          \
+      }                                                                     \
+    }
+
+    BV_STAGE_IN(0, qt0)
+    BV_WRITE_DOT(lds0)
+    __syncthreads();
+
+    int cur = 0;
+    for (int qt = qt0; qt < qtn; ++qt) {
+      const bool have_next = (qt + 1) < qtn;
+      if (have_next) BV_STAGE_IN(cur ^ 1, qt + 1)
+
+      char* qb_ = cur ? lds1 : lds0;
+      char* dob = qb_ + DOTOFF;
+      const float* lse2t = reinterpret_cast<const float*>(qb_ + LSEOFF);
+      // this wave's kv rows need q >= kv0w; skip tiles fully in the past
+      const bool active = qt * 64 + 63 >= kv0w;
+      const bool need_mask = qt * 64 < kv0w + 32;
+
+      union PF { bf16x8 v; unsigned u[4]; } pf[4];
+      if (active) {
+#pragma unroll
+        for (int sub = 0; sub < 2; ++sub) {
+          f32x16 s;
+#pragma unroll
+          for (int r = 0; r < 16; ++r) s[r] = 0.f;
+#pragma unroll
+          for (int kc = 0; kc < 8; ++kc) {
+            bf16x8 a = *reinterpret_cast<const bf16x8*>(
+                &qb_[k_byte(sub * 32 + low, kc * 16 + hi * 8)]);
+            s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, kf[kc].v, s,
+                                                        0, 0, 0);
+          }
+          // P = 2^(s*scale2 - lse2[q]); mask q < kv
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int qr = (r & 3) + 8 * (r >> 2) + 4 * hi;   // q within sub
+            float x = s[r] * scale2 - lse2t[sub * 32 + qr];
+            if (need_mask && qt * 64 + sub * 32 + qr < kvrow) x = -1e30f;
+            s[r] = exp2_raw(x);
+          }
+          // repack P (C[q][kv]) -> B-fragments [k=q chunk][n=kv]
+#pragma unroll
+          for (int cc = 0; cc < 2; ++cc) {
+            const int pb = cc * 8;
+            unsigned a0 = cvt_pk_bf16(s[pb + 0], s[pb + 1]);
+            unsigned b0 = cvt_pk_bf16(s[pb + 4], s[pb + 5]);
+            unsigned a1 = cvt_pk_bf16(s[pb + 2], s[pb + 3]);
+            unsigned b1 = cvt_pk_bf16(s[pb + 6], s[pb + 7]);
+            auto r02 = __builtin_amdgcn_permlane32_swap(a0, b0, false, false);
+            auto r13 = __builtin_amdgcn_permlane32_swap(a1, b1, false, false);
+            PF f;
+            f.u[0] = r02[0];
+            f.u[1] = r13[0];
+            f.u[2] = r02[1];
+            f.u[3] = r13[1];
+            pf[sub * 2 + cc] = f;
+          }
+        }
+      }
+      if (have_next) {
+        char* nb = cur ? lds0 : lds1;
+        BV_WRITE_DOT(nb)
+      }
+      if (active) {
+        // dV^T += dO^T @ P over 4 q chunks x 4 d subtiles
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+#pragma unroll
+          for (int ds = 0; ds < 4; ++ds) {
+            bf16x8 af = *reinterpret_cast<const bf16x8*>(
+                &dob[vt_byte(ds * 32 + low, c * 16 + hi * 8)]);
+            dvacc[ds] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                af, pf[c].v, dvacc[ds], 0, 0, 0);
+          }
+        }
+      }
+      __syncthreads();
+      cur ^= 1;
+    }
+  }
+
+  // epilogue: dv [B,HKV,S,D] contiguous
+  u16* dvrow = dv + (((long)b * n_kv_heads + hkv) * S + kvrow) * ATTN_D;
+#pragma unroll
+  for (int ds = 0; ds < 4; ++ds) {
+#pragma unroll
+    for (int g = 0; g < 4; ++g) {
+      const int d0 = 8 * g + 4 * hi + 32 * ds;
+      const unsigned w0 = cvt_pk_bf16(dvacc[ds][4 * g + 0],
+                                      dvacc[ds][4 * g + 1]);
+      const unsigned w1 = cvt_pk_bf16(dvacc[ds][4 * g + 2],
+                                      dvacc[ds][4 * g + 3]);
+      uint2 wv;
+      wv.x = w0;
+      wv.y = w1;
+      *reinterpret_cast<uint2*>(dvrow + d0) = wv;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// dK kernel: same kv-block walk as dV.
+//   S^T: C[q][kv] = Q @ K^T       (K fragments persistent)
+//   dP:  C[q][kv] = dO @ V^T      (V fragments persistent)
+//   dS = P o (dP - delta[q]) * scale
+//   dK^T += Q^T @ dS: C[d][kv] via A = Q^T image, B = dS repack
+// LDS per buffer: Q row image + dO row image + Q^T image + lse + delta.
+// ---------------------------------------------------------------------------
+
+#define BK_Q_U16 (V6_BN * ATTN_D)
+#define BK_DO_U16 (V6_BN * ATTN_D)
+#define BK_QT_U16 (ATTN_D * (VT_PITCH_B / 2) + 128)
+#define BK_LSE_U16 128
+#define BK_DLT_U16 128
+#define BK_BUF_U16 (BK_Q_U16 + BK_DO_U16 + BK_QT_U16 + BK_LSE_U16 + BK_DLT_U16)
+
+__global__ __launch_bounds__(512) void attn_bwd_dk_kernel(
+    const u16* __restrict__ q, const u16* __restrict__ k,
+    const u16* __restrict__ v, const u16* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    u16* __restrict__ dk,
+    long q_sb, long q_sh, long q_ss,
+    long k_sb, long k_sh, long k_ss,
+    long v_sb, long v_sh, long v_ss,
+    int n_heads, int n_kv_heads, int S, float scale) {
+  const int kvb = blockIdx.x;
+  const int hkv = blockIdx.y;
+  const int b = blockIdx.z;
+  const int gqa_group = n_heads / n_kv_heads;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int lane = tid & 63;
+  const int low = lane & 31;
+  const int hi = lane >> 5;
+  const float scale2 = scale * 1.4426950408889634f;
+
+  __shared__ __attribute__((aligned(16))) u16 lds[2][BK_BUF_U16];
+  char* lds0 = reinterpret_cast<char*>(&lds[0][0]);
+  char* lds1 = reinterpret_cast<char*>(&lds[1][0]);
+  const int DOOFF = BK_Q_U16 * 2;
+  const int QTOFF = (BK_Q_U16 + BK_DO_U16) * 2;
+  const int LSEOFF = (BK_Q_U16 + BK_DO_U16 + BK_QT_U16) * 2;
+  const int DLTOFF = LSEOFF + BK_LSE_U16 * 2;
+
+  const int kv0w = kvb * 256 + wid * 32;
+  const int kvrow = kv0w + low;
+
+  union F8 { bf16x8 v; uint4 u; u16 h[8]; };
+  const u16* kptr = k + (long)b * k_sb + (long)hkv * k_sh + (long)kvrow * k_ss;
+  const u16* vptr = v + (long)b * v_sb + (long)hkv * v_sh + (long)kvrow * v_ss;
+  F8 kf[8], vf[8];
+#pragma unroll
+  for (int kc = 0; kc < 8; ++kc) {
+    kf[kc].u = *reinterpret_cast<const uint4*>(kptr + kc * 16 + hi * 8);
+    vf[kc].u = *reinterpret_cast<const uint4*>(vptr + kc * 16 + hi * 8);
+  }
+
+  f32x16 dkacc[4];
+#pragma unroll
+  for (int ds = 0; ds < 4; ++ds)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dkacc[ds][r] = 0.f;
+
+  const int qt0 = kvb * 4;
+  const int qtn = S / 64;
+  const int wu64 = __builtin_amdgcn_readfirstlane(tid >> 6);
+  uint4 qtst[2];
+
+  for (int g = 0; g < gqa_group; ++g) {
+    const int h = hkv * gqa_group + g;
+    const u16* qbase = q + (long)b * q_sb + (long)h * q_sh;
+    const u16* dobase = dout + (((long)b * n_heads + h) * S) * ATTN_D;
+    const float* lbase = lse + ((long)b * n_heads + h) * S;
+    const float* dbase = delta + ((long)b * n_heads + h) * S;
+
+    const u16* qgp[2];
+    const u16* dgp[2];
+    const u16* qtp[2];
+    {
+      const int idx0 = tid, idx1 = 512 + tid;
+      const int L0 = idx0 * 16, L1 = idx1 * 16;
+      const int r0 = L0 >> 8, r1 = L1 >> 8;
+      const int in0 = (((L0 & 255) ^ ((r0 & 15) << 4)) >> 1);
+      const int in1 = (((L1 & 255) ^ ((r1 & 15) << 4)) >> 1);
+      qgp[0] = qbase + (long)(qt0 * 64 + r0) * q_ss + in0;
+      qgp[1] = qbase + (long)(qt0 * 64 + r1) * q_ss + in1;
+      dgp[0] = dobase + (long)(qt0 * 64 + r0) * ATTN_D + in0;
+      dgp[1] = dobase + (long)(qt0 * 64 + r1) * ATTN_D + in1;
+      const int rp2 = (tid >> 4) * 2;
+      const int c8 = (tid & 15) * 8;
+      qtp[0] = qbase + (long)(qt0 * 64 + rp2) * q_ss + c8;
+      qtp[1] = qbase + (long)(qt0 * 64 + rp2 + 1) * q_ss + c8;
+    }
+    const long qstep = (long)64 * q_ss;
+    const long dostep = (long)64 * ATTN_D;
+
+#define BK_STAGE_IN(BUFI, QT)                                               \
+    {                                                                       \
+      _Pragma("unroll")                                                     \
+      for (int i = 0; i < 2; ++i) {                                         \
+        __builtin_amdgcn_global_load_lds(                                   \
+            (const u32*)qgp[i], (u32*)&lds[BUFI][(i * 512 + wu64 * 64) * 8],\
+            16, 0, 0);                                                      \
+        qgp[i] += qstep;                                                    \
+        __builtin_amdgcn_global_load_lds(                                   \
+            (const u32*)dgp[i],                                             \
+            (u32*)&lds[BUFI][DOOFF / 2 + (i * 512 + wu64 * 64) * 8],        \
+            16, 0, 0);                                                      \
+        dgp[i] += dostep;                                                   \
+      }                                                                     \
+      qtst[0] = *reinterpret_cast<const uint4*>(qtp[0]);                    \
+      qtst[1] = *reinterpret_cast<const uint4*>(qtp[1]);                    \
+      qtp[0] += qstep;                                                      \
+      qtp[1] += qstep;                                                      \
+      if (tid < 64)                                                         \
+        *reinterpret_cast<float*>(&lds[BUFI][LSEOFF / 2 + tid * 2]) =       \
+            lbase[(QT) * 64 + tid] * 1.4426950408889634f;                   \
+      else if (tid < 128)                                                   \
+        *reinterpret_cast<float*>(                                          \
+            &lds[BUFI][DLTOFF / 2 + (tid - 64) * 2]) =                      \
+            dbase[(QT) * 64 + tid - 64];                                    \
+    }
+#define BK_WRITE_QT(BUF)                                                    \
+    {                                                                       \
+      const int rp2_ = (tid >> 4) * 2;                                      \
+      const int c8_ = (tid & 15) * 8;                                       \
+      union { uint4 u4; u16 h[8]; } va_, vb_;                               \
+      va_.u4 = qtst[0];                                                     \
+      vb_.u4 = qtst[1];                                                     \
+      _Pragma("unroll")                                                     \
+      for (int j = 0; j < 8; ++j) {                                         \
+        const u32 pair_ = (u32)va_.h[j] | ((u32)vb_.h[j] << 16);            \
+        *reinterpret_cast<u32*>(&(BUF)[QTOFF + vt_byte(c8_ + j, rp2_)]) =   \
+            pair_;                                                          \
+      }                                                                     \
+    }
+
+    BK_STAGE_IN(0, qt0)
+    BK_WRITE_QT(lds0)
+    __syncthreads();
+
+    int cur = 0;
+    for (int qt = qt0; qt < qtn; ++qt) {
+      const bool have_next = (qt + 1) < qtn;
+      if (have_next) BK_STAGE_IN(cur ^ 1, qt + 1)
+
+      char* qb_ = cur ? lds1 : lds0;
+      char* dob = qb_ + DOOFF;
+      char* qtb = qb_ + QTOFF;
+      const float* lse2t = reinterpret_cast<const float*>(qb_ + LSEOFF);
+      const float* dltt = reinterpret_cast<const float*>(qb_ + DLTOFF);
+      const bool active = qt * 64 + 63 >= kv0w;
+      const bool need_mask = qt * 64 < kv0w + 32;
+
+      union PF { bf16x8 v; unsigned u[4]; } pf[4];
+      if (active) {
+#pragma unroll
+        for (int sub = 0; sub < 2; ++sub) {
+          f32x16 s, dp;
+#pragma unroll
+          for (int r = 0; r < 16; ++r) { s[r] = 0.f; dp[r] = 0.f; }
+#pragma unroll
+          for (int kc = 0; kc < 8; ++kc) {
+            bf16x8 a = *reinterpret_cast<const bf16x8*>(
+                &qb_[k_byte(sub * 32 + low, kc * 16 + hi * 8)]);
+            s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, kf[kc].v, s,
+                                                        0, 0, 0);
+            bf16x8 a2 = *reinterpret_cast<const bf16x8*>(
+                &dob[k_byte(sub * 32 + low, kc * 16 + hi * 8)]);
+            dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a2, vf[kc].v, dp,
+                                                         0, 0, 0);
+          }
+#pragma unroll
+          for (int r = 0; r < 16; ++r) {
+            const int qr = (r & 3) + 8 * (r >> 2) + 4 * hi;
+            float x = s[r] * scale2 - lse2t[sub * 32 + qr];
+            if (need_mask && qt * 64 + sub * 32 + qr < kvrow) x = -1e30f;
+            const float p = exp2_raw(x);
+            s[r] = p * (dp[r] - dltt[sub * 32 + qr]) * scale;
+          }
+#pragma unroll
+          for (int cc = 0; cc < 2; ++cc) {
+            const int pb = cc * 8;
+            unsigned a0 = cvt_pk_bf16(s[pb + 0], s[pb + 1]);
+            unsigned b0 = cvt_pk_bf16(s[pb + 4], s[pb + 5]);
+            unsigned a1 = cvt_pk_bf16(s[pb + 2], s[pb + 3]);
+            unsigned b1 = cvt_pk_bf16(s[pb + 6], s[pb + 7]);
+            auto r02 = __builtin_amdgcn_permlane32_swap(a0, b0, false, false);
+            auto r13 = __builtin_amdgcn_permlane32_swap(a1, b1, false, false);
+            PF f;
+            f.u[0] = r02[0];
+            f.u[1] = r13[0];
+            f.u[2] = r02[1];
+            f.u[3] = r13[1];
+            pf[sub * 2 + cc] = f;
+          }
+        }
+      }
+      if (have_next) {
+        char* nb = cur ? lds0 : lds1;
+        BK_WRITE_QT(nb)
+      }
+      if (active) {
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+#pragma unroll
+          for (int ds = 0; ds < 4; ++ds) {
+            bf16x8 af = *reinterpret_cast<const bf16x8*>(
+                &qtb[vt_byte(ds * 32 + low, c * 16 + hi * 8)]);
+            dkacc[ds] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                af, pf[c].v, dkacc[ds], 0, 0, 0);
+          }
+        }
+      }
+      __syncthreads();
+      cur ^= 1;
+    }
+  }
+
+  u16* dkrow = dk + (((long)b * n_kv_heads + hkv) * S + kvrow) * ATTN_D;
+#pragma unroll
+  for (int ds = 0; ds < 4; ++ds) {
+#pragma unroll
+    for (int g = 0; g < 4; ++g) {
+      const int d0 = 8 * g + 4 * hi + 32 * ds;
+      const unsigned w0 = cvt_pk_bf16(dkacc[ds][4 * g + 0],
+                                      dkacc[ds][4 * g + 1]);
+      const unsigned w1 = cvt_pk_bf16(dkacc[ds][4 * g + 2],
+                                      dkacc[ds][4 * g + 3]);
+      uint2 wv;
+      wv.x = w0;
+      wv.y = w1;
+      *reinterpret_cast<uint2*>(dkrow + d0) = wv;
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// launchers
+// ---------------------------------------------------------------------------
+
+extern "C" int attn_bwd(void* stream, const void* q, const void* k,
+                        const void* v, const void* out, const void* dout,
+                        const void* lse, void* delta,
+                        void* dq, void* dk, void* dv,
+                        long q_sb, long q_sh, long q_ss,
+                        long k_sb, long k_sh, long k_ss,
+                        long v_sb, long v_sh, long v_ss,
+                        int batch, int n_heads, int n_kv_heads, int S,
+                        float scale) {
+  if (S <= 0 || S % 256 != 0 || n_heads <= 0 || batch <= 0 ||
+      n_kv_heads <= 0 || n_heads % n_kv_heads != 0)
+    return -1;
+  hipStream_t st = reinterpret_cast<hipStream_t>(stream);
+  const long n_rows = (long)batch * n_heads * S;
+  hipLaunchKernelGGL(attn_delta_kernel,
+                     dim3((unsigned)((n_rows + 15) / 16)), dim3(256), 0, st,
+                     (const u16*)dout, (const u16*)out, (float*)delta,
+                     n_rows);
+  {
+    dim3 grid(S / 256, n_heads, batch), block(512);
+    hipLaunchKernelGGL(attn_bwd_dq_kernel, grid, block, 0, st,
+                       (const u16*)q, (const u16*)k, (const u16*)v,
+                       (const u16*)dout, (const float*)lse,
+                       (const float*)delta, (u16*)dq,
+                       q_sb, q_sh, q_ss, k_sb, k_sh, k_ss,
+                       v_sb, v_sh, v_ss, n_heads, n_heads / n_kv_heads, S,
+                       scale);
+  }
+  {
+    dim3 grid(S / 256, n_kv_heads, batch), block(512);
+    hipLaunchKernelGGL(attn_bwd_dv_kernel, grid, block, 0, st,
+                       (const u16*)q, (const u16*)k, (const u16*)dout,
+                       (const float*)lse, (u16*)dv,
+                       q_sb, q_sh, q_ss, k_sb, k_sh, k_ss,
+                       n_heads, n_kv_heads, S, scale);
+    hipLaunchKernelGGL(attn_bwd_dk_kernel, grid, block, 0, st,
+                       (const u16*)q, (const u16*)k, (const u16*)v,
+                       (const u16*)dout, (const float*)lse,
+                       (const float*)delta, (u16*)dk,
+                       q_sb, q_sh, q_ss, k_sb, k_sh, k_ss,
+                       v_sb, v_sh, v_ss, n_heads, n_kv_heads, S, scale);
+  }
   return 0;
 }

@@ -80,6 +80,8 @@ def load(require: bool = True) -> Optional[ctypes.CDLL]:
                            l, l, l, l, l, l, l, l, l, i, i, i, f], i)
     _sig(lib.attn_fwd_nw8, [vp, vp, vp, vp, vp, vp,
                             l, l, l, l, l, l, l, l, l, i, i, i, i, f], i)
+    _sig(lib.attn_bwd, [vp, vp, vp, vp, vp, vp, vp, vp, vp, vp, vp,
+                        l, l, l, l, l, l, l, l, l, i, i, i, i, f], i)
     _sig(lib.mfma_probe32, [vp, vp, vp, vp])
     assert lib.hipops_arch_check() == 950
     _lib = lib
