@@ -159,6 +159,9 @@ class _Handler(BaseHTTPRequestHandler):
         try:
             if (m := JOB_RE.match(u.path)):
                 ns, name, _status = m.groups()
+                if _status:
+                    return self._send(
+                        200, self.fake.update_job_status(ns, name, body))
                 return self._send(200, self.fake.update_job(ns, name, body))
             if "/events" in u.path and u.path.startswith("/api/v1/"):
                 ns = u.path.split("/")[4]

@@ -63,8 +63,15 @@ def test_job_status_roundtrip_and_conflict(cluster):
     fake, api = cluster
     fake.create_job(NS, job_manifest())
     job = api.get_job(NS, "httpjob")
+    # the CRD declares a status subresource: a main-resource PUT silently
+    # DROPS .status (real API-server semantics, emulated by the fake)...
     job["status"] = {"phase": "Running"}
-    api.update_job(NS, "httpjob", job)
+    updated = api.update_job(NS, "httpjob", job)
+    assert api.get_job(NS, "httpjob").get("status", {}).get("phase") \
+        != "Running"
+    # ...and only the /status endpoint persists it
+    updated["status"] = {"phase": "Running"}
+    api.update_job_status(NS, "httpjob", updated)
     assert api.get_job(NS, "httpjob")["status"]["phase"] == "Running"
     # stale resourceVersion conflicts
     with pytest.raises(ApiError) as ei:

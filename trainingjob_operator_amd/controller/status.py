@@ -176,6 +176,10 @@ class StatusEngine:
         # status write — stamped here, not per-sync, to avoid API churn
         from ..utils.k8stime import format_time
         job.status.last_reconcile_time = format_time()
+        # the CRD declares a status subresource, so .status MUST go through
+        # update_job_status (a main-resource PUT silently drops it on a real
+        # API server); annotations (the deferred-termination marker) go
+        # through a main-resource update only when they changed
         for attempt in range(retries):
             try:
                 current = self.api.get_job(job.namespace, job.name)
@@ -183,13 +187,16 @@ class StatusEngine:
                 if e.not_found:
                     return
                 raise
-            current["status"] = job.status.to_dict()
-            current.setdefault("metadata", {})["annotations"] = {
-                **(current.get("metadata", {}).get("annotations") or {}),
-                **job.annotations,
-            }
             try:
-                self.api.update_job(job.namespace, job.name, current)
+                cur_ann = current.get("metadata", {}).get("annotations") or {}
+                want_ann = {**cur_ann, **job.annotations}
+                if want_ann != cur_ann:
+                    current.setdefault("metadata", {})["annotations"] = \
+                        want_ann
+                    current = self.api.update_job(job.namespace, job.name,
+                                                  current)
+                current["status"] = job.status.to_dict()
+                self.api.update_job_status(job.namespace, job.name, current)
                 return
             except ApiError as e:
                 if e.conflict and attempt < retries - 1:

@@ -202,6 +202,9 @@ class FakeKubeApi(KubeApi):
                     if namespace is None or ns == namespace]
 
     def update_job(self, namespace, name, job):
+        # real API servers IGNORE .status on a main-resource PUT when the
+        # CRD declares a status subresource (ours does) -- emulate that so
+        # tests catch controllers writing status through the wrong door
         with self._lock:
             key = (namespace, name)
             if key not in self.jobs:
@@ -211,6 +214,10 @@ class FakeKubeApi(KubeApi):
             if new_rv and new_rv != cur["metadata"]["resourceVersion"]:
                 raise ApiError(409, "resourceVersion conflict")
             job = copy.deepcopy(job)
+            if "status" in cur:
+                job["status"] = copy.deepcopy(cur["status"])
+            else:
+                job.pop("status", None)
             meta(job)["resourceVersion"] = str(next(self._rv))
             meta(job)["namespace"] = namespace
             meta(job).setdefault("uid", cur["metadata"]["uid"])
@@ -220,7 +227,23 @@ class FakeKubeApi(KubeApi):
             return copy.deepcopy(job)
 
     def update_job_status(self, namespace, name, job):
-        return self.update_job(namespace, name, job)
+        # /status subresource: applies ONLY .status (spec/metadata of the
+        # stored object are preserved), like a real API server
+        with self._lock:
+            key = (namespace, name)
+            if key not in self.jobs:
+                raise ApiError(404, f"job {name}")
+            cur = self.jobs[key]
+            new_rv = job.get("metadata", {}).get("resourceVersion")
+            if new_rv and new_rv != cur["metadata"]["resourceVersion"]:
+                raise ApiError(409, "resourceVersion conflict")
+            new = copy.deepcopy(cur)
+            new["status"] = copy.deepcopy(job.get("status") or {})
+            meta(new)["resourceVersion"] = str(next(self._rv))
+            self.jobs[key] = new
+            self._record("update_status", "job", namespace, name)
+            self._emit("MODIFIED", "job", new)
+            return copy.deepcopy(new)
 
     def delete_job(self, namespace, name):
         with self._lock:
