@@ -20,6 +20,7 @@ from ..kube import objects as ko
 from ..kube.client import ApiError, KubeApi
 from ..kube.workqueue import RateLimitedQueue
 from .events import EventRecorder
+from .expectations import ControllerExpectations
 from .gc import GarbageCollector
 from .options import OperatorOptions
 from .pods import PodReconciler, node_ready_map
@@ -37,8 +38,11 @@ class TrainingJobController:
         self.options = options or OperatorOptions()
         self.queue = RateLimitedQueue()
         self.recorder = EventRecorder(api)
-        self.pod_reconciler = PodReconciler(api, self.options, self.recorder)
-        self.service_reconciler = ServiceReconciler(api, self.recorder)
+        self.expectations = ControllerExpectations()
+        self.pod_reconciler = PodReconciler(api, self.options, self.recorder,
+                                            self.expectations)
+        self.service_reconciler = ServiceReconciler(api, self.recorder,
+                                                    self.expectations)
         self.status_engine = StatusEngine(api, self.recorder, self._enqueue)
         self.gc = GarbageCollector(api, self.options.namespace)
         self.metrics = metrics
@@ -110,6 +114,19 @@ class TrainingJobController:
         services = self._claim(
             self.api.list_services(job.namespace, selector), job)
 
+        # in-flight-operation gate (reference: controller.go:295,390-404):
+        # the fresh list first settles any expectation it can prove, then
+        # pending creates/deletes suppress this sync (a delayed requeue
+        # keeps progress even if the watch event is lost)
+        # names are kind-prefixed: per-index headless services share the
+        # pod names, so unprefixed sets would mask pod deletions
+        present = {f"pod/{ko.name_of(o)}" for o in pods} | \
+            {f"svc/{ko.name_of(o)}" for o in services}
+        self.expectations.observe_list(job.key, present)
+        if not self.expectations.satisfied(job.key, now):
+            self.queue.add_after(job.key, 1.0)
+            return
+
         ending_phases: Dict[str, str] = {}
         message = ""
         if not job.status.restart_replica_name:
@@ -155,16 +172,34 @@ class TrainingJobController:
                 dict(job.annotations) != original_annotations:
             self.status_engine.persist(job)
 
-    @staticmethod
-    def _claim(objs: List[dict], job: AITrainingJob) -> List[dict]:
-        """Ownership filter: selector-matched objects whose controllerRef is
-        this job (simplified adoption vs the reference's RefManager,
-        pod.go:134-150 — we never adopt label-matching strays, we only
-        manage what we created)."""
+    def _claim(self, objs: List[dict], job: AITrainingJob) -> List[dict]:
+        """Ownership filter with adoption (reference PodControllerRefManager,
+        pod.go:134-150): selector-matched objects with our controllerRef are
+        managed; label-matching STRAYS with no controller at all are adopted
+        by patching our ownerReference onto them (pods only — services are
+        cheap to recreate); objects owned by another controller are left
+        alone. Release-on-label-mismatch cannot trigger here because the
+        list is already selector-filtered (documented divergence)."""
         out = []
         for o in objs:
             ref = ko.controller_ref(o)
             if ref is None:
+                if ko.is_deleting(o) or not job.uid:
+                    continue
+                if o.get("kind") != "Pod" and "spec" not in o:
+                    continue
+                is_pod = "containers" in (o.get("spec") or {})
+                if not is_pod:
+                    continue
+                try:
+                    patched = self.api.patch_pod_metadata(
+                        ko.namespace_of(o), ko.name_of(o),
+                        {"ownerReferences": [ko.gen_owner_reference(job)]})
+                    self.recorder.event(job, "Normal", "AdoptedPod",
+                                        f"adopted stray pod {ko.name_of(o)}")
+                    out.append(patched)
+                except ApiError:
+                    continue
                 continue
             if ref.get("kind") == C.CRD_KIND and ref.get("name") == job.name:
                 if not job.uid or ref.get("uid") in (None, job.uid):
@@ -184,11 +219,20 @@ class TrainingJobController:
         self.ensure_crd()
         threads: List[threading.Thread] = []
 
-        def watcher(watch_fn, to_key):
+        def watcher(watch_fn, to_key, observe=False):
             for evt_type, obj in watch_fn(self.options.namespace or None,
                                           stop):
                 key = to_key(obj)
                 if key:
+                    if observe:
+                        kind = "svc" if obj.get("kind") == "Service" else \
+                            "pod"
+                        if evt_type == "ADDED":
+                            self.expectations.creation_observed(
+                                key, f"{kind}/{ko.name_of(obj)}")
+                        elif evt_type == "DELETED":
+                            self.expectations.deletion_observed(
+                                key, f"{kind}/{ko.name_of(obj)}")
                     self.queue.add(key)
 
         def job_key(obj):
@@ -201,10 +245,10 @@ class TrainingJobController:
                 return f"{ko.namespace_of(obj)}/{ref.get('name')}"
             return None
 
-        for fn, keyer in ((self.api.watch_jobs, job_key),
-                          (self.api.watch_pods, owned_key),
-                          (self.api.watch_services, owned_key)):
-            t = threading.Thread(target=watcher, args=(fn, keyer),
+        for fn, keyer, obs in ((self.api.watch_jobs, job_key, False),
+                               (self.api.watch_pods, owned_key, True),
+                               (self.api.watch_services, owned_key, True)):
+            t = threading.Thread(target=watcher, args=(fn, keyer, obs),
                                  daemon=True)
             t.start()
             threads.append(t)

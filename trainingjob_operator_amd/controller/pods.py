@@ -91,7 +91,9 @@ def rendezvous_epoch(job: AITrainingJob) -> int:
 
 
 class PodReconciler:
-    def __init__(self, api: KubeApi, options, recorder=None):
+    def __init__(self, api: KubeApi, options, recorder=None,
+                 expectations=None):
+        self.expectations = expectations
         self.api = api
         self.options = options
         self.recorder = recorder
@@ -138,9 +140,24 @@ class PodReconciler:
 
         pod = {"apiVersion": "v1", "kind": "Pod", "metadata": m,
                "spec": pspec}
+        # expectation raised before the call, settled on ANY definitive
+        # answer (success ack or clean ApiError) -- our reads are live
+        # LISTs, not an informer cache, so the ack is authoritative; the
+        # guard protects exactly the ambiguous window (a transport error
+        # where the create may or may not have landed), unlike the
+        # reference whose cache lag needs the watch event
+        # (controller.go:390-404, pod.go:489-494)
+        if self.expectations:
+            self.expectations.expect_creation(job.key, f"pod/{m['name']}")
         try:
             self.api.create_pod(job.namespace, pod)
+            if self.expectations:
+                self.expectations.creation_observed(job.key,
+                                                    f"pod/{m['name']}")
         except ApiError as e:
+            if self.expectations:   # clean failure: nothing in flight
+                self.expectations.creation_observed(job.key,
+                                                    f"pod/{m['name']}")
             if not e.already_exists:
                 raise
         if self.recorder:
@@ -150,10 +167,19 @@ class PodReconciler:
     def _delete_pod(self, job: AITrainingJob, pod: dict,
                     force: bool = False) -> None:
         grace = 0 if force else None
+        if self.expectations:
+            self.expectations.expect_deletion(job.key,
+                                              f"pod/{ko.name_of(pod)}")
         try:
             self.api.delete_pod(ko.namespace_of(pod), ko.name_of(pod),
                                 grace_period=grace)
+            if self.expectations:
+                self.expectations.deletion_observed(
+                    job.key, f"pod/{ko.name_of(pod)}")
         except ApiError as e:
+            if self.expectations:
+                self.expectations.deletion_observed(
+                    job.key, f"pod/{ko.name_of(pod)}")
             if not e.not_found:
                 raise
         if self.recorder:

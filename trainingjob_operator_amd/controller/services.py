@@ -35,9 +35,10 @@ def service_slices(services: List[dict], replicas: int) -> List[List[dict]]:
 
 
 class ServiceReconciler:
-    def __init__(self, api: KubeApi, recorder=None):
+    def __init__(self, api: KubeApi, recorder=None, expectations=None):
         self.api = api
         self.recorder = recorder
+        self.expectations = expectations
 
     def reconcile(self, job: AITrainingJob, all_services: List[dict],
                   rtype: str) -> None:
@@ -77,7 +78,19 @@ class ServiceReconciler:
             },
         }
         try:
-            self.api.create_service(job.namespace, svc)
+            if self.expectations:
+                self.expectations.expect_creation(
+                    job.key, f"svc/{svc['metadata']['name']}")
+            try:
+                self.api.create_service(job.namespace, svc)
+                if self.expectations:
+                    self.expectations.creation_observed(
+                        job.key, f"svc/{svc['metadata']['name']}")
+            except ApiError:
+                if self.expectations:
+                    self.expectations.creation_observed(
+                        job.key, f"svc/{svc['metadata']['name']}")
+                raise
         except ApiError as e:
             if not e.already_exists:
                 raise

@@ -51,6 +51,8 @@ class KubeApi:
                   selector: Optional[Dict[str, str]] = None) -> List[dict]: ...
     def delete_pod(self, namespace: str, name: str,
                    grace_period: Optional[int] = None) -> None: ...
+    def patch_pod_metadata(self, namespace: str, name: str,
+                           metadata_patch: dict) -> dict: ...
     def read_pod_log(self, namespace: str, name: str,
                      tail_lines: Optional[int] = None) -> str: ...
 
@@ -175,6 +177,17 @@ class RealKubeApi(KubeApi):
             body["gracePeriodSeconds"] = grace_period
         self._req("DELETE", self._ns_path("pods", namespace) + "/" + name,
                   body or None)
+
+    def patch_pod_metadata(self, namespace, name, metadata_patch):
+        url = (self.base_url + self._ns_path("pods", namespace)
+               + "/" + name)
+        r = self.session.patch(
+            url, json={"metadata": metadata_patch},
+            headers={"Content-Type": "application/merge-patch+json"},
+            timeout=30.0)
+        if r.status_code >= 400:
+            raise ApiError(r.status_code, r.text[:500])
+        return r.json()
 
     def read_pod_log(self, namespace, name, tail_lines=None):
         params = {}

@@ -92,6 +92,22 @@ class FakeKubeApi(KubeApi):
                 out.append(copy.deepcopy(pod))
             return out
 
+    def patch_pod_metadata(self, namespace, name, metadata_patch):
+        # merge-patch on metadata only (adoption patches ownerReferences)
+        with self._lock:
+            key = (namespace, name)
+            if key not in self.pods:
+                raise ApiError(404, f"pod {name}")
+            pod = self.pods[key]
+            for k2, v2 in metadata_patch.items():
+                if isinstance(v2, dict):
+                    pod["metadata"].setdefault(k2, {}).update(v2)
+                else:
+                    pod["metadata"][k2] = copy.deepcopy(v2)
+            self._record("patch", "pod", namespace, name)
+            self._emit("MODIFIED", "pod", pod)
+            return copy.deepcopy(pod)
+
     def delete_pod(self, namespace, name, grace_period=None):
         with self._lock:
             key = (namespace, name)
