@@ -55,6 +55,10 @@ ENV_WORLD_SIZE = "WORLD_SIZE"
 ENV_RANK = "RANK"
 ENV_LOCAL_RANK = "LOCAL_RANK"
 ENV_REND_EPOCH = "TRAININGJOB_RENDEZVOUS_EPOCH"  # bumped on every elastic resize
+ENV_LOCAL_WORLD_SIZE = "LOCAL_WORLD_SIZE"
+ENV_NPROC_PER_NODE = "NPROC_PER_NODE"       # torchrun convention
+ENV_NODE_RANK = "NODE_RANK"
+GPU_RESOURCE = "amd.com/gpu"                # ROCm k8s device plugin
 ENV_MIN_REPLICAS = "TRAININGJOB_MIN_REPLICAS"
 ENV_MAX_REPLICAS = "TRAININGJOB_MAX_REPLICAS"
 DEFAULT_MASTER_PORT = 23456

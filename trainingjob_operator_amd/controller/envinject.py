@@ -37,6 +37,25 @@ def ports_from_spec(spec: ReplicaSpec) -> List[int]:
     return sorted(set(ports))
 
 
+def gpus_per_pod(spec: ReplicaSpec) -> int:
+    """GPUs one pod of this replica requests (amd.com/gpu limits summed over
+    aitj-* containers; ROCm device-plugin contract). 0 -> CPU-only pod."""
+    total = 0
+    for c in (spec.template.get("spec") or {}).get("containers") or []:
+        if not (c.get("name") or "").startswith(C.CONTAINER_PREFIX):
+            continue
+        res = c.get("resources") or {}
+        for kind in ("limits", "requests"):
+            v = (res.get(kind) or {}).get(C.GPU_RESOURCE)
+            if v is not None:
+                try:
+                    total += int(v)
+                except (TypeError, ValueError):
+                    pass
+                break
+    return total
+
+
 def ports_from_container(container: dict) -> List[int]:
     """A single container's own aitj-* ports (reference: pod.go:644-648)."""
     out = []
@@ -80,18 +99,31 @@ def render_env(job: AITrainingJob, rtype: str, index: int,
     ]
 
     # --- MI355X RCCL rendezvous extension (role-local world) ---
+    # Multi-GPU pods (G = amd.com/gpu per pod > 1) host G ranks launched
+    # in-pod by torchrun: the operator injects the NODE-level contract
+    # (NODE_RANK / NPROC_PER_NODE / base RANK = index*G, WORLD_SIZE =
+    # replicas*G) and leaves per-process RANK/LOCAL_RANK assignment to the
+    # in-pod launcher. Single-GPU (or CPU) pods keep the flat per-pod
+    # contract with LOCAL_RANK=0 (one process per pod).
     spec = job.spec.replica_specs[rtype]
     ports = ports_from_spec(spec)
     master_port = ports[0] if ports else C.DEFAULT_MASTER_PORT
+    gpus = gpus_per_pod(spec)
+    per_pod = max(gpus, 1)
     env += [
         {"name": C.ENV_MASTER_ADDR,
          "value": f"{gen_general_name(job.name, rtype, 0)}.{ns}"},
         {"name": C.ENV_MASTER_PORT, "value": str(master_port)},
-        {"name": C.ENV_WORLD_SIZE, "value": str(spec.replicas or 0)},
-        {"name": C.ENV_RANK, "value": str(index)},
-        {"name": C.ENV_LOCAL_RANK, "value": "0"},
+        {"name": C.ENV_WORLD_SIZE,
+         "value": str((spec.replicas or 0) * per_pod)},
+        {"name": C.ENV_RANK, "value": str(index * per_pod)},
+        {"name": C.ENV_NODE_RANK, "value": str(index)},
+        {"name": C.ENV_NPROC_PER_NODE, "value": str(per_pod)},
+        {"name": C.ENV_LOCAL_WORLD_SIZE, "value": str(per_pod)},
         {"name": C.ENV_REND_EPOCH, "value": str(epoch)},
     ]
+    if per_pod == 1:
+        env.append({"name": C.ENV_LOCAL_RANK, "value": "0"})
     if spec.min_replicas is not None:
         env.append({"name": C.ENV_MIN_REPLICAS,
                     "value": str(spec.min_replicas)})

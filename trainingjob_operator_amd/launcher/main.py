@@ -111,7 +111,34 @@ def _saves_ckpt(args, ctx, trainer) -> bool:
     return ctx.is_rank0
 
 
+def _maybe_reexec_torchrun(argv) -> bool:
+    """Multi-GPU pod bootstrap: the operator injects NPROC_PER_NODE > 1
+    (amd.com/gpu > 1 on the pod) plus NODE_RANK and a node-level
+    WORLD_SIZE; the per-process RANK/LOCAL_RANK come from torchrun, so
+    re-exec this launcher under torch.distributed.run once. Single-GPU
+    pods (NPROC_PER_NODE absent or 1, LOCAL_RANK pre-set) skip this."""
+    nproc = int(os.environ.get("NPROC_PER_NODE", "1") or 1)
+    if nproc <= 1 or "LOCAL_RANK" in os.environ:
+        return False
+    nnodes = os.environ.get("TRAININGJOB_NNODES") or str(
+        max(int(os.environ.get("WORLD_SIZE", nproc)) // nproc, 1))
+    cmd = [
+        sys.executable, "-m", "torch.distributed.run",
+        f"--nnodes={nnodes}",
+        f"--node-rank={os.environ.get('NODE_RANK', '0')}",
+        f"--nproc-per-node={nproc}",
+        f"--master-addr={os.environ.get('MASTER_ADDR', '127.0.0.1')}",
+        f"--master-port={os.environ.get('MASTER_PORT', '23456')}",
+        "-m", "trainingjob_operator_amd.launcher.main",
+    ] + list(argv if argv is not None else sys.argv[1:])
+    log.info("multi-GPU pod: re-exec under torchrun (%d procs)", nproc)
+    os.execv(sys.executable, cmd)
+    return True  # unreachable
+
+
 def main(argv=None) -> int:
+    if _maybe_reexec_torchrun(argv):
+        return 0
     ap = argparse.ArgumentParser()
     ap.add_argument("--model", default="llama3-8b")
     ap.add_argument("--steps", type=int, default=1000,
