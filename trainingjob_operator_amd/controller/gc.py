@@ -38,7 +38,9 @@ class GarbageCollector:
                 if ts is not None and now > ts:
                     self._force_delete(ns, name)
                     deleted += 1
-                continue
+                    continue
+                # deletion pending but not expired: fall through so the
+                # orphan check below can still apply (gc:57-73)
             ref = ko.controller_ref(pod)
             if ref is None or ref.get("kind") != C.CRD_KIND:
                 continue
@@ -46,7 +48,9 @@ class GarbageCollector:
                 continue
             node = ko.pod_node(pod)
             if node and node not in node_ready and dts is not None:
-                continue  # dead node, deletion pending (gc:91-106)
+                # orphan mid-deletion on a dead node: wait for the node
+                # verdict before force-finishing it (gc:64-68, checkNode)
+                continue
             self._force_delete(ns, name)
             deleted += 1
         if deleted:

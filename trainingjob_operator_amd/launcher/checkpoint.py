@@ -23,6 +23,16 @@ import torch
 CKPT_PREFIX = "ckpt_step"
 
 
+def _to_host(t: "torch.Tensor") -> "torch.Tensor":
+    """Pinned-staging D2H (~2x a pageable .cpu() copy); the caller
+    synchronizes once after issuing all copies."""
+    if not t.is_cuda:
+        return t.detach().clone()
+    out = torch.empty_like(t, device="cpu", pin_memory=True)
+    out.copy_(t.detach(), non_blocking=True)
+    return out
+
+
 class Checkpointer:
     def __init__(self, directory: str, keep: int = 2):
         self.directory = directory
@@ -36,15 +46,19 @@ class Checkpointer:
         self.wait()
         step = trainer.opt.step_count
         path = os.path.join(self.directory, f"{CKPT_PREFIX}{step:08d}.pt")
-        # device->host snapshot (cheap vs training step; pinned staging)
+        # The device->host snapshot is SYNCHRONOUS on the training thread
+        # (pinned staging + non_blocking issue, then one synchronize): an
+        # overlapped copy would race the next optimizer step mutating
+        # flat_param/p32/m/v in place, and a device-side staging clone of
+        # the fp32 optimizer state is too large to double-buffer at 8B+
+        # scale. Only the disk write runs in the background.
         state = {
             "step": step,
-            "flat_param": trainer.store.flat_param.detach().to(
-                "cpu", non_blocking=False),
+            "flat_param": _to_host(trainer.store.flat_param),
             "opt": {
-                "p32": trainer.opt.p32.detach().cpu(),
-                "m": trainer.opt.m.detach().cpu(),
-                "v": trainer.opt.v.detach().cpu(),
+                "p32": _to_host(trainer.opt.p32),
+                "m": _to_host(trainer.opt.m),
+                "v": _to_host(trainer.opt.v),
                 "step": trainer.opt.step_count,
             },
             "train_config": trainer.cfg.__dict__.copy(),
@@ -55,6 +69,9 @@ class Checkpointer:
             "shapes": dict(getattr(trainer.store, "shapes", {})),
             "time": time.time(),
         }
+        if trainer.store.flat_param.is_cuda:
+            import torch as _t
+            _t.cuda.synchronize()
 
         def write():
             tmp = path + ".tmp"
