@@ -182,3 +182,17 @@ def test_operator_elastic_resize_end_to_end(tmp_path):
     tc.sync_once(f"{NS}/el")
     job = AITrainingJob.from_dict(api.get_job(NS, "el"))
     assert job.status.phase == Phase.SUCCEEDED
+
+
+@pytest.mark.timeout(600)
+def test_operator_path_restart_to_rejoin(tmp_path):
+    """BASELINE config 4 through the CONTROLLER: SIGKILL a worker, the
+    kubelet stand-in reports Failed(137), the operator runs the two-sync
+    restart dance and recreates the world, and the rejoined workers make
+    checkpointed progress. The measured wall time is the controller-path
+    restart-to-rejoin metric (reported in BASELINE.md)."""
+    from trainingjob_operator_amd.launcher.localrun import restart_benchmark
+    times = restart_benchmark(model="llama-tiny", replicas=2, trials=1,
+                              seq_len=32, ckpt_root=str(tmp_path))
+    assert len(times) == 1 and times[0] > 0
+    print(f"controller-path restart-to-rejoin: {times[0]:.2f}s")

@@ -58,8 +58,13 @@ def from_env() -> DistContext:
 
 def init_process_group(ctx: DistContext,
                        timeout_s: float = 600.0) -> DistContext:
-    """init torch.distributed (backend "nccl" IS RCCL on ROCm)."""
-    if ctx.world_size <= 1:
+    """init torch.distributed (backend "nccl" IS RCCL on ROCm).
+
+    A 1-rank world launched under torchrun (RANK present in the env) still
+    initializes the process group: single-rank RCCL communicator init +
+    barriers are the smallest on-silicon proof of the collective path, and
+    it keeps the torchrun-launched code path identical at every N."""
+    if ctx.world_size <= 1 and os.environ.get("RANK") is None:
         return ctx
     if not dist.is_initialized():
         os.environ.setdefault("MASTER_ADDR", ctx.master_addr)
