@@ -172,7 +172,7 @@ def _latest_step(ckdir: str) -> int:
 def restart_benchmark(model: str = "llama-tiny", replicas: int = 2,
                       trials: int = 3, seq_len: int = 32,
                       ckpt_root: Optional[str] = None,
-                      steps: int = 10_000,
+                      steps: int = 10_000, ckpt_every: int = 1,
                       sync_period_s: float = 0.05,
                       timeout_s: float = 300.0) -> List[float]:
     """Measured: SIGKILL(worker 0) -> [operator restart dance + respawn +
@@ -224,14 +224,15 @@ def restart_benchmark(model: str = "llama-tiny", replicas: int = 2,
         kubelet = LocalKubelet(api, ns, [
             "--model", model, "--steps", str(steps),
             "--seq-len", str(seq_len), "--grad-accum", "1",
-            "--micro-batch", "1", "--ckpt-every", "1", "--log-every", "1",
+            "--micro-batch", "1", "--ckpt-every", str(ckpt_every),
+            "--log-every", "1",
             "--ckpt-dir", ckdir,
         ], free_port())
         kubelet.start()
         try:
             # wait for steady-state progress
             deadline = time.monotonic() + timeout_s
-            while _latest_step(ckdir) < 3:
+            while _latest_step(ckdir) < max(3, ckpt_every):
                 if time.monotonic() > deadline:
                     raise TimeoutError("no initial progress")
                 time.sleep(0.02)
@@ -266,9 +267,11 @@ if __name__ == "__main__":
     ap.add_argument("--trials", type=int, default=3)
     ap.add_argument("--seq-len", type=int, default=32)
     ap.add_argument("--replicas", type=int, default=2)
+    ap.add_argument("--ckpt-every", type=int, default=1)
     a = ap.parse_args()
     ts = restart_benchmark(model=a.model, trials=a.trials,
-                           seq_len=a.seq_len, replicas=a.replicas)
+                           seq_len=a.seq_len, replicas=a.replicas,
+                           ckpt_every=a.ckpt_every)
     ts_s = sorted(ts)
     print({"trials": [round(t, 3) for t in ts],
            "p50_s": round(ts_s[len(ts_s) // 2], 3)})
