@@ -53,6 +53,7 @@ class LocalKubelet:
         # (the supervisor's rejoin marker: steps are collective, so rank 0
         # stepping implies every rank rejoined)
         self.first_step: Dict[str, Optional[float]] = {}
+        self.tail: Dict[str, list] = {}      # uid -> last output lines
         self._stop = threading.Event()
         self._thread: Optional[threading.Thread] = None
         self._lock = threading.Lock()
@@ -106,8 +107,12 @@ class LocalKubelet:
         if is_rank0:
             uid = pod["metadata"]["uid"]
 
+            tl = self.tail.setdefault(uid, [])
+
             def reader():
                 for line in proc.stdout:
+                    tl.append(line.rstrip())
+                    del tl[:-60]
                     if " step " in line and " loss " in line and \
                             self.first_step.get(uid) is None:
                         self.first_step[uid] = time.monotonic()
@@ -234,6 +239,9 @@ def restart_benchmark(model: str = "llama-tiny", replicas: int = 2,
             deadline = time.monotonic() + timeout_s
             while _latest_step(ckdir) < max(3, ckpt_every):
                 if time.monotonic() > deadline:
+                    for uid, lines in kubelet.tail.items():
+                        print(f"--- rank0 tail ({kubelet.uid_name.get(uid)})")
+                        print("\n".join(lines[-40:]))
                     raise TimeoutError("no initial progress")
                 time.sleep(0.02)
             rank0_uids = {u for u, n in kubelet.uid_name.items()
