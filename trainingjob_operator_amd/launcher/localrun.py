@@ -265,6 +265,8 @@ def restart_benchmark(model: str = "llama-tiny", replicas: int = 2,
             stop.set()
             ct.join(timeout=10)
             kubelet.stop()
+            import shutil
+            shutil.rmtree(ckdir, ignore_errors=True)   # 1B ckpts are ~17 GB
     return times
 
 
