@@ -230,3 +230,49 @@ def test_launchers_reject_unsupported_shapes():
     assert lib.rope(None, None, None, None, 4, 8, 16, 24, 1.0) != 0
     assert lib.attn_fwd(None, None, None, None, None, None,
                         0, 0, 0, 0, 0, 0, 0, 0, 0, 1, 4, 100, 1.0) != 0
+
+
+def test_rows_gather_scatter_and_moe_combine():
+    """MoE dispatch kernels vs torch indexing (fwd + autograd bwd)."""
+    from trainingjob_operator_amd.ops import gather_rows, moe_combine
+    torch.manual_seed(11)
+    N, H, T, K = 64, 256, 32, 2
+    x = _mk((N, H)).requires_grad_()
+    idx = torch.randint(0, N, (48,), device=DEV)   # duplicates
+    out = gather_rows(x, idx, bijective=False)
+    dy = _mk((48, H))
+    (out.float() * dy.float()).sum().backward()
+    xa = x.detach().float().cpu().requires_grad_()
+    (xa[idx.cpu()] * dy.float().cpu()).sum().backward()
+    assert torch.allclose(out.float().cpu(), xa.detach()[idx.cpu()],
+                          atol=1e-2)
+    assert torch.allclose(x.grad.float().cpu(), xa.grad, atol=2e-2,
+                          rtol=2e-2)
+
+    # bijective path
+    x2 = _mk((N, H)).requires_grad_()
+    perm = torch.randperm(N, device=DEV)
+    out2 = gather_rows(x2, perm, bijective=True)
+    dy2 = _mk((N, H))
+    (out2.float() * dy2.float()).sum().backward()
+    assert torch.allclose(x2.grad.float().cpu()[perm.cpu()],
+                          dy2.float().cpu(), atol=1e-2)
+
+    # fused combine
+    src = _mk((T * K, H)).requires_grad_()
+    inv = torch.randperm(T * K, device=DEV)
+    gates = torch.rand(T, K, device=DEV, dtype=torch.float32) \
+        .requires_grad_()
+    y = moe_combine(src, inv, gates)
+    gy = _mk((T, H))
+    (y.float() * gy.float()).sum().backward()
+
+    sa = src.detach().float().cpu().requires_grad_()
+    ga = gates.detach().cpu().requires_grad_()
+    pair = sa[inv.cpu()] * ga.reshape(-1, 1)
+    ya = pair.reshape(T, K, H).sum(1)
+    (ya * gy.float().cpu()).sum().backward()
+    assert torch.allclose(y.float().cpu(), ya.detach(), atol=2e-2, rtol=1e-2)
+    assert torch.allclose(src.grad.float().cpu(), sa.grad, atol=2e-2,
+                          rtol=2e-2)
+    assert torch.allclose(gates.grad.cpu(), ga.grad, atol=2e-1, rtol=2e-2)

@@ -9,7 +9,9 @@ from .functional import (
     apply_rope,
     fused_cross_entropy,
     fused_rmsnorm,
+    gather_rows,
     make_inv_freq,
+    moe_combine,
     swiglu,
     swiglu_packed,
 )
@@ -17,6 +19,8 @@ from .native import HipOpsUnavailable, available, build_ops, load
 
 __all__ = [
     "apply_rope",
+    "gather_rows",
+    "moe_combine",
     "fused_cross_entropy",
     "fused_rmsnorm",
     "make_inv_freq",

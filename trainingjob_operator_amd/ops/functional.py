@@ -288,3 +288,113 @@ def fused_cross_entropy(logits: torch.Tensor, targets: torch.Tensor,
     """Per-token CE loss vector (fp32, [T]); rows with ignore_index get 0.
     Mean it over the valid-token count for the training loss."""
     return _FusedCE.apply(logits, targets, ignore_index)
+
+
+# ---------------------------------------------------------------------------
+# MoE token-dispatch row ops (HIP on GPU, torch indexing on CPU)
+# ---------------------------------------------------------------------------
+
+class _GatherRows(torch.autograd.Function):
+    """out[i] = x[idx[i]] over bf16 rows. bijective=True promises idx is a
+    permutation (backward is a conflict-free scatter); False accumulates
+    duplicate rows through an fp32 buffer."""
+
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, idx: torch.Tensor, bijective: bool):
+        ctx.bijective = bijective
+        ctx.n_in = x.shape[0]
+        ctx.save_for_backward(idx)
+        if x.is_cuda and x.dtype == torch.bfloat16 and \
+                x.shape[-1] % 8 == 0:
+            lib = _hip()
+            x = x.contiguous()
+            idx = idx.contiguous()
+            out = torch.empty(idx.shape[0], x.shape[1], dtype=x.dtype,
+                              device=x.device)
+            rc = lib.rows_gather(native.stream_ptr(), _ptr(x), _ptr(idx),
+                                 _ptr(out), idx.shape[0], x.shape[1])
+            native.check_rc(rc, "rows_gather", f"H={x.shape[1]}")
+            return out
+        return x[idx]
+
+    @staticmethod
+    def backward(ctx, dout: torch.Tensor):
+        (idx,) = ctx.saved_tensors
+        H = dout.shape[-1]
+        if dout.is_cuda and dout.dtype == torch.bfloat16 and H % 8 == 0:
+            lib = _hip()
+            dout = dout.contiguous()
+            if ctx.bijective:
+                dx = torch.empty(ctx.n_in, H, dtype=dout.dtype,
+                                 device=dout.device)
+                rc = lib.rows_scatter(native.stream_ptr(), _ptr(dout),
+                                      _ptr(idx), _ptr(dx), idx.shape[0], H)
+                native.check_rc(rc, "rows_scatter", f"H={H}")
+            else:
+                acc = torch.zeros(ctx.n_in, H, dtype=torch.float32,
+                                  device=dout.device)
+                rc = lib.rows_scatter_add_f32(native.stream_ptr(),
+                                              _ptr(dout), _ptr(idx),
+                                              _ptr(acc), idx.shape[0], H)
+                native.check_rc(rc, "rows_scatter_add_f32", f"H={H}")
+                dx = acc.to(dout.dtype)
+            return dx, None, None
+        dx = torch.zeros(ctx.n_in, H, dtype=dout.dtype, device=dout.device)
+        dx.index_add_(0, idx, dout)
+        return dx, None, None
+
+
+def gather_rows(x: torch.Tensor, idx: torch.Tensor,
+                bijective: bool = False) -> torch.Tensor:
+    return _GatherRows.apply(x, idx, bijective)
+
+
+class _MoECombine(torch.autograd.Function):
+    """y[t] = sum_j gates[t, j] * src[inv[t*K + j]] — the MoE un-permute +
+    gate + top-k combine in one pass (inv is a permutation of src rows)."""
+
+    @staticmethod
+    def forward(ctx, src: torch.Tensor, inv: torch.Tensor,
+                gates: torch.Tensor):
+        T, K = gates.shape
+        H = src.shape[-1]
+        ctx.save_for_backward(src, inv, gates)
+        if src.is_cuda and src.dtype == torch.bfloat16 and H % 8 == 0:
+            lib = _hip()
+            src = src.contiguous()
+            g32 = gates.reshape(-1).to(torch.float32).contiguous()
+            y = torch.empty(T, H, dtype=src.dtype, device=src.device)
+            rc = lib.moe_combine(native.stream_ptr(), _ptr(src), _ptr(inv),
+                                 _ptr(g32), _ptr(y), T, K, H)
+            native.check_rc(rc, "moe_combine", f"H={H} K={K}")
+            return y
+        pair = src[inv] * gates.reshape(-1, 1).to(src.dtype)
+        return pair.reshape(T, K, H).sum(dim=1)
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        src, inv, gates = ctx.saved_tensors
+        T, K = gates.shape
+        H = src.shape[-1]
+        if dy.is_cuda and dy.dtype == torch.bfloat16 and H % 8 == 0:
+            lib = _hip()
+            dy = dy.contiguous()
+            g32 = gates.reshape(-1).to(torch.float32).contiguous()
+            dsrc = torch.empty_like(src)
+            dg32 = torch.empty(T * K, dtype=torch.float32, device=dy.device)
+            rc = lib.moe_combine_bwd(native.stream_ptr(), _ptr(src),
+                                     _ptr(dy), _ptr(inv), _ptr(g32),
+                                     _ptr(dsrc), _ptr(dg32), T * K, K, H)
+            native.check_rc(rc, "moe_combine_bwd", f"H={H} K={K}")
+            return dsrc, None, dg32.reshape(T, K).to(gates.dtype)
+        dpair = dy.unsqueeze(1).expand(T, K, H).reshape(T * K, H)
+        dsrc = torch.zeros_like(src)
+        dsrc.index_add_(0, inv,
+                        dpair * gates.reshape(-1, 1).to(src.dtype))
+        dg = (src[inv].float() * dpair.float()).sum(-1).reshape(T, K)
+        return dsrc, None, dg.to(gates.dtype)
+
+
+def moe_combine(src: torch.Tensor, inv: torch.Tensor,
+                gates: torch.Tensor) -> torch.Tensor:
+    return _MoECombine.apply(src, inv, gates)

@@ -566,6 +566,125 @@ __global__ void ce_bwd_kernel(const uint4* __restrict__ logits,
 }
 
 // ===========================================================================
+// MoE token-dispatch row ops (bf16 rows, H % 8 == 0).
+//   rows_gather:        out[i] = in[idx[i]]      (fwd of x[idx])
+//   rows_scatter:       out[idx[i]] = in[i]      (bwd when idx is bijective)
+//   rows_scatter_add:   accum_f32[idx[i]] += in[i]  (bwd with duplicate idx)
+//   moe_combine:        y[t] = sum_j gates[t,j] * src[inv[t*K+j]]
+// The torch index chain these replace launches one kernel per op with an
+// index_put_ backward; fused they are single HBM-streaming passes.
+// ===========================================================================
+
+__global__ void rows_gather_kernel(const uint4* __restrict__ in,
+                                   const long* __restrict__ idx,
+                                   uint4* __restrict__ out,
+                                   long n_out, int H8) {
+  const long total = n_out * H8;
+  for (long g = blockIdx.x * (long)blockDim.x + threadIdx.x; g < total;
+       g += gridDim.x * (long)blockDim.x) {
+    const long i = g / H8;
+    const int h = (int)(g % H8);
+    out[g] = in[idx[i] * H8 + h];
+  }
+}
+
+__global__ void rows_scatter_kernel(const uint4* __restrict__ in,
+                                    const long* __restrict__ idx,
+                                    uint4* __restrict__ out,
+                                    long n_in, int H8) {
+  const long total = n_in * H8;
+  for (long g = blockIdx.x * (long)blockDim.x + threadIdx.x; g < total;
+       g += gridDim.x * (long)blockDim.x) {
+    const long i = g / H8;
+    const int h = (int)(g % H8);
+    out[idx[i] * H8 + h] = in[g];
+  }
+}
+
+__global__ void rows_scatter_add_f32_kernel(const uint4* __restrict__ in,
+                                            const long* __restrict__ idx,
+                                            float* __restrict__ accum,
+                                            long n_in, int H8) {
+  const long total = n_in * H8;
+  for (long g = blockIdx.x * (long)blockDim.x + threadIdx.x; g < total;
+       g += gridDim.x * (long)blockDim.x) {
+    const long i = g / H8;
+    const int h = (int)(g % H8);
+    BF8 a; a.v = in[g];
+    float* dst = accum + (idx[i] * (long)H8 + h) * 8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) atomicAdd(&dst[j], bf2f(a.h[j]));
+  }
+}
+
+__global__ void moe_combine_kernel(const uint4* __restrict__ src,
+                                   const long* __restrict__ inv,
+                                   const float* __restrict__ gates,
+                                   uint4* __restrict__ y,
+                                   long T, int K, int H8) {
+  const long total = T * H8;
+  for (long g = blockIdx.x * (long)blockDim.x + threadIdx.x; g < total;
+       g += gridDim.x * (long)blockDim.x) {
+    const long t = g / H8;
+    const int h = (int)(g % H8);
+    float acc[8] = {};
+    for (int j = 0; j < K; ++j) {
+      const long row = inv[t * K + j];
+      const float gj = gates[t * K + j];
+      BF8 a; a.v = src[row * H8 + h];
+#pragma unroll
+      for (int m = 0; m < 8; ++m) acc[m] += gj * bf2f(a.h[m]);
+    }
+    BF8 o;
+#pragma unroll
+    for (int m = 0; m < 8; ++m) o.h[m] = f2bf(acc[m]);
+    y[g] = o.v;
+  }
+}
+
+__global__ void moe_combine_bwd_dsrc_kernel(const uint4* __restrict__ dy,
+                                            const long* __restrict__ inv,
+                                            const float* __restrict__ gates,
+                                            uint4* __restrict__ dsrc,
+                                            long n_pairs, int K, int H8) {
+  const long total = n_pairs * H8;
+  for (long g = blockIdx.x * (long)blockDim.x + threadIdx.x; g < total;
+       g += gridDim.x * (long)blockDim.x) {
+    const long pair = g / H8;
+    const int h = (int)(g % H8);
+    const long row = inv[pair];
+    const float gj = gates[pair];
+    BF8 d; d.v = dy[(pair / K) * H8 + h];
+    BF8 o;
+#pragma unroll
+    for (int m = 0; m < 8; ++m) o.h[m] = f2bf(gj * bf2f(d.h[m]));
+    dsrc[row * H8 + h] = o.v;
+  }
+}
+
+__global__ void moe_combine_bwd_dgate_kernel(const uint4* __restrict__ src,
+                                             const uint4* __restrict__ dy,
+                                             const long* __restrict__ inv,
+                                             float* __restrict__ dgate,
+                                             long n_pairs, int K, int H8) {
+  // one 16-lane group per (t, j) pair
+  const long pair = blockIdx.x * (long)(BLOCK / 16) + (threadIdx.x >> 4);
+  if (pair >= n_pairs) return;
+  const long row = inv[pair];
+  const long t = pair / K;
+  float acc = 0.f;
+  for (int h = threadIdx.x & 15; h < H8; h += 16) {
+    BF8 a; a.v = src[row * H8 + h];
+    BF8 d; d.v = dy[t * H8 + h];
+#pragma unroll
+    for (int m = 0; m < 8; ++m) acc += bf2f(a.h[m]) * bf2f(d.h[m]);
+  }
+#pragma unroll
+  for (int off = 1; off < 16; off <<= 1) acc += __shfl_xor(acc, off, 64);
+  if ((threadIdx.x & 15) == 0) dgate[pair] = acc;
+}
+
+// ===========================================================================
 // Flat-buffer AdamW with optional fused global-norm gradient clipping.
 //   normsq: device pointer to the summed grad L2^2 (or nullptr for no clip);
 //   the kernel derives scale = clip / max(clip, sqrt(normsq)) — no host sync.
@@ -934,6 +1053,72 @@ int ce_bwd(void* stream, const void* logits, const void* targets,
                      (const uint4*)logits, (const int*)targets,
                      (const float*)lse, (const float*)gscale, (uint4*)dlogits,
                      V, ignore_index);
+  return 0;
+}
+
+int rows_gather(void* stream, const void* in, const void* idx, void* out,
+                long n_out, int H) {
+  if (H <= 0 || H % 8 != 0) return -1;
+  if (n_out <= 0) return 0;
+  const int H8 = H / 8;
+  dim3 grid(elementwise_grid(n_out * (long)H8)), block(BLOCK);
+  hipLaunchKernelGGL(rows_gather_kernel, grid, block, 0, STREAM,
+                     (const uint4*)in, (const long*)idx, (uint4*)out,
+                     n_out, H8);
+  return 0;
+}
+
+int rows_scatter(void* stream, const void* in, const void* idx, void* out,
+                 long n_in, int H) {
+  if (H <= 0 || H % 8 != 0) return -1;
+  if (n_in <= 0) return 0;
+  const int H8 = H / 8;
+  dim3 grid(elementwise_grid(n_in * (long)H8)), block(BLOCK);
+  hipLaunchKernelGGL(rows_scatter_kernel, grid, block, 0, STREAM,
+                     (const uint4*)in, (const long*)idx, (uint4*)out,
+                     n_in, H8);
+  return 0;
+}
+
+int rows_scatter_add_f32(void* stream, const void* in, const void* idx,
+                         void* accum, long n_in, int H) {
+  if (H <= 0 || H % 8 != 0) return -1;
+  if (n_in <= 0) return 0;
+  const int H8 = H / 8;
+  dim3 grid(elementwise_grid(n_in * (long)H8)), block(BLOCK);
+  hipLaunchKernelGGL(rows_scatter_add_f32_kernel, grid, block, 0, STREAM,
+                     (const uint4*)in, (const long*)idx, (float*)accum,
+                     n_in, H8);
+  return 0;
+}
+
+int moe_combine(void* stream, const void* src, const void* inv,
+                const void* gates, void* y, long T, int K, int H) {
+  if (H <= 0 || H % 8 != 0 || K <= 0) return -1;
+  if (T <= 0) return 0;
+  const int H8 = H / 8;
+  dim3 grid(elementwise_grid(T * (long)H8)), block(BLOCK);
+  hipLaunchKernelGGL(moe_combine_kernel, grid, block, 0, STREAM,
+                     (const uint4*)src, (const long*)inv,
+                     (const float*)gates, (uint4*)y, T, K, H8);
+  return 0;
+}
+
+int moe_combine_bwd(void* stream, const void* src, const void* dy,
+                    const void* inv, const void* gates, void* dsrc,
+                    void* dgate, long n_pairs, int K, int H) {
+  if (H <= 0 || H % 8 != 0 || K <= 0) return -1;
+  if (n_pairs <= 0) return 0;
+  const int H8 = H / 8;
+  dim3 grid(elementwise_grid(n_pairs * (long)H8)), block(BLOCK);
+  hipLaunchKernelGGL(moe_combine_bwd_dsrc_kernel, grid, block, 0, STREAM,
+                     (const uint4*)dy, (const long*)inv, (const float*)gates,
+                     (uint4*)dsrc, n_pairs, K, H8);
+  long gblocks = (n_pairs + (BLOCK / 16) - 1) / (BLOCK / 16);
+  if (gblocks > (1 << 20)) gblocks = 1 << 20;
+  hipLaunchKernelGGL(moe_combine_bwd_dgate_kernel, dim3((unsigned)gblocks),
+                     block, 0, STREAM, (const uint4*)src, (const uint4*)dy,
+                     (const long*)inv, (float*)dgate, n_pairs, K, H8);
   return 0;
 }
 

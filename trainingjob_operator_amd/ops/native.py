@@ -71,6 +71,11 @@ def load(require: bool = True) -> Optional[ctypes.CDLL]:
     _sig(lib.ce_fwd, [vp, vp, vp, vp, vp, l, i, i], i)
     _sig(lib.ce_bwd, [vp, vp, vp, vp, vp, vp, l, i, i], i)
     _sig(lib.l2normsq, [vp, vp, l, vp, i, vp], i)
+    _sig(lib.rows_gather, [vp, vp, vp, vp, l, i], i)
+    _sig(lib.rows_scatter, [vp, vp, vp, vp, l, i], i)
+    _sig(lib.rows_scatter_add_f32, [vp, vp, vp, vp, l, i], i)
+    _sig(lib.moe_combine, [vp, vp, vp, vp, vp, l, i, i], i)
+    _sig(lib.moe_combine_bwd, [vp, vp, vp, vp, vp, vp, vp, l, i, i], i)
     _sig(lib.adamw_step, [vp, vp, vp, vp, vp, vp, vp, l,
                           f, f, f, f, f, f, f, f, f, vp], i)
     _sig(lib.mfma_probe, [vp, vp, vp, vp])

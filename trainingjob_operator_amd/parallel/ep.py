@@ -180,7 +180,8 @@ class MoEMLP(nn.Module):
         flat_t = (torch.arange(T, device=x.device)
                   .repeat_interleave(self.top_k))
         order = torch.argsort(flat_e, stable=True)
-        send_tokens = xt[flat_t[order]]
+        from ..ops import gather_rows
+        send_tokens = gather_rows(xt, flat_t[order], bijective=False)
         counts = torch.bincount(flat_e, minlength=self.n_experts)
         in_splits = self._expert_owner_splits(counts)
 
@@ -216,25 +217,17 @@ class MoEMLP(nn.Module):
 
         # recv layout: source-rank major, each source's block sorted by
         # local expert. Permute to expert-major (contiguous run per
-        # expert), apply each expert to its slice, permute back — all via
-        # differentiable index ops.
-        sel_per_expert = []
-        src_offsets = torch.cat([
-            torch.zeros(1, dtype=torch.long, device=recv.device),
-            per_src.sum(dim=1).cumsum(0)[:-1]])
-        for le in range(self.experts_per_rank):
-            parts = []
-            for src in range(per_src.shape[0]):
-                base = int(src_offsets[src] + per_src[src, :le].sum())
-                parts.append(torch.arange(base,
-                                          base + int(per_src[src, le]),
-                                          device=recv.device))
-            sel_per_expert.append(
-                torch.cat(parts) if parts else
-                torch.empty(0, dtype=torch.long, device=recv.device))
-        perm = torch.cat(sel_per_expert)
-        grouped = recv[perm]
-        seg_sizes = [int(s.numel()) for s in sel_per_expert]
+        # expert) with ONE device-side stable argsort over per-row expert
+        # labels, apply each expert to its slice, permute back. (Round 1
+        # built per-(expert, src) index ranges in a Python loop — dozens
+        # of tiny kernels and a HOST SYNC per iteration from the int()
+        # casts; this form has a single host sync for the slice sizes.)
+        le_ids = torch.arange(self.experts_per_rank, device=recv.device)
+        block_le = le_ids.repeat(per_src.shape[0])     # src-major blocks
+        le_labels = torch.repeat_interleave(block_le, per_src.reshape(-1))
+        perm = torch.argsort(le_labels, stable=True)
+        grouped = gather_rows(recv, perm, bijective=True)
+        seg_sizes = per_src.sum(dim=0).tolist()        # one host sync
         seg_out = []
         off = 0
         for le, size in enumerate(seg_sizes):
@@ -245,16 +238,16 @@ class MoEMLP(nn.Module):
                       else grouped[:0])
         inv_perm = torch.empty_like(perm)
         inv_perm[perm] = torch.arange(perm.numel(), device=perm.device)
-        outputs = expert_out[inv_perm]
+        outputs = gather_rows(expert_out, inv_perm, bijective=True)
 
         back = _AllToAll.apply(outputs, out_splits, in_splits, self.group)
 
-        # undo the sort, apply gates, combine top-k
+        # undo the sort, apply gates, combine top-k — one fused pass
+        # (gather + gate + sum and their backwards: ops/hip rows kernels)
         inv = torch.empty_like(order)
         inv[order] = torch.arange(order.numel(), device=order.device)
-        pair_out = back[inv]                                  # [T*k, H]
-        gated = pair_out * gates.reshape(-1, 1).to(pair_out.dtype)
-        y = gated.reshape(T, self.top_k, H).sum(dim=1)
+        from ..ops import moe_combine
+        y = moe_combine(back, inv, gates)
         return y.reshape(orig_shape)
 
     @torch.no_grad()
