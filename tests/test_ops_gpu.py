@@ -229,7 +229,7 @@ def test_launchers_reject_unsupported_shapes():
     assert lib.ce_fwd(None, None, None, None, None, 4, 100, -100) != 0
     assert lib.rope(None, None, None, None, 4, 8, 16, 24, 1.0) != 0
     assert lib.attn_fwd(None, None, None, None, None, None,
-                        0, 0, 0, 0, 0, 0, 0, 0, 0, 1, 4, 100, 1.0) != 0
+                        0, 0, 0, 0, 0, 0, 0, 0, 0, 1, 4, 4, 100, 1.0) != 0
 
 
 def test_rows_gather_scatter_and_moe_combine():
