@@ -133,3 +133,25 @@ def test_dense_trainer_rejects_moe_configs():
     import trainingjob_operator_amd.models.moe_llama  # register  # noqa
     with pytest.raises(ValueError, match="MoE config"):
         Trainer(TrainConfig(model="moe-tiny"))
+
+
+def test_adamw_bf16_moments_short_horizon_parity():
+    """bf16 moments track fp32 moments closely over a short horizon (the
+    loss-parity gate for TrainConfig.adamw_bf16_moments)."""
+    import torch
+    from trainingjob_operator_amd.training import TrainConfig, Trainer
+    torch.manual_seed(0)
+    cfg32 = TrainConfig(model="llama-tiny", micro_batch=1, grad_accum=1,
+                        seq_len=32, lr=1e-3)
+    cfg16 = TrainConfig(model="llama-tiny", micro_batch=1, grad_accum=1,
+                        seq_len=32, lr=1e-3, adamw_bf16_moments=True)
+    torch.manual_seed(0)
+    t32 = Trainer(cfg32, device=torch.device("cpu"))
+    torch.manual_seed(0)
+    t16 = Trainer(cfg16, device=torch.device("cpu"))
+    l32 = [float(t32.train_step()) for _ in range(8)]
+    l16 = [float(t16.train_step()) for _ in range(8)]
+    # identical data/init: the loss curves must stay within bf16 noise
+    for a, b in zip(l32, l16):
+        assert abs(a - b) < 0.05, (l32, l16)
+    assert t16.opt.m.dtype == torch.bfloat16

@@ -78,6 +78,8 @@ def load(require: bool = True) -> Optional[ctypes.CDLL]:
     _sig(lib.moe_combine_bwd, [vp, vp, vp, vp, vp, vp, vp, l, i, i], i)
     _sig(lib.adamw_step, [vp, vp, vp, vp, vp, vp, vp, l,
                           f, f, f, f, f, f, f, f, f, vp], i)
+    _sig(lib.adamw_step_bf16mom, [vp, vp, vp, vp, vp, vp, vp, l,
+                                  f, f, f, f, f, f, f, f, f, vp], i)
     _sig(lib.mfma_probe, [vp, vp, vp, vp])
     _sig(lib.attn_fwd, [vp, vp, vp, vp, vp, vp,
                         l, l, l, l, l, l, l, l, l, i, i, i, i, f], i)
