@@ -84,9 +84,21 @@ def main():
     def adamw():
         lib.adamw_step(stream_ptr(), p32.data_ptr(), m.data_ptr(),
                        v.data_ptr(), grad.data_ptr(), pb.data_ptr(), None, n,
-                       1e-4, 0.9, 0.95, 1e-8, 0.1, 0.1, 0.05, 0.0, 1.0)
+                       1e-4, 0.9, 0.95, 1e-8, 0.1, 0.1, 0.05, 0.0, 1.0, None)
     report("adamw(2B params)", timeit(adamw, iters=10),
            n * (3 * 4 * 2 + 2 + 2 + 2))
+
+    # bf16-moment variant: m/v bf16 (20 B/param/step vs 28)
+    m16 = torch.zeros(n, device=DEV, dtype=torch.bfloat16)
+    v16 = torch.zeros(n, device=DEV, dtype=torch.bfloat16)
+
+    def adamw16():
+        lib.adamw_step_bf16mom(stream_ptr(), p32.data_ptr(), m16.data_ptr(),
+                               v16.data_ptr(), grad.data_ptr(), pb.data_ptr(),
+                               None, n, 1e-4, 0.9, 0.95, 1e-8, 0.1, 0.1,
+                               0.05, 0.0, 1.0, None)
+    report("adamw_bf16mom(2B)", timeit(adamw16, iters=10),
+           n * (2 * 4 * 2 + 2 * 2 * 2 + 2 + 2 + 2))
 
 
 if __name__ == "__main__":

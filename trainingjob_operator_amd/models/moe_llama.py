@@ -172,6 +172,14 @@ class MoELlamaModel(nn.Module):
 # Mixtral-class MoE: 8 experts x top-2, ~47B total / ~13B active params.
 # The EP target config: one expert shard per MI355X on an 8-GPU node
 # (launcher --ep 8), or EP x TP for headroom.
+# single-GPU-profileable mid size (~2.6B params: bf16 + fp32 opt state fit
+# comfortably in 288 GB) — the EP/MoE kernel-profiling model
+MOE_MID = MoELlamaConfig(
+    name="moe-mid", vocab_size=32000, hidden_size=2048,
+    intermediate_size=11264, num_layers=8, num_heads=16, num_kv_heads=4,
+    head_dim=128, rope_theta=500000.0, n_experts=8, top_k=2,
+    expert_ff=5632)
+
 MOE_8X7B = MoELlamaConfig(
     name="moe-8x7b", vocab_size=32000, hidden_size=4096,
     intermediate_size=28672, num_layers=32, num_heads=32, num_kv_heads=8,
@@ -183,4 +191,5 @@ MOE_8X7B = MoELlamaConfig(
 from .config import CONFIGS  # noqa: E402
 
 CONFIGS[MOE_TINY.name] = MOE_TINY
+CONFIGS[MOE_MID.name] = MOE_MID
 CONFIGS[MOE_8X7B.name] = MOE_8X7B
