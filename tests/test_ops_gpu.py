@@ -276,3 +276,25 @@ def test_rows_gather_scatter_and_moe_combine():
     assert torch.allclose(src.grad.float().cpu(), sa.grad, atol=2e-2,
                           rtol=2e-2)
     assert torch.allclose(gates.grad.cpu(), ga.grad, atol=2e-1, rtol=2e-2)
+
+
+def test_dispatch_rows_backward_no_atomics():
+    """dispatch_rows: gather fwd; backward = gather-sum via the inverse
+    pair permutation (vs index_add reference)."""
+    from trainingjob_operator_amd.ops import dispatch_rows
+    torch.manual_seed(13)
+    T, K, H = 40, 2, 256
+    x = _mk((T, H)).requires_grad_()
+    flat_t = torch.arange(T, device=DEV).repeat_interleave(K)
+    order = torch.randperm(T * K, device=DEV)
+    inv = torch.empty_like(order)
+    inv[order] = torch.arange(T * K, device=DEV)
+    out = dispatch_rows(x, flat_t[order], inv, K)
+    dy = _mk((T * K, H))
+    (out.float() * dy.float()).sum().backward()
+    xa = x.detach().float().cpu().requires_grad_()
+    (xa[flat_t[order].cpu()] * dy.float().cpu()).sum().backward()
+    assert torch.allclose(out.float().cpu(), xa.detach()[flat_t[order].cpu()],
+                          atol=1e-2)
+    assert torch.allclose(x.grad.float().cpu(), xa.grad, atol=2e-2,
+                          rtol=2e-2)

@@ -9,6 +9,7 @@ from .functional import (
     apply_rope,
     fused_cross_entropy,
     fused_rmsnorm,
+    dispatch_rows,
     gather_rows,
     make_inv_freq,
     moe_combine,
@@ -19,6 +20,7 @@ from .native import HipOpsUnavailable, available, build_ops, load
 
 __all__ = [
     "apply_rope",
+    "dispatch_rows",
     "gather_rows",
     "moe_combine",
     "fused_cross_entropy",

@@ -398,3 +398,48 @@ class _MoECombine(torch.autograd.Function):
 def moe_combine(src: torch.Tensor, inv: torch.Tensor,
                 gates: torch.Tensor) -> torch.Tensor:
     return _MoECombine.apply(src, inv, gates)
+
+
+class _DispatchRows(torch.autograd.Function):
+    """MoE dispatch gather: out[p] = x[tok[p]] where each token appears in
+    exactly K pairs. Backward is an atomic-free gather-sum using the
+    inverse pair permutation: dx[t] = sum_j dout[inv[t*K+j]] (the
+    rows_scatter_add atomic path measured 435 us/call at moe-mid)."""
+
+    @staticmethod
+    def forward(ctx, x, tok_idx, inv_pairs, K: int):
+        ctx.save_for_backward(inv_pairs)
+        ctx.K = K
+        ctx.n_in = x.shape[0]
+        if x.is_cuda and x.dtype == torch.bfloat16 and x.shape[-1] % 8 == 0:
+            lib = _hip()
+            x = x.contiguous()
+            out = torch.empty(tok_idx.shape[0], x.shape[1], dtype=x.dtype,
+                              device=x.device)
+            rc = lib.rows_gather(native.stream_ptr(), _ptr(x), _ptr(tok_idx),
+                                 _ptr(out), tok_idx.shape[0], x.shape[1])
+            native.check_rc(rc, "rows_gather", f"H={x.shape[1]}")
+            return out
+        return x[tok_idx]
+
+    @staticmethod
+    def backward(ctx, dout):
+        (inv_pairs,) = ctx.saved_tensors
+        K = ctx.K
+        H = dout.shape[-1]
+        T = ctx.n_in
+        if dout.is_cuda and dout.dtype == torch.bfloat16 and H % 8 == 0:
+            lib = _hip()
+            dout = dout.contiguous()
+            dx = torch.empty(T, H, dtype=dout.dtype, device=dout.device)
+            rc = lib.moe_combine(native.stream_ptr(), _ptr(dout),
+                                 _ptr(inv_pairs), None, _ptr(dx), T, K, H)
+            native.check_rc(rc, "moe_combine(dispatch-bwd)", f"H={H}")
+            return dx, None, None, None
+        dx = dout[inv_pairs.reshape(-1)].reshape(T, K, H).sum(dim=1)
+        return dx, None, None, None
+
+
+def dispatch_rows(x: torch.Tensor, tok_idx: torch.Tensor,
+                  inv_pairs: torch.Tensor, K: int) -> torch.Tensor:
+    return _DispatchRows.apply(x, tok_idx, inv_pairs, K)

@@ -630,7 +630,7 @@ __global__ void moe_combine_kernel(const uint4* __restrict__ src,
     float acc[8] = {};
     for (int j = 0; j < K; ++j) {
       const long row = inv[t * K + j];
-      const float gj = gates[t * K + j];
+      const float gj = gates ? gates[t * K + j] : 1.0f;
       BF8 a; a.v = src[row * H8 + h];
 #pragma unroll
       for (int m = 0; m < 8; ++m) acc[m] += gj * bf2f(a.h[m]);

@@ -180,8 +180,10 @@ class MoEMLP(nn.Module):
         flat_t = (torch.arange(T, device=x.device)
                   .repeat_interleave(self.top_k))
         order = torch.argsort(flat_e, stable=True)
-        from ..ops import gather_rows
-        send_tokens = gather_rows(xt, flat_t[order], bijective=False)
+        inv = torch.empty_like(order)
+        inv[order] = torch.arange(order.numel(), device=order.device)
+        from ..ops import dispatch_rows, gather_rows, moe_combine
+        send_tokens = dispatch_rows(xt, flat_t[order], inv, self.top_k)
         counts = torch.bincount(flat_e, minlength=self.n_experts)
         in_splits = self._expert_owner_splits(counts)
 
@@ -244,9 +246,6 @@ class MoEMLP(nn.Module):
 
         # undo the sort, apply gates, combine top-k — one fused pass
         # (gather + gate + sum and their backwards: ops/hip rows kernels)
-        inv = torch.empty_like(order)
-        inv[order] = torch.arange(order.numel(), device=order.device)
-        from ..ops import moe_combine
         y = moe_combine(back, inv, gates)
         return y.reshape(orig_shape)
 
