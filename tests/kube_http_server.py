@@ -61,6 +61,24 @@ class _Handler(BaseHTTPRequestHandler):
         self.send_header("Transfer-Encoding", "chunked")
         self.end_headers()
         stop = threading.Event()
+
+        def emit(evt_type, obj):
+            line = json.dumps({"type": evt_type, "object": obj}) + "\n"
+            data = line.encode()
+            self.wfile.write(f"{len(data):x}\r\n".encode() + data
+                             + b"\r\n")
+            self.wfile.flush()
+
+        # resourceVersion-expiry simulation: one ERROR/410 event, like a
+        # real API server whose etcd compacted the requested rv
+        if getattr(self.fake, "expire_next_watch", False):
+            self.fake.expire_next_watch = False
+            try:
+                emit("ERROR", {"kind": "Status", "code": 410,
+                               "reason": "Expired"})
+            except (BrokenPipeError, ConnectionResetError):
+                pass
+            return
         try:
             for evt_type, obj in self.fake._watch_kind(kind, stop):
                 line = json.dumps({"type": evt_type, "object": obj}) + "\n"

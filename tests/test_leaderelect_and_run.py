@@ -156,7 +156,13 @@ def test_run_loop_multiworker_concurrent_torture():
             "apiVersion": C.API_VERSION, "kind": C.CRD_KIND,
             "metadata": {"name": n, "namespace": "default"},
             "spec": {"replicaSpecs": {"trainer": {
-                "replicas": 2, "restartPolicy": "OnFailure",
+                # elastic (Manual) so chaos scale-downs DELETE the
+                # out-of-range pods — on a non-elastic role they are
+                # deliberately left (reference behavior) and the
+                # pods==replicas convergence invariant would never hold
+                "replicas": 2, "minReplicas": 1, "maxReplicas": 3,
+                "edlPolicy": "Manual",
+                "restartPolicy": "OnFailure",
                 "restartScope": "All", "restartLimit": 100,
                 "template": {"spec": {"containers": [{
                     "name": "aitj-main",

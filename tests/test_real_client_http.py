@@ -167,3 +167,39 @@ def test_list_events_over_http(cluster):
     evs = api.list_events("d", involved_name="evjob")
     assert len(evs) == 1 and evs[0]["reason"] == "Test"
     assert api.list_events("d", involved_name="other") == []
+
+
+def test_watch_410_relist_recovery(cluster):
+    """resourceVersion expiry: the server answers a watch with ERROR/410;
+    the client yields a synthetic RELIST, drops its rv, reconnects, and
+    keeps streaming subsequent events (client-go re-list semantics)."""
+    import threading
+
+    fake, api = cluster
+    fake.create_job(NS, job_manifest())
+    events = []
+    stop = threading.Event()
+    fake.expire_next_watch = True
+
+    def consume():
+        for etype, obj in api.watch_jobs(NS, stop):
+            events.append((etype, obj.get("metadata", {}).get("name")))
+            if len(events) >= 2:
+                stop.set()
+                return
+
+    t = threading.Thread(target=consume, daemon=True)
+    t.start()
+    deadline = __import__("time").monotonic() + 10
+    while not events and __import__("time").monotonic() < deadline:
+        __import__("time").sleep(0.05)
+    assert events and events[0][0] == "RELIST", events
+    # post-expiry events still arrive on the reconnected stream
+    j = api.get_job(NS, "httpjob")
+    j["metadata"]["labels"] = {"poke": "1"}
+    api.update_job(NS, "httpjob", j)
+    deadline = __import__("time").monotonic() + 10
+    while len(events) < 2 and __import__("time").monotonic() < deadline:
+        __import__("time").sleep(0.05)
+    stop.set()
+    assert len(events) >= 2 and events[1][1] == "httpjob", events

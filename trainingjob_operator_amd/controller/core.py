@@ -222,6 +222,16 @@ class TrainingJobController:
         def watcher(watch_fn, to_key, observe=False):
             for evt_type, obj in watch_fn(self.options.namespace or None,
                                           stop):
+                if evt_type == "RELIST":
+                    # watch rv expired (410 Gone): re-list so nothing
+                    # that happened during the gap waits for the resync
+                    try:
+                        for j in self.api.list_jobs(
+                                self.options.namespace or None):
+                            self.queue.add(job_key(j))
+                    except Exception:
+                        log.exception("relist after 410 failed")
+                    continue
                 key = to_key(obj)
                 if key:
                     if observe:

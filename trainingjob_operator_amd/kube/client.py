@@ -284,22 +284,49 @@ class RealKubeApi(KubeApi):
     # -- watches ----------------------------------------------------------
     def _watch(self, path: str, stop: threading.Event,
                params: Optional[dict] = None) -> Iterator[tuple]:
+        """Streaming watch with client-go re-list semantics:
+
+        * tracks resourceVersion across reconnects;
+        * BOOKMARK events only advance the rv (requested via
+          allowWatchBookmarks);
+        * 410 Gone — an ERROR event with code 410, or an HTTP 410 on
+          reconnect — means the rv EXPIRED from etcd: drop the rv (the
+          next connect starts from "now") and yield a synthetic
+          ("RELIST", {}) so the consumer re-lists; events between expiry
+          and reconnect are otherwise silently lost.
+        The controller is level-triggered (full LIST per sync + periodic
+        resync), so a missed event only delays work; RELIST removes even
+        that delay."""
         params = dict(params or {})
         params["watch"] = "true"
+        params["allowWatchBookmarks"] = "true"
         while not stop.is_set():
             try:
                 r = self.session.get(self.base_url + path, params=params,
                                      stream=True, timeout=(10, 300))
+                if r.status_code == 410:
+                    params.pop("resourceVersion", None)
+                    yield "RELIST", {}
+                    continue
                 for line in r.iter_lines():
                     if stop.is_set():
                         return
                     if not line:
                         continue
                     evt = json.loads(line)
+                    etype = evt.get("type", "")
                     obj = evt.get("object", {})
-                    params["resourceVersion"] = (
-                        obj.get("metadata", {}).get("resourceVersion", ""))
-                    yield evt.get("type", ""), obj
+                    if etype == "ERROR":
+                        if obj.get("code") == 410:
+                            params.pop("resourceVersion", None)
+                            yield "RELIST", {}
+                        break   # reconnect either way
+                    rv = obj.get("metadata", {}).get("resourceVersion")
+                    if rv:
+                        params["resourceVersion"] = rv
+                    if etype == "BOOKMARK":
+                        continue
+                    yield etype, obj
             except Exception:
                 if stop.is_set():
                     return
