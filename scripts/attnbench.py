@@ -10,7 +10,8 @@ import torch
 import torch.nn.functional as F
 
 from trainingjob_operator_amd.ops.attention import (
-    flash_attention_fwd_only, flash_attention_fwd_nw8, flash_attention_fwd_v5)
+    flash_attention_fwd_only, flash_attention_fwd_nw8, flash_attention_fwd_v5,
+    flash_attention_fwd_v7)
 
 
 def main():
@@ -29,6 +30,8 @@ def main():
             flash_attention_fwd_v5(q, k, v)
         elif which == "nw8":
             flash_attention_fwd_nw8(q, k, v)
+        elif which == "v7":
+            flash_attention_fwd_v7(q, k, v)
         else:
             F.scaled_dot_product_attention(q, k, v, is_causal=True)
 

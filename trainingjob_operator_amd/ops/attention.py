@@ -138,6 +138,26 @@ def flash_attention_fwd_nw8(q, k, v, scale=None):
     return out, lse
 
 
+def flash_attention_fwd_v7(q, k, v, scale=None):
+    """The 3-deep all-glds pipeline forward (A/B candidate)."""
+    if scale is None:
+        scale = 1.0 / math.sqrt(q.shape[-1])
+    B, H, S, D = q.shape
+    HKV = k.shape[1]
+    lib = native.load(require=True)
+    out = torch.empty(B, H, S, D, dtype=torch.bfloat16, device=q.device)
+    lse = torch.empty(B, H, S, dtype=torch.float32, device=q.device)
+    rc = lib.attn_fwd_v7(
+        native.stream_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+        out.data_ptr(), lse.data_ptr(),
+        q.stride(0), q.stride(1), q.stride(2),
+        k.stride(0), k.stride(1), k.stride(2),
+        v.stride(0), v.stride(1), v.stride(2),
+        B, H, HKV, S, scale)
+    native.check_rc(rc, "attn_fwd_v7", f"B={B} H={H} S={S}")
+    return out, lse
+
+
 def flash_attention_fwd_v5(q, k, v, scale=None):
     """The 4-wave v5 forward, kept callable for A/B benchmarking."""
     if scale is None:

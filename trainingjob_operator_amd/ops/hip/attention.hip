@@ -411,6 +411,13 @@ __global__ __launch_bounds__(NW * 64) void attn_fwd_v6_kernel(
   V6_WRITE_V(lds0)
   __syncthreads();
 
+  // T5 static form: the second-dispatched half of the workgroup loses
+  // VALU arbitration to the older half on every segment; one setprio(1)
+  // for it (wave-uniform via readfirstlane) removes its start-of-segment
+  // penalty (guide: -0.8..1.5% cycles, never negative)
+  if (NW == 8 && __builtin_amdgcn_readfirstlane(threadIdx.x) >= NT / 2)
+    __builtin_amdgcn_s_setprio(1);
+
   int cur = 0;
   for (int t = 0; t < n_tiles; ++t) {
     const int kv0 = t * V6_BN;
@@ -562,6 +569,15 @@ __global__ __launch_bounds__(NW * 64) void attn_fwd_v6_kernel(
         m_run * 0.6931471805599453f + __logf(l_run);
 }
 
+__global__ __launch_bounds__(512) void attn_fwd_v7_kernel(
+    const u16* __restrict__ q, const u16* __restrict__ k,
+    const u16* __restrict__ v, u16* __restrict__ out,
+    float* __restrict__ lse,
+    long q_sb, long q_sh, long q_ss,
+    long k_sb, long k_sh, long k_ss,
+    long v_sb, long v_sh, long v_ss,
+    int n_heads, int gqa_group, int S, float scale2);
+
 extern "C" int attn_fwd(void* stream, const void* q, const void* k,
                         const void* v, void* out, void* lse,
                         long q_sb, long q_sh, long q_ss,
@@ -578,8 +594,10 @@ extern "C" int attn_fwd(void* stream, const void* q, const void* k,
     // B1H32S4096. BM=128 covers the S%128 shapes.
     const float scale2 = scale * 1.4426950408889634f;   // fold log2(e)
     if (S % 256 == 0) {
+      // v7: 3-deep all-glds pipeline, single raw barrier per tile
+      // (299 us vs v6's 346 at B1H32S4096; aotriton 425)
       dim3 grid(S / 256, n_heads, batch), block(512);
-      hipLaunchKernelGGL((attn_fwd_v6_kernel<8>), grid, block, 0,
+      hipLaunchKernelGGL(attn_fwd_v7_kernel, grid, block, 0,
                          reinterpret_cast<hipStream_t>(stream),
                          (const u16*)q, (const u16*)k, (const u16*)v,
                          (u16*)out, (float*)lse, q_sb, q_sh, q_ss,
@@ -1418,5 +1436,300 @@ extern "C" int attn_bwd(void* stream, const void* q, const void* k,
                        q_sb, q_sh, q_ss, k_sb, k_sh, k_ss,
                        v_sb, v_sh, v_ss, n_heads, n_kv_heads, S, scale);
   }
+  return 0;
+}
+
+// ===========================================================================
+// v7 forward: v6's math with a 3-deep ALL-glds staging pipeline.
+//
+// v6 parks ~49% of wave cycles at waits/barriers: its __syncthreads()
+// carries a vmcnt(0) drain (glds in flight) and the V register staging
+// serializes a global-load wait into the middle of every tile. v7:
+//   * K AND row-major V arrive by glds into a 3-slot ring (issue runs two
+//     tiles ahead; nothing in the main loop ever waits vmcnt(0));
+//   * the transposed V image is built LDS->LDS (2 b128 reads + 8 paired
+//     b32 writes per thread) from the landed row image — no global
+//     staging registers at all;
+//   * barriers are RAW s_barrier (guide: "3-buffer glds + raw barrier,
+//     counted vmcnt only"), two per tile: one after the counted
+//     vmcnt making slot t+1 visible for the transpose, one at tile end
+//     guarding ring reuse.
+// ===========================================================================
+
+__device__ __forceinline__ void v7_wait_vmcnt4() {
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+}
+__device__ __forceinline__ void v7_wait_vmcnt0() {
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+}
+__device__ __forceinline__ void v7_barrier() {
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+}
+
+#define V7_RING 3
+
+__global__ __launch_bounds__(512) void attn_fwd_v7_kernel(
+    const u16* __restrict__ q, const u16* __restrict__ k,
+    const u16* __restrict__ v, u16* __restrict__ out,
+    float* __restrict__ lse,
+    long q_sb, long q_sh, long q_ss,
+    long k_sb, long k_sh, long k_ss,
+    long v_sb, long v_sh, long v_ss,
+    int n_heads, int gqa_group, int S, float scale2) {
+  constexpr int NT = 512;
+  const int qb = gridDim.x - 1 - blockIdx.x;
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int hkv = h / gqa_group;
+  const int tid = threadIdx.x;
+  const int wid = tid >> 6;
+  const int lane = tid & 63;
+  const int low = lane & 31;
+  const int hi = lane >> 5;
+
+  // ring: K rows (swizzled image) + V rows (swizzled image), both glds;
+  // 2 transposed-V images built locally
+  __shared__ __attribute__((aligned(16))) u16
+      lds[V7_RING * 2 * V6_K_U16 + 2 * V6_V_U16];
+  // slot addressing by arithmetic (pointer arrays spill under pressure)
+#define LDSK(SLOT) (&lds[(SLOT) * V6_K_U16])
+#define LDSVR(SLOT) (&lds[(V7_RING + (SLOT)) * V6_K_U16])
+#define LDSVT(P) (reinterpret_cast<char*>(&lds[2 * V7_RING * V6_K_U16]) \
+                  + (P) * (V6_V_U16 * 2))
+
+  const int q0w = qb * 256 + wid * 32;
+  const int qrow = q0w + low;
+
+  union F8 { bf16x8 v; uint4 u; u16 h[8]; };
+  const u16* qptr = q + (long)b * q_sb + (long)h * q_sh + (long)qrow * q_ss;
+  F8 qf[8];
+#pragma unroll
+  for (int kc = 0; kc < 8; ++kc)
+    qf[kc].u = *reinterpret_cast<const uint4*>(qptr + kc * 16 + hi * 8);
+
+  f32x16 oacc[4];
+#pragma unroll
+  for (int ds = 0; ds < 4; ++ds)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) oacc[ds][r] = 0.f;
+  float m_run = -INFINITY, l_run = 0.f;
+
+  const u16* kbase = k + (long)b * k_sb + (long)hkv * k_sh;
+  const u16* vbase = v + (long)b * v_sb + (long)hkv * v_sh;
+  const int n_tiles = (qb + 1) * 4;
+
+  // glds source pointers (swizzle-inverted, advanced per issue)
+  const u16* kgp[2];
+  const u16* vgp[2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int idx = i * NT + tid;
+    const int L = idx * 16;
+    const int kr = L >> 8;
+    const int kin = (((L & 255) ^ ((kr & 15) << 4)) >> 1);
+    kgp[i] = kbase + (long)kr * k_ss + kin;
+    vgp[i] = vbase + (long)kr * v_ss + kin;
+  }
+  const long kstep = (long)V6_BN * k_ss;
+  const long vstep = (long)V6_BN * v_ss;
+  const int wu64 = __builtin_amdgcn_readfirstlane(tid >> 6);
+  const int lslot = (tid & ~63) * 8;          // wave-uniform u16 base
+
+#define V7_GLDS(SLOT)                                                       \
+  _Pragma("unroll")                                                         \
+  for (int i = 0; i < 2; ++i) {                                             \
+    __builtin_amdgcn_global_load_lds(                                       \
+        (const u32*)kgp[i], (u32*)&LDSK(SLOT)[i * NT * 8 + lslot],          \
+        16, 0, 0);                                                          \
+    kgp[i] += kstep;                                                        \
+    __builtin_amdgcn_global_load_lds(                                       \
+        (const u32*)vgp[i], (u32*)&LDSVR(SLOT)[i * NT * 8 + lslot],         \
+        16, 0, 0);                                                          \
+    vgp[i] += vstep;                                                        \
+  }
+
+  // LDS->LDS transpose of one landed V row-image into a vt image
+  const int t_rp2 = (tid >> 4) * 2;
+  const int t_c8 = (tid & 15) * 8;
+#define V7_TRANSPOSE(SRCSLOT, DST)                                          \
+  {                                                                         \
+    const char* srcb = reinterpret_cast<const char*>(LDSVR(SRCSLOT));       \
+    union { uint4 u4; u16 h[8]; } va_, vb_;                                 \
+    va_.u4 = *reinterpret_cast<const uint4*>(&srcb[k_byte(t_rp2, t_c8)]);   \
+    vb_.u4 = *reinterpret_cast<const uint4*>(                               \
+        &srcb[k_byte(t_rp2 + 1, t_c8)]);                                    \
+    _Pragma("unroll")                                                       \
+    for (int j = 0; j < 8; ++j) {                                           \
+      const u32 pair_ = (u32)va_.h[j] | ((u32)vb_.h[j] << 16);              \
+      *reinterpret_cast<u32*>(&(DST)[vt_byte(t_c8 + j, t_rp2)]) = pair_;    \
+    }                                                                       \
+  }
+
+  // prologue: issue tiles 0 and 1; build Vt[0] once tile 0 lands
+  V7_GLDS(0)
+  if (n_tiles > 1) V7_GLDS(1)
+  if (n_tiles > 1) v7_wait_vmcnt4(); else v7_wait_vmcnt0();
+  v7_barrier();
+  V7_TRANSPOSE(0, LDSVT(0))
+  // Vt[0] visible to every wave at the first in-loop barrier below
+
+  for (int t = 0; t < n_tiles; ++t) {
+    const int kv0 = t * V6_BN;
+    const bool have2 = (t + 2) < n_tiles;
+    if (have2) V7_GLDS((t + 2) % V7_RING)
+
+    char* kb = reinterpret_cast<char*>(LDSK(t % V7_RING));
+    char* vtb = LDSVT(t & 1);
+    const bool active = kv0 <= q0w + 31;
+    const bool need_mask = kv0 + V6_BN > q0w;
+
+    f32x16 s0, s1;
+    union PF { bf16x8 v; unsigned u[4]; } pa[4];
+    float pm = -INFINITY;
+    if (active) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) { s0[r] = 0.f; s1[r] = 0.f; }
+#pragma unroll
+      for (int kc = 0; kc < 8; ++kc) {
+        bf16x8 a0 = *reinterpret_cast<const bf16x8*>(
+            &kb[k_byte(low, kc * 16 + hi * 8)]);
+        s0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, qf[kc].v, s0, 0, 0, 0);
+        bf16x8 a1 = *reinterpret_cast<const bf16x8*>(
+            &kb[k_byte(32 + low, kc * 16 + hi * 8)]);
+        s1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, qf[kc].v, s1, 0, 0, 0);
+      }
+      if (need_mask) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kvr = kv0 + (r & 3) + 8 * (r >> 2) + 4 * hi;
+          float x0 = s0[r] * scale2;
+          if (kvr > qrow) x0 = -1e30f;
+          s0[r] = x0;
+          float x1 = s1[r] * scale2;
+          if (kvr + 32 > qrow) x1 = -1e30f;
+          s1[r] = x1;
+          pm = fmaxf(pm, fmaxf(x0, x1));
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          s0[r] *= scale2;
+          s1[r] *= scale2;
+          pm = fmaxf(pm, fmaxf(s0[r], s1[r]));
+        }
+      }
+      pm = fmaxf(pm, __shfl_xor(pm, 32, 64));
+
+      if (!__all(pm - m_run <= 8.0f)) {
+        const float mn = fmaxf(m_run, pm);
+        const float alpha = (m_run == -INFINITY) ? 0.f : exp2_raw(m_run - mn);
+        m_run = mn;
+        l_run *= alpha;
+#pragma unroll
+        for (int ds = 0; ds < 4; ++ds)
+#pragma unroll
+          for (int r = 0; r < 16; ++r) oacc[ds][r] *= alpha;
+      }
+      float psum = 0.f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const float p0 = exp2_raw(s0[r] - m_run);
+        const float p1 = exp2_raw(s1[r] - m_run);
+        s0[r] = p0;
+        s1[r] = p1;
+        psum += p0 + p1;
+      }
+      psum += __shfl_xor(psum, 32, 64);
+      l_run += psum;
+
+#pragma unroll
+      for (int sub = 0; sub < 2; ++sub) {
+#pragma unroll
+        for (int cc = 0; cc < 2; ++cc) {
+          const int pb = cc * 8;
+          unsigned a0, b0, a1, b1;
+          if (sub == 0) {
+            a0 = cvt_pk_bf16(s0[pb + 0], s0[pb + 1]);
+            b0 = cvt_pk_bf16(s0[pb + 4], s0[pb + 5]);
+            a1 = cvt_pk_bf16(s0[pb + 2], s0[pb + 3]);
+            b1 = cvt_pk_bf16(s0[pb + 6], s0[pb + 7]);
+          } else {
+            a0 = cvt_pk_bf16(s1[pb + 0], s1[pb + 1]);
+            b0 = cvt_pk_bf16(s1[pb + 4], s1[pb + 5]);
+            a1 = cvt_pk_bf16(s1[pb + 2], s1[pb + 3]);
+            b1 = cvt_pk_bf16(s1[pb + 6], s1[pb + 7]);
+          }
+          auto r02 = __builtin_amdgcn_permlane32_swap(a0, b0, false, false);
+          auto r13 = __builtin_amdgcn_permlane32_swap(a1, b1, false, false);
+          PF f;
+          f.u[0] = r02[0];
+          f.u[1] = r13[0];
+          f.u[2] = r02[1];
+          f.u[3] = r13[1];
+          pa[sub * 2 + cc] = f;
+        }
+      }
+
+#pragma unroll
+      for (int c = 0; c < 4; ++c) {
+#pragma unroll
+        for (int ds = 0; ds < 4; ++ds) {
+          bf16x8 vf = *reinterpret_cast<const bf16x8*>(
+              &vtb[vt_byte(ds * 32 + low, c * 16 + hi * 8)]);
+          oacc[ds] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              vf, pa[c].v, oacc[ds], 0, 0, 0);
+        }
+      }
+    }
+    // end-of-tile: wait slot t+1 (counted — t+2 stays in flight), build
+    // the NEXT transposed image (different Vt buffer than this tile's PV
+    // reads, so it needs no pre-barrier), then ONE raw barrier guarding
+    // ring reuse + Vt visibility for tile t+1
+    if (t + 1 < n_tiles) {
+      if (have2) v7_wait_vmcnt4(); else v7_wait_vmcnt0();
+      V7_TRANSPOSE((t + 1) % V7_RING, LDSVT((t + 1) & 1))
+    }
+    v7_barrier();
+  }
+
+  const float inv_l = 1.0f / l_run;
+  u16* orow = out + (((long)b * n_heads + h) * S + qrow) * ATTN_D;
+#pragma unroll
+  for (int ds = 0; ds < 4; ++ds) {
+#pragma unroll
+    for (int g = 0; g < 4; ++g) {
+      const int d0 = 8 * g + 4 * hi + 32 * ds;
+      const unsigned w0 = cvt_pk_bf16(oacc[ds][4 * g + 0] * inv_l,
+                                      oacc[ds][4 * g + 1] * inv_l);
+      const unsigned w1 = cvt_pk_bf16(oacc[ds][4 * g + 2] * inv_l,
+                                      oacc[ds][4 * g + 3] * inv_l);
+      uint2 wv;
+      wv.x = w0;
+      wv.y = w1;
+      *reinterpret_cast<uint2*>(orow + d0) = wv;
+    }
+  }
+  if (hi == 0)
+    lse[((long)b * n_heads + h) * S + qrow] =
+        m_run * 0.6931471805599453f + __logf(l_run);
+}
+
+extern "C" int attn_fwd_v7(void* stream, const void* q, const void* k,
+                           const void* v, void* out, void* lse,
+                           long q_sb, long q_sh, long q_ss,
+                           long k_sb, long k_sh, long k_ss,
+                           long v_sb, long v_sh, long v_ss,
+                           int batch, int n_heads, int n_kv_heads, int S,
+                           float scale) {
+  if (S <= 0 || S % 256 != 0 || n_heads % n_kv_heads != 0) return -1;
+  dim3 grid(S / 256, n_heads, batch), block(512);
+  const float scale2 = scale * 1.4426950408889634f;
+  hipLaunchKernelGGL(attn_fwd_v7_kernel, grid, block, 0,
+                     reinterpret_cast<hipStream_t>(stream),
+                     (const u16*)q, (const u16*)k, (const u16*)v, (u16*)out,
+                     (float*)lse, q_sb, q_sh, q_ss, k_sb, k_sh, k_ss,
+                     v_sb, v_sh, v_ss, n_heads, n_heads / n_kv_heads, S,
+                     scale2);
   return 0;
 }

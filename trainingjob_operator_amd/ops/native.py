@@ -87,6 +87,8 @@ def load(require: bool = True) -> Optional[ctypes.CDLL]:
                            l, l, l, l, l, l, l, l, l, i, i, i, f], i)
     _sig(lib.attn_fwd_nw8, [vp, vp, vp, vp, vp, vp,
                             l, l, l, l, l, l, l, l, l, i, i, i, i, f], i)
+    _sig(lib.attn_fwd_v7, [vp, vp, vp, vp, vp, vp,
+                           l, l, l, l, l, l, l, l, l, i, i, i, i, f], i)
     _sig(lib.attn_bwd, [vp, vp, vp, vp, vp, vp, vp, vp, vp, vp, vp,
                         l, l, l, l, l, l, l, l, l, i, i, i, i, f], i)
     _sig(lib.mfma_probe32, [vp, vp, vp, vp])
