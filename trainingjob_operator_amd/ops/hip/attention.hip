@@ -293,6 +293,22 @@ __device__ __forceinline__ float exp2_raw(float x) {
   return r;
 }
 
+// 3-deep glds pipeline helpers (fwd v7 + the v7-style backward kernels):
+// counted vmcnt only, raw barriers (no vmcnt(0) drain in any main loop)
+__device__ __forceinline__ void v7_wait_vmcnt4() {
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+}
+__device__ __forceinline__ void v7_wait_vmcnt0() {
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+}
+__device__ __forceinline__ void v7_barrier() {
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+}
+
+#define V7_RING 3
+
+
 template <int NW>   // waves per workgroup: 4 (BM=128, 2 WGs/CU) or 8 (BM=256)
 __global__ __launch_bounds__(NW * 64) void attn_fwd_v6_kernel(
     const u16* __restrict__ q, const u16* __restrict__ k,
@@ -721,6 +737,9 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     long k_sb, long k_sh, long k_ss,
     long v_sb, long v_sh, long v_ss,
     int n_heads, int gqa_group, int S, float scale) {
+  // v7-style staging: K rows + V rows by glds into 3-slot rings, K^T
+  // built LDS->LDS at tile end, ONE raw barrier per tile, counted vmcnt.
+  constexpr int NT = 512;
   const int qb = gridDim.x - 1 - blockIdx.x;
   const int h = blockIdx.y;
   const int b = blockIdx.z;
@@ -732,14 +751,18 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
   const int hi = lane >> 5;
   const float scale2 = scale * 1.4426950408889634f;
 
-  __shared__ __attribute__((aligned(16))) u16 lds[2][BQ_BUF_U16];
+  __shared__ __attribute__((aligned(16))) u16
+      lds[V7_RING * 2 * V6_K_U16 + 2 * V6_V_U16];
+#define DQ_K(SLOT) (&lds[(SLOT) * V6_K_U16])
+#define DQ_VR(SLOT) (&lds[(V7_RING + (SLOT)) * V6_K_U16])
+#define DQ_KT(P) (reinterpret_cast<char*>(&lds[2 * V7_RING * V6_K_U16]) \
+                  + (P) * (V6_V_U16 * 2))
 
   const int q0w = qb * 256 + wid * 32;
   const int qrow = q0w + low;
 
   union F8 { bf16x8 v; uint4 u; u16 h[8]; };
   const u16* qptr = q + (long)b * q_sb + (long)h * q_sh + (long)qrow * q_ss;
-  // dO is produced contiguous [B,H,S,D] by autograd (we .contiguous() it)
   const u16* doptr = dout + (((long)b * n_heads + h) * S + qrow) * ATTN_D;
   F8 qf[8], dof[8];
 #pragma unroll
@@ -759,86 +782,66 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
 
   const u16* kbase = k + (long)b * k_sb + (long)hkv * k_sh;
   const u16* vbase = v + (long)b * v_sb + (long)hkv * v_sh;
-  const int n_tiles = (qb + 1) * 4;           // (qb*256+256)/64
+  const int n_tiles = (qb + 1) * 4;
 
-  char* lds0 = reinterpret_cast<char*>(&lds[0][0]);
-  char* lds1 = reinterpret_cast<char*>(&lds[1][0]);
-  const int KTOFF = BQ_K_U16 * 2;
-  const int VROFF = (BQ_K_U16 + BQ_KT_U16) * 2;
-
-  // staging: K rows + V rows by glds (swizzle-inverted source); K^T by
-  // register scatter (kv-pair u32 writes into the vt image)
   const u16* kgp[2];
   const u16* vgp[2];
-  const u16* ktp[2];
 #pragma unroll
   for (int i = 0; i < 2; ++i) {
-    const int idx = i * 512 + tid;
+    const int idx = i * NT + tid;
     const int L = idx * 16;
     const int kr = L >> 8;
     const int kin = (((L & 255) ^ ((kr & 15) << 4)) >> 1);
     kgp[i] = kbase + (long)kr * k_ss + kin;
     vgp[i] = vbase + (long)kr * v_ss + kin;
   }
-  {
-    const int rp2 = (tid >> 4) * 2;
-    const int c8 = (tid & 15) * 8;
-    ktp[0] = kbase + (long)rp2 * k_ss + c8;
-    ktp[1] = kbase + (long)(rp2 + 1) * k_ss + c8;
-  }
   const long kstep = (long)V6_BN * k_ss;
   const long vstep = (long)V6_BN * v_ss;
-  const int wu64 = __builtin_amdgcn_readfirstlane(tid >> 6);
-  uint4 ktst[2];
+  const int lslot = (tid & ~63) * 8;
 
-#define BQ_GLDS(BUFI)                                                       \
+#define DQ_GLDS(SLOT)                                                       \
   _Pragma("unroll")                                                         \
   for (int i = 0; i < 2; ++i) {                                             \
     __builtin_amdgcn_global_load_lds(                                       \
-        (const u32*)kgp[i], (u32*)&lds[BUFI][(i * 512 + wu64 * 64) * 8],    \
+        (const u32*)kgp[i], (u32*)&DQ_K(SLOT)[i * NT * 8 + lslot],          \
         16, 0, 0);                                                          \
     kgp[i] += kstep;                                                        \
     __builtin_amdgcn_global_load_lds(                                       \
-        (const u32*)vgp[i],                                                 \
-        (u32*)&lds[BUFI][VROFF / 2 + (i * 512 + wu64 * 64) * 8],            \
+        (const u32*)vgp[i], (u32*)&DQ_VR(SLOT)[i * NT * 8 + lslot],         \
         16, 0, 0);                                                          \
     vgp[i] += vstep;                                                        \
   }
-#define BQ_LOAD_KT()                                                        \
+
+  const int t_rp2 = (tid >> 4) * 2;
+  const int t_c8 = (tid & 15) * 8;
+#define DQ_TRANSPOSE(SRCSLOT, DST)                                          \
   {                                                                         \
-    ktst[0] = *reinterpret_cast<const uint4*>(ktp[0]);                      \
-    ktst[1] = *reinterpret_cast<const uint4*>(ktp[1]);                      \
-    ktp[0] += kstep;                                                        \
-    ktp[1] += kstep;                                                        \
-  }
-#define BQ_WRITE_KT(BUF)                                                    \
-  {                                                                         \
-    const int rp2_ = (tid >> 4) * 2;                                        \
-    const int c8_ = (tid & 15) * 8;                                         \
+    const char* srcb = reinterpret_cast<const char*>(DQ_K(SRCSLOT));        \
     union { uint4 u4; u16 h[8]; } va_, vb_;                                 \
-    va_.u4 = ktst[0];                                                       \
-    vb_.u4 = ktst[1];                                                       \
+    va_.u4 = *reinterpret_cast<const uint4*>(&srcb[k_byte(t_rp2, t_c8)]);   \
+    vb_.u4 = *reinterpret_cast<const uint4*>(                               \
+        &srcb[k_byte(t_rp2 + 1, t_c8)]);                                    \
     _Pragma("unroll")                                                       \
     for (int j = 0; j < 8; ++j) {                                           \
       const u32 pair_ = (u32)va_.h[j] | ((u32)vb_.h[j] << 16);              \
-      *reinterpret_cast<u32*>(&(BUF)[KTOFF + vt_byte(c8_ + j, rp2_)]) = pair_; \
+      *reinterpret_cast<u32*>(&(DST)[vt_byte(t_c8 + j, t_rp2)]) = pair_;    \
     }                                                                       \
   }
 
-  BQ_GLDS(0)
-  BQ_LOAD_KT()
-  BQ_WRITE_KT(lds0)
-  __syncthreads();
+  DQ_GLDS(0)
+  if (n_tiles > 1) DQ_GLDS(1)
+  if (n_tiles > 1) v7_wait_vmcnt4(); else v7_wait_vmcnt0();
+  v7_barrier();
+  DQ_TRANSPOSE(0, DQ_KT(0))
 
-  int cur = 0;
   for (int t = 0; t < n_tiles; ++t) {
     const int kv0 = t * V6_BN;
-    const bool have_next = (t + 1) < n_tiles;
-    if (have_next) { BQ_GLDS(cur ^ 1) BQ_LOAD_KT() }
+    const bool have2 = (t + 2) < n_tiles;
+    if (have2) DQ_GLDS((t + 2) % V7_RING)
 
-    char* kb = cur ? lds1 : lds0;
-    char* ktb = kb + KTOFF;
-    char* vb = kb + VROFF;
+    char* kb = reinterpret_cast<char*>(DQ_K(t % V7_RING));
+    char* vb = reinterpret_cast<char*>(DQ_VR(t % V7_RING));
+    char* ktb = DQ_KT(t & 1);
     const bool active = kv0 <= q0w + 31;
     const bool need_mask = kv0 + V6_BN > q0w;
 
@@ -846,46 +849,37 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
     if (active) {
 #pragma unroll
       for (int sub = 0; sub < 2; ++sub) {
-        // ---- S^T subtile: C[kv32][q32] ----
-        f32x16 s;
+        f32x16 sx, dp;
 #pragma unroll
-        for (int r = 0; r < 16; ++r) s[r] = 0.f;
+        for (int r = 0; r < 16; ++r) { sx[r] = 0.f; dp[r] = 0.f; }
 #pragma unroll
         for (int kc = 0; kc < 8; ++kc) {
           bf16x8 a = *reinterpret_cast<const bf16x8*>(
               &kb[k_byte(sub * 32 + low, kc * 16 + hi * 8)]);
-          s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qf[kc].v, s, 0, 0, 0);
-        }
-        // ---- dP^T subtile: C[kv32][q32] ----
-        f32x16 dp;
-#pragma unroll
-        for (int r = 0; r < 16; ++r) dp[r] = 0.f;
-#pragma unroll
-        for (int kc = 0; kc < 8; ++kc) {
-          bf16x8 a = *reinterpret_cast<const bf16x8*>(
+          sx = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, qf[kc].v, sx,
+                                                       0, 0, 0);
+          bf16x8 a2 = *reinterpret_cast<const bf16x8*>(
               &vb[k_byte(sub * 32 + low, kc * 16 + hi * 8)]);
-          dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, dof[kc].v, dp,
+          dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a2, dof[kc].v, dp,
                                                        0, 0, 0);
         }
-        // ---- dS = P o (dP - delta) * scale,  P = 2^(s*scale2 - lse2) ----
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
-          float x = s[r] * scale2 - lse2;
+          float x = sx[r] * scale2 - lse2;
           if (need_mask) {
             const int kvr = kv0 + sub * 32 + (r & 3) + 8 * (r >> 2) + 4 * hi;
             if (kvr > qrow) x = -1e30f;
           }
           const float p = exp2_raw(x);
-          s[r] = p * (dp[r] - dlt) * scale;
+          sx[r] = p * (dp[r] - dlt) * scale;
         }
-        // ---- repack dS -> bf16 fragments ----
 #pragma unroll
         for (int cc = 0; cc < 2; ++cc) {
           const int pb = cc * 8;
-          unsigned a0 = cvt_pk_bf16(s[pb + 0], s[pb + 1]);
-          unsigned b0 = cvt_pk_bf16(s[pb + 4], s[pb + 5]);
-          unsigned a1 = cvt_pk_bf16(s[pb + 2], s[pb + 3]);
-          unsigned b1 = cvt_pk_bf16(s[pb + 6], s[pb + 7]);
+          unsigned a0 = cvt_pk_bf16(sx[pb + 0], sx[pb + 1]);
+          unsigned b0 = cvt_pk_bf16(sx[pb + 4], sx[pb + 5]);
+          unsigned a1 = cvt_pk_bf16(sx[pb + 2], sx[pb + 3]);
+          unsigned b1 = cvt_pk_bf16(sx[pb + 6], sx[pb + 7]);
           auto r02 = __builtin_amdgcn_permlane32_swap(a0, b0, false, false);
           auto r13 = __builtin_amdgcn_permlane32_swap(a1, b1, false, false);
           PF f;
@@ -896,13 +890,6 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
           pds[sub * 2 + cc] = f;
         }
       }
-    }
-    if (have_next) {
-      char* nb = cur ? lds0 : lds1;
-      BQ_WRITE_KT(nb)
-    }
-    if (active) {
-      // ---- dQ^T += K^T @ dS over 4 kv chunks x 4 d subtiles ----
 #pragma unroll
       for (int c = 0; c < 4; ++c) {
 #pragma unroll
@@ -914,11 +901,13 @@ __global__ __launch_bounds__(512) void attn_bwd_dq_kernel(
         }
       }
     }
-    __syncthreads();
-    cur ^= 1;
+    if (t + 1 < n_tiles) {
+      if (have2) v7_wait_vmcnt4(); else v7_wait_vmcnt0();
+      DQ_TRANSPOSE((t + 1) % V7_RING, DQ_KT((t + 1) & 1))
+    }
+    v7_barrier();
   }
 
-  // ---- epilogue: dq [B,H,S,D] contiguous ----
   u16* dqrow = dq + (((long)b * n_heads + h) * S + qrow) * ATTN_D;
 #pragma unroll
   for (int ds = 0; ds < 4; ++ds) {
@@ -1455,19 +1444,6 @@ extern "C" int attn_bwd(void* stream, const void* q, const void* k,
 //     vmcnt making slot t+1 visible for the transpose, one at tile end
 //     guarding ring reuse.
 // ===========================================================================
-
-__device__ __forceinline__ void v7_wait_vmcnt4() {
-  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-}
-__device__ __forceinline__ void v7_wait_vmcnt0() {
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-}
-__device__ __forceinline__ void v7_barrier() {
-  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-  __builtin_amdgcn_s_barrier();
-}
-
-#define V7_RING 3
 
 __global__ __launch_bounds__(512) void attn_fwd_v7_kernel(
     const u16* __restrict__ q, const u16* __restrict__ k,
