@@ -292,7 +292,15 @@ def split_ep(full_named: Dict, cfg, ep: int) -> List[Dict]:
 # -- ZeRO-1: per-rank moment shards <-> full --------------------------------
 
 def zero_chunk_bounds(total: int, dp: int) -> List:
-    """The exact shard math training.py uses (aligned ceil-div chunks)."""
+    """The exact shard math training.py uses: the reduce-scatter layout
+    (equal ALIGN-floored chunks, tail on the last rank) when it divides,
+    else the legacy aligned ceil-div chunks."""
+    from ..parallel.flat import ALIGN
+    chunk_lo = (total // dp) // ALIGN * ALIGN
+    if chunk_lo > 0:
+        return [(i * chunk_lo,
+                 (i + 1) * chunk_lo if i < dp - 1 else total)
+                for i in range(dp)]
     chunk = _aligned(-(-total // dp))
     return [(min(i * chunk, total), min((i + 1) * chunk, total))
             for i in range(dp)]
@@ -304,7 +312,14 @@ def merge_zero(states: List[dict]) -> dict:
     rank's chunk and concatenate back to full."""
     full = dict(states[0])
     total = states[0]["flat_param"].numel()
-    bounds = zero_chunk_bounds(total, len(states))
+    # bounds from the ACTUAL shard lengths (works for both the RS layout
+    # and the legacy ceil-div layout — checkpoints carry the truth)
+    lens = [st["opt"]["p32"].numel() for st in states]
+    offs = [0]
+    for n in lens:
+        offs.append(offs[-1] + n)
+    assert offs[-1] == total, f"zero shards sum {offs[-1]} != {total}"
+    bounds = list(zip(offs[:-1], offs[1:]))
     opt = {"step": states[0]["opt"]["step"]}
     for key in ("p32", "m", "v"):
         buf = torch.zeros(total, dtype=torch.float32)
