@@ -204,3 +204,22 @@ def test_attn_native_backward_matches_ref():
             ok = torch.allclose(a.grad.float(), b2.grad.float(), atol=7e-2,
                                 rtol=5e-2)
             assert ok, f"{name} max err {err} at B{B} H{H} HKV{HKV} S{S}"
+
+
+def test_attn_fwd_fallback_shapes():
+    """Dispatch coverage: S%128-only -> NW4 kernel; S%64-only (equal
+    heads) -> v5 kernel. Both against library SDPA."""
+    from trainingjob_operator_amd.ops.attention import flash_attention_fwd_only
+    for B, H, HKV, S in ((1, 4, 2, 384),    # NW4 (S%256!=0, %128==0), GQA
+                         (1, 4, 4, 192)):   # v5 (S%128!=0, %64==0)
+        q = (torch.randn(B, H, S, 128, device=DEV) * 0.5).to(torch.bfloat16)
+        k = (torch.randn(B, HKV, S, 128, device=DEV) * 0.5).to(torch.bfloat16)
+        v = (torch.randn(B, HKV, S, 128, device=DEV) * 0.5).to(torch.bfloat16)
+        if HKV != H and S % 128 != 0:
+            k = k.repeat_interleave(H // HKV, dim=1)
+            v = v.repeat_interleave(H // HKV, dim=1)
+        out, _ = flash_attention_fwd_only(q, k, v)
+        ref = torch.nn.functional.scaled_dot_product_attention(
+            q, k, v, is_causal=True, enable_gqa=k.shape[1] != H)
+        assert torch.allclose(out.float(), ref.float(), atol=3e-2,
+                              rtol=2e-2), f"S={S}"
