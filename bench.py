@@ -64,6 +64,9 @@ def main() -> None:
     ap.add_argument("--bucket-mb", type=int, default=128)
     ap.add_argument("--tp", type=int, default=1,
                     help="tensor-parallel degree (world = dp x tp)")
+    ap.add_argument("--fp8", action="store_true",
+                    help="fp8 (e4m3fn) forward projection GEMMs — opt-in; "
+                         "the headline bench stays bf16")
     ap.add_argument("--graphs", action="store_true",
                     help="EXPERIMENTAL: hipGraph step capture (replay of "
                          "large-seq graphs currently faults in the ROCm "
@@ -100,6 +103,7 @@ def main() -> None:
         checkpoint_activations=args.checkpoint_activations,
         bucket_bytes=args.bucket_mb << 20,
         use_graphs=on_gpu and args.graphs,
+        fp8_projections=on_gpu and args.fp8,
         tp_size=args.tp,
     )
     log(f"[bench] rank {ctx.rank}/{n_gpus} model={model} "

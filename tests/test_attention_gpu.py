@@ -223,3 +223,26 @@ def test_attn_fwd_fallback_shapes():
             q, k, v, is_causal=True, enable_gqa=k.shape[1] != H)
         assert torch.allclose(out.float(), ref.float(), atol=3e-2,
                               rtol=2e-2), f"S={S}"
+
+
+def test_fp8_projections_loss_parity(monkeypatch):
+    """Opt-in fp8 forward projections: short-horizon loss parity vs bf16
+    (the gate for publishing an fp8 config)."""
+    import torch as t
+    from trainingjob_operator_amd.training import TrainConfig, Trainer
+    monkeypatch.delenv("AITJ_FP8_PROJ", raising=False)
+    t.manual_seed(0)
+    cfg = TrainConfig(model="llama-smoke", micro_batch=1, grad_accum=1,
+                      seq_len=512, lr=1e-4)
+    tr_bf = Trainer(cfg)
+    l_bf = [float(tr_bf.train_step()) for _ in range(6)]
+    del tr_bf
+    t.cuda.empty_cache()
+    t.manual_seed(0)
+    cfg8 = TrainConfig(model="llama-smoke", micro_batch=1, grad_accum=1,
+                      seq_len=512, lr=1e-4, fp8_projections=True)
+    tr_f8 = Trainer(cfg8)
+    l_f8 = [float(tr_f8.train_step()) for _ in range(6)]
+    monkeypatch.delenv("AITJ_FP8_PROJ", raising=False)
+    for a, b in zip(l_bf, l_f8):
+        assert abs(a - b) < 0.15, (l_bf, l_f8)

@@ -60,6 +60,15 @@ def _sdpa(q, k, v, enable_gqa=False, is_causal=True):
 
 from ..ops import (apply_rope, fused_cross_entropy, fused_rmsnorm,
                    make_inv_freq, swiglu_packed)
+from ..ops.fp8 import fp8_enabled, fp8_linear
+
+
+def _proj(x, linear):
+    """Projection GEMM: fp8 forward (AITJ_FP8_PROJ=1, gfx950 e4m3fn via
+    hipBLASLt) or the regular bf16 path."""
+    if fp8_enabled() and x.is_cuda:
+        return fp8_linear(x, linear.weight)
+    return linear(x)
 from .config import LlamaConfig
 
 
@@ -79,7 +88,7 @@ class Attention(nn.Module):
     def forward(self, x: torch.Tensor, inv_freq: torch.Tensor) -> torch.Tensor:
         B, S, H = x.shape
         cfg = self.cfg
-        qkv = self.qkv_proj(x)
+        qkv = _proj(x, self.qkv_proj)
         q, k, v = qkv.split([self.q_size, self.kv_size, self.kv_size],
                             dim=-1)
         # q/k go through the RoPE kernel, which wants dense rows
@@ -104,7 +113,7 @@ class Attention(nn.Module):
             o = _sdpa(q, k, v,
                       enable_gqa=cfg.num_heads != cfg.num_kv_heads)
         o = o.transpose(1, 2).reshape(B, S, cfg.num_heads * cfg.head_dim)
-        return self.o_proj(o)
+        return _proj(o, self.o_proj)
 
 
 class MLP(nn.Module):
@@ -117,7 +126,8 @@ class MLP(nn.Module):
         self.down_proj = nn.Linear(FF, H, bias=False)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.down_proj(swiglu_packed(self.gate_up_proj(x)))
+        return _proj(swiglu_packed(_proj(x, self.gate_up_proj)),
+                     self.down_proj)
 
 
 class Block(nn.Module):

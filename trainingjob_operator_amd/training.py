@@ -7,6 +7,7 @@ flat AdamW (optim.py).
 """
 from __future__ import annotations
 
+import os
 from contextlib import nullcontext
 from dataclasses import dataclass
 from typing import Iterator, Optional, Tuple
@@ -54,6 +55,10 @@ class TrainConfig:
     # group (12 bytes/param -> 12/dp); each rank updates its slice, then
     # the updated bf16 params are re-assembled across the group
     zero1: bool = False
+    # fp8 (e4m3fn) projection GEMMs — forward only, per-tensor scales,
+    # backward stays bf16. OPT-IN: the flagship bench dtype stays bf16;
+    # publish fp8 numbers as a separate config.
+    fp8_projections: bool = False
     # bf16 first/second AdamW moments: ~29% less optimizer HBM traffic and
     # half the checkpoint moment bytes. Default OFF until loss-curve parity
     # is validated for the target run (tests cover short-horizon parity).
@@ -113,6 +118,8 @@ def synthetic_batches(cfg: TrainConfig, device: torch.device,
 class Trainer:
     def __init__(self, cfg: TrainConfig, ctx: Optional[DistContext] = None,
                  device: Optional[torch.device] = None):
+        if cfg.fp8_projections:
+            os.environ["AITJ_FP8_PROJ"] = "1"
         from .models.moe_llama import MoELlamaConfig
         if isinstance(cfg.model_config, MoELlamaConfig):
             raise ValueError(
