@@ -67,7 +67,7 @@ def _proj(x, linear):
     """Projection GEMM: fp8 forward (AITJ_FP8_PROJ=1, gfx950 e4m3fn via
     hipBLASLt) or the regular bf16 path."""
     if fp8_enabled() and x.is_cuda:
-        return fp8_linear(x, linear.weight)
+        return fp8_linear(x, linear)
     return linear(x)
 from .config import LlamaConfig
 

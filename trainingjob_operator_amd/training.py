@@ -245,6 +245,9 @@ class Trainer:
                           in_graph_capture=in_graph)
         if self._zero_shards is not None:
             self._zero_allgather_params()
+        if self.cfg.fp8_projections:
+            from .ops import fp8
+            fp8.bump_version()     # weights changed: refresh fp8 images
         self.opt.zero_grad()
         return loss
 
