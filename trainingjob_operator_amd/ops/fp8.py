@@ -28,9 +28,10 @@ def fp8_enabled() -> bool:
 
 def _quantize(t: torch.Tensor):
     """Per-tensor amax scaling to e4m3fn; returns (fp8 tensor, scale)."""
-    amax = t.detach().abs().amax().float().clamp(min=1e-12)
-    scale = amax / _E4M3_MAX
-    q = (t.float() / scale).clamp(-_E4M3_MAX, _E4M3_MAX) \
+    td = t.detach()    # the Function supplies its own backward; a cached
+    amax = td.abs().amax().float().clamp(min=1e-12)   # wq must carry NO
+    scale = amax / _E4M3_MAX                          # autograd history
+    q = (td.float() / scale).clamp(-_E4M3_MAX, _E4M3_MAX) \
         .to(torch.float8_e4m3fn)
     return q, scale
 
