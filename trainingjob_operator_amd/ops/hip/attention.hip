@@ -1379,6 +1379,8 @@ __global__ __launch_bounds__(512) void attn_bwd_dk_kernel(
   }
 }
 
+
+
 // ---------------------------------------------------------------------------
 // launchers
 // ---------------------------------------------------------------------------
