@@ -557,3 +557,69 @@ def test_moe_param_formula_and_8x7b():
     assert 12e9 < MOE_8X7B.active_params < 14e9
     from trainingjob_operator_amd.models.config import CONFIGS
     assert CONFIGS["moe-8x7b"] is MOE_8X7B
+
+
+def _spmoe_worker(rank, world, port, outdir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        import dataclasses
+        import json
+
+        from trainingjob_operator_amd.models.moe_llama import (
+            MOE_TINY, MoELlamaModel,
+        )
+        from trainingjob_operator_amd.parallel.sp import SPMoEModel
+        from trainingjob_operator_amd.parallel.tp import shard_from
+        cfg = dataclasses.replace(MOE_TINY, aux_loss_coef=0.0)
+        torch.manual_seed(44)
+        # the reference must be UNsharded: give it a single-rank ep group
+        # (group=None means the whole world once dist is initialized)
+        selfgroups = [dist.new_group([r]) for r in range(world)]
+        full = MoELlamaModel(cfg, ep_group=selfgroups[rank])
+        spm = SPMoEModel(cfg, group=None, ep_group=None)
+        spm.shard_from_full(full)
+
+        g = torch.Generator().manual_seed(10)
+        tokens = torch.randint(0, cfg.vocab_size, (2, 16), generator=g)
+        loss = spm(tokens, tokens)
+        loss.backward()
+        spm.allreduce_sp_grads()
+        ref = full(tokens, tokens)
+        ref.backward()
+        assert torch.allclose(loss, ref, atol=1e-5), (loss.item(),
+                                                      ref.item())
+        b, fb = spm.blocks[0], full.blocks[0]
+        # TP-sharded attention grads match the full slice
+        assert torch.allclose(
+            b.attn.o_proj.weight.grad,
+            shard_from(fb.attn.o_proj.weight.grad, 1, None), atol=1e-4)
+        # seq-sharded (summed) grads match
+        assert torch.allclose(b.input_norm_weight.grad,
+                              fb.input_norm_weight.grad, atol=1e-4)
+        assert torch.allclose(b.moe.router.weight.grad,
+                              fb.moe.router.weight.grad, atol=1e-4)
+        # this rank's experts got the complete (both shards') grads
+        per = b.moe.experts_per_rank
+        ex = b.moe.experts[0]
+        src = fb.moe.experts[rank * per]
+        assert torch.allclose(ex.down_proj.weight.grad,
+                              src.down_proj.weight.grad, atol=1e-4)
+        with open(os.path.join(outdir, f"spmoe{rank}.json"), "w") as f:
+            json.dump(float(loss), f)
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_sp_moe_matches_unsharded(tmp_path):
+    """SP x EP: seq-sharded attention + EP-dispatched experts (the last
+    cell of the parallelism matrix) vs the unsharded MoE model."""
+    import json
+    port = _free_port()
+    mp.spawn(_spmoe_worker, args=(2, port, str(tmp_path)), nprocs=2,
+             join=True)
+    a = json.load(open(os.path.join(str(tmp_path), "spmoe0.json")))
+    b = json.load(open(os.path.join(str(tmp_path), "spmoe1.json")))
+    assert a == pytest.approx(b)
