@@ -39,24 +39,42 @@ class KVCache:
         self.len = 0
 
 
-def _rope_at(x: torch.Tensor, inv_freq: torch.Tensor,
-             pos0: int) -> torch.Tensor:
-    """Neox half-rotation with explicit positions pos0..pos0+S-1.
-    x: [B, S, n_heads, D]."""
-    B, S, nh, D = x.shape
-    half = D // 2
-    pos = (torch.arange(S, device=x.device) + pos0).float()
+def _rope_tables(inv_freq: torch.Tensor, pos0: int, S: int, device):
+    """cos/sin for positions pos0..pos0+S-1, computed ONCE per step (the
+    round-1 decode rebuilt them per LAYER: ~10 extra launches x n_layers
+    per decoded token)."""
+    pos = (torch.arange(S, device=device) + pos0).float()
     ang = pos[:, None] * inv_freq[None, :].float()       # [S, half]
-    cos = ang.cos()[None, :, None, :]
-    sin = ang.sin()[None, :, None, :]
+    return ang.cos()[None, :, None, :], ang.sin()[None, :, None, :]
+
+
+def _rope_at(x: torch.Tensor, cos: torch.Tensor,
+             sin: torch.Tensor) -> torch.Tensor:
+    """Neox half-rotation with precomputed tables. x: [B, S, n_heads, D]."""
+    half = x.shape[-1] // 2
     xf = x.float()
     x1, x2 = xf[..., :half], xf[..., half:]
     return torch.cat([x1 * cos - x2 * sin,
                       x1 * sin + x2 * cos], dim=-1).to(x.dtype)
 
 
-def _attn_cached(attn, x: torch.Tensor, inv_freq: torch.Tensor,
-                 ck: torch.Tensor, cv: torch.Tensor,
+def _decode_attn(q, kk, vv, n_kv: int):
+    """One-token attention over the cache WITHOUT expanding KV heads:
+    grouped einsum on views (the library GQA path repeat_interleaves the
+    cache -- hundreds of MB per layer at batch 8). q: [B, H, 1, D]."""
+    import math
+    B, H, _, D = q.shape
+    G = H // n_kv
+    qg = q.reshape(B, n_kv, G, D)
+    scores = torch.einsum("bkgd,bksd->bkgs", qg.float(), kk.float())
+    scores = scores / math.sqrt(D)
+    p = torch.softmax(scores, dim=-1)
+    o = torch.einsum("bkgs,bksd->bkgd", p, vv.float())
+    return o.reshape(B, H, 1, D).to(q.dtype)
+
+
+def _attn_cached(attn, x: torch.Tensor, cos: torch.Tensor,
+                 sin: torch.Tensor, ck: torch.Tensor, cv: torch.Tensor,
                  pos0: int) -> torch.Tensor:
     cfg = attn.cfg
     B, S, _ = x.shape
@@ -65,18 +83,19 @@ def _attn_cached(attn, x: torch.Tensor, inv_freq: torch.Tensor,
     q = q.reshape(B, S, cfg.num_heads, cfg.head_dim)
     k = k.reshape(B, S, cfg.num_kv_heads, cfg.head_dim)
     v = v.reshape(B, S, cfg.num_kv_heads, cfg.head_dim)
-    q = _rope_at(q, inv_freq, pos0).transpose(1, 2)      # [B, nh, S, D]
-    k = _rope_at(k, inv_freq, pos0).transpose(1, 2)
+    q = _rope_at(q, cos, sin).transpose(1, 2)            # [B, nh, S, D]
+    k = _rope_at(k, cos, sin).transpose(1, 2)
     v = v.transpose(1, 2)
     ck[:, :, pos0:pos0 + S] = k
     cv[:, :, pos0:pos0 + S] = v
     kk = ck[:, :, :pos0 + S]
     vv = cv[:, :, :pos0 + S]
     gqa = cfg.num_heads != cfg.num_kv_heads
-    # prefill (S>1, empty cache) is causal among the new tokens; a decode
-    # step (S==1) attends to everything cached
-    o = _sdpa(q, kk, vv, enable_gqa=gqa) if S > 1 else \
-        _sdpa(q, kk, vv, enable_gqa=gqa, is_causal=False)
+    if S > 1:
+        # prefill: causal among the new tokens
+        o = _sdpa(q, kk, vv, enable_gqa=gqa)
+    else:
+        o = _decode_attn(q, kk, vv, cfg.num_kv_heads)
     o = o.transpose(1, 2).reshape(B, S, cfg.num_heads * cfg.head_dim)
     return attn.o_proj(o)
 
@@ -89,11 +108,13 @@ def _forward_cached(model: LlamaModel, tokens: torch.Tensor,
     cfg = model.cfg
     pos0 = cache.len
     x = model.embed(tokens)
+    cos, sin = _rope_tables(model.inv_freq, pos0, tokens.shape[1],
+                            tokens.device)
     residual = None
     for li, blk in enumerate(model.blocks):
         normed, residual = fused_rmsnorm(x, blk.input_norm_weight,
                                          residual, cfg.norm_eps)
-        attn_out = _attn_cached(blk.attn, normed, model.inv_freq,
+        attn_out = _attn_cached(blk.attn, normed, cos, sin,
                                 cache.k[li], cache.v[li], pos0)
         normed, residual = fused_rmsnorm(attn_out,
                                          blk.post_attn_norm_weight,
@@ -141,8 +162,11 @@ def generate(model: LlamaModel, tokens: torch.Tensor,
         out = tokens
         done = torch.zeros(B, dtype=torch.bool, device=tokens.device)
         for _ in range(max_new_tokens):
-            nxt = _sample(logits[:, -1].cpu(), temperature, top_k,
-                          gen).to(tokens.device)
+            if temperature <= 0.0:
+                nxt = logits[:, -1].argmax(dim=-1)   # greedy: on device,
+            else:                                    # no per-token sync
+                nxt = _sample(logits[:, -1].cpu(), temperature, top_k,
+                              gen).to(tokens.device)
             if eos_token is not None:
                 nxt = torch.where(done, torch.full_like(nxt, eos_token),
                                   nxt)
