@@ -58,6 +58,25 @@ def _rope_at(x: torch.Tensor, cos: torch.Tensor,
                       x1 * sin + x2 * cos], dim=-1).to(x.dtype)
 
 
+def _rope_positions(x: torch.Tensor, inv_freq: torch.Tensor, S: int,
+                    pos0: int) -> torch.Tensor:
+    """RoPE at positions pos0 + (flat % S): the native kernel on GPU
+    (one launch vs ~8 torch ops), torch fallback on CPU.
+    x: [B, S, n_heads, D] bf16-contiguous for the kernel path."""
+    if x.is_cuda and x.dtype == torch.bfloat16 and x.shape[-1] % 16 == 0:
+        from ..ops import native
+        lib = native.load(require=True)
+        xt = x.reshape(-1, x.shape[-2], x.shape[-1]).contiguous()
+        out = torch.empty_like(xt)
+        rc = lib.rope_at(native.stream_ptr(), xt.data_ptr(),
+                         out.data_ptr(), inv_freq.data_ptr(),
+                         xt.shape[0], xt.shape[1], S, xt.shape[2], 1.0,
+                         pos0)
+        native.check_rc(rc, "rope_at", f"D={x.shape[-1]}")
+        return out.reshape(x.shape)
+    return None
+
+
 def _decode_attn(q, kk, vv, n_kv: int):
     """One-token attention over the cache WITHOUT expanding KV heads:
     grouped einsum on views (the library GQA path repeat_interleaves the
@@ -83,8 +102,15 @@ def _attn_cached(attn, x: torch.Tensor, cos: torch.Tensor,
     q = q.reshape(B, S, cfg.num_heads, cfg.head_dim)
     k = k.reshape(B, S, cfg.num_kv_heads, cfg.head_dim)
     v = v.reshape(B, S, cfg.num_kv_heads, cfg.head_dim)
-    q = _rope_at(q, cos, sin).transpose(1, 2)            # [B, nh, S, D]
-    k = _rope_at(k, cos, sin).transpose(1, 2)
+    qr = _rope_positions(q, attn._inv_freq_dec, S, pos0) \
+        if hasattr(attn, "_inv_freq_dec") else None
+    if qr is not None:
+        q = qr.transpose(1, 2)
+        k = _rope_positions(k, attn._inv_freq_dec, S,
+                            pos0).transpose(1, 2)
+    else:
+        q = _rope_at(q, cos, sin).transpose(1, 2)        # [B, nh, S, D]
+        k = _rope_at(k, cos, sin).transpose(1, 2)
     v = v.transpose(1, 2)
     ck[:, :, pos0:pos0 + S] = k
     cv[:, :, pos0:pos0 + S] = v
@@ -108,10 +134,17 @@ def _forward_cached(model: LlamaModel, tokens: torch.Tensor,
     cfg = model.cfg
     pos0 = cache.len
     x = model.embed(tokens)
-    cos, sin = _rope_tables(model.inv_freq, pos0, tokens.shape[1],
-                            tokens.device)
+    use_kernel = tokens.is_cuda and model.inv_freq.dtype == torch.float32
+    cos = sin = None
+    if not use_kernel:
+        cos, sin = _rope_tables(model.inv_freq, pos0, tokens.shape[1],
+                                tokens.device)
     residual = None
     for li, blk in enumerate(model.blocks):
+        if use_kernel:
+            blk.attn._inv_freq_dec = model.inv_freq
+        elif hasattr(blk.attn, "_inv_freq_dec"):
+            del blk.attn._inv_freq_dec
         normed, residual = fused_rmsnorm(x, blk.input_norm_weight,
                                          residual, cfg.norm_eps)
         attn_out = _attn_cached(blk.attn, normed, cos, sin,

@@ -349,7 +349,7 @@ __global__ void rmsnorm_dw_reduce_kernel(const float* __restrict__ dw_partial,
 __global__ void rope_kernel(const uint4* __restrict__ x, uint4* __restrict__ out,
                             const float* __restrict__ inv_freq,
                             long total_vec, int vec_per_half, int n_heads,
-                            int S, int D, float sign) {
+                            int S, int D, float sign, int pos0) {
   const int HALF = D / 2;
   for (long g = blockIdx.x * (long)blockDim.x + threadIdx.x; g < total_vec;
        g += gridDim.x * (long)blockDim.x) {
@@ -359,7 +359,7 @@ __global__ void rope_kernel(const uint4* __restrict__ x, uint4* __restrict__ out
     const int i0 = (int)(g % per_head) * 8;          // dim offset in [0, HALF)
     const long t = head_g / n_heads;                 // token index
     const int h = (int)(head_g % n_heads);
-    const int pos = (int)(t % S);
+    const int pos = pos0 + (int)(t % S);
     const long base = (t * n_heads + h) * (D / 8);   // vec8 index of head start
     BF8 a; a.v = x[base + i0 / 8];
     BF8 b; b.v = x[base + (HALF + i0) / 8];
@@ -1030,7 +1030,24 @@ int rope(void* stream, const void* x, void* out, const void* inv_freq,
   dim3 grid(elementwise_grid(total_vec)), block(BLOCK);
   hipLaunchKernelGGL(rope_kernel, grid, block, 0, STREAM, (const uint4*)x,
                      (uint4*)out, (const float*)inv_freq, total_vec,
-                     vec_per_half, n_heads, S, D, sign);
+                     vec_per_half, n_heads, S, D, sign, 0);
+  return 0;
+}
+
+// decode variant: every token sits at position pos0 + (t % S) — a KV-cached
+// step passes S = new-token count and the cache fill as pos0
+int rope_at(void* stream, const void* x, void* out, const void* inv_freq,
+            long n_tokens, int n_heads, int S, int D, float sign,
+            int pos0) {
+  if (D <= 0 || D % 16 != 0 || n_heads <= 0 || S <= 0 || pos0 < 0)
+    return -1;
+  if (n_tokens <= 0) return 0;
+  const int vec_per_half = (D / 2) / 8;
+  const long total_vec = n_tokens * (long)n_heads * vec_per_half;
+  dim3 grid(elementwise_grid(total_vec)), block(BLOCK);
+  hipLaunchKernelGGL(rope_kernel, grid, block, 0, STREAM, (const uint4*)x,
+                     (uint4*)out, (const float*)inv_freq, total_vec,
+                     vec_per_half, n_heads, S, D, sign, pos0);
   return 0;
 }
 
