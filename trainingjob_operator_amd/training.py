@@ -173,7 +173,10 @@ class Trainer:
             from .parallel.flat import ALIGN, _aligned
             dp, r = self.topo.dp_size, self.topo.dp_rank
             chunk_lo = (self.store.total // dp) // ALIGN * ALIGN
-            if cfg.zero1_rs and chunk_lo > 0:
+            # the RS layout leaves non-shard grad regions UNreduced, so it
+            # cannot combine with the TP-aware clip (which reads the full
+            # flat grad); zero1+tp falls back to the broadcast layout
+            if cfg.zero1_rs and chunk_lo > 0 and self.topo.tp_size == 1:
                 # equal chunks + tail on the last rank: the equal region
                 # [0, dp*chunk) goes through ONE reduce_scatter_tensor /
                 # all_gather_into_tensor pair on RCCL; the small tail
