@@ -298,3 +298,34 @@ def test_dispatch_rows_backward_no_atomics():
                           atol=1e-2)
     assert torch.allclose(x.grad.float().cpu(), xa.grad, atol=2e-2,
                           rtol=2e-2)
+
+
+def test_gemv_bf16_matches_fp32():
+    """Decode GEMV (wave-per-row, hip/ops.hip gemv_bf16) vs an fp32
+    matmul on the decode projection shapes (llama3-8b qkv/o/gate_up/down
+    plus a small odd-M case)."""
+    from trainingjob_operator_amd.ops import decode_linear
+    torch.manual_seed(17)
+    for M, K in [(6144, 4096), (4096, 4096), (1026, 512), (512, 14336)]:
+        for N in (1, 3, 8):
+            w = _mk((M, K))
+            x = _mk((N, K))
+            y = decode_linear(x, w)
+            ref = x.float() @ w.float().t()
+            err = (y.float() - ref).abs().max().item()
+            tol = ref.abs().max().item() * 2e-2 + 2e-2
+            assert err < tol, f"M={M} K={K} N={N}: {err} vs {tol}"
+
+
+def test_gemv_bf16_fallback_shapes():
+    """Unsupported shapes (K % 512 != 0, > 8 rows) fall back to the
+    library GEMM and stay correct."""
+    from trainingjob_operator_amd.ops import decode_linear
+    torch.manual_seed(18)
+    for M, K, N in [(256, 320, 2), (256, 512, 17)]:
+        w = _mk((M, K))
+        x = _mk((N, K))
+        y = decode_linear(x, w)
+        ref = x.float() @ w.float().t()
+        assert (y.float() - ref).abs().max().item() < \
+            ref.abs().max().item() * 2e-2 + 2e-2

@@ -1,0 +1,145 @@
+"""hipGraph-captured one-token decode step for dense Llama serving.
+
+The eager decode step issues ~650 kernels per token (projections, rope,
+cache writes, attention einsums, norms) and the MI355X sits ~80% idle in
+launch gaps (profiles/r02_serving_profile.md). This module captures the
+whole step in one hipGraph and replays it per token. Everything the step
+needs per token lives on the DEVICE so a replay takes no host input:
+
+  * `cur`   — the token ids, copied in before each replay;
+  * `pos_t` — the fill position as a device tensor: the RoPE angles and
+    the attention mask are computed from it in-graph, the KV write is an
+    `index_copy_` with it, and the graph increments it at the end, so
+    consecutive replays decode consecutive positions;
+  * attention runs over the cache's full `max_len` with an additive
+    `-inf` mask beyond `pos_t` (fixed shapes are what make the step
+    capturable; the masked tail costs bandwidth, not correctness).
+
+Opt-in via AITJ_DECODE_GRAPH=1 (generate() checks it): hipGraph capture
+of a large training step hung on ROCm 7.2 (ROADMAP.md), so the default
+stays on the eager path and the graph path is validated explicitly under
+a timeout in scripts/graphcheck.py rather than in the driver-run suite.
+
+Reference parity note: the reference operator has no serving path at all
+ — this is framework-side capability its users get on
+top.
+"""
+from __future__ import annotations
+
+import torch
+
+from ..ops import decode_linear, fused_rmsnorm, swiglu_packed
+from .generate import KVCache, _mlp_cached
+from .llama import LlamaModel
+
+
+def _rope_dev(x: torch.Tensor, cos: torch.Tensor,
+              sin: torch.Tensor) -> torch.Tensor:
+    """Neox half-rotation, tables shaped [1, 1, 1, half]; x [B,1,nh,D]."""
+    half = x.shape[-1] // 2
+    xf = x.float()
+    x1, x2 = xf[..., :half], xf[..., half:]
+    return torch.cat([x1 * cos - x2 * sin,
+                      x1 * sin + x2 * cos], dim=-1).to(x.dtype)
+
+
+def _step_static(model: LlamaModel, cur: torch.Tensor, cache: KVCache,
+                 pos_t: torch.Tensor, arange: torch.Tensor) -> torch.Tensor:
+    """One decode step with device-resident position: returns logits
+    [B, V]. Every shape is independent of the position, so the whole
+    call is stream-capturable."""
+    import math
+    cfg = model.cfg
+    B = cur.shape[0]
+    x = model.embed(cur)                                   # [B, 1, H]
+    ang = pos_t.float()[:, None] * model.inv_freq[None, :].float()
+    cos = ang.cos()[None, :, None, :]                      # [1,1,1,half]
+    sin = ang.sin()[None, :, None, :]
+    # additive mask over the whole cache: -inf beyond the current position
+    mask = torch.where(arange[None, None, None, :] > pos_t[0],
+                       float("-inf"), 0.0)
+    residual = None
+    for li, blk in enumerate(model.blocks):
+        attn = blk.attn
+        normed, residual = fused_rmsnorm(x, blk.input_norm_weight,
+                                         residual, cfg.norm_eps)
+        qkv = decode_linear(normed, attn.qkv_proj.weight)
+        q, k, v = qkv.split([attn.q_size, attn.kv_size, attn.kv_size],
+                            dim=-1)
+        q = q.reshape(B, 1, cfg.num_heads, cfg.head_dim)
+        k = k.reshape(B, 1, cfg.num_kv_heads, cfg.head_dim)
+        v = v.reshape(B, 1, cfg.num_kv_heads, cfg.head_dim)
+        q = _rope_dev(q, cos, sin).transpose(1, 2)         # [B, nh, 1, D]
+        k = _rope_dev(k, cos, sin).transpose(1, 2)
+        v = v.transpose(1, 2)
+        cache.k[li].index_copy_(2, pos_t, k)
+        cache.v[li].index_copy_(2, pos_t, v)
+        kk, vv = cache.k[li], cache.v[li]                  # full max_len
+        G = cfg.num_heads // cfg.num_kv_heads
+        qg = q.reshape(B, cfg.num_kv_heads, G, cfg.head_dim)
+        scores = torch.einsum("bkgd,bksd->bkgs", qg.float(), kk.float())
+        scores = scores / math.sqrt(cfg.head_dim) + mask
+        p = torch.softmax(scores, dim=-1)
+        o = torch.einsum("bkgs,bksd->bkgd", p, vv.float())
+        o = o.reshape(B, 1, cfg.num_heads * cfg.head_dim).to(x.dtype)
+        attn_out = decode_linear(o, attn.o_proj.weight)
+        normed, residual = fused_rmsnorm(attn_out,
+                                         blk.post_attn_norm_weight,
+                                         residual, cfg.norm_eps)
+        x = _mlp_cached(blk.mlp, normed)
+    normed, _ = fused_rmsnorm(x, model.final_norm_weight, residual,
+                              cfg.norm_eps)
+    logits = decode_linear(normed, model.lm_head.weight)
+    pos_t.add_(1)                                          # in-graph advance
+    return logits.reshape(B, -1)
+
+
+class GraphedDecoder:
+    """Captures `_step_static` once and replays it per token.
+
+    Usage (generate() drives this): prefill eagerly, then
+        dec = GraphedDecoder(model, cache, batch)
+        dec.prime()                     # warmup + capture at cache.len
+        logits = dec.step(tokens)       # per decoded token
+    The warmup steps write scratch K/V at positions >= cache.len; real
+    replays overwrite those same positions, so the cache stays correct.
+    """
+
+    def __init__(self, model: LlamaModel, cache: KVCache, batch: int):
+        dev = model.embed.weight.device
+        self.model, self.cache = model, cache
+        self.cur = torch.zeros(batch, dtype=torch.long, device=dev)
+        self.pos_t = torch.zeros(1, dtype=torch.long, device=dev)
+        self.arange = torch.arange(cache.max_len, device=dev)
+        self.graph: torch.cuda.CUDAGraph | None = None
+        self.logits: torch.Tensor | None = None
+
+    def _run(self) -> torch.Tensor:
+        return _step_static(self.model, self.cur[:, None], self.cache,
+                            self.pos_t, self.arange)
+
+    def prime(self) -> None:
+        assert self.cache.len + 2 < self.cache.max_len, \
+            "graph warmup needs 2 free cache slots"
+        self.pos_t.fill_(self.cache.len)
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(2):                 # warmup (eager, side stream)
+                self._run()
+        torch.cuda.current_stream().wait_stream(side)
+        self.pos_t.fill_(self.cache.len)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self.logits = self._run()
+        self.graph = g
+        # capture records without executing, but reset defensively: the
+        # first replay must decode at exactly cache.len
+        self.pos_t.fill_(self.cache.len)
+
+    def step(self, tokens: torch.Tensor) -> torch.Tensor:
+        """tokens [B] -> logits [B, V]; advances the cache position."""
+        self.cur.copy_(tokens)
+        self.graph.replay()
+        self.cache.len += 1
+        return self.logits

@@ -14,11 +14,12 @@ fused decode attention) is round-2 serving work — see ROADMAP.md.
 """
 from __future__ import annotations
 
+import os
 from typing import List, Optional
 
 import torch
 
-from ..ops import fused_rmsnorm
+from ..ops import decode_linear, fused_rmsnorm, swiglu_packed
 from .config import LlamaConfig
 from .llama import LlamaModel, _sdpa
 
@@ -97,7 +98,7 @@ def _attn_cached(attn, x: torch.Tensor, cos: torch.Tensor,
                  pos0: int) -> torch.Tensor:
     cfg = attn.cfg
     B, S, _ = x.shape
-    qkv = attn.qkv_proj(x)
+    qkv = decode_linear(x, attn.qkv_proj.weight)
     q, k, v = qkv.split([attn.q_size, attn.kv_size, attn.kv_size], dim=-1)
     q = q.reshape(B, S, cfg.num_heads, cfg.head_dim)
     k = k.reshape(B, S, cfg.num_kv_heads, cfg.head_dim)
@@ -123,7 +124,16 @@ def _attn_cached(attn, x: torch.Tensor, cos: torch.Tensor,
     else:
         o = _decode_attn(q, kk, vv, cfg.num_kv_heads)
     o = o.transpose(1, 2).reshape(B, S, cfg.num_heads * cfg.head_dim)
-    return attn.o_proj(o)
+    return decode_linear(o, attn.o_proj.weight)
+
+
+
+def _mlp_cached(mlp, x: torch.Tensor) -> torch.Tensor:
+    """SwiGLU MLP through the decode GEMV (decode_linear falls back to the
+    library GEMM above 8 rows, so prefill takes the normal path)."""
+    gu = decode_linear(x, mlp.gate_up_proj.weight)
+    return decode_linear(swiglu_packed(gu.contiguous()),
+                         mlp.down_proj.weight)
 
 
 @torch.no_grad()
@@ -154,11 +164,12 @@ def _forward_cached(model: LlamaModel, tokens: torch.Tensor,
                                          residual, cfg.norm_eps)
         # dense blocks carry .mlp; MoE blocks carry .moe (routing works
         # per token, so single-token decode steps route normally)
-        x = blk.mlp(normed) if hasattr(blk, "mlp") else blk.moe(normed)
+        x = _mlp_cached(blk.mlp, normed) if hasattr(blk, "mlp") \
+            else blk.moe(normed)
     cache.len = pos0 + tokens.shape[1]
     normed, _ = fused_rmsnorm(x, model.final_norm_weight, residual,
                               cfg.norm_eps)
-    return model.lm_head(normed)
+    return decode_linear(normed, model.lm_head.weight)
 
 
 def _sample(logits: torch.Tensor, temperature: float, top_k: int,
@@ -191,14 +202,22 @@ def generate(model: LlamaModel, tokens: torch.Tensor,
         gen = None
         if seed is not None:
             gen = torch.Generator(device="cpu").manual_seed(seed)
-        logits = _forward_cached(model, tokens, cache)
+        last = _forward_cached(model, tokens, cache)[:, -1]
+        # opt-in hipGraph decode (see decode_graph.py for why not default)
+        dec = None
+        if (os.environ.get("AITJ_DECODE_GRAPH") == "1" and tokens.is_cuda
+                and max_new_tokens > 2
+                and all(hasattr(b, "mlp") for b in model.blocks)):
+            from .decode_graph import GraphedDecoder
+            dec = GraphedDecoder(model, cache, B)
+            dec.prime()
         out = tokens
         done = torch.zeros(B, dtype=torch.bool, device=tokens.device)
         for _ in range(max_new_tokens):
             if temperature <= 0.0:
-                nxt = logits[:, -1].argmax(dim=-1)   # greedy: on device,
+                nxt = last.argmax(dim=-1)            # greedy: on device,
             else:                                    # no per-token sync
-                nxt = _sample(logits[:, -1].cpu(), temperature, top_k,
+                nxt = _sample(last.cpu(), temperature, top_k,
                               gen).to(tokens.device)
             if eos_token is not None:
                 nxt = torch.where(done, torch.full_like(nxt, eos_token),
@@ -207,7 +226,8 @@ def generate(model: LlamaModel, tokens: torch.Tensor,
             out = torch.cat([out, nxt[:, None]], dim=1)
             if eos_token is not None and bool(done.all()):
                 break
-            logits = _forward_cached(model, nxt[:, None], cache)
+            last = dec.step(nxt) if dec is not None else \
+                _forward_cached(model, nxt[:, None], cache)[:, -1]
         return out
     finally:
         if was_training:

@@ -239,6 +239,27 @@ def swiglu_packed(gu: torch.Tensor) -> torch.Tensor:
     return _SwiGLUPacked.apply(gu)
 
 
+def decode_linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
+    """y = x @ W^T for decode-sized x (<= 8 total rows): the hand-written
+    wave-per-row GEMV (hip/ops.hip gemv_bf16) instead of hipBLASLt, which
+    leaves most of HBM3E idle on these skinny shapes. Inference-only (no
+    autograd); falls back to F.linear off-GPU or on unsupported shapes."""
+    K = x.shape[-1]
+    rows = x.numel() // K
+    if (x.is_cuda and x.dtype == torch.bfloat16
+            and weight.dtype == torch.bfloat16 and 1 <= rows <= 8
+            and K % 512 == 0 and weight.stride(-1) == 1):
+        lib = _hip()
+        x2 = x.reshape(rows, K).contiguous()
+        M = weight.shape[0]
+        y = torch.empty(rows, M, dtype=torch.bfloat16, device=x.device)
+        rc = lib.gemv_bf16(native.stream_ptr(), _ptr(weight), _ptr(x2),
+                           _ptr(y), M, K, rows)
+        native.check_rc(rc, "gemv_bf16", f"M={M} K={K} N={rows}")
+        return y.reshape(*x.shape[:-1], M)
+    return torch.nn.functional.linear(x, weight)
+
+
 # ---------------------------------------------------------------------------
 # Fused cross-entropy
 # ---------------------------------------------------------------------------

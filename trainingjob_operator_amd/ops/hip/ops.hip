@@ -915,6 +915,53 @@ static inline int elementwise_grid(long nvec) {
 
 #define STREAM reinterpret_cast<hipStream_t>(stream)
 
+
+// ---------------------------------------------------------------------------
+// Decode GEMV: y[n][M] = x[n][K] @ W[M][K]^T for n <= 8 tokens (bf16 in/
+// out, fp32 accumulate). At decode batch sizes the projection GEMMs are
+// pure weight streams (W is read once per token, x is KBs); hipBLASLt's
+// skinny-GEMM kernels leave most of HBM3E idle on these shapes, so this
+// is a bandwidth kernel: one wave per output row, lanes stride the row in
+// 16-byte pieces (the whole wave reads 1 KB per iteration, coalesced), x
+// re-reads hit L1, wave-reduce via shfl_xor, lane 0 writes. N is a
+// template parameter so the per-token accumulators stay in registers.
+// ---------------------------------------------------------------------------
+template <int N>
+__global__ __launch_bounds__(256) void gemv_bf16_kernel(
+    const u16* __restrict__ w, const u16* __restrict__ x,
+    u16* __restrict__ y, int M, int K) {
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int m = blockIdx.x * 4 + wid;
+  if (m >= M) return;
+  const u16* wrow = w + (long)m * K;
+  float acc[N];
+#pragma unroll
+  for (int n = 0; n < N; ++n) acc[n] = 0.f;
+  union V8 { uint4 u; u16 h[8]; };
+#pragma unroll 4
+  for (int c0 = lane * 8; c0 < K; c0 += 64 * 8) {
+    V8 wv;
+    wv.u = *reinterpret_cast<const uint4*>(wrow + c0);
+#pragma unroll
+    for (int n = 0; n < N; ++n) {
+      V8 xv;
+      xv.u = *reinterpret_cast<const uint4*>(x + (long)n * K + c0);
+      float s = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) s += bf2f(wv.h[j]) * bf2f(xv.h[j]);
+      acc[n] += s;
+    }
+  }
+#pragma unroll
+  for (int n = 0; n < N; ++n) {
+#pragma unroll
+    for (int off = 32; off; off >>= 1)
+      acc[n] += __shfl_xor(acc[n], off, 64);
+    if (lane == 0) y[(long)n * M + m] = f2bf(acc[n]);
+  }
+}
+
 extern "C" {
 
 int hipops_arch_check() {
@@ -1229,6 +1276,27 @@ int adamw_step(void* stream, void* p32, void* m, void* v, const void* grad,
                      (const float*)normsq, n4, lr, beta1, beta2, eps,
                      weight_decay, bc1, bc2, clip, pre_scale,
                      (const float*)bc_dev);
+  return 0;
+}
+
+
+// decode GEMV (gemv_bf16_kernel): rows of W must be contiguous; x/y are
+// [N, K] / [N, M] row-major. K % 512 keeps every lane on whole 16-byte
+// pieces; callers fall back to the library GEMM otherwise.
+int gemv_bf16(void* stream, const void* w, const void* x, void* y,
+              int M, int K, int N) {
+  if (M <= 0 || K <= 0 || (K % 512) != 0 || N < 1 || N > 8) return -1;
+  dim3 grid((unsigned)((M + 3) / 4)), block(256);
+  switch (N) {
+#define GEMV_CASE(NN) \
+    case NN: \
+      hipLaunchKernelGGL((gemv_bf16_kernel<NN>), grid, block, 0, STREAM, \
+                         (const u16*)w, (const u16*)x, (u16*)y, M, K); \
+      break;
+    GEMV_CASE(1) GEMV_CASE(2) GEMV_CASE(3) GEMV_CASE(4)
+    GEMV_CASE(5) GEMV_CASE(6) GEMV_CASE(7) GEMV_CASE(8)
+#undef GEMV_CASE
+  }
   return 0;
 }
 
