@@ -28,19 +28,26 @@ from __future__ import annotations
 
 import torch
 
-from ..ops import decode_linear, fused_rmsnorm, swiglu_packed
+from ..ops import decode_attention, decode_linear, fused_rmsnorm
 from .generate import KVCache, _mlp_cached
 from .llama import LlamaModel
 
 
-def _rope_dev(x: torch.Tensor, cos: torch.Tensor,
-              sin: torch.Tensor) -> torch.Tensor:
-    """Neox half-rotation, tables shaped [1, 1, 1, half]; x [B,1,nh,D]."""
-    half = x.shape[-1] // 2
-    xf = x.float()
-    x1, x2 = xf[..., :half], xf[..., half:]
-    return torch.cat([x1 * cos - x2 * sin,
-                      x1 * sin + x2 * cos], dim=-1).to(x.dtype)
+def _rope_dev(x: torch.Tensor, inv_freq: torch.Tensor,
+              pos_t: torch.Tensor) -> torch.Tensor:
+    """Neox half-rotation at a DEVICE-resident position (one kernel per
+    tensor instead of ~10 elementwise launches; graph-replayable because
+    the kernel dereferences pos_t). x: [B, 1, nh, D] bf16."""
+    from ..ops import native
+    B, S, nh, D = x.shape
+    lib = native.load(require=True)
+    xt = x.reshape(-1, nh, D).contiguous()
+    out = torch.empty_like(xt)
+    rc = lib.rope_at_dev(native.stream_ptr(), xt.data_ptr(),
+                         out.data_ptr(), inv_freq.data_ptr(),
+                         xt.shape[0], nh, 1, D, 1.0, pos_t.data_ptr())
+    native.check_rc(rc, "rope_at_dev", f"D={D}")
+    return out.reshape(x.shape)
 
 
 def _step_static(model: LlamaModel, cur: torch.Tensor, cache: KVCache,
@@ -52,12 +59,7 @@ def _step_static(model: LlamaModel, cur: torch.Tensor, cache: KVCache,
     cfg = model.cfg
     B = cur.shape[0]
     x = model.embed(cur)                                   # [B, 1, H]
-    ang = pos_t.float()[:, None] * model.inv_freq[None, :].float()
-    cos = ang.cos()[None, :, None, :]                      # [1,1,1,half]
-    sin = ang.sin()[None, :, None, :]
-    # additive mask over the whole cache: -inf beyond the current position
-    mask = torch.where(arange[None, None, None, :] > pos_t[0],
-                       float("-inf"), 0.0)
+    mask = None          # built lazily: only the einsum fallback needs it
     residual = None
     for li, blk in enumerate(model.blocks):
         attn = blk.attn
@@ -69,18 +71,26 @@ def _step_static(model: LlamaModel, cur: torch.Tensor, cache: KVCache,
         q = q.reshape(B, 1, cfg.num_heads, cfg.head_dim)
         k = k.reshape(B, 1, cfg.num_kv_heads, cfg.head_dim)
         v = v.reshape(B, 1, cfg.num_kv_heads, cfg.head_dim)
-        q = _rope_dev(q, cos, sin).transpose(1, 2)         # [B, nh, 1, D]
-        k = _rope_dev(k, cos, sin).transpose(1, 2)
+        q = _rope_dev(q, model.inv_freq, pos_t).transpose(1, 2)
+        k = _rope_dev(k, model.inv_freq, pos_t).transpose(1, 2)
         v = v.transpose(1, 2)
         cache.k[li].index_copy_(2, pos_t, k)
         cache.v[li].index_copy_(2, pos_t, v)
         kk, vv = cache.k[li], cache.v[li]                  # full max_len
-        G = cfg.num_heads // cfg.num_kv_heads
-        qg = q.reshape(B, cfg.num_kv_heads, G, cfg.head_dim)
-        scores = torch.einsum("bkgd,bksd->bkgs", qg.float(), kk.float())
-        scores = scores / math.sqrt(cfg.head_dim) + mask
-        p = torch.softmax(scores, dim=-1)
-        o = torch.einsum("bkgs,bksd->bkgd", p, vv.float())
+        scale = 1.0 / math.sqrt(cfg.head_dim)
+        o = decode_attention(q, kk, vv, pos_t, scale)
+        if o is None:                         # einsum fallback (odd GQA)
+            if mask is None:
+                mask = torch.where(
+                    arange[None, None, None, :] > pos_t[0],
+                    float("-inf"), 0.0)
+            G = cfg.num_heads // cfg.num_kv_heads
+            qg = q.reshape(B, cfg.num_kv_heads, G, cfg.head_dim)
+            scores = torch.einsum("bkgd,bksd->bkgs", qg.float(),
+                                  kk.float())
+            scores = scores * scale + mask
+            p = torch.softmax(scores, dim=-1)
+            o = torch.einsum("bkgs,bksd->bkgd", p, vv.float())
         o = o.reshape(B, 1, cfg.num_heads * cfg.head_dim).to(x.dtype)
         attn_out = decode_linear(o, attn.o_proj.weight)
         normed, residual = fused_rmsnorm(attn_out,

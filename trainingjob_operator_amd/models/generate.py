@@ -78,12 +78,20 @@ def _rope_positions(x: torch.Tensor, inv_freq: torch.Tensor, S: int,
     return None
 
 
-def _decode_attn(q, kk, vv, n_kv: int):
-    """One-token attention over the cache WITHOUT expanding KV heads:
-    grouped einsum on views (the library GQA path repeat_interleaves the
-    cache -- hundreds of MB per layer at batch 8). q: [B, H, 1, D]."""
+def _decode_attn(q, kk, vv, n_kv: int, ck=None, cv=None, pos_t=None):
+    """One-token attention over the cache WITHOUT expanding KV heads.
+    With the full cache buffers + a device position tensor, the fused
+    flash-decoding kernel (ops.decode_attention, 2 launches) runs;
+    otherwise a grouped einsum on views (the library GQA path
+    repeat_interleaves the cache -- hundreds of MB per layer at batch
+    8). q: [B, H, 1, D]."""
     import math
     B, H, _, D = q.shape
+    if ck is not None and pos_t is not None and q.is_cuda:
+        from ..ops import decode_attention
+        o = decode_attention(q, ck, cv, pos_t, 1.0 / math.sqrt(D))
+        if o is not None:
+            return o
     G = H // n_kv
     qg = q.reshape(B, n_kv, G, D)
     scores = torch.einsum("bkgd,bksd->bkgs", qg.float(), kk.float())
@@ -122,7 +130,8 @@ def _attn_cached(attn, x: torch.Tensor, cos: torch.Tensor,
         # prefill: causal among the new tokens
         o = _sdpa(q, kk, vv, enable_gqa=gqa)
     else:
-        o = _decode_attn(q, kk, vv, cfg.num_kv_heads)
+        o = _decode_attn(q, kk, vv, cfg.num_kv_heads, ck, cv,
+                         getattr(attn, "_pos_dec", None))
     o = o.transpose(1, 2).reshape(B, S, cfg.num_heads * cfg.head_dim)
     return decode_linear(o, attn.o_proj.weight)
 
@@ -150,11 +159,17 @@ def _forward_cached(model: LlamaModel, tokens: torch.Tensor,
         cos, sin = _rope_tables(model.inv_freq, pos0, tokens.shape[1],
                                 tokens.device)
     residual = None
+    pos_dec = None
+    if tokens.shape[1] == 1 and tokens.is_cuda:
+        # device-side fill position for the fused decode-attention kernel
+        pos_dec = torch.full((1,), pos0, dtype=torch.int64,
+                             device=tokens.device)
     for li, blk in enumerate(model.blocks):
         if use_kernel:
             blk.attn._inv_freq_dec = model.inv_freq
         elif hasattr(blk.attn, "_inv_freq_dec"):
             del blk.attn._inv_freq_dec
+        blk.attn._pos_dec = pos_dec
         normed, residual = fused_rmsnorm(x, blk.input_norm_weight,
                                          residual, cfg.norm_eps)
         attn_out = _attn_cached(blk.attn, normed, cos, sin,

@@ -7,6 +7,7 @@ numerics tests in tests/test_ops_gpu.py.
 """
 from .functional import (
     apply_rope,
+    decode_attention,
     decode_linear,
     fused_cross_entropy,
     fused_rmsnorm,
@@ -21,6 +22,7 @@ from .native import HipOpsUnavailable, available, build_ops, load
 
 __all__ = [
     "apply_rope",
+    "decode_attention",
     "decode_linear",
     "dispatch_rows",
     "gather_rows",

@@ -239,6 +239,35 @@ def swiglu_packed(gu: torch.Tensor) -> torch.Tensor:
     return _SwiGLUPacked.apply(gu)
 
 
+def decode_attention(q: torch.Tensor, kc: torch.Tensor,
+                     vc: torch.Tensor, pos_t: torch.Tensor,
+                     scale: float) -> Optional[torch.Tensor]:
+    """Fused one-token GQA attention over the KV cache (hip/ops.hip
+    attn_decode: split-L flash-decoding partials + log-sum-exp combine,
+    2 launches instead of the ~10-kernel einsum chain). The valid length
+    is read from pos_t ON DEVICE, so the call replays inside a hipGraph.
+    Returns [B, H, 1, D], or None when the shape is unsupported
+    (caller falls back to the einsum path). Inference-only."""
+    B, H, S, D = q.shape
+    n_kv, Lmax = kc.shape[1], kc.shape[2]
+    G = H // n_kv if H % n_kv == 0 else 0
+    if not (q.is_cuda and q.dtype == torch.bfloat16 and D == 128
+            and S == 1 and G in (1, 2, 4, 8) and kc.is_contiguous()
+            and vc.is_contiguous() and pos_t.dtype == torch.int64):
+        return None
+    lib = _hip()
+    q2 = q.reshape(B, H, D).contiguous()
+    n_chunk = (Lmax + 127) // 128
+    partial = torch.empty(B, n_kv, n_chunk, G, 130, dtype=torch.float32,
+                          device=q.device)
+    o = torch.empty(B, H, D, dtype=torch.bfloat16, device=q.device)
+    rc = lib.attn_decode(native.stream_ptr(), _ptr(q2), _ptr(kc),
+                         _ptr(vc), _ptr(pos_t), _ptr(partial), _ptr(o),
+                         B, H, n_kv, Lmax, n_chunk, scale)
+    native.check_rc(rc, "attn_decode", f"H={H} n_kv={n_kv} Lmax={Lmax}")
+    return o[:, :, None, :]
+
+
 def decode_linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     """y = x @ W^T for decode-sized x (<= 8 total rows): the hand-written
     wave-per-row GEMV (hip/ops.hip gemv_bf16) instead of hipBLASLt, which

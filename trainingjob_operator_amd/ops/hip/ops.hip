@@ -349,7 +349,8 @@ __global__ void rmsnorm_dw_reduce_kernel(const float* __restrict__ dw_partial,
 __global__ void rope_kernel(const uint4* __restrict__ x, uint4* __restrict__ out,
                             const float* __restrict__ inv_freq,
                             long total_vec, int vec_per_half, int n_heads,
-                            int S, int D, float sign, int pos0) {
+                            int S, int D, float sign, int pos0,
+                            const long* __restrict__ pos_dev) {
   const int HALF = D / 2;
   for (long g = blockIdx.x * (long)blockDim.x + threadIdx.x; g < total_vec;
        g += gridDim.x * (long)blockDim.x) {
@@ -359,7 +360,9 @@ __global__ void rope_kernel(const uint4* __restrict__ x, uint4* __restrict__ out
     const int i0 = (int)(g % per_head) * 8;          // dim offset in [0, HALF)
     const long t = head_g / n_heads;                 // token index
     const int h = (int)(head_g % n_heads);
-    const int pos = pos0 + (int)(t % S);
+    // pos_dev: device-resident base position (hipGraph decode replays
+    // shift position without host input); nullptr on the training path
+    const int pos = pos0 + (int)(t % S) + (pos_dev ? (int)*pos_dev : 0);
     const long base = (t * n_heads + h) * (D / 8);   // vec8 index of head start
     BF8 a; a.v = x[base + i0 / 8];
     BF8 b; b.v = x[base + (HALF + i0) / 8];
@@ -962,6 +965,140 @@ __global__ __launch_bounds__(256) void gemv_bf16_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// Fused decode attention (flash-decoding): one token's GQA attention over
+// the KV cache in TWO kernels instead of the ~10-kernel einsum/softmax/
+// convert chain per layer. Stage 1 tiles the cache length into 128-row
+// chunks — grid (n_kv, n_chunk, B), 128 threads — and emits per-chunk
+// unnormalized partials (chunk max m, sum l, acc[D]); stage 2 merges the
+// chunks with the standard log-sum-exp combine. The fill position is read
+// from DEVICE memory (pos_dev) so the kernels replay inside a hipGraph
+// with no host input; rows beyond *pos_dev are masked to -inf.
+// Phase A: lane <-> cache row (q rows staged in LDS, broadcast reads);
+// Phase B: lane <-> d column (v reads coalesce per row). fp32 everywhere
+// between the bf16 loads and the bf16 store.
+// ---------------------------------------------------------------------------
+
+#define ADEC_CHUNK 128
+#define ADEC_D 128
+
+template <int G>
+__global__ __launch_bounds__(128) void attn_decode_partial_kernel(
+    const u16* __restrict__ q,       // [B, n_kv*G, D]
+    const u16* __restrict__ kc,      // [B, n_kv, Lmax, D]
+    const u16* __restrict__ vc,
+    const long* __restrict__ pos_dev,
+    float* __restrict__ partial,     // [B, n_kv, n_chunk, G, D+2]
+    int n_kv, int Lmax, int n_chunk, float scale) {
+  const int hkv = blockIdx.x;
+  const int chunk = blockIdx.y;
+  const int b = blockIdx.z;
+  const int tid = threadIdx.x;
+  const long pos = *pos_dev;               // rows 0..pos are valid
+  __shared__ float sp[ADEC_CHUNK][G];      // scores, then p
+  __shared__ u16 qs[G][ADEC_D];
+  for (int i = tid; i < G * ADEC_D; i += 128)
+    qs[i / ADEC_D][i % ADEC_D] =
+        q[((long)b * n_kv * G + hkv * G + i / ADEC_D) * ADEC_D
+          + i % ADEC_D];
+  __syncthreads();
+
+  const long row = (long)chunk * ADEC_CHUNK + tid;
+  float s[G];
+#pragma unroll
+  for (int g = 0; g < G; ++g) s[g] = 0.f;
+  if (row <= pos && row < Lmax) {
+    const u16* krow = kc + (((long)b * n_kv + hkv) * Lmax + row) * ADEC_D;
+    union V8 { uint4 u; u16 h[8]; };
+#pragma unroll 4
+    for (int c = 0; c < ADEC_D; c += 8) {
+      V8 kv8;
+      kv8.u = *reinterpret_cast<const uint4*>(krow + c);
+#pragma unroll
+      for (int g = 0; g < G; ++g)
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          s[g] += bf2f(qs[g][c + j]) * bf2f(kv8.h[j]);
+    }
+#pragma unroll
+    for (int g = 0; g < G; ++g) sp[tid][g] = s[g] * scale;
+  } else {
+#pragma unroll
+    for (int g = 0; g < G; ++g) sp[tid][g] = -1e30f;
+  }
+  __syncthreads();
+
+  // chunk max per g (every lane scans the 128 LDS values: broadcast
+  // reads, no reduction tree needed at this size), then p = exp(s - m)
+  float m[G], p[G];
+#pragma unroll
+  for (int g = 0; g < G; ++g) m[g] = -1e30f;
+  for (int r = 0; r < ADEC_CHUNK; ++r)
+#pragma unroll
+    for (int g = 0; g < G; ++g) m[g] = fmaxf(m[g], sp[r][g]);
+#pragma unroll
+  for (int g = 0; g < G; ++g) {
+    p[g] = (sp[tid][g] > -1e30f) ? __expf(sp[tid][g] - m[g]) : 0.f;
+  }
+  __syncthreads();
+#pragma unroll
+  for (int g = 0; g < G; ++g) sp[tid][g] = p[g];
+  __syncthreads();
+
+  // phase B: lane <-> d; acc[g] = sum_row p[row][g] * v[row][d]
+  float acc[G];
+#pragma unroll
+  for (int g = 0; g < G; ++g) acc[g] = 0.f;
+  const long lim = pos + 1 - (long)chunk * ADEC_CHUNK;   // valid rows here
+  const int nrow = lim < 0 ? 0 : (lim < ADEC_CHUNK ? (int)lim : ADEC_CHUNK);
+  const u16* vbase = vc
+      + (((long)b * n_kv + hkv) * Lmax + (long)chunk * ADEC_CHUNK) * ADEC_D
+      + tid;
+  for (int r = 0; r < nrow; ++r) {
+    const float vv = bf2f(vbase[(long)r * ADEC_D]);
+#pragma unroll
+    for (int g = 0; g < G; ++g) acc[g] += sp[r][g] * vv;
+  }
+
+  float* out = partial
+      + ((((long)b * n_kv + hkv) * n_chunk + chunk) * G) * (ADEC_D + 2);
+#pragma unroll
+  for (int g = 0; g < G; ++g)
+    out[(long)g * (ADEC_D + 2) + tid] = acc[g];
+  if (tid < G) {            // lane g writes its head's chunk m and l
+    float l = 0.f;
+    for (int r = 0; r < ADEC_CHUNK; ++r) l += sp[r][tid];
+    out[(long)tid * (ADEC_D + 2) + ADEC_D] = m[tid];
+    out[(long)tid * (ADEC_D + 2) + ADEC_D + 1] = l;
+  }
+}
+
+// grid (B * H), 128 threads: merge the chunks for one (b, head).
+__global__ __launch_bounds__(128) void attn_decode_combine_kernel(
+    const float* __restrict__ partial, u16* __restrict__ o,
+    int n_kv, int G, int n_chunk) {
+  const int head = blockIdx.x;          // b * (n_kv*G) + hkv*G + g
+  const int H = n_kv * G;
+  const int b = head / H;
+  const int hkv = (head % H) / G;
+  const int g = head % G;
+  const int tid = threadIdx.x;          // <-> d
+  const float* base = partial
+      + (((long)b * n_kv + hkv) * n_chunk * G + g) * (ADEC_D + 2);
+  float M = -1e30f;
+  for (int c = 0; c < n_chunk; ++c)
+    M = fmaxf(M, base[(long)c * G * (ADEC_D + 2) + ADEC_D]);
+  float L = 0.f, od = 0.f;
+  for (int c = 0; c < n_chunk; ++c) {
+    const float* pc = base + (long)c * G * (ADEC_D + 2);
+    const float w = __expf(pc[ADEC_D] - M);
+    L += w * pc[ADEC_D + 1];
+    od += w * pc[tid];
+  }
+  o[(long)head * ADEC_D + tid] = f2bf(od / fmaxf(L, 1e-30f));
+}
+
 extern "C" {
 
 int hipops_arch_check() {
@@ -1077,7 +1214,7 @@ int rope(void* stream, const void* x, void* out, const void* inv_freq,
   dim3 grid(elementwise_grid(total_vec)), block(BLOCK);
   hipLaunchKernelGGL(rope_kernel, grid, block, 0, STREAM, (const uint4*)x,
                      (uint4*)out, (const float*)inv_freq, total_vec,
-                     vec_per_half, n_heads, S, D, sign, 0);
+                     vec_per_half, n_heads, S, D, sign, 0, nullptr);
   return 0;
 }
 
@@ -1094,9 +1231,26 @@ int rope_at(void* stream, const void* x, void* out, const void* inv_freq,
   dim3 grid(elementwise_grid(total_vec)), block(BLOCK);
   hipLaunchKernelGGL(rope_kernel, grid, block, 0, STREAM, (const uint4*)x,
                      (uint4*)out, (const float*)inv_freq, total_vec,
-                     vec_per_half, n_heads, S, D, sign, pos0);
+                     vec_per_half, n_heads, S, D, sign, pos0, nullptr);
   return 0;
 }
+
+// rope_at with the base position read from DEVICE memory
+int rope_at_dev(void* stream, const void* x, void* out, const void* inv_freq,
+            long n_tokens, int n_heads, int S, int D, float sign,
+            const void* pos_dev) {
+  if (D <= 0 || D % 16 != 0 || n_heads <= 0 || S <= 0 || !pos_dev)
+    return -1;
+  if (n_tokens <= 0) return 0;
+  const int vec_per_half = (D / 2) / 8;
+  const long total_vec = n_tokens * (long)n_heads * vec_per_half;
+  dim3 grid(elementwise_grid(total_vec)), block(BLOCK);
+  hipLaunchKernelGGL(rope_kernel, grid, block, 0, STREAM, (const uint4*)x,
+                     (uint4*)out, (const float*)inv_freq, total_vec,
+                     vec_per_half, n_heads, S, D, sign, 0, (const long*)pos_dev);
+  return 0;
+}
+
 
 int swiglu_fwd(void* stream, const void* g, const void* u, void* out, long n) {
   if (n % 8 != 0) return -1;
@@ -1297,6 +1451,38 @@ int gemv_bf16(void* stream, const void* w, const void* x, void* y,
     GEMV_CASE(5) GEMV_CASE(6) GEMV_CASE(7) GEMV_CASE(8)
 #undef GEMV_CASE
   }
+  return 0;
+}
+
+
+// fused decode attention: q [B, H, D=128] bf16, caches [B, n_kv, Lmax,
+// 128] bf16 contiguous, pos_dev a device scalar (long) = last valid
+// cache row, partial fp32 [B, n_kv, n_chunk, G, 130] scratch, o [B, H,
+// 128] bf16. n_chunk must be (Lmax + 127) / 128.
+int attn_decode(void* stream, const void* q, const void* kc,
+                const void* vc, const void* pos_dev, void* partial,
+                void* o, int B, int H, int n_kv, int Lmax, int n_chunk,
+                float scale) {
+  if (B <= 0 || n_kv <= 0 || H % n_kv != 0 ||
+      n_chunk != (Lmax + ADEC_CHUNK - 1) / ADEC_CHUNK)
+    return -1;
+  const int G = H / n_kv;
+  dim3 grid(n_kv, n_chunk, B), block(128);
+  switch (G) {
+#define ADEC_CASE(GG) \
+    case GG: \
+      hipLaunchKernelGGL((attn_decode_partial_kernel<GG>), grid, block, \
+                         0, STREAM, (const u16*)q, (const u16*)kc, \
+                         (const u16*)vc, (const long*)pos_dev, \
+                         (float*)partial, n_kv, Lmax, n_chunk, scale); \
+      break;
+    ADEC_CASE(1) ADEC_CASE(2) ADEC_CASE(4) ADEC_CASE(8)
+#undef ADEC_CASE
+    default: return -1;
+  }
+  hipLaunchKernelGGL(attn_decode_combine_kernel, dim3((unsigned)(B * H)),
+                     block, 0, STREAM, (const float*)partial, (u16*)o,
+                     n_kv, G, n_chunk);
   return 0;
 }
 
