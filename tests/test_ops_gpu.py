@@ -356,3 +356,53 @@ def test_decode_attention_matches_fp32():
         ref = ref.reshape(B, H, 1, D)
         err = (o.float() - ref).abs().max().item()
         assert err < 2e-2, f"B{B} H{H} kv{n_kv} L{Lmax} pos{pos}: {err}"
+
+
+def test_rope_cache_matches_reference():
+    """Fused decode rope+cache-write (rope_cache) vs the torch Neox
+    rotation: q roped to a buffer, k roped into cache[pos], v copied."""
+    from trainingjob_operator_amd.ops import make_inv_freq, native
+    lib = native.load(require=True)
+    torch.manual_seed(23)
+    B, nh, nkv, Lmax, D, pos = 2, 8, 2, 96, 64, 37
+    inv_freq = make_inv_freq(D, 10000.0, device=DEV)
+    qkv = _mk((B, (nh + 2 * nkv) * D))
+    kc = torch.zeros(B, nkv, Lmax, D, dtype=torch.bfloat16, device=DEV)
+    vc = torch.zeros_like(kc)
+    q = torch.empty(B, nh, D, dtype=torch.bfloat16, device=DEV)
+    pos_t = torch.full((1,), pos, dtype=torch.int64, device=DEV)
+    rc = lib.rope_cache(native.stream_ptr(), qkv.data_ptr(), q.data_ptr(),
+                        kc.data_ptr(), vc.data_ptr(), inv_freq.data_ptr(),
+                        pos_t.data_ptr(), B, nh, nkv, Lmax, D)
+    native.check_rc(rc, "rope_cache", "test")
+    torch.cuda.synchronize()
+
+    def rope_ref(x):                       # [B, n, D] fp32 at position pos
+        half = D // 2
+        ang = pos * inv_freq.float()
+        c, s = ang.cos(), ang.sin()
+        x1, x2 = x[..., :half], x[..., half:]
+        return torch.cat([x1 * c - x2 * s, x1 * s + x2 * c], dim=-1)
+
+    f = qkv.float().reshape(B, nh + 2 * nkv, D)
+    qr, kr, vr = f[:, :nh], f[:, nh:nh + nkv], f[:, nh + nkv:]
+    assert (q.float() - rope_ref(qr)).abs().max().item() < 2e-2
+    assert (kc[:, :, pos].float() - rope_ref(kr)).abs().max().item() < 2e-2
+    assert (vc[:, :, pos].float() - vr).abs().max().item() < 1e-6
+    assert kc[:, :, pos + 1].abs().max().item() == 0   # only row pos written
+
+
+def test_gemv_swiglu_matches_fp32():
+    """Fused decode GEMV+SwiGLU vs fp32 silu(x@Wg)*(x@Wu)."""
+    from trainingjob_operator_amd.ops import decode_swiglu
+    torch.manual_seed(29)
+    for F, K, N in [(14336, 4096, 1), (512, 512, 3), (1024, 1024, 8)]:
+        w = _mk((2 * F, K)) * 0.05
+        x = _mk((N, K)) * 0.05
+        y = decode_swiglu(x, w)
+        assert y is not None
+        g = x.float() @ w.float()[:F].t()
+        u = x.float() @ w.float()[F:].t()
+        ref = torch.nn.functional.silu(g) * u
+        err = (y.float() - ref).abs().max().item()
+        assert err < ref.abs().max().item() * 2e-2 + 2e-2, (F, K, N, err)

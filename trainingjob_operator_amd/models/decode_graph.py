@@ -28,26 +28,9 @@ from __future__ import annotations
 
 import torch
 
-from ..ops import decode_attention, decode_linear, fused_rmsnorm
+from ..ops import decode_attention, decode_linear, fused_rmsnorm, native
 from .generate import KVCache, _mlp_cached
 from .llama import LlamaModel
-
-
-def _rope_dev(x: torch.Tensor, inv_freq: torch.Tensor,
-              pos_t: torch.Tensor) -> torch.Tensor:
-    """Neox half-rotation at a DEVICE-resident position (one kernel per
-    tensor instead of ~10 elementwise launches; graph-replayable because
-    the kernel dereferences pos_t). x: [B, 1, nh, D] bf16."""
-    from ..ops import native
-    B, S, nh, D = x.shape
-    lib = native.load(require=True)
-    xt = x.reshape(-1, nh, D).contiguous()
-    out = torch.empty_like(xt)
-    rc = lib.rope_at_dev(native.stream_ptr(), xt.data_ptr(),
-                         out.data_ptr(), inv_freq.data_ptr(),
-                         xt.shape[0], nh, 1, D, 1.0, pos_t.data_ptr())
-    native.check_rc(rc, "rope_at_dev", f"D={D}")
-    return out.reshape(x.shape)
 
 
 def _step_static(model: LlamaModel, cur: torch.Tensor, cache: KVCache,
@@ -66,16 +49,19 @@ def _step_static(model: LlamaModel, cur: torch.Tensor, cache: KVCache,
         normed, residual = fused_rmsnorm(x, blk.input_norm_weight,
                                          residual, cfg.norm_eps)
         qkv = decode_linear(normed, attn.qkv_proj.weight)
-        q, k, v = qkv.split([attn.q_size, attn.kv_size, attn.kv_size],
-                            dim=-1)
-        q = q.reshape(B, 1, cfg.num_heads, cfg.head_dim)
-        k = k.reshape(B, 1, cfg.num_kv_heads, cfg.head_dim)
-        v = v.reshape(B, 1, cfg.num_kv_heads, cfg.head_dim)
-        q = _rope_dev(q, model.inv_freq, pos_t).transpose(1, 2)
-        k = _rope_dev(k, model.inv_freq, pos_t).transpose(1, 2)
-        v = v.transpose(1, 2)
-        cache.k[li].index_copy_(2, pos_t, k)
-        cache.v[li].index_copy_(2, pos_t, v)
+        # one fused launch: rope q -> contiguous buffer, rope k ->
+        # cache[pos], copy v -> cache[pos] (vs 2 rope + 2 index_copy)
+        lib = native.load(require=True)
+        q = torch.empty(B, cfg.num_heads, cfg.head_dim,
+                        dtype=torch.bfloat16, device=qkv.device)
+        rc = lib.rope_cache(
+            native.stream_ptr(), qkv.contiguous().data_ptr(),
+            q.data_ptr(), cache.k[li].data_ptr(), cache.v[li].data_ptr(),
+            model.inv_freq.data_ptr(), pos_t.data_ptr(),
+            B, cfg.num_heads, cfg.num_kv_heads, cache.max_len,
+            cfg.head_dim)
+        native.check_rc(rc, "rope_cache", f"D={cfg.head_dim}")
+        q = q[:, :, None, :]                               # [B, nh, 1, D]
         kk, vv = cache.k[li], cache.v[li]                  # full max_len
         scale = 1.0 / math.sqrt(cfg.head_dim)
         o = decode_attention(q, kk, vv, pos_t, scale)

@@ -138,11 +138,15 @@ def _attn_cached(attn, x: torch.Tensor, cos: torch.Tensor,
 
 
 def _mlp_cached(mlp, x: torch.Tensor) -> torch.Tensor:
-    """SwiGLU MLP through the decode GEMV (decode_linear falls back to the
+    """SwiGLU MLP through the decode GEMV path: fused GEMV+SwiGLU when
+    supported, else decode_linear + swiglu (both fall back to the
     library GEMM above 8 rows, so prefill takes the normal path)."""
-    gu = decode_linear(x, mlp.gate_up_proj.weight)
-    return decode_linear(swiglu_packed(gu.contiguous()),
-                         mlp.down_proj.weight)
+    from ..ops import decode_swiglu
+    h = decode_swiglu(x, mlp.gate_up_proj.weight)
+    if h is None:
+        gu = decode_linear(x, mlp.gate_up_proj.weight)
+        h = swiglu_packed(gu.contiguous())
+    return decode_linear(h, mlp.down_proj.weight)
 
 
 @torch.no_grad()

@@ -9,6 +9,7 @@ from .functional import (
     apply_rope,
     decode_attention,
     decode_linear,
+    decode_swiglu,
     fused_cross_entropy,
     fused_rmsnorm,
     dispatch_rows,
@@ -24,6 +25,7 @@ __all__ = [
     "apply_rope",
     "decode_attention",
     "decode_linear",
+    "decode_swiglu",
     "dispatch_rows",
     "gather_rows",
     "moe_combine",

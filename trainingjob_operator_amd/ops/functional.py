@@ -268,6 +268,30 @@ def decode_attention(q: torch.Tensor, kc: torch.Tensor,
     return o[:, :, None, :]
 
 
+def decode_swiglu(x: torch.Tensor,
+                  gate_up_weight: torch.Tensor) -> Optional[torch.Tensor]:
+    """silu(x @ Wg^T) * (x @ Wu^T) over the packed [2F, K] gate|up weight
+    in ONE kernel (hip/ops.hip gemv_swiglu): same weight bytes as the
+    plain decode GEMV, no 2F-wide intermediate, no separate swiglu
+    launch. Returns [..., F], or None on unsupported shapes (caller
+    falls back to decode_linear + swiglu_packed). Inference-only."""
+    K = x.shape[-1]
+    rows = x.numel() // K
+    F = gate_up_weight.shape[0] // 2
+    if not (x.is_cuda and x.dtype == torch.bfloat16
+            and gate_up_weight.dtype == torch.bfloat16 and 1 <= rows <= 8
+            and K % 512 == 0 and gate_up_weight.stride(-1) == 1
+            and gate_up_weight.shape[0] % 2 == 0):
+        return None
+    lib = _hip()
+    x2 = x.reshape(rows, K).contiguous()
+    y = torch.empty(rows, F, dtype=torch.bfloat16, device=x.device)
+    rc = lib.gemv_swiglu(native.stream_ptr(), _ptr(gate_up_weight),
+                         _ptr(x2), _ptr(y), F, K, rows)
+    native.check_rc(rc, "gemv_swiglu", f"F={F} K={K} N={rows}")
+    return y.reshape(*x.shape[:-1], F)
+
+
 def decode_linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     """y = x @ W^T for decode-sized x (<= 8 total rows): the hand-written
     wave-per-row GEMV (hip/ops.hip gemv_bf16) instead of hipBLASLt, which

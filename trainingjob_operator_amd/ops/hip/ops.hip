@@ -1099,6 +1099,145 @@ __global__ __launch_bounds__(128) void attn_decode_combine_kernel(
   o[(long)head * ADEC_D + tid] = f2bf(od / fmaxf(L, 1e-30f));
 }
 
+
+// GEMV tuning variants (gemv_bf16_ab): UNROLL = in-flight W loads per
+// wave, NT = nontemporal W loads (decode weights are a pure stream, so
+// bypassing L2 avoids thrashing it), W1 = 64-thread workgroups (grid =
+// one wave per row -> 4x more workgroups for small M).
+template <int N, int UNROLL, bool NT, bool W1>
+__global__ __launch_bounds__(W1 ? 64 : 256) void gemv_bf16_v_kernel(
+    const u16* __restrict__ w, const u16* __restrict__ x,
+    u16* __restrict__ y, int M, int K) {
+  const int wid = W1 ? 0 : (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  const int m = W1 ? blockIdx.x : blockIdx.x * 4 + wid;
+  if (m >= M) return;
+  const u16* wrow = w + (long)m * K;
+  float acc[N];
+#pragma unroll
+  for (int n = 0; n < N; ++n) acc[n] = 0.f;
+  typedef __attribute__((ext_vector_type(4))) unsigned int u32x4_;
+  union V8 { uint4 u; u32x4_ e; u16 h[8]; };
+#pragma unroll UNROLL
+  for (int c0 = lane * 8; c0 < K; c0 += 64 * 8) {
+    V8 wv;
+    if (NT)
+      wv.e = __builtin_nontemporal_load(
+          reinterpret_cast<const u32x4_*>(wrow + c0));
+    else
+      wv.u = *reinterpret_cast<const uint4*>(wrow + c0);
+#pragma unroll
+    for (int n = 0; n < N; ++n) {
+      V8 xv;
+      xv.u = *reinterpret_cast<const uint4*>(x + (long)n * K + c0);
+      float s = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) s += bf2f(wv.h[j]) * bf2f(xv.h[j]);
+      acc[n] += s;
+    }
+  }
+#pragma unroll
+  for (int n = 0; n < N; ++n) {
+#pragma unroll
+    for (int off = 32; off; off >>= 1)
+      acc[n] += __shfl_xor(acc[n], off, 64);
+    if (lane == 0) y[(long)n * M + m] = f2bf(acc[n]);
+  }
+}
+
+
+// ---------------------------------------------------------------------------
+// Fused decode rope + KV-cache write: consumes the packed qkv GEMV output
+// [B, nh*D + 2*nkv*D] and in ONE launch ropes q into a contiguous
+// [B, nh, D] buffer, ropes k into cache[pos], and copies v into cache
+// [pos] — replacing 2 rope launches + 2 index_copy launches per layer in
+// the captured decode step. pos comes from device memory (graph-safe).
+// One 32-lane-...: one workgroup of D/2 threads per head; lane d holds
+// the (d, d+D/2) Neox pair.
+// ---------------------------------------------------------------------------
+__global__ void rope_cache_kernel(
+    const u16* __restrict__ qkv, u16* __restrict__ qout,
+    u16* __restrict__ kcache, u16* __restrict__ vcache,
+    const float* __restrict__ inv_freq, const long* __restrict__ pos_dev,
+    int nh, int nkv, int Lmax, int D) {
+  const int hx = blockIdx.x;             // 0..nh-1 q, ..+nkv k, ..+nkv v
+  const int b = blockIdx.y;
+  const int d = threadIdx.x;             // 0 .. D/2-1
+  const long pos = *pos_dev;
+  const int half = D / 2;
+  const u16* src = qkv + ((long)b * (nh + 2 * nkv)) * D + (long)hx * D;
+  const float x1 = bf2f(src[d]), x2 = bf2f(src[d + half]);
+  u16* dst;
+  if (hx < nh) {
+    dst = qout + ((long)b * nh + hx) * D;
+  } else if (hx < nh + nkv) {
+    dst = kcache + (((long)b * nkv + (hx - nh)) * Lmax + pos) * D;
+  } else {
+    vcache[(((long)b * nkv + (hx - nh - nkv)) * Lmax + pos) * D + d] =
+        src[d];
+    vcache[(((long)b * nkv + (hx - nh - nkv)) * Lmax + pos) * D + d
+           + half] = src[d + half];
+    return;
+  }
+  float c, sn;
+  __sincosf((float)pos * inv_freq[d], &sn, &c);
+  dst[d] = f2bf(x1 * c - x2 * sn);
+  dst[d + half] = f2bf(x1 * sn + x2 * c);
+}
+
+
+// ---------------------------------------------------------------------------
+// Decode GEMV + SwiGLU: y[n][m] = silu(x @ Wg[m]) * (x @ Wu[m]) where the
+// packed weight is [2F, K] = [gate | up] rows (the training layout fed to
+// swiglu_packed). One wave computes BOTH dots for its m — same W bytes as
+// the plain GEMV, one launch instead of two and no 2F-wide intermediate.
+// ---------------------------------------------------------------------------
+template <int N>
+__global__ __launch_bounds__(256) void gemv_swiglu_kernel(
+    const u16* __restrict__ w, const u16* __restrict__ x,
+    u16* __restrict__ y, int F, int K) {
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int m = blockIdx.x * 4 + wid;
+  if (m >= F) return;
+  const u16* grow = w + (long)m * K;
+  const u16* urow = w + (long)(m + F) * K;
+  float ag[N], au[N];
+#pragma unroll
+  for (int n = 0; n < N; ++n) ag[n] = au[n] = 0.f;
+  union V8 { uint4 u; u16 h[8]; };
+#pragma unroll 4
+  for (int c0 = lane * 8; c0 < K; c0 += 64 * 8) {
+    V8 gv, uv;
+    gv.u = *reinterpret_cast<const uint4*>(grow + c0);
+    uv.u = *reinterpret_cast<const uint4*>(urow + c0);
+#pragma unroll
+    for (int n = 0; n < N; ++n) {
+      V8 xv;
+      xv.u = *reinterpret_cast<const uint4*>(x + (long)n * K + c0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float xf = bf2f(xv.h[j]);
+        ag[n] += bf2f(gv.h[j]) * xf;
+        au[n] += bf2f(uv.h[j]) * xf;
+      }
+    }
+  }
+#pragma unroll
+  for (int n = 0; n < N; ++n) {
+#pragma unroll
+    for (int off = 32; off; off >>= 1) {
+      ag[n] += __shfl_xor(ag[n], off, 64);
+      au[n] += __shfl_xor(au[n], off, 64);
+    }
+    if (lane == 0) {
+      const float g = ag[n];
+      const float sg = g / (1.f + __expf(-g));   // silu
+      y[(long)n * F + m] = f2bf(sg * au[n]);
+    }
+  }
+}
+
 extern "C" {
 
 int hipops_arch_check() {
@@ -1440,6 +1579,24 @@ int adamw_step(void* stream, void* p32, void* m, void* v, const void* grad,
 int gemv_bf16(void* stream, const void* w, const void* x, void* y,
               int M, int K, int N) {
   if (M <= 0 || K <= 0 || (K % 512) != 0 || N < 1 || N > 8) return -1;
+  if (M >= 65536) {
+    // lm_head-sized: one wave per WG + nontemporal W loads measured
+    // 7.4 vs 6.4 TB/s (scripts/gemvbench.py) — the 1 GB matrix only
+    // thrashes L2, and the bigger grid fills the 8 XCDs
+    dim3 grid((unsigned)M), block(64);
+    switch (N) {
+#define GEMV_CASE(NN) \
+      case NN: \
+        hipLaunchKernelGGL((gemv_bf16_v_kernel<NN, 8, true, true>), grid, \
+                           block, 0, STREAM, (const u16*)w, \
+                           (const u16*)x, (u16*)y, M, K); \
+        break;
+      GEMV_CASE(1) GEMV_CASE(2) GEMV_CASE(3) GEMV_CASE(4)
+      GEMV_CASE(5) GEMV_CASE(6) GEMV_CASE(7) GEMV_CASE(8)
+#undef GEMV_CASE
+    }
+    return 0;
+  }
   dim3 grid((unsigned)((M + 3) / 4)), block(256);
   switch (N) {
 #define GEMV_CASE(NN) \
@@ -1483,6 +1640,78 @@ int attn_decode(void* stream, const void* q, const void* kc,
   hipLaunchKernelGGL(attn_decode_combine_kernel, dim3((unsigned)(B * H)),
                      block, 0, STREAM, (const float*)partial, (u16*)o,
                      n_kv, G, n_chunk);
+  return 0;
+}
+
+// within-box A/B of GEMV variants (N=1 only): 0=unroll4 (shipped
+// baseline layout), 1=unroll8, 2=unroll8+nontemporal, 3=wave-per-WG
+// grid M + unroll8 + nontemporal.
+int gemv_bf16_ab(void* stream, int which, const void* w, const void* x,
+                 void* y, int M, int K) {
+  if (K % 512 != 0) return -1;
+  dim3 b256(256), b64(64);
+  dim3 g4((unsigned)((M + 3) / 4)), g1((unsigned)M);
+  switch (which) {
+    case 0:
+      hipLaunchKernelGGL((gemv_bf16_v_kernel<1, 4, false, false>), g4,
+                         b256, 0, STREAM, (const u16*)w, (const u16*)x,
+                         (u16*)y, M, K);
+      break;
+    case 1:
+      hipLaunchKernelGGL((gemv_bf16_v_kernel<1, 8, false, false>), g4,
+                         b256, 0, STREAM, (const u16*)w, (const u16*)x,
+                         (u16*)y, M, K);
+      break;
+    case 2:
+      hipLaunchKernelGGL((gemv_bf16_v_kernel<1, 8, true, false>), g4,
+                         b256, 0, STREAM, (const u16*)w, (const u16*)x,
+                         (u16*)y, M, K);
+      break;
+    case 3:
+      hipLaunchKernelGGL((gemv_bf16_v_kernel<1, 8, true, true>), g1,
+                         b64, 0, STREAM, (const u16*)w, (const u16*)x,
+                         (u16*)y, M, K);
+      break;
+    default: return -1;
+  }
+  return 0;
+}
+
+
+// fused decode rope + cache write (rope_cache_kernel): qkv packed
+// [B, (nh+2*nkv)*D] bf16 (q | k | v), caches [B, nkv, Lmax, D]
+// contiguous, pos_dev device long scalar. D even, D/2 <= 1024.
+int rope_cache(void* stream, const void* qkv, void* qout, void* kcache,
+               void* vcache, const void* inv_freq, const void* pos_dev,
+               int B, int nh, int nkv, int Lmax, int D) {
+  if (D <= 0 || D % 16 != 0 || D / 2 > 1024 || nh <= 0 || nkv <= 0 ||
+      B <= 0 || !pos_dev)
+    return -1;
+  dim3 grid((unsigned)(nh + 2 * nkv), (unsigned)B), block(D / 2);
+  hipLaunchKernelGGL(rope_cache_kernel, grid, block, 0, STREAM,
+                     (const u16*)qkv, (u16*)qout, (u16*)kcache,
+                     (u16*)vcache, (const float*)inv_freq,
+                     (const long*)pos_dev, nh, nkv, Lmax, D);
+  return 0;
+}
+
+
+// decode GEMV+SwiGLU over the packed [2F, K] gate|up weight.
+int gemv_swiglu(void* stream, const void* w, const void* x, void* y,
+                int F, int K, int N) {
+  if (F <= 0 || K <= 0 || (K % 512) != 0 || N < 1 || N > 8) return -1;
+  dim3 grid((unsigned)((F + 3) / 4)), block(256);
+  switch (N) {
+#define GSW_CASE(NN) \
+    case NN: \
+      hipLaunchKernelGGL((gemv_swiglu_kernel<NN>), grid, block, 0, \
+                         STREAM, (const u16*)w, (const u16*)x, (u16*)y, \
+                         F, K); \
+      break;
+    GSW_CASE(1) GSW_CASE(2) GSW_CASE(3) GSW_CASE(4)
+    GSW_CASE(5) GSW_CASE(6) GSW_CASE(7) GSW_CASE(8)
+#undef GSW_CASE
+  }
   return 0;
 }
 

@@ -67,8 +67,10 @@ def load(require: bool = True) -> Optional[ctypes.CDLL]:
     _sig(lib.rope_at, [vp, vp, vp, vp, l, i, i, i, f, i], i)
     _sig(lib.rope_at_dev, [vp, vp, vp, vp, l, i, i, i, f, vp], i)
     _sig(lib.gemv_bf16, [vp, vp, vp, vp, i, i, i], i)
+    _sig(lib.gemv_swiglu, [vp, vp, vp, vp, i, i, i], i)
     _sig(lib.attn_decode, [vp, vp, vp, vp, vp, vp, vp,
                            i, i, i, i, i, f], i)
+    _sig(lib.rope_cache, [vp, vp, vp, vp, vp, vp, vp, i, i, i, i, i], i)
     _sig(lib.swiglu_fwd, [vp, vp, vp, vp, l], i)
     _sig(lib.swiglu_bwd, [vp, vp, vp, vp, vp, vp, l], i)
     _sig(lib.swiglu_packed_fwd, [vp, vp, vp, l, i], i)
