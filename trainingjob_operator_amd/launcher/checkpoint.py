@@ -128,3 +128,26 @@ class Checkpointer:
         trainer.opt.step_count = int(opt["step"])
         trainer.step_count = int(state["step"])
         return trainer.step_count
+
+
+def load_model_only(directory: str, model) -> Optional[int]:
+    """Serving-side weight load: copy the model's named parameters out of
+    the latest flat checkpoint (ignores optimizer state; works from a
+    checkpoint taken at any world size because DP replicates the flat
+    stream). Returns the checkpoint step, or None if none found."""
+    ck = Checkpointer(directory)
+    path = ck.latest()
+    if path is None:
+        return None
+    state = torch.load(path, map_location="cpu", weights_only=False)
+    flat = state["flat_param"]
+    offsets, shapes = state["offsets"], state["shapes"]
+    named = dict(model.named_parameters())
+    with torch.no_grad():
+        for name, (off, numel) in offsets.items():
+            p = named.get(name)
+            if p is None:
+                continue
+            p.copy_(flat[off:off + numel]
+                    .view(shapes.get(name, p.shape)).to(p.dtype))
+    return state["step"]
