@@ -15,10 +15,11 @@ def main():
     prompt_len = int(sys.argv[3]) if len(sys.argv) > 3 else 512
     new_tokens = int(sys.argv[4]) if len(sys.argv) > 4 else 128
     from trainingjob_operator_amd.models.config import CONFIGS
-    from trainingjob_operator_amd.models.generate import generate
-    from trainingjob_operator_amd.training import build_model
+    from trainingjob_operator_amd.models.generate import (
+        build_inference_model, generate,
+    )
     cfg = CONFIGS[model]
-    m = build_model(cfg, torch.device("cuda:0"))
+    m = build_inference_model(cfg, torch.device("cuda:0"))
     g = torch.Generator().manual_seed(1)
     prompt = torch.randint(0, cfg.vocab_size, (batch, prompt_len),
                            generator=g).to("cuda:0")

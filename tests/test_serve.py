@@ -97,3 +97,20 @@ def test_serve_checkpoint_load(tmp_path):
     for name, p in model.named_parameters():
         assert torch.equal(p.detach().to(torch.float32),
                            trained[name].detach().to(torch.float32)), name
+
+
+def test_serve_moe_model():
+    """The endpoint serves the MoE family too (build_inference_model
+    accepts both; routing runs per token in decode)."""
+    from trainingjob_operator_amd.launcher.serve import create_app
+    from trainingjob_operator_amd.models.config import CONFIGS
+    from trainingjob_operator_amd.models.generate import (
+        build_inference_model,
+    )
+    torch.manual_seed(11)
+    model = build_inference_model(CONFIGS["moe-tiny"], torch.device("cpu"))
+    c = TestClient(create_app(model, "moe-tiny"))
+    r = c.post("/generate", json={"prompt_tokens": [[1, 2, 3]],
+                                  "max_new_tokens": 4})
+    assert r.status_code == 200, r.text
+    assert len(r.json()["tokens"][0]) == 7

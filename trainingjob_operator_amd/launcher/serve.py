@@ -107,7 +107,7 @@ def main(argv=None):
     import uvicorn
 
     from ..models.config import CONFIGS
-    from ..training import build_model
+    from ..models.generate import build_inference_model
 
     ap = argparse.ArgumentParser()
     ap.add_argument("--model", required=True, choices=sorted(CONFIGS))
@@ -123,12 +123,11 @@ def main(argv=None):
     if args.graph:
         os.environ["AITJ_DECODE_GRAPH"] = "1"
     device = torch.device(args.device)
-    model = build_model(CONFIGS[args.model], device)
+    model = build_inference_model(CONFIGS[args.model], device)
     if args.ckpt_dir:
         from .checkpoint import load_model_only
         step = load_model_only(args.ckpt_dir, model)
         print(f"loaded checkpoint step {step} from {args.ckpt_dir}")
-    model.eval()
     app = create_app(model, args.model)
     uvicorn.run(app, host=args.host, port=args.port, log_level="info")
 

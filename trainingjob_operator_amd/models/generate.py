@@ -24,6 +24,25 @@ from .config import LlamaConfig
 from .llama import LlamaModel, _sdpa
 
 
+def build_inference_model(cfg, device):
+    """Build any registered config (dense LlamaModel or MoELlamaModel)
+    in inference form: bf16 weights, fp32 rope frequencies, eval mode.
+    The training-side build_model() deliberately rejects MoE configs
+    (those go through EPTrainer); serving and genbench accept both."""
+    from ..ops import make_inv_freq
+    from .moe_llama import MoELlamaConfig, MoELlamaModel
+    if isinstance(cfg, MoELlamaConfig):
+        with torch.device(device):
+            model = MoELlamaModel(cfg)
+        model = model.to(torch.bfloat16)
+    else:
+        from ..training import build_model
+        return build_model(cfg, device).eval()
+    model.inv_freq = make_inv_freq(cfg.head_dim, cfg.rope_theta,
+                                   device=device)
+    return model.eval()
+
+
 class KVCache:
     """Per-layer [B, n_kv, max_len, D] key/value buffers + fill length."""
 

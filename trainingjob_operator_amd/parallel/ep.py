@@ -96,6 +96,15 @@ class Expert(nn.Module):
         self.down_proj = nn.Linear(ff, hidden, bias=False)
 
     def forward(self, x):
+        if not torch.is_grad_enabled() and x.is_cuda and x.shape[0] <= 8:
+            # decode-sized slices (<= batch*top_k rows) stream the expert
+            # weights through the wave-per-row GEMV instead of hipBLASLt
+            # (decode_linear falls back itself on unsupported shapes)
+            from ..ops import decode_linear
+            g = decode_linear(x, self.gate_proj.weight)
+            u = decode_linear(x, self.up_proj.weight)
+            return decode_linear(swiglu(g.contiguous(), u.contiguous()),
+                                 self.down_proj.weight)
         return self.down_proj(swiglu(self.gate_proj(x).contiguous(),
                                      self.up_proj(x).contiguous()))
 
