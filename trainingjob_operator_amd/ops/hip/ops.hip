@@ -1238,6 +1238,86 @@ __global__ __launch_bounds__(256) void gemv_swiglu_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// Indirect (routed) decode GEMV for MoE: the expert id comes from a
+// DEVICE tensor and the weight base from a device-resident pointer
+// table, so b1 decode routing never synchronizes to the host (the
+// grouped path costs one D2H per layer per token for the slice sizes).
+// Each (token, k) routed pair is an independent grid column: pair n
+// reads x row n/xdiv and expert table[idx[n]]'s weights.
+//   gemv_moe_swiglu: y[n] = silu(Wg[e_n] @ x) * (Wu[e_n] @ x)
+//   gemv_moe:        y[n] = W[e_n] @ x[n]
+// Same wave-per-row layout as gemv_bf16; weight bytes scale with the
+// pair count, so callers gate this to decode-sized T*k.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void gemv_moe_swiglu_kernel(
+    const unsigned long long* __restrict__ table_g,
+    const unsigned long long* __restrict__ table_u,
+    const long* __restrict__ idx, const u16* __restrict__ x,
+    u16* __restrict__ y, int F, int K, int xdiv) {
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int m = blockIdx.x * 4 + wid;
+  const int n = blockIdx.y;
+  if (m >= F) return;
+  const long e = idx[n];
+  const u16* grow = reinterpret_cast<const u16*>(table_g[e]) + (long)m * K;
+  const u16* urow = reinterpret_cast<const u16*>(table_u[e]) + (long)m * K;
+  const u16* xrow = x + (long)(n / xdiv) * K;
+  float ag = 0.f, au = 0.f;
+  union V8 { uint4 u; u16 h[8]; };
+#pragma unroll 4
+  for (int c0 = lane * 8; c0 < K; c0 += 64 * 8) {
+    V8 gv, uv, xv;
+    gv.u = *reinterpret_cast<const uint4*>(grow + c0);
+    uv.u = *reinterpret_cast<const uint4*>(urow + c0);
+    xv.u = *reinterpret_cast<const uint4*>(xrow + c0);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float xf = bf2f(xv.h[j]);
+      ag += bf2f(gv.h[j]) * xf;
+      au += bf2f(uv.h[j]) * xf;
+    }
+  }
+#pragma unroll
+  for (int off = 32; off; off >>= 1) {
+    ag += __shfl_xor(ag, off, 64);
+    au += __shfl_xor(au, off, 64);
+  }
+  if (lane == 0) {
+    const float sg = ag / (1.f + __expf(-ag));
+    y[(long)n * F + m] = f2bf(sg * au);
+  }
+}
+
+__global__ __launch_bounds__(256) void gemv_moe_kernel(
+    const unsigned long long* __restrict__ table,
+    const long* __restrict__ idx, const u16* __restrict__ x,
+    u16* __restrict__ y, int M, int K, int xdiv) {
+  const int wid = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  const int m = blockIdx.x * 4 + wid;
+  const int n = blockIdx.y;
+  if (m >= M) return;
+  const long e = idx[n];
+  const u16* wrow = reinterpret_cast<const u16*>(table[e]) + (long)m * K;
+  const u16* xrow = x + (long)(n / xdiv) * K;
+  float acc = 0.f;
+  union V8 { uint4 u; u16 h[8]; };
+#pragma unroll 4
+  for (int c0 = lane * 8; c0 < K; c0 += 64 * 8) {
+    V8 wv, xv;
+    wv.u = *reinterpret_cast<const uint4*>(wrow + c0);
+    xv.u = *reinterpret_cast<const uint4*>(xrow + c0);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc += bf2f(wv.h[j]) * bf2f(xv.h[j]);
+  }
+#pragma unroll
+  for (int off = 32; off; off >>= 1) acc += __shfl_xor(acc, off, 64);
+  if (lane == 0) y[(long)n * M + m] = f2bf(acc);
+}
+
 extern "C" {
 
 int hipops_arch_check() {
@@ -1712,6 +1792,35 @@ int gemv_swiglu(void* stream, const void* w, const void* x, void* y,
     GSW_CASE(5) GSW_CASE(6) GSW_CASE(7) GSW_CASE(8)
 #undef GSW_CASE
   }
+  return 0;
+}
+
+
+// routed decode GEMVs (tables = device u64 arrays of expert weight base
+// pointers; idx = device long per routed pair). xdiv folds the
+// pair->token mapping: the swiglu stage reads x[n / top_k], the down
+// stage reads its own pair-major input (xdiv 1).
+int gemv_moe_swiglu(void* stream, const void* table_g, const void* table_u,
+                    const void* idx, const void* x, void* y,
+                    int F, int K, int N, int xdiv) {
+  if (F <= 0 || K <= 0 || (K % 512) != 0 || N < 1 || N > 64 || xdiv < 1)
+    return -1;
+  dim3 grid((unsigned)((F + 3) / 4), (unsigned)N), block(256);
+  hipLaunchKernelGGL(gemv_moe_swiglu_kernel, grid, block, 0, STREAM,
+                     (const unsigned long long*)table_g,
+                     (const unsigned long long*)table_u,
+                     (const long*)idx, (const u16*)x, (u16*)y, F, K, xdiv);
+  return 0;
+}
+
+int gemv_moe(void* stream, const void* table, const void* idx,
+             const void* x, void* y, int M, int K, int N, int xdiv) {
+  if (M <= 0 || K <= 0 || (K % 512) != 0 || N < 1 || N > 64 || xdiv < 1)
+    return -1;
+  dim3 grid((unsigned)((M + 3) / 4), (unsigned)N), block(256);
+  hipLaunchKernelGGL(gemv_moe_kernel, grid, block, 0, STREAM,
+                     (const unsigned long long*)table,
+                     (const long*)idx, (const u16*)x, (u16*)y, M, K, xdiv);
   return 0;
 }
 

@@ -68,6 +68,8 @@ def load(require: bool = True) -> Optional[ctypes.CDLL]:
     _sig(lib.rope_at_dev, [vp, vp, vp, vp, l, i, i, i, f, vp], i)
     _sig(lib.gemv_bf16, [vp, vp, vp, vp, i, i, i], i)
     _sig(lib.gemv_swiglu, [vp, vp, vp, vp, i, i, i], i)
+    _sig(lib.gemv_moe_swiglu, [vp, vp, vp, vp, vp, vp, i, i, i, i], i)
+    _sig(lib.gemv_moe, [vp, vp, vp, vp, vp, i, i, i, i], i)
     _sig(lib.attn_decode, [vp, vp, vp, vp, vp, vp, vp,
                            i, i, i, i, i, f], i)
     _sig(lib.rope_cache, [vp, vp, vp, vp, vp, vp, vp, i, i, i, i, i], i)

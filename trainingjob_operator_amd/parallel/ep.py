@@ -163,10 +163,77 @@ class MoEMLP(nn.Module):
         per_rank = counts.view(-1, self.experts_per_rank).sum(dim=1)
         return [int(c) for c in per_rank]
 
+    def _decode_tables(self, device):
+        """Device-resident expert weight pointer tables for the routed
+        decode GEMVs (built lazily, invalidated if the weights move)."""
+        key = self.experts[0].gate_proj.weight.data_ptr()
+        cached = getattr(self, "_ptr_tables", None)
+        if cached is not None and cached[0] == key:
+            return cached[1:]
+        tg = torch.tensor([e.gate_proj.weight.data_ptr()
+                           for e in self.experts], dtype=torch.int64,
+                          device=device)
+        tu = torch.tensor([e.up_proj.weight.data_ptr()
+                           for e in self.experts], dtype=torch.int64,
+                          device=device)
+        td = torch.tensor([e.down_proj.weight.data_ptr()
+                           for e in self.experts], dtype=torch.int64,
+                          device=device)
+        self._ptr_tables = (key, tg, tu, td)
+        return tg, tu, td
+
+    def _decode_forward(self, xt: torch.Tensor):
+        """Sync-free decode path (single process, <= 8 routed pairs):
+        router topk feeds DEVICE index tensors straight into the routed
+        GEMV kernels via expert pointer tables — no bincount/argsort/
+        all-to-all and no D2H slice sizes. Weight bytes scale with the
+        pair count, so the grouped path stays better at prefill sizes.
+        Returns None when unsupported (caller falls back)."""
+        H = xt.shape[-1]
+        ex = self.experts[0]
+        ff = ex.gate_proj.weight.shape[0]
+        if not (xt.is_cuda and xt.dtype == torch.bfloat16
+                and H % 512 == 0 and ff % 512 == 0
+                and type(ex).__name__ == "Expert"):
+            return None
+        from ..ops import native
+        lib = native.load(require=True)
+        T = xt.shape[0]
+        logits = nn.functional.linear(xt.float(),
+                                      self.router.weight.float())
+        probs = torch.softmax(logits, dim=-1)
+        gates, idx = probs.topk(self.top_k, dim=-1)
+        gates = gates / gates.sum(dim=-1, keepdim=True)
+        self.aux_loss = xt.new_zeros(())        # inference: unused
+        tg, tu, td = self._decode_tables(xt.device)
+        N = T * self.top_k
+        idxf = idx.reshape(-1).to(torch.int64)
+        xc = xt.contiguous()
+        y = torch.empty(N, ff, dtype=torch.bfloat16, device=xt.device)
+        rc = lib.gemv_moe_swiglu(native.stream_ptr(), tg.data_ptr(),
+                                 tu.data_ptr(), idxf.data_ptr(),
+                                 xc.data_ptr(), y.data_ptr(),
+                                 ff, H, N, self.top_k)
+        native.check_rc(rc, "gemv_moe_swiglu", f"ff={ff} H={H} N={N}")
+        z = torch.empty(N, H, dtype=torch.bfloat16, device=xt.device)
+        rc = lib.gemv_moe(native.stream_ptr(), td.data_ptr(),
+                          idxf.data_ptr(), y.data_ptr(), z.data_ptr(),
+                          H, ff, N, 1)
+        native.check_rc(rc, "gemv_moe", f"H={H} ff={ff} N={N}")
+        out = (z.reshape(T, self.top_k, H).float()
+               * gates[..., None]).sum(dim=1)
+        return out.to(xt.dtype)
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         orig_shape = x.shape
         H = orig_shape[-1]
         xt = x.reshape(-1, H)
+        if (not torch.is_grad_enabled() and self.tp_group is None
+                and _group_size(self.group) == 1
+                and xt.shape[0] * self.top_k <= 16):
+            fast = self._decode_forward(xt)
+            if fast is not None:
+                return fast.reshape(orig_shape)
         if self.tp_group is not None and _group_size(self.tp_group) > 1:
             # the ONE f-op for the whole MoE under EP x TP: forward
             # identity, backward sums the tp peers' partial input grads
