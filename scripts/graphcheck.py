@@ -38,21 +38,25 @@ def tokens_per_sec(model, batch, prompt_len, new_tokens):
 
 def check():
     from trainingjob_operator_amd.models.config import CONFIGS
-    from trainingjob_operator_amd.models.generate import generate
-    from trainingjob_operator_amd.training import build_model
-    cfg = CONFIGS["llama-smoke"]
-    m = build_model(cfg, torch.device("cuda:0"))
-    g = torch.Generator().manual_seed(3)
-    prompt = torch.randint(0, cfg.vocab_size, (2, 16),
-                           generator=g).to("cuda:0")
-    os.environ["AITJ_DECODE_GRAPH"] = "0"
-    ref = generate(m, prompt, max_new_tokens=24)
-    os.environ["AITJ_DECODE_GRAPH"] = "1"
-    got = generate(m, prompt, max_new_tokens=24)
-    match = (ref == got).float().mean().item()
-    print(f"token agreement graph vs eager: {match:.3f} "
-          f"({ref.shape[1]} positions)")
-    assert match > 0.95, (ref, got)
+    from trainingjob_operator_amd.models.generate import (
+        build_inference_model, generate,
+    )
+    for name in ("llama-smoke", "moe-mid"):
+        cfg = CONFIGS[name]
+        m = build_inference_model(cfg, torch.device("cuda:0"))
+        g = torch.Generator().manual_seed(3)
+        prompt = torch.randint(0, cfg.vocab_size, (2, 16),
+                               generator=g).to("cuda:0")
+        os.environ["AITJ_DECODE_GRAPH"] = "0"
+        ref = generate(m, prompt, max_new_tokens=24)
+        os.environ["AITJ_DECODE_GRAPH"] = "1"
+        got = generate(m, prompt, max_new_tokens=24)
+        match = (ref == got).float().mean().item()
+        print(f"{name}: token agreement graph vs eager: {match:.3f} "
+              f"({ref.shape[1]} positions)")
+        assert match > 0.95, (name, ref, got)
+        del m
+        torch.cuda.empty_cache()
     print("graphcheck OK")
 
 

@@ -82,7 +82,11 @@ def _step_static(model: LlamaModel, cur: torch.Tensor, cache: KVCache,
         normed, residual = fused_rmsnorm(attn_out,
                                          blk.post_attn_norm_weight,
                                          residual, cfg.norm_eps)
-        x = _mlp_cached(blk.mlp, normed)
+        # MoE blocks are capturable because the routed decode path keeps
+        # routing on device (parallel/ep.py _decode_forward): topk ->
+        # pointer-table GEMVs, fixed shapes, no host syncs
+        x = _mlp_cached(blk.mlp, normed) if hasattr(blk, "mlp") \
+            else blk.moe(normed)
     normed, _ = fused_rmsnorm(x, model.final_norm_weight, residual,
                               cfg.norm_eps)
     logits = decode_linear(normed, model.lm_head.weight)

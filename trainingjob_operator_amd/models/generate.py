@@ -243,9 +243,20 @@ def generate(model: LlamaModel, tokens: torch.Tensor,
         last = _forward_cached(model, tokens, cache)[:, -1]
         # opt-in hipGraph decode (see decode_graph.py for why not default)
         dec = None
+        def _moe_graphable(moe):
+            # must guarantee the ROUTED (sync-free) decode path engages,
+            # or capture would hit the grouped path's D2H slice sizes
+            ex = moe.experts[0]
+            return (moe.group is None and moe.tp_group is None
+                    and B * moe.top_k <= 16
+                    and type(ex).__name__ == "Expert"
+                    and ex.gate_proj.weight.shape[1] % 512 == 0
+                    and ex.gate_proj.weight.shape[0] % 512 == 0)
+
+        graph_ok = all(hasattr(b, "mlp") or _moe_graphable(b.moe)
+                       for b in model.blocks)
         if (os.environ.get("AITJ_DECODE_GRAPH") == "1" and tokens.is_cuda
-                and max_new_tokens > 2
-                and all(hasattr(b, "mlp") for b in model.blocks)):
+                and max_new_tokens > 2 and graph_ok):
             from .decode_graph import GraphedDecoder
             dec = GraphedDecoder(model, cache, B)
             dec.prime()
