@@ -68,8 +68,49 @@ def bench():
               f"{ms:.2f} ms/token")
 
 
+def session():
+    """DecodeSession: capture once, serve many requests (correctness on
+    llama-smoke + amortized timing on llama3-8b)."""
+    from trainingjob_operator_amd.models.config import CONFIGS
+    from trainingjob_operator_amd.models.decode_graph import DecodeSession
+    from trainingjob_operator_amd.models.generate import (
+        build_inference_model, generate,
+    )
+    cfg = CONFIGS["llama-smoke"]
+    m = build_inference_model(cfg, torch.device("cuda:0"))
+    ses = DecodeSession(m, batch=2, max_len=256)
+    g = torch.Generator().manual_seed(5)
+    for i in range(3):                     # 3 requests, one capture
+        prompt = torch.randint(0, cfg.vocab_size, (2, 12 + i),
+                               generator=g).to("cuda:0")
+        got = ses.generate(prompt, max_new_tokens=16)
+        os.environ["AITJ_DECODE_GRAPH"] = "0"
+        ref = generate(m, prompt, max_new_tokens=16)
+        match = (ref == got).float().mean().item()
+        print(f"request {i}: agreement {match:.3f}")
+        assert match > 0.95
+    del m
+    torch.cuda.empty_cache()
+
+    cfg = CONFIGS["llama3-8b"]
+    m = build_inference_model(cfg, torch.device("cuda:0"))
+    ses = DecodeSession(m, batch=1, max_len=1024)
+    prompt = torch.randint(0, cfg.vocab_size, (1, 512),
+                           generator=g).to("cuda:0")
+    ses.generate(prompt, max_new_tokens=8)           # capture + warmup
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    out = ses.generate(prompt, max_new_tokens=128)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    print(f"llama3-8b b1 session decode (no capture in-request): "
+          f"{128 / dt:.1f} tok/s, {dt * 1e3 / 128:.2f} ms/token")
+
+
 if __name__ == "__main__":
     if len(sys.argv) > 1 and sys.argv[1] == "bench":
         bench()
+    elif len(sys.argv) > 1 and sys.argv[1] == "session":
+        session()
     else:
         check()

@@ -53,6 +53,26 @@ def create_app(model, model_name: str = "model"):
     app = FastAPI(title="trainingjob-operator-amd serving", version="1.0")
     lock = threading.Lock()          # one generation at a time per GPU
     device = next(model.parameters()).device
+    sessions = {}                    # batch -> DecodeSession (graph mode)
+
+    def _session(batch: int, need_len: int):
+        """Reuse a captured-graph session per batch size (the capture is
+        position-independent; see models/decode_graph.DecodeSession).
+        Returns None when the graph path is off or doesn't fit."""
+        import os
+        if os.environ.get("AITJ_DECODE_GRAPH") != "1" \
+                or device.type != "cuda":
+            return None
+        ses = sessions.get(batch)
+        if ses is not None and need_len <= ses.max_len:
+            return ses
+        if ses is None and len(sessions) < 4:
+            from ..models.decode_graph import DecodeSession
+            max_len = max(need_len + 8, 2048)
+            ses = DecodeSession(model, batch, max_len)
+            sessions[batch] = ses
+            return ses
+        return None                  # fits neither cache: eager path
 
     @app.get("/healthz")
     def healthz():
@@ -87,10 +107,19 @@ def create_app(model, model_name: str = "model"):
                               device=device)
         with lock:
             t0 = time.perf_counter()
-            out = generate(model, prompt,
-                           max_new_tokens=req.max_new_tokens,
-                           temperature=req.temperature, top_k=req.top_k,
-                           eos_token=req.eos_token, seed=req.seed)
+            ses = _session(prompt.shape[0],
+                           prompt.shape[1] + req.max_new_tokens)
+            if ses is not None:
+                out = ses.generate(prompt, req.max_new_tokens,
+                                   temperature=req.temperature,
+                                   top_k=req.top_k,
+                                   eos_token=req.eos_token, seed=req.seed)
+            else:
+                out = generate(model, prompt,
+                               max_new_tokens=req.max_new_tokens,
+                               temperature=req.temperature,
+                               top_k=req.top_k,
+                               eos_token=req.eos_token, seed=req.seed)
             if device.type == "cuda":
                 torch.cuda.synchronize()
             dt = time.perf_counter() - t0
