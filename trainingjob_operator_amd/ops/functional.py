@@ -252,12 +252,12 @@ def decode_attention(q: torch.Tensor, kc: torch.Tensor,
     n_kv, Lmax = kc.shape[1], kc.shape[2]
     G = H // n_kv if H % n_kv == 0 else 0
     if not (q.is_cuda and q.dtype == torch.bfloat16 and D == 128
-            and S == 1 and G in (1, 2, 4, 8) and kc.is_contiguous()
+            and S == 1 and G >= 1 and kc.is_contiguous()
             and vc.is_contiguous() and pos_t.dtype == torch.int64):
         return None
     lib = _hip()
     q2 = q.reshape(B, H, D).contiguous()
-    n_chunk = (Lmax + 127) // 128
+    n_chunk = (Lmax + 63) // 64
     partial = torch.empty(B, n_kv, n_chunk, G, 130, dtype=torch.float32,
                           device=q.device)
     o = torch.empty(B, H, D, dtype=torch.bfloat16, device=q.device)

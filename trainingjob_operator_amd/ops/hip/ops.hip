@@ -980,97 +980,86 @@ __global__ __launch_bounds__(256) void gemv_bf16_kernel(
 // between the bf16 loads and the bf16 store.
 // ---------------------------------------------------------------------------
 
-#define ADEC_CHUNK 128
+#define ADEC_CHUNK 64
 #define ADEC_D 128
 
-template <int G>
-__global__ __launch_bounds__(128) void attn_decode_partial_kernel(
+// grid (n_kv * G, n_chunk, B), ONE 64-thread wave per (kv head, q head,
+// 64-row cache chunk): at decode batch sizes the whole chip is idle, so
+// fill is everything — splitting the GQA heads apart and halving the
+// chunk gives 8x the workgroups of the first version (which ran 40 WGs
+// at b1 and was latency-bound at 18 us) for a trivial re-read of the
+// tiny K rows. Phase A: lane <-> cache row; phase B: lane <-> d pair.
+__global__ __launch_bounds__(64) void attn_decode_partial_kernel(
     const u16* __restrict__ q,       // [B, n_kv*G, D]
     const u16* __restrict__ kc,      // [B, n_kv, Lmax, D]
     const u16* __restrict__ vc,
     const long* __restrict__ pos_dev,
     float* __restrict__ partial,     // [B, n_kv, n_chunk, G, D+2]
-    int n_kv, int Lmax, int n_chunk, float scale) {
-  const int hkv = blockIdx.x;
+    int n_kv, int G, int Lmax, int n_chunk, float scale) {
+  const int hkv = blockIdx.x / G;
+  const int g = blockIdx.x % G;
   const int chunk = blockIdx.y;
   const int b = blockIdx.z;
   const int tid = threadIdx.x;
   const long pos = *pos_dev;               // rows 0..pos are valid
-  __shared__ float sp[ADEC_CHUNK][G];      // scores, then p
-  __shared__ u16 qs[G][ADEC_D];
-  for (int i = tid; i < G * ADEC_D; i += 128)
-    qs[i / ADEC_D][i % ADEC_D] =
-        q[((long)b * n_kv * G + hkv * G + i / ADEC_D) * ADEC_D
-          + i % ADEC_D];
+  __shared__ float sp[ADEC_CHUNK];         // scores, then p
+  __shared__ u16 qs[ADEC_D];
+  reinterpret_cast<u32*>(qs)[tid] = reinterpret_cast<const u32*>(
+      q + ((long)b * n_kv * G + hkv * G + g) * ADEC_D)[tid];
   __syncthreads();
 
   const long row = (long)chunk * ADEC_CHUNK + tid;
-  float s[G];
-#pragma unroll
-  for (int g = 0; g < G; ++g) s[g] = 0.f;
   if (row <= pos && row < Lmax) {
     const u16* krow = kc + (((long)b * n_kv + hkv) * Lmax + row) * ADEC_D;
+    float sx = 0.f;
     union V8 { uint4 u; u16 h[8]; };
 #pragma unroll 4
     for (int c = 0; c < ADEC_D; c += 8) {
       V8 kv8;
       kv8.u = *reinterpret_cast<const uint4*>(krow + c);
 #pragma unroll
-      for (int g = 0; g < G; ++g)
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          s[g] += bf2f(qs[g][c + j]) * bf2f(kv8.h[j]);
+      for (int j = 0; j < 8; ++j) sx += bf2f(qs[c + j]) * bf2f(kv8.h[j]);
     }
-#pragma unroll
-    for (int g = 0; g < G; ++g) sp[tid][g] = s[g] * scale;
+    sp[tid] = sx * scale;
   } else {
-#pragma unroll
-    for (int g = 0; g < G; ++g) sp[tid][g] = -1e30f;
+    sp[tid] = -1e30f;
   }
   __syncthreads();
 
-  // chunk max per g (every lane scans the 128 LDS values: broadcast
-  // reads, no reduction tree needed at this size), then p = exp(s - m)
-  float m[G], p[G];
-#pragma unroll
-  for (int g = 0; g < G; ++g) m[g] = -1e30f;
-  for (int r = 0; r < ADEC_CHUNK; ++r)
-#pragma unroll
-    for (int g = 0; g < G; ++g) m[g] = fmaxf(m[g], sp[r][g]);
-#pragma unroll
-  for (int g = 0; g < G; ++g) {
-    p[g] = (sp[tid][g] > -1e30f) ? __expf(sp[tid][g] - m[g]) : 0.f;
-  }
+  // chunk max (every lane scans the 64 LDS values: broadcast reads),
+  // then p = exp(s - m) written back over s
+  float m = -1e30f;
+  for (int r = 0; r < ADEC_CHUNK; ++r) m = fmaxf(m, sp[r]);
+  const float pv = (sp[tid] > -1e30f) ? __expf(sp[tid] - m) : 0.f;
   __syncthreads();
-#pragma unroll
-  for (int g = 0; g < G; ++g) sp[tid][g] = p[g];
+  sp[tid] = pv;
   __syncthreads();
 
-  // phase B: lane <-> d; acc[g] = sum_row p[row][g] * v[row][d]
-  float acc[G];
-#pragma unroll
-  for (int g = 0; g < G; ++g) acc[g] = 0.f;
-  const long lim = pos + 1 - (long)chunk * ADEC_CHUNK;   // valid rows here
+  // phase B: lane <-> d pair; v reads coalesce per row (64 x u32)
+  const long lim = pos + 1 - (long)chunk * ADEC_CHUNK;
   const int nrow = lim < 0 ? 0 : (lim < ADEC_CHUNK ? (int)lim : ADEC_CHUNK);
   const u16* vbase = vc
       + (((long)b * n_kv + hkv) * Lmax + (long)chunk * ADEC_CHUNK) * ADEC_D
-      + tid;
+      + tid * 2;
+  float a0 = 0.f, a1 = 0.f;
   for (int r = 0; r < nrow; ++r) {
-    const float vv = bf2f(vbase[(long)r * ADEC_D]);
-#pragma unroll
-    for (int g = 0; g < G; ++g) acc[g] += sp[r][g] * vv;
+    const float p = sp[r];
+    union { u32 w; u16 h[2]; } vv;
+    vv.w = *reinterpret_cast<const u32*>(vbase + (long)r * ADEC_D);
+    a0 += p * bf2f(vv.h[0]);
+    a1 += p * bf2f(vv.h[1]);
   }
 
   float* out = partial
-      + ((((long)b * n_kv + hkv) * n_chunk + chunk) * G) * (ADEC_D + 2);
-#pragma unroll
-  for (int g = 0; g < G; ++g)
-    out[(long)g * (ADEC_D + 2) + tid] = acc[g];
-  if (tid < G) {            // lane g writes its head's chunk m and l
+      + ((((long)b * n_kv + hkv) * n_chunk + chunk) * G + g)
+      * (ADEC_D + 2);
+  out[tid * 2] = a0;
+  out[tid * 2 + 1] = a1;
+  if (tid == 0) {
     float l = 0.f;
-    for (int r = 0; r < ADEC_CHUNK; ++r) l += sp[r][tid];
-    out[(long)tid * (ADEC_D + 2) + ADEC_D] = m[tid];
-    out[(long)tid * (ADEC_D + 2) + ADEC_D + 1] = l;
+    for (int r = 0; r < ADEC_CHUNK; ++r) l += sp[r];
+    out[ADEC_D] = m;
+    out[ADEC_D + 1] = l;
   }
 }
 
@@ -1704,21 +1693,13 @@ int attn_decode(void* stream, const void* q, const void* kc,
       n_chunk != (Lmax + ADEC_CHUNK - 1) / ADEC_CHUNK)
     return -1;
   const int G = H / n_kv;
-  dim3 grid(n_kv, n_chunk, B), block(128);
-  switch (G) {
-#define ADEC_CASE(GG) \
-    case GG: \
-      hipLaunchKernelGGL((attn_decode_partial_kernel<GG>), grid, block, \
-                         0, STREAM, (const u16*)q, (const u16*)kc, \
-                         (const u16*)vc, (const long*)pos_dev, \
-                         (float*)partial, n_kv, Lmax, n_chunk, scale); \
-      break;
-    ADEC_CASE(1) ADEC_CASE(2) ADEC_CASE(4) ADEC_CASE(8)
-#undef ADEC_CASE
-    default: return -1;
-  }
+  dim3 grid((unsigned)(n_kv * G), (unsigned)n_chunk, (unsigned)B);
+  hipLaunchKernelGGL(attn_decode_partial_kernel, grid, dim3(64), 0,
+                     STREAM, (const u16*)q, (const u16*)kc,
+                     (const u16*)vc, (const long*)pos_dev,
+                     (float*)partial, n_kv, G, Lmax, n_chunk, scale);
   hipLaunchKernelGGL(attn_decode_combine_kernel, dim3((unsigned)(B * H)),
-                     block, 0, STREAM, (const float*)partial, (u16*)o,
+                     dim3(128), 0, STREAM, (const float*)partial, (u16*)o,
                      n_kv, G, n_chunk);
   return 0;
 }
