@@ -114,3 +114,12 @@ def test_serve_moe_model():
                                   "max_new_tokens": 4})
     assert r.status_code == 200, r.text
     assert len(r.json()["tokens"][0]) == 7
+
+
+def test_metrics_endpoint(client):
+    client.post("/generate", json={"prompt_tokens": [[1, 2]],
+                                   "max_new_tokens": 2})
+    r = client.get("/metrics")
+    assert r.status_code == 200
+    assert "aitj_serve_requests_total" in r.text
+    assert "aitj_serve_tokens_total" in r.text

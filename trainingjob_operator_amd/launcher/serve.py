@@ -54,6 +54,21 @@ def create_app(model, model_name: str = "model"):
     lock = threading.Lock()          # one generation at a time per GPU
     device = next(model.parameters()).device
     sessions = {}                    # batch -> DecodeSession (graph mode)
+    try:                             # prometheus (same registry the
+        from prometheus_client import (  # operator's metrics use)
+            Counter, Histogram, generate_latest,
+        )
+        mx = {
+            "requests": Counter("aitj_serve_requests_total",
+                                "Generation requests", ["status"]),
+            "tokens": Counter("aitj_serve_tokens_total",
+                              "Tokens decoded"),
+            "latency": Histogram("aitj_serve_request_seconds",
+                                 "Request wall time",
+                                 buckets=(.05, .2, .5, 1., 2., 5., 15.)),
+        }
+    except Exception:                # prometheus_client optional
+        mx, generate_latest = None, None
 
     def _session(batch: int, need_len: int):
         """Reuse a captured-graph session per batch size (the capture is
@@ -89,8 +104,17 @@ def create_app(model, model_name: str = "model"):
             "graph_decode": os.environ.get("AITJ_DECODE_GRAPH") == "1",
         }
 
+    @app.get("/metrics")
+    def metrics():
+        from fastapi.responses import PlainTextResponse
+        if generate_latest is None:
+            raise HTTPException(404, "prometheus_client not installed")
+        return PlainTextResponse(generate_latest().decode())
+
     @app.post("/generate")
     def gen(req: GenerateRequest):
+        if mx:
+            mx["requests"].labels(status="received").inc()
         if not req.prompt_tokens or not all(req.prompt_tokens):
             raise HTTPException(400, "prompt_tokens must be non-empty")
         lens = {len(p) for p in req.prompt_tokens}
@@ -124,6 +148,10 @@ def create_app(model, model_name: str = "model"):
                 torch.cuda.synchronize()
             dt = time.perf_counter() - t0
         n_new = out.shape[1] - prompt.shape[1]
+        if mx:
+            mx["requests"].labels(status="ok").inc()
+            mx["tokens"].inc(out.shape[0] * n_new)
+            mx["latency"].observe(dt)
         return {"tokens": out.tolist(),
                 "decode_tok_s": round(out.shape[0] * n_new / dt, 1)}
 
