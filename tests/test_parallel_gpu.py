@@ -159,3 +159,16 @@ def test_operator_drives_gpu_training(tmp_path):
         stop.set()
         t.join(timeout=10)
         kubelet.stop()
+
+
+def test_cp_trainer_single_gpu():
+    """Context-parallel trainer, world 1 (the ring degenerates to local
+    flash-style accumulation; multi-rank rings are gloo-covered)."""
+    from trainingjob_operator_amd.parallel.cp import CPTrainer
+    from trainingjob_operator_amd.training import TrainConfig
+    cfg = TrainConfig(model="llama-tiny", micro_batch=2, grad_accum=2,
+                      seq_len=64, lr=1e-3)
+    tr = CPTrainer(cfg, device=DEV)
+    losses = [float(tr.train_step()) for _ in range(3)]
+    assert all(l == l for l in losses), losses
+    assert tr.step_count == 3

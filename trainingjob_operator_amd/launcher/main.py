@@ -179,6 +179,10 @@ def main(argv=None) -> int:
                                            "") == "1",
                     help="shard the LM head + CE over the tp group "
                          "(requires --tp > 1)")
+    ap.add_argument("--cp", type=int, default=int(os.environ.get(
+        "AITJ_CP_SIZE", "0")),
+                    help="context parallelism: shard the sequence cp-ways "
+                         "with ring attention (world = dp x cp)")
     ap.add_argument("--sp", action="store_true",
                     default=os.environ.get("TRAININGJOB_SP", "") == "1",
                     help="sequence parallelism on top of --tp (Megatron "
@@ -244,6 +248,11 @@ def main(argv=None) -> int:
         assert args.tp > 1 and args.pp == 1 and not args.ep, \
             "--vocab-parallel is a pure-TP option"
         assert not args.sp, "vocab-parallel with --sp is roadmap"
+    if args.cp > 1:
+        assert args.tp == 1 and args.pp == 1 and not args.ep \
+            and not args.zero1, \
+            "--cp composes with dp only (tp/pp/ep grids: roadmap)"
+        assert args.seq_len % args.cp == 0, "--cp needs seq_len % cp == 0"
     if args.sp:
         assert args.tp > 1, "--sp requires --tp > 1"
         assert args.seq_len % args.tp == 0, "--sp needs seq_len % tp == 0"
@@ -284,6 +293,14 @@ def main(argv=None) -> int:
         if args.tp > 1:
             sub += f"_tp{grid.tp_rank}"
         ckpt = Checkpointer(os.path.join(args.ckpt_dir, sub))
+    elif args.cp > 1:
+        from ..parallel.cp import CPTrainer
+        import torch
+        dev = torch.device(f"cuda:{ctx.local_rank}"
+                           if torch.cuda.is_available() else "cpu")
+        trainer = CPTrainer(cfg, cp_size=args.cp, device=dev)
+        # weights replicated everywhere: one rank-0 stream, like DP
+        ckpt = Checkpointer(args.ckpt_dir)
     elif args.ep:
         from ..parallel.ep import EPTrainer
         import torch
