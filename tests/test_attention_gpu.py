@@ -246,3 +246,21 @@ def test_fp8_projections_loss_parity(monkeypatch):
     monkeypatch.delenv("AITJ_FP8_PROJ", raising=False)
     for a, b in zip(l_bf, l_f8):
         assert abs(a - b) < 0.15, (l_bf, l_f8)
+
+
+def test_prompt_lookup_identical_on_gpu():
+    """Speculative verify exercises the mid-cache offset-causal mask and
+    the fused decode path on silicon; output must equal plain greedy."""
+    import torch
+    from trainingjob_operator_amd.models.config import CONFIGS
+    from trainingjob_operator_amd.models.generate import (
+        generate, generate_lookup,
+    )
+    from trainingjob_operator_amd.training import build_model
+    torch.manual_seed(3)
+    m = build_model(CONFIGS["llama-smoke"], torch.device("cuda:0"))
+    prompt = torch.tensor([[11, 12, 13, 11, 12, 13, 11, 12]],
+                          device="cuda:0")
+    ref = generate(m, prompt, max_new_tokens=24)
+    got = generate_lookup(m, prompt, max_new_tokens=24, lookup_k=6)
+    assert torch.equal(ref, got), (ref, got)

@@ -104,3 +104,31 @@ def test_moe_generate_matches_uncached():
             nxt = moe(ref)[:, -1].argmax(dim=-1)
             ref = torch.cat([ref, nxt[:, None]], dim=1)
     assert torch.equal(out, ref)
+
+
+def test_prompt_lookup_identical_to_greedy(model):
+    """Prompt-lookup speculative decode must emit EXACTLY the greedy
+    sequence — speculation changes the forward count, never the
+    output — on repetitive (draft-accepting) and random prompts."""
+    from trainingjob_operator_amd.models.generate import generate_lookup
+    rep = torch.tensor([[5, 6, 7, 5, 6, 7, 5, 6]])
+    g = torch.Generator().manual_seed(9)
+    rnd = torch.randint(0, 512, (1, 8), generator=g)
+    for prompt in (rep, rnd):
+        ref = generate(model, prompt, max_new_tokens=16)
+        for k in (2, 4, 8):
+            got = generate_lookup(model, prompt, max_new_tokens=16,
+                                  lookup_k=k)
+            assert torch.equal(ref, got), (k, ref, got)
+
+
+def test_prompt_lookup_eos(model):
+    from trainingjob_operator_amd.models.generate import generate_lookup
+    rep = torch.tensor([[5, 6, 7, 5, 6, 7, 5, 6]])
+    ref = generate(model, rep, max_new_tokens=16)
+    eos = int(ref[0, 12])
+    ref_e = generate(model, rep, max_new_tokens=16, eos_token=eos)
+    got = generate_lookup(model, rep, max_new_tokens=16, lookup_k=4,
+                          eos_token=eos)
+    assert torch.equal(ref_e[0, :got.shape[1]], got[0])
+    assert int(got[0, -1]) == eos

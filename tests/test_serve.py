@@ -123,3 +123,13 @@ def test_metrics_endpoint(client):
     assert r.status_code == 200
     assert "aitj_serve_requests_total" in r.text
     assert "aitj_serve_tokens_total" in r.text
+
+
+def test_generate_prompt_lookup(client):
+    base = client.post("/generate", json={
+        "prompt_tokens": [[5, 6, 7, 5, 6, 7, 5, 6]],
+        "max_new_tokens": 12}).json()["tokens"]
+    spec = client.post("/generate", json={
+        "prompt_tokens": [[5, 6, 7, 5, 6, 7, 5, 6]],
+        "max_new_tokens": 12, "prompt_lookup": 4}).json()["tokens"]
+    assert base == spec

@@ -49,6 +49,10 @@ def create_app(model, model_name: str = "model"):
         top_k: int = 0
         eos_token: Optional[int] = None
         seed: Optional[int] = None
+        # greedy batch-1 only: propose continuations from earlier n-gram
+        # matches in the context, verify in one forward (output identical
+        # to plain greedy; wins on repetitive text)
+        prompt_lookup: int = 0
 
     app = FastAPI(title="trainingjob-operator-amd serving", version="1.0")
     lock = threading.Lock()          # one generation at a time per GPU
@@ -131,9 +135,16 @@ def create_app(model, model_name: str = "model"):
                               device=device)
         with lock:
             t0 = time.perf_counter()
-            ses = _session(prompt.shape[0],
-                           prompt.shape[1] + req.max_new_tokens)
-            if ses is not None:
+            use_lookup = (req.prompt_lookup > 0 and req.temperature <= 0
+                          and prompt.shape[0] == 1)
+            ses = None if use_lookup else _session(
+                prompt.shape[0], prompt.shape[1] + req.max_new_tokens)
+            if use_lookup:
+                from ..models.generate import generate_lookup
+                out = generate_lookup(model, prompt, req.max_new_tokens,
+                                      lookup_k=req.prompt_lookup,
+                                      eos_token=req.eos_token)
+            elif ses is not None:
                 out = ses.generate(prompt, req.max_new_tokens,
                                    temperature=req.temperature,
                                    top_k=req.top_k,

@@ -146,8 +146,17 @@ def _attn_cached(attn, x: torch.Tensor, cos: torch.Tensor,
     vv = cv[:, :, :pos0 + S]
     gqa = cfg.num_heads != cfg.num_kv_heads
     if S > 1:
-        # prefill: causal among the new tokens
-        o = _sdpa(q, kk, vv, enable_gqa=gqa)
+        if pos0 == 0:
+            # prefill: causal among the new tokens
+            o = _sdpa(q, kk, vv, enable_gqa=gqa)
+        else:
+            # multi-token continuation mid-cache (speculative verify):
+            # is_causal aligns TOP-LEFT in torch SDPA, which would hide
+            # the cache — build the offset (bottom-right) causal mask
+            mask = torch.ones(S, pos0 + S, dtype=torch.bool,
+                              device=q.device).tril(pos0)
+            o = torch.nn.functional.scaled_dot_product_attention(
+                q, kk, vv, attn_mask=mask, enable_gqa=gqa)
     else:
         o = _decode_attn(q, kk, vv, cfg.num_kv_heads, ck, cv,
                          getattr(attn, "_pos_dec", None))
@@ -278,6 +287,77 @@ def generate(model: LlamaModel, tokens: torch.Tensor,
             last = dec.step(nxt) if dec is not None else \
                 _forward_cached(model, nxt[:, None], cache)[:, -1]
         return out
+    finally:
+        if was_training:
+            model.train()
+
+
+def _ngram_proposal(ctx: list, k: int) -> list:
+    """Prompt-lookup draft: find the latest earlier occurrence of the
+    current 3-gram (then 2-gram) suffix and propose the k tokens that
+    followed it. Free speculation — no draft model — that pays off on
+    repetitive text (code, retrieval contexts) and costs one batched
+    verify forward otherwise."""
+    L = len(ctx)
+    for n in (3, 2):
+        if L < n + 1:
+            continue
+        tail = ctx[L - n:]
+        # search latest match of the n-gram strictly before the suffix
+        for i in range(L - n - 1, -1, -1):
+            if ctx[i:i + n] == tail:
+                prop = ctx[i + n:i + n + k]
+                if prop:
+                    return prop
+    return []
+
+
+@torch.no_grad()
+def generate_lookup(model, tokens: torch.Tensor, max_new_tokens: int,
+                    lookup_k: int = 8,
+                    eos_token: Optional[int] = None) -> torch.Tensor:
+    """Greedy generation with prompt-lookup speculative decoding
+    (batch 1): propose up to lookup_k continuation tokens from an
+    earlier occurrence of the current n-gram, verify them in ONE
+    cached forward, accept the matching prefix plus the bonus token
+    from the first mismatch, and roll the cache back over the rest.
+    Output is IDENTICAL to plain greedy generate() — speculation only
+    changes how many forwards it takes (tests/test_generate.py)."""
+    assert tokens.shape[0] == 1, "prompt-lookup decode is batch-1"
+    was_training = model.training
+    model.eval()
+    try:
+        B, S0 = tokens.shape
+        cache = KVCache(model.cfg, B, S0 + max_new_tokens + lookup_k + 1,
+                        tokens.device, model.embed.weight.dtype)
+        logits = _forward_cached(model, tokens, cache)
+        cur = int(logits[0, -1].argmax())
+        ctx = tokens[0].tolist() + [cur]
+        produced = 1
+        while produced < max_new_tokens and                 (eos_token is None or cur != eos_token):
+            room = max_new_tokens - produced
+            prop = _ngram_proposal(ctx, min(lookup_k, room)) if room > 1                 else []
+            step = torch.tensor([[cur] + prop], device=tokens.device)
+            pos0 = cache.len
+            logits = _forward_cached(model, step, cache)
+            nxt = logits[0].argmax(dim=-1).tolist()   # true tokens
+            accepted = 0
+            for j, pj in enumerate(prop):
+                if pj == nxt[j] and accepted + 1 < room:
+                    accepted += 1
+                else:
+                    break
+            emitted = nxt[:accepted + 1]              # accepted + bonus
+            # roll back cache rows written for rejected draft tokens
+            cache.len = pos0 + 1 + accepted
+            ctx.extend(emitted)
+            produced += len(emitted)
+            cur = emitted[-1]
+            if eos_token is not None and eos_token in emitted:
+                ctx = ctx[:len(ctx) - len(emitted)
+                          + emitted.index(eos_token) + 1]
+                break
+        return torch.tensor([ctx], device=tokens.device)
     finally:
         if was_training:
             model.train()
