@@ -324,6 +324,8 @@ def generate_lookup(model, tokens: torch.Tensor, max_new_tokens: int,
     Output is IDENTICAL to plain greedy generate() — speculation only
     changes how many forwards it takes (tests/test_generate.py)."""
     assert tokens.shape[0] == 1, "prompt-lookup decode is batch-1"
+    if max_new_tokens <= 0:
+        return tokens
     was_training = model.training
     model.eval()
     try:
