@@ -180,3 +180,22 @@ def test_cp_composes_with_dp():
     assert results[0] == results[1]
     assert results[2] == results[3]
     assert results[0][1] != results[2][1]
+
+
+def test_cp_checkpoint_is_a_dp_checkpoint(tmp_path):
+    """CP replicates weights with the same module names/order as the
+    dense model, so its flat checkpoint IS a DP checkpoint: save from
+    CPTrainer, resume in the plain Trainer (and back) with no reshard."""
+    import torch as t
+    from trainingjob_operator_amd.launcher.checkpoint import Checkpointer
+    from trainingjob_operator_amd.parallel.cp import CPTrainer
+    from trainingjob_operator_amd.training import TrainConfig, Trainer
+    cfg = TrainConfig(model="llama-tiny", micro_batch=1, grad_accum=1,
+                      seq_len=32, lr=1e-3)
+    cp = CPTrainer(cfg, device="cpu")
+    cp.train_step()
+    Checkpointer(str(tmp_path)).save_async(cp, blocking=True)
+    dense = Trainer(cfg, device=t.device("cpu"))
+    assert Checkpointer(str(tmp_path)).load_latest(dense) == 1
+    assert t.equal(dense.store.flat_param, cp.store.flat_param)
+    dense.train_step()          # resumed state actually trains
