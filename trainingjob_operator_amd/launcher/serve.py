@@ -85,13 +85,15 @@ def create_app(model, model_name: str = "model"):
         ses = sessions.get(batch)
         if ses is not None and need_len <= ses.max_len:
             return ses
-        if ses is None and len(sessions) < 4:
+        if ses is not None:          # longer request: rebuild bigger
+            del sessions[batch]
+        if len(sessions) < 4:
             from ..models.decode_graph import DecodeSession
             max_len = max(need_len + 8, 2048)
             ses = DecodeSession(model, batch, max_len)
             sessions[batch] = ses
             return ses
-        return None                  # fits neither cache: eager path
+        return None                  # session pool full: eager path
 
     @app.get("/healthz")
     def healthz():
