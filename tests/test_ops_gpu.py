@@ -339,7 +339,8 @@ def test_decode_attention_matches_fp32():
     from trainingjob_operator_amd.ops import decode_attention
     torch.manual_seed(21)
     for B, H, n_kv, Lmax, pos in [(1, 32, 8, 640, 517), (2, 8, 8, 256, 255),
-                                  (2, 16, 4, 384, 127), (1, 8, 1, 128, 0)]:
+                                  (2, 16, 4, 384, 127), (1, 8, 1, 128, 0),
+                                  (1, 8, 4, 200, 150)]:   # Lmax % 64 != 0
         D = 128
         q = _mk((B, H, 1, D))
         kc = _mk((B, n_kv, Lmax, D))
