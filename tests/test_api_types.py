@@ -190,3 +190,29 @@ def test_k8stime_roundtrip():
     assert parse_time(None) is None
     assert parse_time("") is None
     assert parse_time(12.5) == 12.5
+
+
+def test_example_manifests_validate():
+    """Every shipped example AITrainingJob must parse, default, and pass
+    the validator (keeps manifests/ honest as validation evolves)."""
+    import glob
+    import os
+
+    import yaml
+
+    from trainingjob_operator_amd.api.defaults import set_defaults
+    from trainingjob_operator_amd.api.types import AITrainingJob
+    from trainingjob_operator_amd.api.validation import validate
+    root = os.path.join(os.path.dirname(__file__), "..", "manifests",
+                        "examples")
+    found = 0
+    for f in sorted(glob.glob(os.path.join(root, "*.yaml"))):
+        with open(f) as fh:
+            d = yaml.safe_load(fh)
+        if d.get("kind") != "AITrainingJob":
+            continue
+        job = AITrainingJob.from_dict(d)
+        set_defaults(job)
+        assert validate(job) == [], f
+        found += 1
+    assert found >= 8
