@@ -12,6 +12,20 @@ def pytest_configure(config):
     )
 
 
+def pytest_collection_finish(session):
+    """Backstop timeout on every test (pytest-timeout): the spawn-heavy
+    gloo tests have a rare rendezvous flake (ROADMAP §13) — a test that
+    wedges should FAIL loudly, never hang a CI/driver run. 900 s is ~15x
+    the slowest legitimate test; explicit @pytest.mark.timeout wins."""
+    try:
+        import pytest_timeout  # noqa: F401
+    except ImportError:
+        return
+    for item in session.items:
+        if item.get_closest_marker("timeout") is None:
+            item.add_marker(pytest.mark.timeout(900))
+
+
 def pytest_collection_modifyitems(config, items):
     """Skip gpu-marked tests automatically when no GPU is present."""
     try:
