@@ -1511,7 +1511,6 @@ __global__ __launch_bounds__(512) void attn_fwd_v7_kernel(
   }
   const long kstep = (long)V6_BN * k_ss;
   const long vstep = (long)V6_BN * v_ss;
-  const int wu64 = __builtin_amdgcn_readfirstlane(tid >> 6);
   const int lslot = (tid & ~63) * 8;          // wave-uniform u16 base
 
 #define V7_GLDS(SLOT)                                                       \
