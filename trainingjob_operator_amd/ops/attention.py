@@ -1,18 +1,20 @@
-"""Native MFMA flash-attention forward (hip/attention.hip) with the aten
-flash backward for training.
+"""Native MFMA flash attention (hip/attention.hip): hand-written CDNA4
+forward AND backward, the default SDPA path for training.
 
-Two hand-written CDNA4 forwards:
-  * v6 (S % 256 == 0): 8-wave swapped-operand kernel - 32x32x16 MFMA,
-    in-register softmax (q row lane-local on both products), defer-rescale,
-    double-buffered K/V staging, native GQA. The production path.
+Forwards:
+  * v7 (S % 256 == 0, the production path): 8-wave swapped-operand
+    32x32x16 MFMA, in-register softmax, 3-slot all-glds K/V ring with
+    ONE raw barrier per tile, native GQA — 299 us vs the aotriton
+    library's 425 at the flagship shape.
   * v5 (S % 64 == 0 fallback): 4-wave 16x16x32 kernel with the LDS P
     round-trip; requires equal head counts (GQA expanded by the caller).
 
-The forward emits exactly what aten::_scaled_dot_product_flash_attention_
-backward consumes (O + logsumexp at natural-log scale), so the hand-written
-forward drops into autograd with the library backward until the native
-backward lands. GQA head counts pass straight through (the aten backward
-reduces dk/dv to the kv heads).
+Backward (S % 256 == 0): delta + dq + dv + dk kernels (fwd+bwd 2858 us
+vs aotriton's 4140 — profiles/attention_kernel_notes.md). The forward
+emits O + logsumexp at natural-log scale — exactly what
+aten::_scaled_dot_product_flash_attention_backward consumes — so the
+aten backward remains a drop-in fallback (AITJ_ATTN_BWD=aten, and the
+automatic path for S % 256 != 0).
 """
 from __future__ import annotations
 
