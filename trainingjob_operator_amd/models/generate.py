@@ -5,12 +5,15 @@ Training is the flagship workload, but a framework its users can switch
 to needs an eval/inference path: greedy or temperature/top-k sampling
 with a per-layer KV cache (prefill once, then one-token decode steps).
 
-The decode path reuses the training modules' weights and the fused
-rmsnorm/SwiGLU ops; RoPE uses an explicit-position rotation here (the
-training kernel derives positions from flat indices, which is wrong for
-a 1-token step at position p) — matching ops/reference.py rope_rotate
-semantics (Neox half-rotation). Decode-side kernel fusion (paged KV,
-fused decode attention) is round-2 serving work — see ROADMAP.md.
+The decode path reuses the training modules' weights and runs the
+hand-written decode stack: wave-per-row GEMV projections
+(ops.decode_linear / decode_swiglu), fused flash-decoding attention
+over the cache (ops.decode_attention, device-resident length), native
+explicit-position RoPE, and — opt-in via AITJ_DECODE_GRAPH=1 — the
+hipGraph-captured step in models/decode_graph.py. Prompt-lookup
+speculative decoding (generate_lookup) emits greedy-identical output
+with fewer forwards. See docs/SERVING.md and
+profiles/r02_serving_profile.md for the measured arc.
 """
 from __future__ import annotations
 
